@@ -1,0 +1,234 @@
+"""Validating admission for grove.io resources.
+
+Behavior parity with /root/reference/operator/internal/webhook/admission/pcs/validation/
+podcliqueset.go (structure checks, 45-char combined-name budget :44,1024-1040, startup DAG
+acyclicity :464-489, PCSG membership rules :337-404,499-519, minAvailable/scaleConfig
+constraints :573-589, immutable-field update rules :643-702). Fresh implementation.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from . import constants as c
+from .defaulting import parse_duration_seconds
+from ..kubecore.store import invalid
+
+Obj = Dict[str, Any]
+
+MAX_COMBINED_RESOURCE_NAME_LENGTH = 45
+
+
+def _err(path: str, msg: str):
+    return invalid(f"{path}: {msg}")
+
+
+def _check_name_budget(pcs_name: str, pcsg_name: str, pclq_name: str) -> None:
+    total = len(pcs_name) + len(pcsg_name) + len(pclq_name)
+    if total > MAX_COMBINED_RESOURCE_NAME_LENGTH:
+        if pcsg_name:
+            raise invalid(
+                f"combined resource name length {total} exceeds 45-character limit required "
+                f"for pod naming. Consider shortening: PodCliqueSet '{pcs_name}', "
+                f"PodCliqueScalingGroup '{pcsg_name}', or PodClique '{pclq_name}'")
+        raise invalid(
+            f"combined resource name length {total} exceeds 45-character limit required "
+            f"for pod naming. Consider shortening: PodCliqueSet '{pcs_name}' or "
+            f"PodClique '{pclq_name}'")
+
+
+def _check_startup_dag(cliques: List[Obj], startup_type: str) -> None:
+    names = [cl.get("name") for cl in cliques]
+    name_set = set(names)
+    if startup_type != c.STARTUP_EXPLICIT:
+        for cl in cliques:
+            if cl.get("spec", {}).get("startsAfter"):
+                raise _err("spec.template.cliques",
+                           "startsAfter may only be set with CliqueStartupTypeExplicit")
+        return
+    deps = {}
+    for cl in cliques:
+        sa = cl.get("spec", {}).get("startsAfter") or []
+        for d in sa:
+            if d not in name_set:
+                raise _err("spec.template.cliques",
+                           f"clique {cl.get('name')!r} startsAfter references unknown clique {d!r}")
+            if d == cl.get("name"):
+                raise _err("spec.template.cliques",
+                           f"clique {cl.get('name')!r} cannot start after itself")
+        deps[cl.get("name")] = list(sa)
+    # cycle check (iterative DFS, colors)
+    WHITE, GRAY, BLACK = 0, 1, 2
+    color = {n: WHITE for n in names}
+
+    def visit(start: str) -> None:
+        stack = [(start, iter(deps.get(start, ())))]
+        color[start] = GRAY
+        while stack:
+            node, it = stack[-1]
+            adv = next(it, None)
+            if adv is None:
+                color[node] = BLACK
+                stack.pop()
+            elif color[adv] == GRAY:
+                raise _err("spec.template.cliques", "startsAfter dependencies form a cycle")
+            elif color[adv] == WHITE:
+                color[adv] = GRAY
+                stack.append((adv, iter(deps.get(adv, ()))))
+
+    for n in names:
+        if color[n] == WHITE:
+            visit(n)
+
+
+def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
+    name = pcs.get("metadata", {}).get("name", "")
+    spec = pcs.get("spec") or {}
+    if spec.get("replicas", 0) < 0:
+        raise _err("spec.replicas", "must be >= 0")
+    ustrat = (spec.get("updateStrategy") or {}).get("type")
+    if ustrat not in (None, c.UPDATE_ROLLING_RECREATE, c.UPDATE_ON_DELETE):
+        raise _err("spec.updateStrategy.type", f"unsupported strategy {ustrat!r}")
+    tmpl = spec.get("template") or {}
+    cliques = tmpl.get("cliques") or []
+    if not cliques:
+        raise _err("spec.template.cliques", "at least one PodClique is required")
+
+    startup = tmpl.get("cliqueStartupType", c.STARTUP_ANY_ORDER)
+    if startup not in (c.STARTUP_ANY_ORDER, c.STARTUP_IN_ORDER, c.STARTUP_EXPLICIT):
+        raise _err("spec.template.cliqueStartupType", f"unsupported startup type {startup!r}")
+
+    seen = set()
+    for cl in cliques:
+        cn = cl.get("name")
+        if not cn:
+            raise _err("spec.template.cliques", "clique name is required")
+        if cn in seen:
+            raise _err("spec.template.cliques", f"duplicate clique name {cn!r}")
+        seen.add(cn)
+        cs = cl.get("spec") or {}
+        if not cs.get("roleName"):
+            raise _err(f"spec.template.cliques[{cn}].spec.roleName", "roleName is required")
+        reps = cs.get("replicas", 1)
+        if reps < 0:
+            raise _err(f"spec.template.cliques[{cn}].spec.replicas", "must be >= 0")
+        ma = cs.get("minAvailable")
+        if ma is not None:
+            if ma < 1:
+                raise _err(f"spec.template.cliques[{cn}].spec.minAvailable", "must be >= 1")
+            if ma > reps:
+                raise _err(f"spec.template.cliques[{cn}].spec.minAvailable",
+                           "must not be greater than replicas")
+        podspec = cs.get("podSpec") or {}
+        if not podspec.get("containers"):
+            raise _err(f"spec.template.cliques[{cn}].spec.podSpec.containers",
+                       "at least one container is required")
+        for ctr in podspec.get("containers", []):
+            for ev in ctr.get("env") or []:
+                if str(ev.get("name", "")).startswith("GROVE_"):
+                    raise _err(f"spec.template.cliques[{cn}].spec.podSpec",
+                               f"env var {ev.get('name')!r} uses the reserved GROVE_ prefix")
+        asc = cs.get("autoScalingConfig")
+        if asc is not None:
+            _validate_scale_config(asc, cs.get("minAvailable", reps),
+                                   f"spec.template.cliques[{cn}].spec.autoScalingConfig")
+
+    sg_member_cliques: set = set()
+    sg_names: set = set()
+    for sg in tmpl.get("podCliqueScalingGroups") or []:
+        sgn = sg.get("name")
+        if not sgn:
+            raise _err("spec.template.podCliqueScalingGroups", "scaling group name is required")
+        if sgn in sg_names:
+            raise _err("spec.template.podCliqueScalingGroups", f"duplicate scaling group {sgn!r}")
+        sg_names.add(sgn)
+        members = sg.get("cliqueNames") or []
+        if not members:
+            raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].cliqueNames",
+                       "at least one clique name is required")
+        for mn in members:
+            if mn not in seen:
+                raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].cliqueNames",
+                           f"unknown clique {mn!r}")
+            if mn in sg_member_cliques:
+                raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].cliqueNames",
+                           f"clique {mn!r} belongs to more than one scaling group")
+            sg_member_cliques.add(mn)
+            _check_name_budget(name, sgn, mn)
+        reps = sg.get("replicas", 1)
+        ma = sg.get("minAvailable", 1)
+        if ma < 1:
+            raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].minAvailable", "must be >= 1")
+        if ma > reps:
+            raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].minAvailable",
+                       "must not be greater than replicas")
+        if sg.get("scaleConfig") is not None:
+            _validate_scale_config(sg["scaleConfig"], ma,
+                                   f"spec.template.podCliqueScalingGroups[{sgn}].scaleConfig")
+        # scaling-group members must not have individual autoscaling
+        for cl in cliques:
+            if cl.get("name") in members and (cl.get("spec") or {}).get("autoScalingConfig"):
+                raise _err(f"spec.template.cliques[{cl['name']}].spec.autoScalingConfig",
+                           "cliques in a scaling group cannot define individual autoscaling")
+
+    for cl in cliques:
+        if cl.get("name") not in sg_member_cliques:
+            _check_name_budget(name, "", cl.get("name", ""))
+
+    _check_startup_dag(cliques, startup)
+
+    td = tmpl.get("terminationDelay")
+    if td is not None and parse_duration_seconds(td) <= 0:
+        raise _err("spec.template.terminationDelay", "must be greater than 0")
+
+    if old is not None:
+        _validate_pcs_update(pcs, old)
+
+
+def _validate_scale_config(sc: Obj, min_available: int, path: str) -> None:
+    maxr = sc.get("maxReplicas")
+    if maxr is None:
+        raise _err(f"{path}.maxReplicas", "maxReplicas is required")
+    minr = sc.get("minReplicas")
+    if minr is not None:
+        if minr < 1:
+            raise _err(f"{path}.minReplicas", "must be >= 1")
+        if maxr < minr:
+            raise _err(f"{path}.maxReplicas", "must be >= minReplicas")
+        if minr < min_available:
+            raise _err(f"{path}.minReplicas", "must not be less than minAvailable")
+
+
+def _validate_pcs_update(new: Obj, old: Obj) -> None:
+    """Immutable-field rules (validation/podcliqueset.go:643-702,967-1010)."""
+    nt = (new.get("spec") or {}).get("template") or {}
+    ot = (old.get("spec") or {}).get("template") or {}
+    if nt.get("cliqueStartupType") != ot.get("cliqueStartupType"):
+        raise _err("spec.template.cliqueStartupType", "field is immutable")
+    new_names = [cl.get("name") for cl in nt.get("cliques") or []]
+    old_names = [cl.get("name") for cl in ot.get("cliques") or []]
+    if new_names != old_names:
+        raise _err("spec.template.cliques", "clique names cannot be added, removed or reordered")
+    new_sgs = {sg.get("name"): sg.get("cliqueNames") for sg in nt.get("podCliqueScalingGroups") or []}
+    old_sgs = {sg.get("name"): sg.get("cliqueNames") for sg in ot.get("podCliqueScalingGroups") or []}
+    if new_sgs.keys() != old_sgs.keys():
+        raise _err("spec.template.podCliqueScalingGroups", "scaling groups cannot be added or removed")
+    for k in new_sgs:
+        if new_sgs[k] != old_sgs[k]:
+            raise _err(f"spec.template.podCliqueScalingGroups[{k}].cliqueNames", "field is immutable")
+
+
+def validate_clustertopologybinding(ctb: Obj, old: Optional[Obj] = None) -> None:
+    levels = (ctb.get("spec") or {}).get("levels") or []
+    if not levels:
+        raise _err("spec.levels", "at least one topology level is required")
+    seen_d, seen_k = set(), set()
+    for lv in levels:
+        d, k = lv.get("domain"), lv.get("nodeLabelKey") or lv.get("key")
+        if not d or not k:
+            raise _err("spec.levels", "each level requires domain and nodeLabelKey")
+        if d in seen_d:
+            raise _err("spec.levels", f"duplicate domain {d!r}")
+        if k in seen_k:
+            raise _err("spec.levels", f"duplicate node label key {k!r}")
+        seen_d.add(d)
+        seen_k.add(k)

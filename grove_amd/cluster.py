@@ -1,0 +1,221 @@
+"""Cluster facade: wires store + admission + controllers + scheduler + kubelets.
+
+This is the in-process equivalent of the reference's deployment unit (operator manager +
+webhooks + external scheduler + kubelet/KWOK cluster, cmd/main.go:44): one object that
+tests, the benchmark harness, and the GPU node agent all drive the same way.
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Any, Callable, Dict, List, Optional
+
+import yaml as _yaml
+
+from .api import constants as c
+from .api.defaulting import default_podcliqueset, default_podclique, default_pcsg
+from .api.validation import validate_podcliqueset, validate_clustertopologybinding
+from .controllers.manager import Controller, Manager, Result
+from .controllers.podclique import PodCliqueReconciler
+from .controllers.podcliqueset import PodCliqueSetReconciler
+from .controllers.podcliquescalinggroup import PCSGReconciler
+from .kubecore.store import Store, Obj, ApiError
+from .kubelet.virtual import VirtualKubelet, make_virtual_node
+from .scheduler.plugin import GangScheduler
+from .utils import conditions as cond
+
+log = logging.getLogger("grove.cluster")
+
+
+class Cluster:
+    def __init__(self,
+                 scheduler_name: str = c.SCHEDULER_AMD_GANG,
+                 concurrent_syncs: int = 4,
+                 startup_latency_s: float = 0.0,
+                 ready_latency_s: float = 0.0,
+                 pod_payload: Optional[Callable[[Obj], None]] = None,
+                 use_native_scheduler: Optional[bool] = None):
+        self.store = Store()
+        self.scheduler_name = scheduler_name
+
+        # admission (webhook parity: defaulting before validation)
+        self.store.register_mutator(c.KIND_PCS, default_podcliqueset)
+        self.store.register_validator(c.KIND_PCS, validate_podcliqueset)
+        self.store.register_mutator(c.KIND_PCLQ, default_podclique)
+        self.store.register_mutator(c.KIND_PCSG, default_pcsg)
+        self.store.register_validator(c.KIND_CTB, validate_clustertopologybinding)
+
+        self.manager = Manager(self.store)
+        self.pcs_rec = PodCliqueSetReconciler(self.store, scheduler_name)
+        self.pclq_rec = PodCliqueReconciler(self.store, scheduler_name)
+        self.pcsg_rec = PCSGReconciler(self.store, scheduler_name)
+        self.scheduler = GangScheduler(self.store, scheduler_name,
+                                       use_native=use_native_scheduler)
+        self.kubelet = VirtualKubelet(self.store,
+                                      startup_latency_s=startup_latency_s,
+                                      ready_latency_s=ready_latency_s,
+                                      payload=pod_payload)
+
+        m = self.manager
+        self.c_pcs = m.add_controller(Controller(
+            "podcliqueset", self.pcs_rec.reconcile, workers=concurrent_syncs))
+        self.c_pclq = m.add_controller(Controller(
+            "podclique", self.pclq_rec.reconcile, workers=concurrent_syncs))
+        self.c_pcsg = m.add_controller(Controller(
+            "podcliquescalinggroup", self.pcsg_rec.reconcile, workers=concurrent_syncs))
+        self.c_sched = m.add_controller(Controller(
+            "gang-scheduler", lambda ns, n: self.scheduler.reconcile(ns, n) or Result.DONE,
+            workers=1))
+        self.c_kubelet = m.add_controller(Controller(
+            "kubelet", self.kubelet.reconcile, workers=max(2, concurrent_syncs)))
+
+        self._wire_watches()
+        self._started = False
+
+    # ------------------------------------------------------------------ watches
+    def _wire_watches(self) -> None:
+        m = self.manager
+
+        def on_pcs(ev: str, obj: Obj, _old) -> None:
+            md = obj["metadata"]
+            self.c_pcs.enqueue(md.get("namespace", "default"), md["name"])
+
+        def on_pclq(ev: str, obj: Obj, _old) -> None:
+            md = obj["metadata"]
+            ns = md.get("namespace", "default")
+            self.c_pclq.enqueue(ns, md["name"])
+            pcs = md.get("labels", {}).get(c.LABEL_PART_OF)
+            if pcs:
+                self.c_pcs.enqueue(ns, pcs)
+            pcsg = md.get("labels", {}).get(c.LABEL_PCSG)
+            if pcsg:
+                self.c_pcsg.enqueue(ns, pcsg)
+
+        def on_pcsg(ev: str, obj: Obj, _old) -> None:
+            md = obj["metadata"]
+            ns = md.get("namespace", "default")
+            self.c_pcsg.enqueue(ns, md["name"])
+            pcs = md.get("labels", {}).get(c.LABEL_PART_OF)
+            if pcs:
+                self.c_pcs.enqueue(ns, pcs)
+
+        def on_pod(ev: str, obj: Obj, _old) -> None:
+            md = obj["metadata"]
+            ns = md.get("namespace", "default")
+            pclq = md.get("labels", {}).get(c.LABEL_PODCLIQUE)
+            if pclq:
+                self.c_pclq.enqueue(ns, pclq)
+            self.c_sched.enqueue("", "pass")
+            if obj.get("spec", {}).get("nodeName"):
+                self.c_kubelet.enqueue(ns, md["name"])
+
+        def on_podgang(ev: str, obj: Obj, _old) -> None:
+            md = obj["metadata"]
+            ns = md.get("namespace", "default")
+            self.c_sched.enqueue("", "pass")
+            # gate-removal re-check for every member clique
+            for group in (obj.get("spec") or {}).get("podGroups") or []:
+                self.c_pclq.enqueue(ns, group.get("name", ""))
+            base = md.get("labels", {}).get(c.LABEL_BASE_PODGANG)
+            if base:
+                # scaled gangs unblock when base gets scheduled
+                pass
+
+        def on_node(ev: str, obj: Obj, _old) -> None:
+            self.c_sched.enqueue("", "pass")
+
+        m.watch(c.KIND_PCS, on_pcs)
+        m.watch(c.KIND_PCLQ, on_pclq)
+        m.watch(c.KIND_PCSG, on_pcsg)
+        m.watch("Pod", on_pod)
+        m.watch(c.KIND_PODGANG, on_podgang)
+        m.watch("Node", on_node)
+
+    # ------------------------------------------------------------------ lifecycle
+    def start(self) -> "Cluster":
+        if not self._started:
+            self.manager.start()
+            self._started = True
+        return self
+
+    def stop(self) -> None:
+        if self._started:
+            self.manager.stop()
+            self._started = False
+
+    def __enter__(self) -> "Cluster":
+        return self.start()
+
+    def __exit__(self, *exc) -> None:
+        self.stop()
+
+    # ------------------------------------------------------------------ node helpers
+    def add_virtual_nodes(self, count: int, gpus: int = 0, prefix: str = "node",
+                          **kw) -> List[str]:
+        names = []
+        for i in range(count):
+            name = f"{prefix}-{i}"
+            self.store.create(make_virtual_node(name, gpus=gpus, **kw))
+            names.append(name)
+        return names
+
+    # ------------------------------------------------------------------ apply / wait
+    def apply(self, manifest) -> List[Obj]:
+        """Apply YAML text / dict / list of dicts. Create-or-update semantics."""
+        if isinstance(manifest, str):
+            docs = [d for d in _yaml.safe_load_all(manifest) if d]
+        elif isinstance(manifest, dict):
+            docs = [manifest]
+        else:
+            docs = list(manifest)
+        out = []
+        for doc in docs:
+            kind = doc.get("kind")
+            md = doc.get("metadata", {})
+            cur = self.store.try_get(kind, md.get("namespace"), md.get("name", ""))
+            if cur is None:
+                out.append(self.store.create(doc))
+            else:
+                doc = dict(doc)
+                doc.setdefault("metadata", {})["resourceVersion"] = \
+                    cur["metadata"]["resourceVersion"]
+                out.append(self.store.update(doc))
+        return out
+
+    def delete_pcs(self, name: str, namespace: str = "default") -> None:
+        self.store.delete(c.KIND_PCS, namespace, name)
+        self.c_pcs.enqueue(namespace, name)
+
+    def wait_for(self, predicate: Callable[[], bool], timeout: float = 30.0,
+                 poll: float = 0.01, desc: str = "condition") -> None:
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if predicate():
+                return
+            time.sleep(poll)
+        raise TimeoutError(f"timed out waiting for {desc}")
+
+    def wait_pcs_available(self, name: str, namespace: str = "default",
+                           timeout: float = 30.0, min_available: Optional[int] = None)\
+            -> Obj:
+        def ok() -> bool:
+            pcs = self.store.try_get(c.KIND_PCS, namespace, name)
+            if pcs is None:
+                return False
+            want = min_available if min_available is not None \
+                else int(pcs["spec"].get("replicas", 0))
+            return int((pcs.get("status") or {}).get("availableReplicas", 0)) >= want
+        self.wait_for(ok, timeout, desc=f"PCS {name} available")
+        return self.store.get(c.KIND_PCS, namespace, name)
+
+    def wait_pods_ready(self, selector: Dict[str, str], count: int,
+                        namespace: str = "default", timeout: float = 30.0) -> None:
+        def ok() -> bool:
+            pods = self.store.list("Pod", namespace, selector)
+            return sum(1 for p in pods if cond.pod_is_ready(p)) >= count
+        self.wait_for(ok, timeout, desc=f"{count} ready pods for {selector}")
+
+    def wait_deleted(self, kind: str, name: str, namespace: str = "default",
+                     timeout: float = 30.0) -> None:
+        self.wait_for(lambda: self.store.try_get(kind, namespace, name) is None,
+                      timeout, desc=f"{kind} {name} deleted")

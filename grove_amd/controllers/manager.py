@@ -1,0 +1,238 @@
+"""Controller manager: informer-driven workqueues with per-controller concurrency.
+
+Parity role: controller-runtime manager + workqueue machinery the reference builds in
+operator/internal/controller/manager.go:55 and register.go:34. Fresh design: each
+Controller owns a rate-limited dedup workqueue fed by store watches through mapping
+functions; N worker threads call reconcile(key) with requeue/backoff semantics.
+"""
+from __future__ import annotations
+
+import heapq
+import logging
+import threading
+import time
+import traceback
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+from ..kubecore.store import Store, Obj
+
+log = logging.getLogger("grove")
+
+Key = Tuple[str, str]  # (namespace, name)
+
+
+class Result:
+    """Reconcile outcome (mirrors the reference's ReconcileStepResult vocabulary,
+    controller/common/flow.go:29-113)."""
+
+    __slots__ = ("requeue_after", "requeue")
+
+    def __init__(self, requeue: bool = False, requeue_after: Optional[float] = None):
+        self.requeue = requeue
+        self.requeue_after = requeue_after
+
+    DONE: "Result"
+
+
+Result.DONE = Result()
+
+
+class WorkQueue:
+    """Dedup + delayed workqueue with exponential per-item backoff."""
+
+    def __init__(self, base_delay: float = 0.005, max_delay: float = 30.0):
+        self._cond = threading.Condition()
+        self._queue: List[Key] = []
+        self._queued: set = set()
+        self._processing: set = set()
+        self._dirty: set = set()
+        self._delayed: List[Tuple[float, int, Key]] = []
+        self._seq = 0
+        self._failures: Dict[Key, int] = {}
+        self._base_delay = base_delay
+        self._max_delay = max_delay
+        self._shutdown = False
+
+    def add(self, key: Key) -> None:
+        with self._cond:
+            if key in self._processing:
+                self._dirty.add(key)
+                return
+            if key not in self._queued:
+                self._queued.add(key)
+                self._queue.append(key)
+                self._cond.notify()
+
+    def add_after(self, key: Key, delay: float) -> None:
+        if delay <= 0:
+            self.add(key)
+            return
+        with self._cond:
+            self._seq += 1
+            heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, key))
+            self._cond.notify()
+
+    def add_rate_limited(self, key: Key) -> None:
+        with self._cond:
+            n = self._failures.get(key, 0)
+            self._failures[key] = n + 1
+        self.add_after(key, min(self._base_delay * (2 ** n), self._max_delay))
+
+    def forget(self, key: Key) -> None:
+        with self._cond:
+            self._failures.pop(key, None)
+
+    def get(self, timeout: float = 0.2) -> Optional[Key]:
+        deadline = time.monotonic() + timeout
+        with self._cond:
+            while True:
+                now = time.monotonic()
+                while self._delayed and self._delayed[0][0] <= now:
+                    _, _, k = heapq.heappop(self._delayed)
+                    if k not in self._queued and k not in self._processing:
+                        self._queued.add(k)
+                        self._queue.append(k)
+                    elif k in self._processing:
+                        self._dirty.add(k)
+                if self._queue:
+                    k = self._queue.pop(0)
+                    self._queued.discard(k)
+                    self._processing.add(k)
+                    return k
+                if self._shutdown:
+                    return None
+                wait = deadline - now
+                if self._delayed:
+                    wait = min(wait, self._delayed[0][0] - now)
+                if wait <= 0:
+                    return None
+                self._cond.wait(wait)
+
+    def done(self, key: Key) -> None:
+        with self._cond:
+            self._processing.discard(key)
+            if key in self._dirty:
+                self._dirty.discard(key)
+                if key not in self._queued:
+                    self._queued.add(key)
+                    self._queue.append(key)
+                    self._cond.notify()
+
+    def shut_down(self) -> None:
+        with self._cond:
+            self._shutdown = True
+            self._cond.notify_all()
+
+    def __len__(self) -> int:
+        with self._cond:
+            return len(self._queue) + len(self._delayed)
+
+
+class Controller:
+    def __init__(self, name: str, reconcile: Callable[[str, str], Result], workers: int = 2):
+        self.name = name
+        self.reconcile = reconcile
+        self.workers = workers
+        self.queue = WorkQueue()
+        self._threads: List[threading.Thread] = []
+        self._stop = threading.Event()
+
+    def enqueue(self, namespace: str, name: str) -> None:
+        self.queue.add((namespace, name))
+
+    def enqueue_after(self, namespace: str, name: str, delay: float) -> None:
+        self.queue.add_after((namespace, name), delay)
+
+    def start(self) -> None:
+        for i in range(self.workers):
+            t = threading.Thread(target=self._worker, name=f"{self.name}-{i}", daemon=True)
+            t.start()
+            self._threads.append(t)
+
+    def _worker(self) -> None:
+        while not self._stop.is_set():
+            item = self.queue.get()
+            if item is None:
+                if self._stop.is_set():
+                    return
+                continue
+            ns, name = item
+            try:
+                res = self.reconcile(ns, name)
+                self.queue.forget(item)
+                if res is not None and res.requeue_after is not None:
+                    self.queue.add_after(item, res.requeue_after)
+                elif res is not None and res.requeue:
+                    self.queue.add(item)
+            except Exception:
+                log.debug("reconcile %s %s/%s failed:\n%s", self.name, ns, name,
+                          traceback.format_exc())
+                self.queue.add_rate_limited(item)
+            finally:
+                self.queue.done(item)
+
+    def stop(self) -> None:
+        self._stop.set()
+        self.queue.shut_down()
+
+
+class Manager:
+    """Owns the store watches and dispatches events to controllers via map functions."""
+
+    def __init__(self, store: Store):
+        self.store = store
+        self.controllers: List[Controller] = []
+        self._watch_threads: List[threading.Thread] = []
+        self._watches: List[Any] = []
+        self._stop = threading.Event()
+
+    def add_controller(self, ctrl: Controller) -> Controller:
+        self.controllers.append(ctrl)
+        return ctrl
+
+    def watch(self, kind: str, handler: Callable[[str, Obj, Optional[Obj]], None],
+              seed: bool = True) -> None:
+        """handler(event_type, obj, old_obj) — old_obj currently unused (None)."""
+        w = self.store.watch(kind, seed=seed)
+        self._watches.append(w)
+
+        def run() -> None:
+            while not self._stop.is_set():
+                try:
+                    import queue as _q
+                    ev, obj = w.queue.get(timeout=0.2)
+                except Exception:
+                    continue
+                try:
+                    handler(ev, obj, None)
+                except Exception:
+                    log.debug("watch handler for %s failed:\n%s", kind, traceback.format_exc())
+
+        t = threading.Thread(target=run, name=f"watch-{kind}-{len(self._watch_threads)}",
+                             daemon=True)
+        self._watch_threads.append(t)
+
+    def start(self) -> None:
+        for t in self._watch_threads:
+            t.start()
+        for ctrl in self.controllers:
+            ctrl.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        for w in self._watches:
+            w.stop()
+        for ctrl in self.controllers:
+            ctrl.stop()
+
+    def wait_idle(self, timeout: float = 30.0, settle: float = 0.05) -> bool:
+        """Best-effort: wait until all queues are empty and stay empty for `settle`."""
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if all(len(c.queue) == 0 for c in self.controllers):
+                time.sleep(settle)
+                if all(len(c.queue) == 0 for c in self.controllers):
+                    return True
+            else:
+                time.sleep(0.01)
+        return False

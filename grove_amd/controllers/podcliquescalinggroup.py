@@ -1,0 +1,248 @@
+"""PodCliqueScalingGroup controller.
+
+Behavior parity: operator/internal/controller/podcliquescalinggroup/ — member PodClique
+creation per replica index (components/podclique/sync.go:55-199), scale-in deletion,
+PCSG-replica-scoped gang recycle (sync.go:127), availability status + MinAvailableBreached /
+GangTerminationInProgress conditions (reconcilestatus.go), PCSG-scoped rolling update
+(components/rollingupdate.go). Fresh implementation.
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Any, Dict, List, Optional
+
+from ..api import constants as c
+from ..api import namegen
+from ..api.defaulting import parse_duration_seconds
+from ..kubecore.store import Store, Obj, ApiError
+from ..utils import conditions as cond
+from ..utils.hashing import pod_template_hash
+from . import builders
+from .manager import Result
+from .podcliqueset import _iso_to_epoch
+
+log = logging.getLogger("grove.pcsg")
+
+
+class PCSGReconciler:
+    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG):
+        self.store = store
+        self.scheduler_name = scheduler_name
+
+    def reconcile(self, namespace: str, name: str) -> Result:
+        pcsg = self.store.try_get(c.KIND_PCSG, namespace, name)
+        if pcsg is None:
+            return Result.DONE
+        if pcsg["metadata"].get("deletionTimestamp"):
+            return self._reconcile_delete(pcsg)
+        pcs = self._find_pcs(pcsg)
+        if pcs is None:
+            return Result(requeue_after=0.1)
+        res = self._sync_member_pclqs(pcs, pcsg)
+        recycle_wait = self._replica_recycle(pcs, pcsg)
+        self._reconcile_status(namespace, name)
+        if recycle_wait is not None:
+            return Result(requeue_after=recycle_wait)
+        return res
+
+    # ------------------------------------------------------------------ helpers
+    def _find_pcs(self, pcsg: Obj) -> Optional[Obj]:
+        pcs_name = pcsg["metadata"]["labels"].get(c.LABEL_PART_OF)
+        if not pcs_name:
+            return None
+        return self.store.try_get(c.KIND_PCS, pcsg["metadata"].get("namespace"), pcs_name)
+
+    def _member_pclqs(self, pcsg: Obj) -> List[Obj]:
+        return self.store.list(c.KIND_PCLQ, pcsg["metadata"].get("namespace"),
+                               {c.LABEL_PCSG: pcsg["metadata"]["name"]})
+
+    @staticmethod
+    def _sg_config(pcs: Obj, pcsg: Obj) -> Optional[Obj]:
+        pcs_replica = int(pcsg["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX, 0))
+        sg_name = namegen.extract_scaling_group_name(
+            pcsg["metadata"]["name"], pcs["metadata"]["name"], pcs_replica)
+        for sg in pcs["spec"]["template"].get("podCliqueScalingGroups") or []:
+            if sg["name"] == sg_name:
+                return sg
+        return None
+
+    # ------------------------------------------------------------------ delete
+    def _reconcile_delete(self, pcsg: Obj) -> Result:
+        ns, name = pcsg["metadata"].get("namespace"), pcsg["metadata"]["name"]
+        remaining = 0
+        for q in self._member_pclqs(pcsg):
+            remaining += 1
+            try:
+                self.store.delete(c.KIND_PCLQ, ns, q["metadata"]["name"])
+            except ApiError:
+                pass
+        if remaining:
+            return Result(requeue_after=0.02)
+
+        def rm(o: Obj) -> None:
+            o["metadata"]["finalizers"] = [
+                f for f in o["metadata"].get("finalizers", []) if f != c.FINALIZER_PCSG]
+        try:
+            self.store.patch(c.KIND_PCSG, ns, name, rm)
+        except ApiError:
+            pass
+        return Result.DONE
+
+    # ------------------------------------------------------------------ spec
+    def _sync_member_pclqs(self, pcs: Obj, pcsg: Obj) -> Result:
+        ns = pcsg["metadata"].get("namespace", "default")
+        sg_fqn = pcsg["metadata"]["name"]
+        pcs_replica = int(pcsg["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX, 0))
+        replicas = int(pcsg["spec"].get("replicas", 1))
+        min_avail = int(pcsg["spec"].get("minAvailable", 1))
+        member_names = pcsg["spec"].get("cliqueNames") or []
+        cliques = {cl["name"]: cl for cl in pcs["spec"]["template"].get("cliques") or []}
+        base_pg = namegen.base_podgang_name(pcs["metadata"]["name"], pcs_replica)
+
+        existing = {q["metadata"]["name"]: q for q in self._member_pclqs(pcsg)}
+        expected: set = set()
+        for j in range(replicas):
+            pg_name = namegen.podgang_name_for_pclq_in_pcsg(
+                pcs["metadata"]["name"], pcs_replica, sg_fqn, min_avail, j)
+            for mn in member_names:
+                cl = cliques.get(mn)
+                if cl is None:
+                    continue
+                fqn = namegen.podclique_name(sg_fqn, j, mn)
+                expected.add(fqn)
+                cur = existing.get(fqn)
+                if cur is None:
+                    obj = builders.build_podclique(
+                        pcs, pcs_replica, cl, owner=pcsg,
+                        pcsg_name=sg_fqn, pcsg_replica=j,
+                        podgang_name=pg_name,
+                        base_podgang_name=base_pg if j >= min_avail else None)
+                    obj["spec"]["updateStrategy"] = (
+                        pcs["spec"].get("updateStrategy") or {}).get(
+                        "type", c.UPDATE_ROLLING_RECREATE)
+                    try:
+                        self.store.create(obj)
+                    except ApiError:
+                        pass
+                    continue
+                if cur["metadata"].get("deletionTimestamp"):
+                    continue
+                new_hash = pod_template_hash(
+                    mn, cl["spec"].get("podSpec", {}),
+                    pcs["spec"]["template"].get("priorityClassName", ""))
+                if cur["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) != new_hash \
+                        and self._replica_selected_for_update(pcs, pcs_replica):
+                    def upd(o: Obj) -> None:
+                        o["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] = new_hash
+                        o["spec"]["podSpec"] = cl["spec"].get("podSpec", {})
+                    try:
+                        self.store.patch(c.KIND_PCLQ, ns, fqn, upd)
+                    except ApiError:
+                        pass
+        # scale-in: delete member PCLQs beyond current replicas
+        for fqn, q in existing.items():
+            if fqn not in expected:
+                try:
+                    self.store.delete(c.KIND_PCLQ, ns, fqn)
+                except ApiError:
+                    pass
+        return Result.DONE
+
+    @staticmethod
+    def _replica_selected_for_update(pcs: Obj, r: int) -> bool:
+        prog = (pcs.get("status") or {}).get("rollingUpdateProgress")
+        if prog is None:
+            return True
+        cu = prog.get("currentlyUpdating")
+        return cu is not None and int(cu.get("replicaIndex", -1)) == r
+
+    # ------------------------------------------------------------------ replica recycle
+    def _replica_recycle(self, pcs: Obj, pcsg: Obj) -> Optional[float]:
+        """PCSG-replica-scoped gang recycle (sync.go:127): a scaled replica whose member
+        cliques breached MinAvailable past terminationDelay is deleted and recreated.
+        Fresh cliques have everScheduled=False so the loop cannot re-fire (WasPCLQ-
+        EverScheduled parity)."""
+        ns = pcsg["metadata"].get("namespace", "default")
+        delay = parse_duration_seconds(
+            pcs["spec"]["template"].get("terminationDelay", "4h"))
+        now = time.time()
+        next_wait: Optional[float] = None
+        by_replica: Dict[int, List[Obj]] = {}
+        for q in self._member_pclqs(pcsg):
+            js = q["metadata"]["labels"].get(c.LABEL_PCSG_REPLICA_INDEX, "")
+            if js.isdigit():
+                by_replica.setdefault(int(js), []).append(q)
+        for j, qs in by_replica.items():
+            breach_since: Optional[float] = None
+            for q in qs:
+                if not (q.get("status") or {}).get("everScheduled"):
+                    breach_since = None
+                    break
+                bc = cond.get_condition(q, c.COND_MIN_AVAILABLE_BREACHED)
+                if bc and bc.get("status") == "True":
+                    ts = _iso_to_epoch(bc.get("lastTransitionTime", ""))
+                    breach_since = ts if breach_since is None else min(breach_since, ts)
+            if breach_since is None:
+                continue
+            remaining = delay - (now - breach_since)
+            if remaining > 0:
+                next_wait = remaining if next_wait is None else min(next_wait, remaining)
+                continue
+            log.info("PCSG %s/%s recycling replica %d", ns, pcsg["metadata"]["name"], j)
+            for q in qs:
+                try:
+                    self.store.delete(c.KIND_PCLQ, ns, q["metadata"]["name"])
+                except ApiError:
+                    pass
+        return next_wait
+
+    # ------------------------------------------------------------------ status
+    def _reconcile_status(self, namespace: str, name: str) -> None:
+        pcsg = self.store.try_get(c.KIND_PCSG, namespace, name)
+        if pcsg is None or pcsg["metadata"].get("deletionTimestamp"):
+            return
+        replicas = int(pcsg["spec"].get("replicas", 1))
+        min_avail = int(pcsg["spec"].get("minAvailable", 1))
+        members = pcsg["spec"].get("cliqueNames") or []
+        by_replica: Dict[int, List[Obj]] = {}
+        for q in self._member_pclqs(pcsg):
+            js = q["metadata"]["labels"].get(c.LABEL_PCSG_REPLICA_INDEX, "")
+            if js.isdigit():
+                by_replica.setdefault(int(js), []).append(q)
+
+        sched = avail = 0
+        for j in range(replicas):
+            qs = by_replica.get(j, [])
+            if len(qs) < len(members):
+                continue
+            if all(int((q.get("status") or {}).get("scheduledReplicas", 0))
+                   >= int(q["spec"].get("minAvailable", 1)) for q in qs):
+                sched += 1
+            if all(int((q.get("status") or {}).get("readyReplicas", 0))
+                   >= int(q["spec"].get("minAvailable", 1)) for q in qs):
+                avail += 1
+
+        def upd(o: Obj) -> None:
+            st = o.setdefault("status", {})
+            st["replicas"] = replicas
+            st["scheduledReplicas"] = sched
+            st["availableReplicas"] = avail
+            st["observedGeneration"] = o["metadata"].get("generation")
+            ever = bool(st.get("everAvailable")) or avail >= min_avail
+            st["everAvailable"] = ever
+            if ever and avail < min_avail:
+                cond.set_condition(o, c.COND_MIN_AVAILABLE_BREACHED, True,
+                                   c.REASON_INSUFFICIENT_AVAILABLE_PCSG_REPLICAS)
+            else:
+                changed = cond.set_condition(o, c.COND_MIN_AVAILABLE_BREACHED, False,
+                                             c.REASON_SUFFICIENT_AVAILABLE_PCSG_REPLICAS)
+                # recovery clears the PCS-level gang-termination latch
+                if cond.condition_true(o, c.COND_GANG_TERMINATION_IN_PROGRESS) \
+                        and avail >= min_avail:
+                    cond.set_condition(o, c.COND_GANG_TERMINATION_IN_PROGRESS, False,
+                                       c.REASON_GANG_TERMINATION_ACTIVE)
+        try:
+            self.store.patch(c.KIND_PCSG, namespace, name, upd, status=True)
+        except ApiError:
+            pass

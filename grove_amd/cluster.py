@@ -90,6 +90,13 @@ class Cluster:
             pcsg = md.get("labels", {}).get(c.LABEL_PCSG)
             if pcsg:
                 self.c_pcsg.enqueue(ns, pcsg)
+            # A base-gang member's status change (scheduledReplicas) can unblock gate
+            # removal for every scaled-gang PCLQ anchored to that base gang
+            # (pod/syncflow.go:386-424 watch-mapping parity).
+            pg = md.get("labels", {}).get(c.LABEL_PODGANG)
+            if pg and not md.get("labels", {}).get(c.LABEL_BASE_PODGANG):
+                for q in self.store.list(c.KIND_PCLQ, ns, {c.LABEL_BASE_PODGANG: pg}):
+                    self.c_pclq.enqueue(ns, q["metadata"]["name"])
 
         def on_pcsg(ev: str, obj: Obj, _old) -> None:
             md = obj["metadata"]

@@ -197,6 +197,8 @@ def build_pod(pcs: Obj, pclq: Obj, pod_index: int, scheduler_name: str,
             "generateName": f"{pclq_name}-",
             "namespace": namespace,
             "labels": labels,
+            # clique-template annotations ride on the PCLQ and flow to its pods
+            "annotations": dict(pclq["metadata"].get("annotations") or {}),
             "ownerReferences": [owner_reference(pclq)],
         },
         "spec": pod_spec,

@@ -243,6 +243,8 @@ class Store:
             if m.get("resourceVersion") and m["resourceVersion"] != cur_m["resourceVersion"]:
                 raise conflict(kind, m["name"])
             if status_only:
+                if cur.get("status", {}) == obj.get("status", {}):
+                    return copy.deepcopy(cur)  # no-op: no rv bump, no event
                 new = copy.deepcopy(cur)
                 new["status"] = copy.deepcopy(obj.get("status", {}))
             else:
@@ -262,6 +264,9 @@ class Store:
                     new["status"] = copy.deepcopy(cur.get("status", {}))
                 if cur_m.get("deletionTimestamp"):
                     new["metadata"]["deletionTimestamp"] = cur_m["deletionTimestamp"]
+                new["metadata"]["resourceVersion"] = cur_m["resourceVersion"]
+                if new == cur:
+                    return copy.deepcopy(cur)  # no-op update: no rv bump, no event
                 if new.get("spec") != cur.get("spec"):
                     new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
             new["metadata"]["resourceVersion"] = self._next_rv()

@@ -1,0 +1,160 @@
+// _sched — C++ gang-placement core (pybind11, CPU).
+//
+// Native implementation of grove_amd/scheduler/placement.py (same algorithm, same
+// results — cross-checked by tests/test_placement_native.py). This is the hot path of
+// the control plane at scale: one call per pending PodGang per scheduling pass, over
+// potentially hundreds of nodes. Filter = capacity (cpu/mem/gpu/pods); Score = xGMI
+// packing (one 8×MI355X hive ≈153 GB/s per-GPU ring bandwidth; cross-node placements
+// are NIC-bound and score low).
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <optional>
+#include <string>
+#include <tuple>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+constexpr double kXgmiLinkGBps = 153.0;
+constexpr double kXgmiPeerLinks = 7.0;
+constexpr double kNicGBps = 50.0;
+
+struct Node {
+  std::string name;
+  int64_t cpu_milli;
+  double mem_bytes;
+  std::vector<int> gpu_ids;
+  int64_t pods;
+};
+
+struct Pod {
+  std::string name;
+  int64_t cpu_milli;
+  double mem_bytes;
+  int gpus;
+};
+
+bool fits(const Node& n, const Pod& p) {
+  return n.cpu_milli >= p.cpu_milli && n.mem_bytes >= p.mem_bytes &&
+         (int)n.gpu_ids.size() >= p.gpus && n.pods >= 1;
+}
+
+std::vector<int> take(Node& n, const Pod& p) {
+  n.cpu_milli -= p.cpu_milli;
+  n.mem_bytes -= p.mem_bytes;
+  n.pods -= 1;
+  std::vector<int> out(n.gpu_ids.begin(), n.gpu_ids.begin() + p.gpus);
+  n.gpu_ids.erase(n.gpu_ids.begin(), n.gpu_ids.begin() + p.gpus);
+  return out;
+}
+
+double placement_score(size_t n_nodes_used, int total_gpus) {
+  if (total_gpus <= 1) return kXgmiLinkGBps * kXgmiPeerLinks;
+  if (n_nodes_used <= 1) return kXgmiLinkGBps;
+  return kNicGBps / (2.0 * (double)(n_nodes_used - 1));
+}
+
+using Assignment = std::tuple<std::string, std::string, std::vector<int>>;
+using NodeState = std::tuple<std::string, int64_t, double, std::vector<int>, int64_t>;
+
+// Returns (assignments, score, consumed-node-states) or None.
+py::object place_gang(std::vector<NodeState> node_states,
+                      std::vector<std::tuple<std::string, int64_t, double, int>> pod_specs) {
+  std::vector<Node> nodes;
+  nodes.reserve(node_states.size());
+  for (auto& t : node_states)
+    nodes.push_back(Node{std::get<0>(t), std::get<1>(t), std::get<2>(t),
+                         std::get<3>(t), std::get<4>(t)});
+  std::vector<Pod> pods;
+  pods.reserve(pod_specs.size());
+  int total_gpus = 0;
+  for (auto& t : pod_specs) {
+    pods.push_back(Pod{std::get<0>(t), std::get<1>(t), std::get<2>(t), std::get<3>(t)});
+    total_gpus += std::get<3>(t);
+  }
+  // sort pods by descending gpu then cpu demand (stable for determinism)
+  std::vector<size_t> order(pods.size());
+  for (size_t i = 0; i < order.size(); ++i) order[i] = i;
+  std::stable_sort(order.begin(), order.end(), [&](size_t a, size_t b) {
+    if (pods[a].gpus != pods[b].gpus) return pods[a].gpus > pods[b].gpus;
+    return pods[a].cpu_milli > pods[b].cpu_milli;
+  });
+
+  auto emit = [&](std::vector<Node>& state, std::vector<Assignment>& assignments,
+                  double score) -> py::object {
+    std::vector<NodeState> consumed;
+    consumed.reserve(state.size());
+    for (auto& n : state)
+      consumed.emplace_back(n.name, n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods);
+    return py::make_tuple(assignments, score, consumed);
+  };
+
+  // Phase 1: single-node best-fit (leaves fewest free GPUs behind).
+  int best = -1;
+  size_t best_left = SIZE_MAX;
+  for (size_t i = 0; i < nodes.size(); ++i) {
+    Node trial = nodes[i];
+    bool ok = true;
+    for (size_t oi : order) {
+      if (!fits(trial, pods[oi])) { ok = false; break; }
+      take(trial, pods[oi]);
+    }
+    if (ok && trial.gpu_ids.size() < best_left) {
+      best = (int)i;
+      best_left = trial.gpu_ids.size();
+    }
+  }
+  if (best >= 0) {
+    std::vector<Assignment> assignments;
+    for (size_t oi : order) {
+      auto ids = take(nodes[best], pods[oi]);
+      assignments.emplace_back(pods[oi].name, nodes[best].name, ids);
+    }
+    return emit(nodes, assignments, placement_score(1, total_gpus));
+  }
+
+  // Phase 2: minimal spread, first-fit-decreasing preferring already-used nodes with
+  // the most free GPUs.
+  std::vector<Node> state = nodes;  // work on a copy; rollback = return None
+  std::vector<Assignment> assignments;
+  std::vector<char> used(state.size(), 0);
+  size_t used_count = 0;
+  for (size_t oi : order) {
+    int pick = -1;
+    for (int pass = 0; pass < 2 && pick < 0; ++pass) {
+      // pass 0: only already-used nodes; pass 1: any node (largest free GPUs first)
+      int best_i = -1;
+      long best_key = -1;
+      for (size_t i = 0; i < state.size(); ++i) {
+        if (pass == 0 && !used[i]) continue;
+        if (!fits(state[i], pods[oi])) continue;
+        long key = (long)state[i].gpu_ids.size() * 1000000 + state[i].cpu_milli / 1000;
+        if (key > best_key) { best_key = key; best_i = (int)i; }
+      }
+      pick = best_i;
+    }
+    if (pick < 0) return py::none();
+    auto ids = take(state[pick], pods[oi]);
+    assignments.emplace_back(pods[oi].name, state[pick].name, ids);
+    if (!used[pick]) { used[pick] = 1; ++used_count; }
+  }
+  return emit(state, assignments, placement_score(used_count, total_gpus));
+}
+
+double score_only(size_t n_nodes, int total_gpus) {
+  return placement_score(n_nodes, total_gpus);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_sched, m) {
+  m.doc() = "grove_amd native gang-placement core (xGMI-aware Filter/Score)";
+  m.def("place_gang", &place_gang, py::arg("nodes"), py::arg("pods"),
+        "All-or-nothing gang placement; returns (assignments, score, consumed) or None");
+  m.def("placement_score", &score_only, py::arg("n_nodes_used"), py::arg("total_gpus"));
+}

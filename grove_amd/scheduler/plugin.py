@@ -71,6 +71,8 @@ class GangScheduler:
         for p in pods:
             if p["metadata"].get("deletionTimestamp"):
                 continue
+            if (p.get("status") or {}).get("phase") in ("Succeeded", "Failed"):
+                continue  # terminal pods release their resources
             if p.get("spec", {}).get("nodeName"):
                 bound.append(p)
                 continue

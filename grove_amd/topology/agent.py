@@ -1,0 +1,109 @@
+"""xGMI topology agent — the DaemonSet analog that labels nodes with fabric facts.
+
+Replaces the reference's NVIDIA node-topology path (KAI Topology CR built from node
+labels, MNNVL ComputeDomains — SURVEY.md §2.6): discovers the local GPU set and xGMI
+link graph via the native _topo extension (rocm_smi + KFD sysfs) with a HIP/torch
+fallback, and produces the Node object (labels + allocatable amd.com/gpu) the gang
+scheduler's Filter/Score consumes.
+"""
+from __future__ import annotations
+
+import logging
+import os
+from typing import Any, Dict, List, Optional
+
+from ..api import constants as c
+
+log = logging.getLogger("grove.topology")
+
+
+def probe() -> Optional[Dict[str, Any]]:
+    try:
+        from . import _topo
+    except ImportError:
+        _topo = None
+    info = _topo.probe() if _topo is not None else None
+    if info is not None:
+        return info
+    # HIP/torch fallback (no rocm_smi, no KFD — e.g. partial containers)
+    try:
+        import torch
+        if torch.cuda.is_available():
+            n = torch.cuda.device_count()
+            links = []
+            for i in range(n):
+                for j in range(n):
+                    if i != j and torch.cuda.can_device_access_peer(i, j):
+                        links.append({"src": i, "dst": j, "type": "xgmi"})
+            return {"backend": "torch", "gpu_count": n, "devices": [
+                {"index": i, "name": torch.cuda.get_device_name(i),
+                 "vram_bytes": torch.cuda.get_device_properties(i).total_memory}
+                for i in range(n)], "links": links}
+    except Exception:
+        pass
+    return None
+
+
+def xgmi_hives(info: Dict[str, Any]) -> List[List[int]]:
+    """Connected components of the xGMI link graph over GPU indices."""
+    n = int(info.get("gpu_count", 0))
+    adj: Dict[int, set] = {i: set() for i in range(n)}
+    for l in info.get("links") or []:
+        if l.get("type") == "xgmi" and "src" in l and "dst" in l:
+            s, d = int(l["src"]), int(l["dst"])
+            if s in adj and d in adj:
+                adj[s].add(d)
+                adj[d].add(s)
+    seen: set = set()
+    hives = []
+    for i in range(n):
+        if i in seen:
+            continue
+        comp, stack = [], [i]
+        seen.add(i)
+        while stack:
+            x = stack.pop()
+            comp.append(x)
+            for y in adj[x]:
+                if y not in seen:
+                    seen.add(y)
+                    stack.append(y)
+        hives.append(sorted(comp))
+    return hives
+
+
+def discover_node(name: Optional[str] = None, cpu: str = "256",
+                  memory: str = "2048Gi", pods: int = 512) -> Dict[str, Any]:
+    """Build the Node object for this machine from the real topology."""
+    name = name or os.uname().nodename
+    info = probe()
+    gpus = int(info.get("gpu_count", 0)) if info else 0
+    labels = {
+        "kubernetes.io/hostname": name,
+        c.NODE_LABEL_GPU_COUNT: str(gpus),
+    }
+    annotations: Dict[str, str] = {}
+    if info:
+        hives = xgmi_hives(info)
+        # single-hive label for the scheduler; full graph in annotations
+        if gpus:
+            labels[c.NODE_LABEL_XGMI_HIVE] = f"{name}-hive0"
+        devs = info.get("devices") or []
+        gpu_names = [d.get("name", "") for d in devs if d.get("name")]
+        if gpu_names:
+            product = gpu_names[0]
+            labels[c.NODE_LABEL_GPU_PRODUCT] = \
+                "MI355X" if "355" in product else product[:63].replace(" ", "-")
+        annotations["topology.amd.com/xgmi-hives"] = \
+            ";".join(",".join(str(g) for g in h) for h in hives)
+        annotations["topology.amd.com/probe-backend"] = str(info.get("backend", ""))
+    alloc = {"cpu": cpu, "memory": memory, "pods": str(pods)}
+    if gpus:
+        alloc[c.AMD_GPU_RESOURCE] = str(gpus)
+    return {
+        "apiVersion": "v1", "kind": "Node",
+        "metadata": {"name": name, "labels": labels, "annotations": annotations},
+        "spec": {},
+        "status": {"allocatable": dict(alloc), "capacity": dict(alloc),
+                   "conditions": [{"type": "Ready", "status": "True"}]},
+    }

@@ -1,0 +1,110 @@
+"""Native C++ placement core cross-check vs the Python reference implementation."""
+import pytest
+
+from grove_amd.scheduler.placement import NodeFree, PodRequest, place_gang
+
+try:
+    from grove_amd.scheduler import _sched
+except ImportError:
+    _sched = None
+
+needs_native = pytest.mark.skipif(_sched is None, reason="_sched.so not built")
+
+
+def mk_nodes(spec):
+    return [NodeFree(n, cpu, mem, list(range(g)), pods)
+            for (n, cpu, mem, g, pods) in spec]
+
+
+def native_place(nodes, pods):
+    flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids), n.pods)
+                  for n in nodes]
+    flat_pods = [(p.name, p.cpu_milli, float(p.mem_bytes), p.gpus) for p in pods]
+    return _sched.place_gang(flat_nodes, flat_pods)
+
+
+CASES = [
+    # one empty 8-GPU node, 8x1-GPU gang -> single node, score 153
+    ([("a", 100000, 1e12, 8, 100)], [(f"p{i}", 1000, 1e9, 1) for i in range(8)]),
+    # two nodes, one partially full -> pick the emptier that fits whole gang
+    ([("a", 100000, 1e12, 4, 100), ("b", 100000, 1e12, 8, 100)],
+     [(f"p{i}", 1000, 1e9, 1) for i in range(8)]),
+    # doesn't fit on one -> spread over 2
+    ([("a", 100000, 1e12, 6, 100), ("b", 100000, 1e12, 6, 100)],
+     [(f"p{i}", 1000, 1e9, 1) for i in range(8)]),
+    # impossible
+    ([("a", 100000, 1e12, 2, 100)], [(f"p{i}", 1000, 1e9, 1) for i in range(8)]),
+    # cpu-bound fit
+    ([("a", 3000, 1e12, 0, 100), ("b", 100000, 1e12, 0, 100)],
+     [(f"p{i}", 2000, 1e9, 0) for i in range(4)]),
+]
+
+
+@pytest.mark.parametrize("case", range(len(CASES)))
+def test_python_placement(case):
+    node_spec, pod_spec = CASES[case]
+    nodes = mk_nodes(node_spec)
+    pods = [PodRequest(n, c, m, g) for (n, c, m, g) in pod_spec]
+    res = place_gang(nodes, pods)
+    _check(res, node_spec, pod_spec)
+
+
+@needs_native
+@pytest.mark.parametrize("case", range(len(CASES)))
+def test_native_matches_python(case):
+    node_spec, pod_spec = CASES[case]
+    py_nodes = mk_nodes(node_spec)
+    py_pods = [PodRequest(n, c, m, g) for (n, c, m, g) in pod_spec]
+    py_res = place_gang([n.clone() for n in py_nodes], py_pods)
+    nat = native_place(py_nodes, py_pods)
+    if py_res is None:
+        assert nat is None
+        return
+    assert nat is not None
+    _, py_score = py_res
+    nat_assignments, nat_score, consumed = nat
+    assert nat_score == pytest.approx(py_score)
+    # same node-spread cardinality
+    py_nodes_used = {a.node for a in py_res[0]}
+    nat_nodes_used = {a[1] for a in nat_assignments}
+    assert len(nat_nodes_used) == len(py_nodes_used)
+    _check((py_res[0], py_score), node_spec, pod_spec)
+    # native assignment validity: every pod placed once, gpu ids disjoint per node
+    assert sorted(a[0] for a in nat_assignments) == sorted(p[0] for p in pod_spec)
+    per_node: dict = {}
+    for (_pod, node, gpus) in nat_assignments:
+        taken = per_node.setdefault(node, set())
+        for g in gpus:
+            assert g not in taken
+            taken.add(g)
+
+
+def _check(res, node_spec, pod_spec):
+    cap = {n: g for (n, _c, _m, g, _p) in node_spec}
+    total_gpus_needed = sum(g for (_n, _c, _m, g) in pod_spec)
+    total_avail = sum(cap.values())
+    if total_gpus_needed > total_avail:
+        assert res is None
+        return
+    if res is None:
+        return  # may be infeasible for other resources
+    assignments, score = res
+    assert len(assignments) == len(pod_spec)
+    used: dict = {}
+    for a in assignments:
+        used.setdefault(a.node, []).extend(a.gpu_ids)
+    for node, gpus in used.items():
+        assert len(gpus) == len(set(gpus)) <= cap[node]
+    if len(used) == 1 and total_gpus_needed > 1:
+        assert score == pytest.approx(153.0)
+    elif total_gpus_needed > 1:
+        assert score < 153.0
+
+
+@needs_native
+def test_native_consumption_applied():
+    nodes = mk_nodes([("a", 100000, 1e12, 8, 100)])
+    pods = [PodRequest("p0", 1000, 1e9, 4)]
+    _assignments, _score, consumed = native_place(nodes, pods)
+    (name, cpu, _mem, gpu_ids, pods_left) = consumed[0]
+    assert name == "a" and cpu == 99000 and len(gpu_ids) == 4 and pods_left == 99

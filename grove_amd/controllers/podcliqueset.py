@@ -280,11 +280,27 @@ class PodCliqueSetReconciler:
                  self.store.list(c.KIND_PCSG, ns, {c.LABEL_PART_OF: name})}
         by_replica: Dict[int, List[Obj]] = {}
         for q in pclqs:
+            # PCS-scope constituents are STANDALONE cliques + the PCSGs themselves;
+            # PCSG member cliques are judged through their PCSG's MinAvailableBreached
+            # (getPCSReplicaDeletionWork parity) and recycled at PCSG scope.
+            if q["metadata"]["labels"].get(c.LABEL_COMPONENT) != c.COMPONENT_PCS_PODCLIQUE:
+                continue
             ridx = q["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX)
             if ridx is not None and ridx.isdigit():
                 by_replica.setdefault(int(ridx), []).append(q)
+        for g in pcsgs.values():
+            ridx = g["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX)
+            if ridx is not None and ridx.isdigit():
+                by_replica.setdefault(int(ridx), [])
 
+        cu = ((pcs.get("status") or {}).get("rollingUpdateProgress") or {}) \
+            .get("currentlyUpdating") or {}
+        updating_replica = int(cu.get("replicaIndex", -1))
         for ridx, constituents in by_replica.items():
+            # suppression: a replica mid-rolling-update dips below MinAvailable by
+            # design; never gang-terminate it while it is the one being updated
+            if ridx == updating_replica:
+                continue
             # suppression: in-flight termination for this replica's PCSGs
             replica_pcsgs = [g for g in pcsgs.values()
                              if g["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX)
@@ -368,6 +384,8 @@ class PodCliqueSetReconciler:
                 qst = q.get("status") or {}
                 if int(qst.get("updatedReplicas", 0)) < int(q["spec"].get("replicas", 1)):
                     return False
+                if int(qst.get("replicas", 0)) != int(q["spec"].get("replicas", 1)):
+                    return False  # outdated pods still draining
                 if int(qst.get("readyReplicas", 0)) < int(q["spec"].get("minAvailable", 1)):
                     return False
             return True

@@ -151,6 +151,8 @@ class PCSGReconciler:
 
     @staticmethod
     def _replica_selected_for_update(pcs: Obj, r: int) -> bool:
+        if (pcs["spec"].get("updateStrategy") or {}).get("type") == c.UPDATE_ON_DELETE:
+            return True  # OnDelete: spec propagates immediately, pods wait for the user
         prog = (pcs.get("status") or {}).get("rollingUpdateProgress")
         if prog is None:
             return True

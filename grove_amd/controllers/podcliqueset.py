@@ -220,6 +220,8 @@ class PodCliqueSetReconciler:
         # only non-scaled fields drift-corrected here.
 
     def _replica_selected_for_update(self, pcs: Obj, r: int) -> bool:
+        if (pcs["spec"].get("updateStrategy") or {}).get("type") == c.UPDATE_ON_DELETE:
+            return True  # OnDelete: spec propagates immediately, pods wait for the user
         prog = (pcs.get("status") or {}).get("rollingUpdateProgress")
         if prog is None:
             return True  # no update in flight → initial create path
@@ -374,19 +376,32 @@ class PodCliqueSetReconciler:
             return ""
 
         def replica_updated(r: int) -> bool:
+            # Judged from the pods themselves, not PCLQ status: a freshly patched PCLQ
+            # still carries pre-patch status counts (informer-staleness class the
+            # reference handles with its expectations store, expect/expectations.go).
             qs = replica_pclqs(r)
             if not qs:
                 return False
+            from ..utils import conditions as _cond
             for q in qs:
                 if q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) \
                         != pclq_expected_hash(q):
                     return False
-                qst = q.get("status") or {}
-                if int(qst.get("updatedReplicas", 0)) < int(q["spec"].get("replicas", 1)):
+                want = int(q["spec"].get("replicas", 1))
+                min_avail = int(q["spec"].get("minAvailable", 1))
+                pods = self.store.list("Pod", ns, {
+                    c.LABEL_PODCLIQUE: q["metadata"]["name"]})
+                expected_hash = q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH)
+                if len(pods) != want:
                     return False
-                if int(qst.get("replicas", 0)) != int(q["spec"].get("replicas", 1)):
-                    return False  # outdated pods still draining
-                if int(qst.get("readyReplicas", 0)) < int(q["spec"].get("minAvailable", 1)):
+                n_ready = 0
+                for p in pods:
+                    if p["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) \
+                            != expected_hash:
+                        return False  # outdated pod still draining
+                    if _cond.pod_is_ready(p):
+                        n_ready += 1
+                if n_ready < min_avail:
                     return False
             return True
 

@@ -19,6 +19,9 @@ from .controllers.manager import Controller, Manager, Result
 from .controllers.podclique import PodCliqueReconciler
 from .controllers.podcliqueset import PodCliqueSetReconciler
 from .controllers.podcliquescalinggroup import PCSGReconciler
+from .controllers.podgang import PodGangReconciler
+from .controllers.clustertopology import ClusterTopologyReconciler
+from .scheduler.backends import Registry
 from .kubecore.store import Store, Obj, ApiError
 from .kubelet.virtual import VirtualKubelet, make_virtual_node
 from .scheduler.plugin import GangScheduler
@@ -46,9 +49,12 @@ class Cluster:
         self.store.register_validator(c.KIND_CTB, validate_clustertopologybinding)
 
         self.manager = Manager(self.store)
+        self.registry = Registry(self.store, default=scheduler_name)
         self.pcs_rec = PodCliqueSetReconciler(self.store, scheduler_name)
         self.pclq_rec = PodCliqueReconciler(self.store, scheduler_name)
         self.pcsg_rec = PCSGReconciler(self.store, scheduler_name)
+        self.podgang_rec = PodGangReconciler(self.store, self.registry)
+        self.ctb_rec = ClusterTopologyReconciler(self.store, self.registry)
         self.scheduler = GangScheduler(self.store, scheduler_name,
                                        use_native=use_native_scheduler)
         self.kubelet = VirtualKubelet(self.store,
@@ -63,6 +69,10 @@ class Cluster:
             "podclique", self.pclq_rec.reconcile, workers=concurrent_syncs))
         self.c_pcsg = m.add_controller(Controller(
             "podcliquescalinggroup", self.pcsg_rec.reconcile, workers=concurrent_syncs))
+        self.c_podgang = m.add_controller(Controller(
+            "podgang", self.podgang_rec.reconcile, workers=2))
+        self.c_ctb = m.add_controller(Controller(
+            "clustertopology", self.ctb_rec.reconcile, workers=1))
         self.c_sched = m.add_controller(Controller(
             "gang-scheduler", lambda ns, n: self.scheduler.reconcile(ns, n) or Result.DONE,
             workers=1))
@@ -120,6 +130,8 @@ class Cluster:
             md = obj["metadata"]
             ns = md.get("namespace", "default")
             self.c_sched.enqueue("", "pass")
+            if ev != "DELETED":
+                self.c_podgang.enqueue(ns, md["name"])
             # gate-removal re-check for every member clique
             for group in (obj.get("spec") or {}).get("podGroups") or []:
                 self.c_pclq.enqueue(ns, group.get("name", ""))
@@ -131,6 +143,14 @@ class Cluster:
         def on_node(ev: str, obj: Obj, _old) -> None:
             self.c_sched.enqueue("", "pass")
 
+        def on_ctb(ev: str, obj: Obj, _old) -> None:
+            self.c_ctb.enqueue("", obj["metadata"]["name"])
+            # topology translation feeds PodGang specs: re-sync every PCS
+            for p in self.store.list(c.KIND_PCS):
+                self.c_pcs.enqueue(p["metadata"].get("namespace", "default"),
+                                   p["metadata"]["name"])
+
+        m.watch(c.KIND_CTB, on_ctb)
         m.watch(c.KIND_PCS, on_pcs)
         m.watch(c.KIND_PCLQ, on_pclq)
         m.watch(c.KIND_PCSG, on_pcsg)

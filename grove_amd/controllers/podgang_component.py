@@ -21,12 +21,46 @@ Obj = Dict[str, Any]
 
 
 class ExpectedGang:
-    __slots__ = ("name", "base", "groups")
+    __slots__ = ("name", "base", "groups", "constraint", "group_configs")
 
-    def __init__(self, name: str, base: Optional[str], groups: List[Dict[str, Any]]):
+    def __init__(self, name: str, base: Optional[str], groups: List[Dict[str, Any]],
+                 constraint: Optional[Dict[str, Any]] = None,
+                 group_configs: Optional[List[Dict[str, Any]]] = None):
         self.name = name
         self.base = base      # base podgang name for scaled gangs, else None
         self.groups = groups  # [{"name": pclq_fqn, "minReplicas": int, "replicas": int}]
+        self.constraint = constraint          # translated gang-level TopologyConstraint
+        self.group_configs = group_configs or []  # TopologyConstraintGroupConfig[]
+
+
+def domain_key_map(store: Store) -> Dict[str, str]:
+    """CTB levels → {domain: nodeLabelKey}, with native defaults for host/xgmi-hive
+    (components/podgang/syncflow.go:349-380 translation parity)."""
+    m = {"xgmi-hive": c.NODE_LABEL_XGMI_HIVE, "host": "kubernetes.io/hostname"}
+    for ctb in store.list(c.KIND_CTB):
+        for lv in (ctb.get("spec") or {}).get("levels") or []:
+            key = lv.get("nodeLabelKey") or lv.get("key")
+            if lv.get("domain") and key:
+                m[lv["domain"]] = key
+        break  # one CTB is the source of truth
+    return m
+
+
+def translate_constraint(tc: Optional[Dict[str, Any]],
+                         dmap: Dict[str, str]) -> Optional[Dict[str, Any]]:
+    """PCS-style TopologyConstraint (pack.required/preferred domains, deprecated
+    packDomain) → PodGang-style packConstraint holding node label KEYS."""
+    if not tc:
+        return None
+    pack = tc.get("pack") or {}
+    required = pack.get("required") or tc.get("packDomain")
+    preferred = pack.get("preferred")
+    out: Dict[str, str] = {}
+    if required and required in dmap:
+        out["required"] = dmap[required]
+    if preferred and preferred in dmap:
+        out["preferred"] = dmap[preferred]
+    return {"packConstraint": out} if out else None
 
 
 def compute_expected_podgangs(store: Store, pcs: Obj) -> List[ExpectedGang]:
@@ -38,47 +72,78 @@ def compute_expected_podgangs(store: Store, pcs: Obj) -> List[ExpectedGang]:
     sg_cfgs = tmpl.get("podCliqueScalingGroups") or []
     sg_members = {m for sg in sg_cfgs for m in (sg.get("cliqueNames") or [])}
     clique_by_name = {cl["name"]: cl for cl in cliques}
+    dmap = domain_key_map(store)
+    gang_tc = translate_constraint(tmpl.get("topologyConstraint"), dmap)
     out: List[ExpectedGang] = []
+
+    def clique_tc(mn: str) -> Optional[Dict[str, Any]]:
+        return translate_constraint(
+            clique_by_name[mn].get("topologyConstraint")
+            or clique_by_name[mn].get("spec", {}).get("topologyConstraint"), dmap)
 
     for r in range(int(pcs["spec"].get("replicas", 0))):
         base_name = namegen.base_podgang_name(pcs_name, r)
         groups: List[Dict[str, Any]] = []
+        base_group_cfgs: List[Dict[str, Any]] = []
         for cl in cliques:
             if cl["name"] in sg_members:
                 continue
             spec = cl.get("spec", {})
+            fqn = namegen.podclique_name(pcs_name, r, cl["name"])
             groups.append({
-                "name": namegen.podclique_name(pcs_name, r, cl["name"]),
+                "name": fqn,
                 "minReplicas": int(spec.get("minAvailable", spec.get("replicas", 1))),
                 "replicas": int(spec.get("replicas", 1)),
             })
+            tc = clique_tc(cl["name"])
+            if tc:
+                base_group_cfgs.append({"name": fqn, "podGroupNames": [fqn],
+                                        "topologyConstraint": tc})
         for sg in sg_cfgs:
             sg_fqn = namegen.pcsg_name(pcs_name, r, sg["name"])
+            sg_tc = translate_constraint(sg.get("topologyConstraint"), dmap)
             # live PCSG replica count (HPA may have scaled it)
             pcsg = store.try_get(c.KIND_PCSG, ns, sg_fqn)
             replicas = int((pcsg or {}).get("spec", {}).get("replicas", sg.get("replicas", 1)))
             min_avail = int((pcsg or {}).get("spec", {}).get(
                 "minAvailable", sg.get("minAvailable", 1)))
             for j in range(min(min_avail, replicas)):
+                sub_names = []
                 for mn in sg.get("cliqueNames") or []:
                     spec = clique_by_name[mn].get("spec", {})
+                    fqn = namegen.podclique_name(sg_fqn, j, mn)
+                    sub_names.append(fqn)
                     groups.append({
-                        "name": namegen.podclique_name(sg_fqn, j, mn),
+                        "name": fqn,
                         "minReplicas": int(spec.get("minAvailable", spec.get("replicas", 1))),
                         "replicas": int(spec.get("replicas", 1)),
                     })
+                if sg_tc:
+                    # each PCSG replica packs within its own domain (parent subgroup,
+                    # kai/backend.go:260 hierarchical-subgroup parity)
+                    base_group_cfgs.append({"name": f"{sg_fqn}-{j}",
+                                            "podGroupNames": sub_names,
+                                            "topologyConstraint": sg_tc})
             for j in range(min_avail, replicas):
                 sg_groups = []
+                sub_cfgs: List[Dict[str, Any]] = []
                 for mn in sg.get("cliqueNames") or []:
                     spec = clique_by_name[mn].get("spec", {})
+                    fqn = namegen.podclique_name(sg_fqn, j, mn)
                     sg_groups.append({
-                        "name": namegen.podclique_name(sg_fqn, j, mn),
+                        "name": fqn,
                         "minReplicas": int(spec.get("minAvailable", spec.get("replicas", 1))),
                         "replicas": int(spec.get("replicas", 1)),
                     })
+                    tc = clique_tc(mn)
+                    if tc:
+                        sub_cfgs.append({"name": fqn, "podGroupNames": [fqn],
+                                         "topologyConstraint": tc})
                 out.append(ExpectedGang(
-                    namegen.scaled_podgang_name(sg_fqn, j - min_avail), base_name, sg_groups))
-        out.append(ExpectedGang(base_name, None, groups))
+                    namegen.scaled_podgang_name(sg_fqn, j - min_avail), base_name,
+                    sg_groups, constraint=sg_tc or gang_tc, group_configs=sub_cfgs))
+        out.append(ExpectedGang(base_name, None, groups, constraint=gang_tc,
+                                group_configs=base_group_cfgs))
     return out
 
 
@@ -125,6 +190,10 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
                                          base_podgang=gang.base,
                                          priority_class=priority_class)
             obj["spec"]["podGroups"] = groups_spec
+            if gang.constraint:
+                obj["spec"]["topologyConstraint"] = gang.constraint
+            if gang.group_configs:
+                obj["spec"]["topologyConstraintGroupConfigs"] = gang.group_configs
             cond.set_condition(obj, c.PODGANG_COND_INITIALIZED, False, "PendingPodCreation")
             try:
                 cur = store.create(obj)
@@ -135,7 +204,12 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
 
         def upd(o: Obj) -> None:
             o["spec"]["podGroups"] = groups_spec
-        if cur["spec"].get("podGroups") != groups_spec:
+            if gang.constraint:
+                o["spec"]["topologyConstraint"] = gang.constraint
+            if gang.group_configs:
+                o["spec"]["topologyConstraintGroupConfigs"] = gang.group_configs
+        if cur["spec"].get("podGroups") != groups_spec \
+                or cur["spec"].get("topologyConstraint") != gang.constraint:
             try:
                 cur = store.patch(c.KIND_PODGANG, ns, gang.name, upd)
             except ApiError:

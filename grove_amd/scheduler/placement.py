@@ -97,6 +97,59 @@ def placement_score(n_nodes_used: int, total_gpus: int) -> float:
     return NIC_GBPS / (2.0 * (n_nodes_used - 1))
 
 
+def snapshot(nodes: List[NodeFree]):
+    return [(n, n.cpu_milli, n.mem_bytes, list(n.gpu_ids), n.pods) for n in nodes]
+
+
+def restore(snap) -> None:
+    for n, cpu, mem, gpus, pods_ in snap:
+        n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods = cpu, mem, gpus, pods_
+
+
+def place_gang_constrained(nodes: List[NodeFree], pods: List[PodRequest],
+                           required_key: Optional[str] = None,
+                           preferred_key: Optional[str] = None,
+                           place_fn=None,
+                           ) -> Optional[Tuple[List[Assignment], float]]:
+    """place_gang under a topology pack constraint (scheduler.grove.io
+    TopologyPackConstraint semantics, podgang.go:101-118): with required_key, every pod
+    of the gang must land in ONE domain (nodes sharing a value of that label; unlabeled
+    nodes are ineligible). preferred_key narrows best-effort within the required domain,
+    falling back to the full domain when the narrower pack cannot fit."""
+    if place_fn is None:
+        place_fn = place_gang
+    if required_key is None and preferred_key is None:
+        return place_fn(nodes, pods)
+
+    def domains(pool: List[NodeFree], key: str) -> List[List[NodeFree]]:
+        by: Dict[str, List[NodeFree]] = {}
+        for n in pool:
+            v = n.labels.get(key)
+            if v is not None:
+                by.setdefault(v, []).append(n)
+        # try densest domains first (most free GPUs, then cpu)
+        return sorted(by.values(),
+                      key=lambda ns: (-sum(len(n.gpu_ids) for n in ns),
+                                      -sum(n.cpu_milli for n in ns)))
+
+    pools = domains(nodes, required_key) if required_key is not None else [nodes]
+    if not pools:
+        return None
+    # pass 1: a preferred-domain pack anywhere beats any fallback placement
+    if preferred_key is not None:
+        for pool in pools:
+            for sub in domains(pool, preferred_key):
+                res = place_fn(sub, pods)
+                if res is not None:
+                    return res
+    # pass 2: fall back to the full (required) domain
+    for pool in pools:
+        res = place_fn(pool, pods)
+        if res is not None:
+            return res
+    return None
+
+
 def place_gang(nodes: List[NodeFree], pods: List[PodRequest],
                spread: bool = False) -> Optional[Tuple[List[Assignment], float]]:
     """All-or-nothing gang placement. Returns (assignments, score) or None.

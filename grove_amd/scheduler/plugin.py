@@ -192,8 +192,7 @@ class GangScheduler:
                 return  # not admittable yet
             chosen.extend(have)  # place everything available, all-or-nothing on the mins
 
-        reqs = [self._pod_request(p) for p in chosen]
-        result = self._place(nodes, reqs)
+        result = self._place_gang_pods(nodes, pg, chosen)
         if result is None and len(chosen) > sum(int(g.get("minReplicas", 0))
                                                 for g in groups):
             # fall back to the gang minimum only
@@ -202,8 +201,7 @@ class GangScheduler:
                 have = by_clique.get(g["name"], [])
                 chosen2.extend(have[: int(g.get("minReplicas", 0))])
             chosen = chosen2
-            reqs = [self._pod_request(p) for p in chosen]
-            result = self._place(nodes, reqs)
+            result = self._place_gang_pods(nodes, pg, chosen)
         if result is None:
             return  # Permit rollback: nothing bound
         assignments, score = result
@@ -227,6 +225,61 @@ class GangScheduler:
             if p and p.get("spec", {}).get("nodeName"):
                 n += 1
         return n
+
+    def _place_gang_pods(self, nodes: List[NodeFree], pg: Obj, chosen: List[Obj]):
+        """Constraint-aware all-or-nothing placement of a gang's pods.
+
+        Applies the gang-level packConstraint plus topologyConstraintGroupConfigs
+        (each config's pods must pack within one domain of ITS key) with full rollback
+        when any part cannot place."""
+        from .placement import place_gang_constrained, snapshot, restore
+
+        spec = pg.get("spec") or {}
+        tc = (spec.get("topologyConstraint") or {}).get("packConstraint") or {}
+        req_key, pref_key = tc.get("required"), tc.get("preferred")
+        cfgs = spec.get("topologyConstraintGroupConfigs") or []
+        if not cfgs:
+            return place_gang_constrained(nodes, [self._pod_request(p) for p in chosen],
+                                          req_key, pref_key, place_fn=self._place)
+        snap = snapshot(nodes)
+        by_clique: Dict[str, List[Obj]] = {}
+        for p in chosen:
+            by_clique.setdefault(
+                p["metadata"]["labels"].get(c.LABEL_PODCLIQUE, ""), []).append(p)
+        claimed: set = set()
+        all_assignments: List[Assignment] = []
+        worst_score = float("inf")
+        for cfg in cfgs:
+            sub_pods: List[Obj] = []
+            for gname in cfg.get("podGroupNames") or []:
+                sub_pods.extend(by_clique.get(gname, []))
+            if not sub_pods:
+                continue
+            sub_tc = ((cfg.get("topologyConstraint") or {})
+                      .get("packConstraint") or {})
+            res = place_gang_constrained(
+                nodes, [self._pod_request(p) for p in sub_pods],
+                sub_tc.get("required") or req_key,
+                sub_tc.get("preferred") or pref_key, place_fn=self._place)
+            if res is None:
+                restore(snap)
+                return None
+            assignments, score = res
+            all_assignments.extend(assignments)
+            worst_score = min(worst_score, score)
+            claimed.update(p["metadata"]["name"] for p in sub_pods)
+        rest = [p for p in chosen if p["metadata"]["name"] not in claimed]
+        if rest:
+            res = place_gang_constrained(
+                nodes, [self._pod_request(p) for p in rest], req_key, pref_key,
+                place_fn=self._place)
+            if res is None:
+                restore(snap)
+                return None
+            assignments, score = res
+            all_assignments.extend(assignments)
+            worst_score = min(worst_score, score)
+        return all_assignments, (worst_score if worst_score != float("inf") else 0.0)
 
     def _place(self, nodes: List[NodeFree], reqs: List[PodRequest]):
         if self.use_native and _native is not None:

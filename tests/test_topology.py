@@ -1,0 +1,146 @@
+"""Topology-aware scheduling tests (reference TAS1-18 parity, e2e/tests/topology_test.go):
+CTB → SchedulerTopology sync + drift condition, constraint translation to node-label
+keys, and pack-constraint enforcement in the gang scheduler."""
+import time
+
+import pytest
+
+from grove_amd.api import constants as c
+from grove_amd.scheduler.backends import KIND_SCHEDULER_TOPOLOGY
+from grove_amd.utils import conditions as cond
+
+
+CTB = {
+    "apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+    "metadata": {"name": "cluster-topology"},
+    "spec": {"levels": [
+        {"domain": "zone", "nodeLabelKey": "topology.kubernetes.io/zone"},
+        {"domain": "rack", "nodeLabelKey": "topology.kubernetes.io/rack"},
+        {"domain": "host", "nodeLabelKey": "kubernetes.io/hostname"},
+    ]},
+}
+
+
+def _pcs(name, gang_pods, gpus_per_pod=1, constraint=None):
+    tmpl = {"cliques": [{"name": "w", "spec": {
+        "roleName": "w", "replicas": gang_pods, "minAvailable": gang_pods,
+        "podSpec": {"containers": [{"name": "m", "image": "img",
+                                    "resources": {"requests": {
+                                        "cpu": "1",
+                                        c.AMD_GPU_RESOURCE: str(gpus_per_pod)}}}]}}}]}
+    if constraint:
+        tmpl["topologyConstraint"] = constraint
+    return {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+            "metadata": {"name": name},
+            "spec": {"replicas": 1, "template": tmpl}}
+
+
+def test_ctb_sync_creates_scheduler_topology(cluster):
+    cluster.store.create(CTB)
+
+    def synced():
+        st = cluster.store.try_get(KIND_SCHEDULER_TOPOLOGY, None, "cluster-topology")
+        return st is not None
+    cluster.wait_for(synced, timeout=10, desc="SchedulerTopology created")
+    st = cluster.store.get(KIND_SCHEDULER_TOPOLOGY, None, "cluster-topology")
+    domains = [lv["domain"] for lv in st["spec"]["levels"]]
+    assert domains == ["zone", "rack", "host", "xgmi-hive"]
+    assert st["spec"]["levels"][-1]["nodeLabelKey"] == c.NODE_LABEL_XGMI_HIVE
+
+    def in_sync():
+        ctb = cluster.store.get(c.KIND_CTB, None, "cluster-topology")
+        d = cond.get_condition(ctb, c.COND_SCHEDULER_TOPOLOGY_DRIFT)
+        return d is not None and d["status"] == "False"
+    cluster.wait_for(in_sync, timeout=10, desc="drift condition False")
+
+
+def test_ctb_drift_repaired(cluster):
+    cluster.store.create(CTB)
+    cluster.wait_for(lambda: cluster.store.try_get(
+        KIND_SCHEDULER_TOPOLOGY, None, "cluster-topology") is not None, timeout=10)
+    # external mutation drifts the backend topology resource
+    cluster.store.patch(KIND_SCHEDULER_TOPOLOGY, None, "cluster-topology",
+                        lambda o: o["spec"].update(levels=[]))
+    cluster.c_ctb.enqueue("", "cluster-topology")
+
+    def repaired():
+        st = cluster.store.get(KIND_SCHEDULER_TOPOLOGY, None, "cluster-topology")
+        return len(st["spec"]["levels"]) == 4
+    cluster.wait_for(repaired, timeout=10, desc="auto-managed topology repaired")
+
+
+def test_ctb_validation_rejects_duplicates(cluster):
+    bad = {"apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+           "metadata": {"name": "bad"},
+           "spec": {"levels": [
+               {"domain": "rack", "nodeLabelKey": "a"},
+               {"domain": "rack", "nodeLabelKey": "b"}]}}
+    from grove_amd.kubecore.store import ApiError
+    with pytest.raises(ApiError):
+        cluster.store.create(bad)
+
+
+def test_pack_constraint_translated_to_podgang(cluster):
+    cluster.store.create(CTB)
+    cluster.add_virtual_nodes(1, gpus=8,
+                              labels={"topology.kubernetes.io/rack": "rack-0"})
+    cluster.apply(_pcs("tc1", 2, constraint={"pack": {"required": "rack",
+                                                      "preferred": "host"}}))
+    cluster.wait_pcs_available("tc1", timeout=20)
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "tc1-0")
+    pc = pg["spec"]["topologyConstraint"]["packConstraint"]
+    assert pc == {"required": "topology.kubernetes.io/rack",
+                  "preferred": "kubernetes.io/hostname"}
+    # amd backend stamps the topology-name discovery annotation (bridge controller)
+    assert pg["metadata"]["annotations"][c.ANNOTATION_TOPOLOGY_NAME] == \
+        "cluster-topology"
+
+
+def test_required_pack_enforced(cluster):
+    """8-GPU gang with required rack packing: racks with insufficient capacity are
+    rejected; the gang lands entirely in the one rack that fits."""
+    cluster.store.create(CTB)
+    # rack-a: two nodes with 4 free GPUs each (8 total but split); rack-b: one 8-GPU node
+    for i in range(2):
+        cluster.add_virtual_nodes(1, gpus=4, prefix=f"a{i}",
+                                  labels={"topology.kubernetes.io/rack": "rack-a"})
+    cluster.add_virtual_nodes(1, gpus=8, prefix="b",
+                              labels={"topology.kubernetes.io/rack": "rack-b"})
+    cluster.apply(_pcs("tc2", 8, constraint={"pack": {"required": "rack",
+                                                      "preferred": "host"}}))
+    cluster.wait_pcs_available("tc2", timeout=20)
+    pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "tc2"})
+    nodes = {p["spec"]["nodeName"] for p in pods}
+    assert nodes == {"b-0"}  # preferred host-pack within required rack → one node
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "tc2-0")
+    assert pg["status"]["placementScore"] == pytest.approx(c.XGMI_LINK_GBPS)
+
+
+def test_required_pack_unsatisfiable_blocks_gang(cluster):
+    cluster.store.create(CTB)
+    for i in range(3):
+        cluster.add_virtual_nodes(1, gpus=4, prefix=f"n{i}",
+                                  labels={"topology.kubernetes.io/rack": f"rack-{i}"})
+    cluster.apply(_pcs("tc3", 8, constraint={"pack": {"required": "rack"}}))
+    time.sleep(1.0)
+    pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "tc3"})
+    assert all(not p["spec"].get("nodeName") for p in pods)
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "tc3-0")
+    assert not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED)
+    # capacity appears in one rack → gang goes through
+    cluster.add_virtual_nodes(1, gpus=8, prefix="big",
+                              labels={"topology.kubernetes.io/rack": "rack-big"})
+    cluster.wait_pcs_available("tc3", timeout=20)
+
+
+def test_preferred_pack_falls_back(cluster):
+    """Preferred host-pack falls back to rack spread when no single host fits."""
+    cluster.store.create(CTB)
+    for i in range(2):
+        cluster.add_virtual_nodes(1, gpus=4, prefix=f"r{i}",
+                                  labels={"topology.kubernetes.io/rack": "rack-x"})
+    cluster.apply(_pcs("tc4", 8, constraint={"pack": {"required": "rack",
+                                                      "preferred": "host"}}))
+    cluster.wait_pcs_available("tc4", timeout=20)
+    pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "tc4"})
+    assert {p["spec"]["nodeName"] for p in pods} == {"r0-0", "r1-0"}

@@ -87,6 +87,10 @@ class TestGangTermination:
         cluster.apply(_pcs("gt4", termination_delay="300ms",
                            cliques=(("w", 1, 1),), sg=sg))
         cluster.wait_pcs_available("gt4", timeout=20)
+        # PCS availability needs only minAvailable PCSG replicas; wait until the
+        # SCALED replica's pod is ready too, else the kill races the first schedule
+        # and the never-scheduled suppression (by design) blocks the recycle.
+        cluster.wait_pods_ready({c.LABEL_PART_OF: "gt4"}, 2, timeout=20)
         scaled_pclq = "gt4-0-sg-1-w"
         old_uid = cluster.store.get(c.KIND_PCLQ, "default", scaled_pclq)["metadata"]["uid"]
         for n in cluster.store.list("Node"):

@@ -58,7 +58,7 @@ class PodCliqueReconciler:
     def _owned_pods(self, pclq: Obj) -> List[Obj]:
         return self.store.list(
             "Pod", pclq["metadata"].get("namespace"),
-            {c.LABEL_PODCLIQUE: pclq["metadata"]["name"]})
+            {c.LABEL_PODCLIQUE: pclq["metadata"]["name"]}, copy_objects=False)
 
     def _reconcile_spec(self, pclq: Obj) -> Result:
         ns = pclq["metadata"].get("namespace")

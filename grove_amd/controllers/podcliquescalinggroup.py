@@ -55,7 +55,8 @@ class PCSGReconciler:
 
     def _member_pclqs(self, pcsg: Obj) -> List[Obj]:
         return self.store.list(c.KIND_PCLQ, pcsg["metadata"].get("namespace"),
-                               {c.LABEL_PCSG: pcsg["metadata"]["name"]})
+                               {c.LABEL_PCSG: pcsg["metadata"]["name"]},
+                               copy_objects=False)
 
     @staticmethod
     def _sg_config(pcs: Obj, pcsg: Obj) -> Optional[Obj]:

@@ -277,9 +277,10 @@ class PodCliqueSetReconciler:
         next_wait: Optional[float] = None
         now = time.time()
 
-        pclqs = self.store.list(c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name})
+        pclqs = self.store.list(c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name}, copy_objects=False)
         pcsgs = {g["metadata"]["name"]: g for g in
-                 self.store.list(c.KIND_PCSG, ns, {c.LABEL_PART_OF: name})}
+                 self.store.list(c.KIND_PCSG, ns, {c.LABEL_PART_OF: name},
+                                 copy_objects=False)}
         by_replica: Dict[int, List[Obj]] = {}
         for q in pclqs:
             # PCS-scope constituents are STANDALONE cliques + the PCSGs themselves;
@@ -365,7 +366,8 @@ class PodCliqueSetReconciler:
 
         def replica_pclqs(r: int) -> List[Obj]:
             return self.store.list(c.KIND_PCLQ, ns, {
-                c.LABEL_PART_OF: name, c.LABEL_PCS_REPLICA_INDEX: str(r)})
+                c.LABEL_PART_OF: name, c.LABEL_PCS_REPLICA_INDEX: str(r)},
+                copy_objects=False)
 
         def pclq_expected_hash(q: Obj) -> str:
             cl_name = q["metadata"]["name"].rsplit("-", 1)[-1]
@@ -390,7 +392,7 @@ class PodCliqueSetReconciler:
                 want = int(q["spec"].get("replicas", 1))
                 min_avail = int(q["spec"].get("minAvailable", 1))
                 pods = self.store.list("Pod", ns, {
-                    c.LABEL_PODCLIQUE: q["metadata"]["name"]})
+                    c.LABEL_PODCLIQUE: q["metadata"]["name"]}, copy_objects=False)
                 expected_hash = q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH)
                 if len(pods) != want:
                     return False
@@ -470,8 +472,8 @@ class PodCliqueSetReconciler:
         sg_members = {m for sg in tmpl.get("podCliqueScalingGroups") or []
                       for m in (sg.get("cliqueNames") or [])}
 
-        pclqs = self.store.list(c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name})
-        pcsgs = self.store.list(c.KIND_PCSG, ns, {c.LABEL_PART_OF: name})
+        pclqs = self.store.list(c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name}, copy_objects=False)
+        pcsgs = self.store.list(c.KIND_PCSG, ns, {c.LABEL_PART_OF: name}, copy_objects=False)
         pclq_by_replica: Dict[int, List[Obj]] = {}
         for q in pclqs:
             if q["metadata"]["labels"].get(c.LABEL_COMPONENT) != c.COMPONENT_PCS_PODCLIQUE:

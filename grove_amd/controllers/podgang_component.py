@@ -154,7 +154,7 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
     expected = compute_expected_podgangs(store, pcs)
     expected_names = {g.name for g in expected}
 
-    existing = store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: pcs_name})
+    existing = store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)
     for pg in existing:
         if pg["metadata"]["name"] not in expected_names:
             try:
@@ -163,7 +163,7 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
                 pass
 
     # index pods once per sync pass: podgang label -> podclique label -> [pod names]
-    pods = store.list("Pod", ns, {c.LABEL_PART_OF: pcs_name})
+    pods = store.list("Pod", ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)
     by_gang_clique: Dict[Tuple[str, str], List[str]] = {}
     for p in pods:
         lbl = p["metadata"].get("labels", {})

@@ -88,12 +88,31 @@ CLUSTER_SCOPED = {"ClusterTopologyBinding", "Node", "Namespace"}
 
 
 class _KindTable:
-    __slots__ = ("objects", "watchers")
+    __slots__ = ("objects", "watchers", "label_indexes")
 
     def __init__(self) -> None:
         # key: (namespace, name) -> obj ; cluster-scoped use namespace ""
         self.objects: Dict[Tuple[str, str], Obj] = {}
         self.watchers: List["Watch"] = []
+        # label_key -> label_value -> set of object keys (inverted index for the
+        # selector scans that dominate reconcile cost at 1000-pod scale)
+        self.label_indexes: Dict[str, Dict[str, set]] = {}
+
+    def index_add(self, key: Tuple[str, str], obj: Obj) -> None:
+        labels = obj.get("metadata", {}).get("labels") or {}
+        for lk, idx in self.label_indexes.items():
+            v = labels.get(lk)
+            if v is not None:
+                idx.setdefault(v, set()).add(key)
+
+    def index_remove(self, key: Tuple[str, str], obj: Obj) -> None:
+        labels = obj.get("metadata", {}).get("labels") or {}
+        for lk, idx in self.label_indexes.items():
+            v = labels.get(lk)
+            if v is not None:
+                s = idx.get(v)
+                if s is not None:
+                    s.discard(key)
 
 
 class Watch:
@@ -193,6 +212,7 @@ class Store:
             m["generation"] = 1
             m["creationTimestamp"] = now_iso()
             tbl.objects[key] = obj
+            tbl.index_add(key, obj)
             self._uid_index[m["uid"]] = (kind, ns, m["name"])
             self._notify(tbl, ADDED, obj)
         return copy.deepcopy(obj)
@@ -212,20 +232,47 @@ class Store:
         except ApiError:
             return None
 
+    # Label keys kept in the inverted index (built lazily on first indexed list()).
+    INDEXED_LABELS = frozenset((
+        "grove.io/podclique", "grove.io/podgang", "grove.io/base-podgang",
+        "grove.io/podcliquescalinggroup", "app.kubernetes.io/part-of",
+    ))
+
     def list(self, kind: str, namespace: Optional[str] = None,
              label_selector: Optional[Dict[str, str]] = None,
-             filter_fn: Optional[Callable[[Obj], bool]] = None) -> List[Obj]:
+             filter_fn: Optional[Callable[[Obj], bool]] = None,
+             copy_objects: bool = True) -> List[Obj]:
+        """List objects. copy_objects=False returns direct references for read-only
+        consumers (status rollups, the scheduler pass) — callers MUST NOT mutate."""
         with self._lock:
             tbl = self._table(kind)
+            candidates = None
+            if label_selector:
+                for lk in label_selector:
+                    if lk in self.INDEXED_LABELS:
+                        idx = tbl.label_indexes.get(lk)
+                        if idx is None:
+                            idx = {}
+                            for key, obj in tbl.objects.items():
+                                v = (obj.get("metadata", {}).get("labels") or {}).get(lk)
+                                if v is not None:
+                                    idx.setdefault(v, set()).add(key)
+                            tbl.label_indexes[lk] = idx
+                        keys = idx.get(label_selector[lk], set())
+                        candidates = [(k, tbl.objects[k]) for k in keys
+                                      if k in tbl.objects]
+                        break
+            if candidates is None:
+                candidates = list(tbl.objects.items())
             out = []
-            for (ns, _name), obj in tbl.objects.items():
+            for (ns, _name), obj in candidates:
                 if namespace is not None and ns != namespace:
                     continue
                 if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
                     continue
                 if filter_fn is not None and not filter_fn(obj):
                     continue
-                out.append(copy.deepcopy(obj))
+                out.append(copy.deepcopy(obj) if copy_objects else obj)
             return out
 
     def _apply_update(self, obj: Obj, status_only: bool) -> Obj:
@@ -270,6 +317,9 @@ class Store:
                 if new.get("spec") != cur.get("spec"):
                     new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
             new["metadata"]["resourceVersion"] = self._next_rv()
+            if not status_only and new["metadata"].get("labels") != cur_m.get("labels"):
+                tbl.index_remove(key, cur)
+                tbl.index_add(key, new)
             tbl.objects[key] = new
             self._notify(tbl, MODIFIED, new)
             # finalizer removal on a deleting object may allow actual deletion
@@ -336,6 +386,7 @@ class Store:
             obj = tbl.objects.pop((ns, name), None)
             if obj is None:
                 return
+            tbl.index_remove((ns, name), obj)
             uid = obj["metadata"].get("uid")
             if uid:
                 self._uid_index.pop(uid, None)

@@ -64,7 +64,7 @@ class GangScheduler:
         nodes = self._build_node_views()
         if not nodes:
             return
-        pods = self.store.list("Pod")
+        pods = self.store.list("Pod", copy_objects=False)
         bound: List[Obj] = []
         pending_by_gang: Dict[Tuple[str, str], List[Obj]] = {}
         pending_single: List[Obj] = []
@@ -118,7 +118,7 @@ class GangScheduler:
     # ------------------------------------------------------------------ node views
     def _build_node_views(self) -> Dict[str, NodeFree]:
         out: Dict[str, NodeFree] = {}
-        for n in self.store.list("Node"):
+        for n in self.store.list("Node", copy_objects=False):
             if (n.get("spec") or {}).get("unschedulable"):
                 continue
             alloc = (n.get("status") or {}).get("allocatable") or {}
@@ -350,7 +350,7 @@ class GangScheduler:
 
     # ------------------------------------------------------------------ ready rollup
     def _rollup_ready(self) -> None:
-        for pg in self.store.list(c.KIND_PODGANG):
+        for pg in self.store.list(c.KIND_PODGANG, copy_objects=False):
             if not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
                 continue
             if cond.condition_true(pg, c.PODGANG_COND_READY):

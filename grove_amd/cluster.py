@@ -21,6 +21,7 @@ from .controllers.podcliqueset import PodCliqueSetReconciler
 from .controllers.podcliquescalinggroup import PCSGReconciler
 from .controllers.podgang import PodGangReconciler
 from .controllers.clustertopology import ClusterTopologyReconciler
+from .controllers.hpa import HPAReconciler
 from .scheduler.backends import Registry
 from .kubecore.store import Store, Obj, ApiError
 from .kubelet.virtual import VirtualKubelet, make_virtual_node
@@ -37,7 +38,9 @@ class Cluster:
                  startup_latency_s: float = 0.0,
                  ready_latency_s: float = 0.0,
                  pod_payload: Optional[Callable[[Obj], None]] = None,
-                 use_native_scheduler: Optional[bool] = None):
+                 use_native_scheduler: Optional[bool] = None,
+                 enable_authorizer: bool = True,
+                 auto_xgmi_domain: bool = False):
         self.store = Store()
         self.scheduler_name = scheduler_name
 
@@ -47,14 +50,20 @@ class Cluster:
         self.store.register_mutator(c.KIND_PCLQ, default_podclique)
         self.store.register_mutator(c.KIND_PCSG, default_pcsg)
         self.store.register_validator(c.KIND_CTB, validate_clustertopologybinding)
+        if enable_authorizer:
+            from .api.authorization import Authorizer
+            self.authorizer = Authorizer(self.store)
+            self.authorizer.register()
 
         self.manager = Manager(self.store)
         self.registry = Registry(self.store, default=scheduler_name)
-        self.pcs_rec = PodCliqueSetReconciler(self.store, scheduler_name)
+        self.pcs_rec = PodCliqueSetReconciler(self.store, scheduler_name,
+                                              auto_xgmi_domain=auto_xgmi_domain)
         self.pclq_rec = PodCliqueReconciler(self.store, scheduler_name)
         self.pcsg_rec = PCSGReconciler(self.store, scheduler_name)
         self.podgang_rec = PodGangReconciler(self.store, self.registry)
         self.ctb_rec = ClusterTopologyReconciler(self.store, self.registry)
+        self.hpa_rec = HPAReconciler(self.store)
         self.scheduler = GangScheduler(self.store, scheduler_name,
                                        use_native=use_native_scheduler)
         self.kubelet = VirtualKubelet(self.store,
@@ -73,6 +82,8 @@ class Cluster:
             "podgang", self.podgang_rec.reconcile, workers=2))
         self.c_ctb = m.add_controller(Controller(
             "clustertopology", self.ctb_rec.reconcile, workers=1))
+        self.c_hpa = m.add_controller(Controller(
+            "hpa", self.hpa_rec.reconcile, workers=1))
         self.c_sched = m.add_controller(Controller(
             "gang-scheduler", lambda ns, n: self.scheduler.reconcile(ns, n) or Result.DONE,
             workers=1))
@@ -150,6 +161,12 @@ class Cluster:
                 self.c_pcs.enqueue(p["metadata"].get("namespace", "default"),
                                    p["metadata"]["name"])
 
+        def on_hpa(ev: str, obj: Obj, _old) -> None:
+            if ev == "ADDED":
+                md = obj["metadata"]
+                self.c_hpa.enqueue(md.get("namespace", "default"), md["name"])
+
+        m.watch("HorizontalPodAutoscaler", on_hpa)
         m.watch(c.KIND_CTB, on_ctb)
         m.watch(c.KIND_PCS, on_pcs)
         m.watch(c.KIND_PCLQ, on_pclq)

@@ -168,6 +168,15 @@ def build_pod(pcs: Obj, pclq: Obj, pod_index: int, scheduler_name: str,
         ctr.setdefault("env", [])
         ctr["env"] = env + ctr["env"]
 
+    claim_refs = pclq["spec"].get("resourceClaims") or []
+    if claim_refs:
+        pod_spec["resourceClaims"] = [
+            {"name": r["name"], "resourceClaimName": r["resourceClaimName"]}
+            for r in claim_refs]
+        for ctr in pod_spec.get("containers", []):
+            ctr.setdefault("resources", {}).setdefault("claims", [])
+            ctr["resources"]["claims"].extend({"name": r["name"]} for r in claim_refs)
+
     pod_spec["schedulingGates"] = [{"name": c.POD_GANG_SCHEDULING_GATE}]
     pod_spec["schedulerName"] = scheduler_name
     pod_spec["hostname"] = namegen.pod_hostname(pclq_name, pod_index)

@@ -21,6 +21,7 @@ from ..utils.hashing import pod_template_hash
 from . import builders
 from .manager import Result
 from .podcliqueset import _iso_to_epoch
+from . import resourceclaims
 
 log = logging.getLogger("grove.pcsg")
 
@@ -101,6 +102,10 @@ class PCSGReconciler:
         cliques = {cl["name"]: cl for cl in pcs["spec"]["template"].get("cliques") or []}
         base_pg = namegen.base_podgang_name(pcs["metadata"]["name"], pcs_replica)
 
+        sg_cfg = self._sg_config(pcs, pcsg) or {}
+        sg_claims = resourceclaims.pcsg_claims(pcs, sg_cfg, sg_fqn, replicas)
+        resourceclaims.ensure_claims(
+            self.store, [(cl0, e) for (cl0, e, _j) in sg_claims])
         existing = {q["metadata"]["name"]: q for q in self._member_pclqs(pcsg)}
         expected: set = set()
         for j in range(replicas):
@@ -119,6 +124,11 @@ class PCSGReconciler:
                         pcsg_name=sg_fqn, pcsg_replica=j,
                         podgang_name=pg_name,
                         base_podgang_name=base_pg if j >= min_avail else None)
+                    refs = resourceclaims.claim_refs_for_clique(
+                        [(cl0, e) for (cl0, e, jj) in sg_claims
+                         if jj is None or jj == j], mn)
+                    if refs:
+                        obj["spec"]["resourceClaims"] = refs
                     obj["spec"]["updateStrategy"] = (
                         pcs["spec"].get("updateStrategy") or {}).get(
                         "type", c.UPDATE_ROLLING_RECREATE)

@@ -19,9 +19,11 @@ from ..api.defaulting import parse_duration_seconds
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
 from ..utils.hashing import pcs_generation_hash, pod_template_hash
+from ..utils import errors as groveerr
 from . import builders
 from .manager import Result
 from .podgang_component import sync_podgangs
+from . import resourceclaims
 
 log = logging.getLogger("grove.podcliqueset")
 
@@ -34,12 +36,30 @@ def _iso_to_epoch(ts: str) -> float:
 
 
 class PodCliqueSetReconciler:
-    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG):
+    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG,
+                 auto_xgmi_domain: bool = False):
         self.store = store
         self.scheduler_name = scheduler_name
+        self.auto_xgmi_domain = auto_xgmi_domain
 
     # ------------------------------------------------------------------ entry
     def reconcile(self, namespace: str, name: str) -> Result:
+        """Top-level reconcile; errors are recorded to status.lastErrors with their
+        ERR_* code (reconcileerrorrecorder.go parity) and re-raised for backoff."""
+        try:
+            res = self._reconcile(namespace, name)
+            groveerr.clear_last_errors(self.store, c.KIND_PCS, namespace, name)
+            return res
+        except groveerr.GroveError as e:
+            groveerr.record_last_error(self.store, c.KIND_PCS, namespace, name,
+                                       e.code, e.message)
+            raise
+        except Exception as e:
+            groveerr.record_last_error(self.store, c.KIND_PCS, namespace, name,
+                                       groveerr.ERR_RECONCILE, str(e))
+            raise
+
+    def _reconcile(self, namespace: str, name: str) -> Result:
         pcs = self.store.try_get(c.KIND_PCS, namespace, name)
         if pcs is None:
             return Result.DONE
@@ -145,15 +165,19 @@ class PodCliqueSetReconciler:
                     pass
         self._sync_hpas(pcs)
 
-        # ---- G2: standalone PodCliques per replica
+        # ---- G2: standalone PodCliques per replica (+ shared ResourceClaims)
         expected_pclqs: Set[str] = set()
         for r in range(replicas):
+            claims = resourceclaims.pcs_claims_for_replica(
+                self.store, pcs, r, self.auto_xgmi_domain)
+            resourceclaims.ensure_claims(self.store, claims)
             for cl in tmpl.get("cliques") or []:
                 if cl["name"] in sg_members:
                     continue
                 fqn = namegen.podclique_name(name, r, cl["name"])
                 expected_pclqs.add(fqn)
-                self._sync_pclq(pcs, r, cl, fqn, owner=pcs)
+                refs = resourceclaims.claim_refs_for_clique(claims, cl["name"])
+                self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs)
         # GC excess standalone PCLQs (scale-in / replica removal)
         for pclq in self.store.list(c.KIND_PCLQ, ns, {
                 c.LABEL_PART_OF: name, c.LABEL_COMPONENT: c.COMPONENT_PCS_PODCLIQUE}):
@@ -186,13 +210,16 @@ class PodCliqueSetReconciler:
         sync_podgangs(self.store, pcs, self.scheduler_name)
         return Result.DONE
 
-    def _sync_pclq(self, pcs: Obj, r: int, clique_tmpl: Obj, fqn: str, owner: Obj) -> None:
+    def _sync_pclq(self, pcs: Obj, r: int, clique_tmpl: Obj, fqn: str, owner: Obj,
+                   claim_refs=None) -> None:
         ns = pcs["metadata"].get("namespace", "default")
         cur = self.store.try_get(c.KIND_PCLQ, ns, fqn)
         if cur is None:
             obj = builders.build_podclique(pcs, r, clique_tmpl, owner)
             obj["spec"]["updateStrategy"] = (pcs["spec"].get("updateStrategy") or {}).get(
                 "type", c.UPDATE_ROLLING_RECREATE)
+            if claim_refs:
+                obj["spec"]["resourceClaims"] = claim_refs
             try:
                 self.store.create(obj)
             except ApiError:
@@ -209,7 +236,11 @@ class PodCliqueSetReconciler:
             def upd(o: Obj) -> None:
                 o["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] = new_hash
                 o["spec"]["podSpec"] = clique_tmpl["spec"].get("podSpec", {})
-                for f in ("replicas", "minAvailable"):
+                # HPA-aware replica preservation (podclique.go:284): never stomp the
+                # replicas of an autoscaled clique
+                fields = ("minAvailable",) if clique_tmpl["spec"].get(
+                    "autoScalingConfig") else ("replicas", "minAvailable")
+                for f in fields:
                     if f in clique_tmpl["spec"]:
                         o["spec"][f] = clique_tmpl["spec"][f]
             try:

@@ -1,0 +1,139 @@
+"""Resource sharing — DRA ResourceClaims, MI355X-native.
+
+Parity source: operator/internal/resourceclaim/{reconcile,resolve,naming}.go +
+components/resourceclaim/ + pod injection (podclique/components/pod/pod.go:206-269):
+named ResourceClaimTemplates declared on the PCS template are materialized as
+ResourceClaim objects at PCS scope (AllReplicas = one claim for the whole set,
+PerReplica = one claim per PCS replica) or PCSG scope, filtered to child cliques, and
+referenced from pod spec.resourceClaims.
+
+MI355X replacement for the reference's MNNVL/ComputeDomain path (mnnvl/):
+`auto_xgmi_domain` creates one xGMI-domain claim per PCS replica (deviceClass
+xgmi.amd.com) injected into every GPU-requesting clique, expressing "this gang shares
+one xGMI hive" to DRA-aware schedulers — single-node 8×MI355X needs no cross-node
+fabric CRD.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Tuple
+
+from ..api import constants as c
+from ..kubecore.store import Store, Obj, ApiError, owner_reference
+
+XGMI_DEVICE_CLASS = "xgmi.amd.com"
+XGMI_TEMPLATE_NAME = "xgmi-domain"
+
+ClaimRef = Dict[str, str]  # {"name": template, "resourceClaimName": claim}
+
+
+def build_resource_claim(pcs: Obj, name: str, spec: Obj) -> Obj:
+    return {
+        "apiVersion": "resource.k8s.io/v1beta1", "kind": "ResourceClaim",
+        "metadata": {
+            "name": name,
+            "namespace": pcs["metadata"].get("namespace", "default"),
+            "labels": {
+                c.LABEL_MANAGED_BY: c.LABEL_MANAGED_BY_VALUE,
+                c.LABEL_PART_OF: pcs["metadata"]["name"],
+                c.LABEL_COMPONENT: c.COMPONENT_RESOURCE_CLAIM,
+            },
+            "ownerReferences": [owner_reference(pcs)],
+        },
+        "spec": spec,
+    }
+
+
+def resolve_templates(pcs: Obj) -> Dict[str, Obj]:
+    out: Dict[str, Obj] = {}
+    for t in pcs["spec"]["template"].get("resourceClaimTemplates") or []:
+        if t.get("name"):
+            out[t["name"]] = t.get("spec") or {}
+    return out
+
+
+def _filter_allows(entry: Obj, clique_name: str) -> bool:
+    filt = entry.get("filter") or {}
+    allowed = filt.get("childCliqueNames")
+    return allowed is None or clique_name in allowed
+
+
+def pcs_claims_for_replica(store: Store, pcs: Obj, r: int,
+                           auto_xgmi_domain: bool) -> List[Tuple[Obj, Obj]]:
+    """Returns [(claim_object, sharing_entry)] to ensure for PCS replica r."""
+    pcs_name = pcs["metadata"]["name"]
+    templates = resolve_templates(pcs)
+    out: List[Tuple[Obj, Obj]] = []
+    for entry in pcs["spec"]["template"].get("resourceSharing") or []:
+        tname = entry.get("templateRef") or entry.get("name")
+        spec = templates.get(tname)
+        if spec is None:
+            continue
+        scope = entry.get("scope", "AllReplicas")
+        claim_name = f"{pcs_name}-{tname}" if scope == "AllReplicas" \
+            else f"{pcs_name}-{r}-{tname}"
+        out.append((build_resource_claim(pcs, claim_name, spec), entry))
+    if auto_xgmi_domain and _pcs_requests_gpus(pcs):
+        entry = {"templateRef": XGMI_TEMPLATE_NAME, "scope": "PerReplica",
+                 "filter": {"childCliqueNames": _gpu_cliques(pcs)}}
+        claim = build_resource_claim(
+            pcs, f"{pcs_name}-{r}-{XGMI_TEMPLATE_NAME}",
+            {"devices": {"requests": [{"name": "xgmi-hive",
+                                       "deviceClassName": XGMI_DEVICE_CLASS}]}})
+        out.append((claim, entry))
+    return out
+
+
+def _gpu_cliques(pcs: Obj) -> List[str]:
+    names = []
+    for cl in pcs["spec"]["template"].get("cliques") or []:
+        for ctr in (cl.get("spec", {}).get("podSpec", {}).get("containers") or []):
+            req = ((ctr.get("resources") or {}).get("requests") or {})
+            lim = ((ctr.get("resources") or {}).get("limits") or {})
+            if c.AMD_GPU_RESOURCE in req or c.AMD_GPU_RESOURCE in lim:
+                names.append(cl["name"])
+                break
+    return names
+
+
+def _pcs_requests_gpus(pcs: Obj) -> bool:
+    return bool(_gpu_cliques(pcs))
+
+
+def ensure_claims(store: Store, claims: List[Tuple[Obj, Obj]]) -> None:
+    for claim, _entry in claims:
+        try:
+            store.create(claim)
+        except ApiError as e:
+            if e.reason != "AlreadyExists":
+                raise
+
+
+def claim_refs_for_clique(claims: List[Tuple[Obj, Obj]],
+                          clique_name: str) -> List[ClaimRef]:
+    refs: List[ClaimRef] = []
+    for claim, entry in claims:
+        if _filter_allows(entry, clique_name):
+            tname = entry.get("templateRef") or entry.get("name")
+            refs.append({"name": tname,
+                         "resourceClaimName": claim["metadata"]["name"]})
+    return refs
+
+
+def pcsg_claims(pcs: Obj, sg_cfg: Obj, sg_fqn: str, replicas: int
+                ) -> List[Tuple[Obj, Obj, Optional[int]]]:
+    """PCSG-level sharing: [(claim, entry, replica_index_or_None_for_all)]."""
+    templates = resolve_templates(pcs)
+    out: List[Tuple[Obj, Obj, Optional[int]]] = []
+    for entry in sg_cfg.get("resourceSharing") or []:
+        tname = entry.get("templateRef") or entry.get("name")
+        spec = templates.get(tname)
+        if spec is None:
+            continue
+        if entry.get("scope", "AllReplicas") == "AllReplicas":
+            out.append((build_resource_claim(pcs, f"{sg_fqn}-{tname}", spec),
+                        entry, None))
+        else:
+            for j in range(replicas):
+                out.append((build_resource_claim(pcs, f"{sg_fqn}-{j}-{tname}", spec),
+                            entry, j))
+    return out

@@ -151,6 +151,7 @@ class Store:
         # admission chains keyed by kind
         self._mutators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
         self._validators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
+        self._delete_validators: Dict[str, List[Callable[[Obj], None]]] = {}
         self.events: List[Obj] = []
         self._events_lock = threading.Lock()
 
@@ -160,6 +161,9 @@ class Store:
 
     def register_validator(self, kind: str, fn: Callable[[Obj, Optional[Obj]], None]) -> None:
         self._validators.setdefault(kind, []).append(fn)
+
+    def register_delete_validator(self, kind: str, fn: Callable[[Obj], None]) -> None:
+        self._delete_validators.setdefault(kind, []).append(fn)
 
     # ------------------------------------------------------------------ internals
     def _table(self, kind: str) -> _KindTable:
@@ -357,6 +361,8 @@ class Store:
             obj = tbl.objects.get((ns, name))
             if obj is None:
                 raise not_found(kind, name)
+            for fn in self._delete_validators.get(kind, ()):
+                fn(obj)
             m = obj["metadata"]
             if m.get("finalizers"):
                 if not m.get("deletionTimestamp"):

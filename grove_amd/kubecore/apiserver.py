@@ -1,0 +1,246 @@
+"""HTTP API surface over the Store (fastapi) — the kube-apiserver stand-in.
+
+Gives the control plane a real wire interface: the initc waiter (grove_amd/initc.py),
+external tooling and multi-process deployments talk to the store over HTTP exactly like
+the reference's components talk to the apiserver. Supports CRUD, the status
+subresource, label selectors, and ndjson watch streams.
+
+Paths (kube-style):
+  GET/POST   /apis/{group}/{version}/namespaces/{ns}/{plural}
+  GET/PUT/DELETE /apis/{group}/{version}/namespaces/{ns}/{plural}/{name}
+  PUT        .../{name}/status
+  GET        ...?labelSelector=k=v,k2=v2
+  GET        ...?watch=true   (ndjson stream of {"type", "object"})
+Core v1 kinds use /api/v1/... ; cluster-scoped kinds omit the namespaces segment.
+"""
+import json
+import queue
+import threading
+from typing import Any, Dict, Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
+
+from .store import Store, ApiError
+
+# plural -> kind for everything the stack serves
+PLURALS: Dict[str, str] = {
+    "podcliquesets": "PodCliqueSet",
+    "podcliques": "PodClique",
+    "podcliquescalinggroups": "PodCliqueScalingGroup",
+    "podgangs": "PodGang",
+    "clustertopologybindings": "ClusterTopologyBinding",
+    "schedulertopologies": "SchedulerTopology",
+    "pods": "Pod",
+    "services": "Service",
+    "secrets": "Secret",
+    "serviceaccounts": "ServiceAccount",
+    "roles": "Role",
+    "rolebindings": "RoleBinding",
+    "horizontalpodautoscalers": "HorizontalPodAutoscaler",
+    "resourceclaims": "ResourceClaim",
+    "nodes": "Node",
+}
+CLUSTER_SCOPED_PLURALS = {"clustertopologybindings", "schedulertopologies", "nodes"}
+
+
+def parse_selector(sel: Optional[str]) -> Optional[Dict[str, str]]:
+    if not sel:
+        return None
+    out = {}
+    for part in sel.split(","):
+        if "=" in part:
+            k, v = part.split("=", 1)
+            out[k.strip()] = v.strip()
+    return out
+
+
+def build_app(store: Store):
+    app = FastAPI(title="grove-amd apiserver")
+
+    def err(e: ApiError):
+        return JSONResponse(status_code=e.code, content={
+            "kind": "Status", "status": "Failure", "reason": e.reason,
+            "message": e.message, "code": e.code})
+
+    def kind_of(plural: str) -> str:
+        kind = PLURALS.get(plural)
+        if kind is None:
+            raise ApiError(404, "NotFound", f"unknown resource {plural!r}")
+        return kind
+
+    async def handle_list_or_watch(request: Request, plural: str,
+                                   ns: Optional[str]):
+        kind = kind_of(plural)
+        params = request.query_params
+        if params.get("watch") in ("true", "1"):
+            w = store.watch(kind, seed=params.get("seed", "true") in ("true", "1"))
+
+            def stream():
+                try:
+                    while True:
+                        try:
+                            ev, obj = w.queue.get(timeout=1.0)
+                        except queue.Empty:
+                            yield "\n"  # keepalive
+                            continue
+                        yield json.dumps({"type": ev, "object": obj}) + "\n"
+                finally:
+                    w.stop()
+            return StreamingResponse(stream(), media_type="application/x-ndjson")
+        items = store.list(kind, ns, parse_selector(params.get("labelSelector")))
+        return JSONResponse({"kind": f"{kind}List", "items": items})
+
+    # ---- namespaced ----
+    @app.get("/apis/{group}/{version}/namespaces/{ns}/{plural}")
+    @app.get("/api/{version}/namespaces/{ns}/{plural}")
+    async def list_ns(request: Request, plural: str, ns: str,
+                      group: str = "", version: str = "v1"):
+        try:
+            return await handle_list_or_watch(request, plural, ns)
+        except ApiError as e:
+            return err(e)
+
+    @app.post("/apis/{group}/{version}/namespaces/{ns}/{plural}")
+    @app.post("/api/{version}/namespaces/{ns}/{plural}")
+    async def create_ns(request: Request, plural: str, ns: str,
+                        group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            obj.setdefault("metadata", {})["namespace"] = ns
+            return JSONResponse(store.create(obj), status_code=201)
+        except ApiError as e:
+            return err(e)
+
+    @app.get("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
+    @app.get("/api/{version}/namespaces/{ns}/{plural}/{name}")
+    async def get_ns(plural: str, ns: str, name: str,
+                     group: str = "", version: str = "v1"):
+        try:
+            return JSONResponse(store.get(kind_of(plural), ns, name))
+        except ApiError as e:
+            return err(e)
+
+    @app.put("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
+    @app.put("/api/{version}/namespaces/{ns}/{plural}/{name}")
+    async def update_ns(request: Request, plural: str, ns: str, name: str,
+                        group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            obj.setdefault("metadata", {})["namespace"] = ns
+            obj["metadata"]["name"] = name
+            return JSONResponse(store.update(obj))
+        except ApiError as e:
+            return err(e)
+
+    @app.put("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}/status")
+    @app.put("/api/{version}/namespaces/{ns}/{plural}/{name}/status")
+    async def update_status_ns(request: Request, plural: str, ns: str, name: str,
+                               group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            obj.setdefault("metadata", {})["namespace"] = ns
+            obj["metadata"]["name"] = name
+            return JSONResponse(store.update_status(obj))
+        except ApiError as e:
+            return err(e)
+
+    @app.delete("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
+    @app.delete("/api/{version}/namespaces/{ns}/{plural}/{name}")
+    async def delete_ns(plural: str, ns: str, name: str,
+                        group: str = "", version: str = "v1"):
+        try:
+            store.delete(kind_of(plural), ns, name)
+            return JSONResponse({"kind": "Status", "status": "Success"})
+        except ApiError as e:
+            return err(e)
+
+    # ---- cluster-scoped ----
+    @app.get("/apis/{group}/{version}/{plural}")
+    @app.get("/api/{version}/{plural}")
+    async def list_cluster(request: Request, plural: str,
+                           group: str = "", version: str = "v1"):
+        try:
+            return await handle_list_or_watch(request, plural, None)
+        except ApiError as e:
+            return err(e)
+
+    @app.post("/apis/{group}/{version}/{plural}")
+    @app.post("/api/{version}/{plural}")
+    async def create_cluster(request: Request, plural: str,
+                             group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            return JSONResponse(store.create(obj), status_code=201)
+        except ApiError as e:
+            return err(e)
+
+    @app.get("/apis/{group}/{version}/{plural}/{name}")
+    @app.get("/api/{version}/{plural}/{name}")
+    async def get_cluster(plural: str, name: str,
+                          group: str = "", version: str = "v1"):
+        try:
+            return JSONResponse(store.get(kind_of(plural), None, name))
+        except ApiError as e:
+            return err(e)
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"status": "ok"}
+
+    @app.get("/readyz")
+    async def readyz():
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    async def metrics():
+        lines = ["# TYPE grove_store_objects gauge"]
+        for kind, n in store.stats().items():
+            lines.append(f'grove_store_objects{{kind="{kind}"}} {n}')
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    return app
+
+
+class ApiServer:
+    """uvicorn in a background thread."""
+
+    def __init__(self, store: Store, host: str = "127.0.0.1", port: int = 8081):
+        self.store = store
+        self.host = host
+        self.port = port
+        self._server = None
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> "ApiServer":
+        import uvicorn
+        app = build_app(self.store)
+        config = uvicorn.Config(app, host=self.host, port=self.port,
+                                log_level="warning", lifespan="off")
+        self._server = uvicorn.Server(config)
+        self._thread = threading.Thread(target=self._server.run, daemon=True)
+        self._thread.start()
+        import time
+        import urllib.request
+        for _ in range(200):
+            try:
+                urllib.request.urlopen(
+                    f"http://{self.host}:{self.port}/healthz", timeout=0.2)
+                return self
+            except Exception:
+                time.sleep(0.05)
+        raise RuntimeError("apiserver failed to start")
+
+    @property
+    def url(self) -> str:
+        return f"http://{self.host}:{self.port}"
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.should_exit = True
+        if self._thread is not None:
+            self._thread.join(timeout=5)

@@ -1,0 +1,129 @@
+"""HTTP apiserver + grove-initc waiter + operator config tests."""
+import json
+import threading
+import time
+import urllib.request
+
+import pytest
+
+from grove_amd.api import constants as c
+
+
+@pytest.fixture()
+def served_cluster(cluster):
+    from grove_amd.kubecore.apiserver import ApiServer
+    api = ApiServer(cluster.store, port=18133)
+    api.start()
+    yield cluster, api
+    api.stop()
+
+
+def _get(url):
+    with urllib.request.urlopen(url, timeout=5) as r:
+        return json.loads(r.read())
+
+
+def test_apiserver_crud_and_selectors(served_cluster, simple1_yaml):
+    cluster, api = served_cluster
+    cluster.add_virtual_nodes(2)
+    # create a PCS over HTTP
+    import yaml as _yaml
+    pcs = list(_yaml.safe_load_all(simple1_yaml))[0]
+    req = urllib.request.Request(
+        f"{api.url}/apis/grove.io/v1alpha1/namespaces/default/podcliquesets",
+        data=json.dumps(pcs).encode(), method="POST",
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=5) as r:
+        assert r.status == 201
+    cluster.wait_pcs_available("simple1", timeout=20)
+    # list pods with label selector over HTTP
+    out = _get(f"{api.url}/api/v1/namespaces/default/pods"
+               f"?labelSelector={c.LABEL_PODCLIQUE}=simple1-0-pca")
+    assert len(out["items"]) == 3
+    # get a single PCLQ
+    q = _get(f"{api.url}/apis/grove.io/v1alpha1/namespaces/default/"
+             f"podcliques/simple1-0-pca")
+    assert q["status"]["readyReplicas"] == 3
+    # cluster-scoped list
+    nodes = _get(f"{api.url}/api/v1/nodes")
+    assert len(nodes["items"]) == 2
+    # health + metrics
+    assert _get(f"{api.url}/healthz")["status"] == "ok"
+    with urllib.request.urlopen(f"{api.url}/metrics", timeout=5) as r:
+        assert b"grove_store_objects" in r.read()
+    # 404 surface
+    try:
+        _get(f"{api.url}/apis/grove.io/v1alpha1/namespaces/default/podcliques/nope")
+        assert False
+    except urllib.error.HTTPError as e:
+        assert e.code == 404
+
+
+def test_initc_waits_for_parent_clique(served_cluster):
+    cluster, api = served_cluster
+    from grove_amd.initc import wait_for_parents
+    cluster.add_virtual_nodes(1)
+    pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+           "metadata": {"name": "ic"},
+           "spec": {"replicas": 1, "template": {
+               "cliqueStartupType": c.STARTUP_EXPLICIT,
+               "cliques": [
+                   {"name": "a", "spec": {"roleName": "a", "replicas": 2,
+                                          "podSpec": {"containers": [{"name": "m",
+                                                                     "image": "i"}]}}},
+                   {"name": "b", "spec": {"roleName": "b", "startsAfter": ["a"],
+                                          "podSpec": {"containers": [{"name": "m",
+                                                                     "image": "i"}]}}},
+               ]}}}
+    results = {}
+
+    def waiter():
+        results["ok"] = wait_for_parents(
+            "default", "ic-0", [("ic-0-a", 2)], server=api.url, timeout=20,
+            poll=0.05)
+    t = threading.Thread(target=waiter)
+    t.start()
+    time.sleep(0.2)
+    assert "ok" not in results  # blocked before workload exists
+    cluster.store.create(pcs)
+    t.join(timeout=25)
+    assert results.get("ok") is True
+
+
+def test_initc_cli_parse():
+    from grove_amd.initc import parse_podcliques
+    assert parse_podcliques(["x-0-a:2", "x-0-b:1"]) == [("x-0-a", 2), ("x-0-b", 1)]
+    with pytest.raises(ValueError):
+        parse_podcliques(["nope"])
+
+
+def test_operator_config_load(tmp_path):
+    from grove_amd.config import load_configuration
+    from grove_amd.kubecore.store import ApiError
+    p = tmp_path / "cfg.yaml"
+    p.write_text("""
+client: {qps: 200, burst: 300}
+controllers:
+  podCliqueSet: {concurrentSyncs: 8}
+servers:
+  api: {enabled: true, port: 9999}
+authorizer: {enabled: false}
+network: {autoXGMIDomainEnabled: true}
+scheduler: {default: amd-gang-scheduler}
+logLevel: debug
+""")
+    cfg = load_configuration(str(p))
+    assert cfg.client_qps == 200 and cfg.concurrent_syncs("podCliqueSet") == 8
+    assert cfg.api_server.enabled and cfg.api_server.port == 9999
+    assert not cfg.authorizer_enabled and cfg.auto_xgmi_domain_enabled
+    # invalid scheduler rejected
+    p.write_text("scheduler: {default: volcano}")
+    with pytest.raises(ApiError):
+        load_configuration(str(p))
+
+
+def test_default_config():
+    from grove_amd.config import default_configuration
+    cfg = default_configuration()
+    assert cfg.default_scheduler == c.SCHEDULER_AMD_GANG
+    assert cfg.authorizer_enabled and cfg.topology_aware_scheduling_enabled

@@ -144,7 +144,7 @@ class Cluster:
             if ev != "DELETED":
                 self.c_podgang.enqueue(ns, md["name"])
             # gate-removal re-check for every member clique
-            for group in (obj.get("spec") or {}).get("podGroups") or []:
+            for group in (obj.get("spec") or {}).get("podgroups") or []:
                 self.c_pclq.enqueue(ns, group.get("name", ""))
             base = md.get("labels", {}).get(c.LABEL_BASE_PODGANG)
             if base:

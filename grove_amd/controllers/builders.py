@@ -346,7 +346,7 @@ def build_podgang(pcs: Obj, name: str, scheduler_name: str,
     }
     if base_podgang:
         labels[c.LABEL_BASE_PODGANG] = base_podgang
-    spec: Dict[str, Any] = {"podGroups": []}
+    spec: Dict[str, Any] = {"podgroups": []}
     if priority_class:
         spec["priorityClassName"] = priority_class
     return {

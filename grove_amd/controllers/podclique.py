@@ -162,7 +162,7 @@ class PodCliqueReconciler:
         if podgang is None:
             return
         refs = set()
-        for group in (podgang.get("spec") or {}).get("podGroups") or []:
+        for group in (podgang.get("spec") or {}).get("podgroups") or []:
             for ref in group.get("podReferences") or []:
                 refs.add(ref.get("name") if isinstance(ref, dict) else ref)
 
@@ -196,7 +196,7 @@ class PodCliqueReconciler:
         base = self.store.try_get(c.KIND_PODGANG, ns, base_name)
         if base is None:
             return False
-        groups = (base.get("spec") or {}).get("podGroups") or []
+        groups = (base.get("spec") or {}).get("podgroups") or []
         if not groups:
             return False
         for group in groups:
@@ -235,6 +235,8 @@ class PodCliqueReconciler:
             st["scheduleGatedReplicas"] = n_gated
             st["updatedReplicas"] = n_updated
             st["observedGeneration"] = o["metadata"].get("generation")
+            st["currentPodTemplateHash"] = tmpl_hash
+            st["hpaPodSelector"] = f"{c.LABEL_PODCLIQUE}={name}"
             # PodCliqueScheduled (reconcilestatus.go:282)
             if n_sched >= min_avail:
                 cond.set_condition(o, c.COND_PODCLIQUE_SCHEDULED, True,

@@ -164,7 +164,7 @@ class PCSGReconciler:
     def _replica_selected_for_update(pcs: Obj, r: int) -> bool:
         if (pcs["spec"].get("updateStrategy") or {}).get("type") == c.UPDATE_ON_DELETE:
             return True  # OnDelete: spec propagates immediately, pods wait for the user
-        prog = (pcs.get("status") or {}).get("rollingUpdateProgress")
+        prog = (pcs.get("status") or {}).get("updateProgress")
         if prog is None:
             return True
         cu = prog.get("currentlyUpdating")
@@ -242,6 +242,7 @@ class PCSGReconciler:
             st["scheduledReplicas"] = sched
             st["availableReplicas"] = avail
             st["observedGeneration"] = o["metadata"].get("generation")
+            st["selector"] = f"{c.LABEL_PCSG}={name}"
             ever = bool(st.get("everAvailable")) or avail >= min_avail
             st["everAvailable"] = ever
             if ever and avail < min_avail:

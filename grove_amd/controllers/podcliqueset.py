@@ -124,7 +124,7 @@ class PodCliqueSetReconciler:
         def upd(o: Obj) -> None:
             s = o.setdefault("status", {})
             if s.get("currentGenerationHash") and s.get("currentGenerationHash") != new_hash:
-                s["rollingUpdateProgress"] = {
+                s["updateProgress"] = {
                     "updateStartedAt": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
                     "currentlyUpdating": None,
                     "updatedReplicas": [],
@@ -253,7 +253,7 @@ class PodCliqueSetReconciler:
     def _replica_selected_for_update(self, pcs: Obj, r: int) -> bool:
         if (pcs["spec"].get("updateStrategy") or {}).get("type") == c.UPDATE_ON_DELETE:
             return True  # OnDelete: spec propagates immediately, pods wait for the user
-        prog = (pcs.get("status") or {}).get("rollingUpdateProgress")
+        prog = (pcs.get("status") or {}).get("updateProgress")
         if prog is None:
             return True  # no update in flight → initial create path
         cu = prog.get("currentlyUpdating")
@@ -327,7 +327,7 @@ class PodCliqueSetReconciler:
             if ridx is not None and ridx.isdigit():
                 by_replica.setdefault(int(ridx), [])
 
-        cu = ((pcs.get("status") or {}).get("rollingUpdateProgress") or {}) \
+        cu = ((pcs.get("status") or {}).get("updateProgress") or {}) \
             .get("currentlyUpdating") or {}
         updating_replica = int(cu.get("replicaIndex", -1))
         for ridx, constituents in by_replica.items():
@@ -384,7 +384,7 @@ class PodCliqueSetReconciler:
         ns = pcs["metadata"].get("namespace", "default")
         name = pcs["metadata"]["name"]
         st = pcs.get("status") or {}
-        prog = st.get("rollingUpdateProgress")
+        prog = st.get("updateProgress")
         if not prog or prog.get("updateEndedAt"):
             return
         if (pcs["spec"].get("updateStrategy") or {}).get("type") == c.UPDATE_ON_DELETE:
@@ -444,12 +444,12 @@ class PodCliqueSetReconciler:
             if not replica_updated(r):
                 return  # still updating this replica
             def done(o: Obj) -> None:
-                p = o["status"].get("rollingUpdateProgress") or {}
+                p = o["status"].get("updateProgress") or {}
                 ur = p.setdefault("updatedReplicas", [])
                 if r not in ur:
                     ur.append(r)
                 p["currentlyUpdating"] = None
-                o["status"]["rollingUpdateProgress"] = p
+                o["status"]["updateProgress"] = p
             try:
                 self.store.patch(c.KIND_PCS, ns, name, done, status=True)
             except ApiError:
@@ -462,10 +462,10 @@ class PodCliqueSetReconciler:
                    and not replica_updated(r)]
         if not pending:
             def finish(o: Obj) -> None:
-                p = o["status"].get("rollingUpdateProgress") or {}
+                p = o["status"].get("updateProgress") or {}
                 p["updateEndedAt"] = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
                 p["currentlyUpdating"] = None
-                o["status"]["rollingUpdateProgress"] = p
+                o["status"]["updateProgress"] = p
             try:
                 self.store.patch(c.KIND_PCS, ns, name, finish, status=True)
             except ApiError:
@@ -481,12 +481,12 @@ class PodCliqueSetReconciler:
         target = sorted(pending, key=order)[0]
 
         def select(o: Obj) -> None:
-            p = o["status"].get("rollingUpdateProgress") or {}
+            p = o["status"].get("updateProgress") or {}
             p["currentlyUpdating"] = {
                 "replicaIndex": target,
                 "startedAt": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
             }
-            o["status"]["rollingUpdateProgress"] = p
+            o["status"]["updateProgress"] = p
         try:
             self.store.patch(c.KIND_PCS, ns, name, select, status=True)
         except ApiError:
@@ -536,14 +536,14 @@ class PodCliqueSetReconciler:
 
         # rolling-update bookkeeping for updatedReplicas count
         st = pcs.get("status") or {}
-        prog = st.get("rollingUpdateProgress")
+        prog = st.get("updateProgress")
         updated = len((prog or {}).get("updatedReplicas") or []) if prog else replicas
 
         # per-gang phase rollup (podcliqueset.go PodGangStatus)
         gang_statuses: List[Dict[str, Any]] = []
         for pg in self.store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: name}):
             phase = "Pending"
-            groups = (pg.get("spec") or {}).get("podGroups") or []
+            groups = (pg.get("spec") or {}).get("podgroups") or []
             if groups:
                 pclq_of = {q["metadata"]["name"]: q for q in pclqs}
                 sched = all(

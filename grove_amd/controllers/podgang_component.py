@@ -189,7 +189,7 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
             obj = builders.build_podgang(pcs, gang.name, scheduler_name,
                                          base_podgang=gang.base,
                                          priority_class=priority_class)
-            obj["spec"]["podGroups"] = groups_spec
+            obj["spec"]["podgroups"] = groups_spec
             if gang.constraint:
                 obj["spec"]["topologyConstraint"] = gang.constraint
             if gang.group_configs:
@@ -203,12 +203,12 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
                     continue
 
         def upd(o: Obj) -> None:
-            o["spec"]["podGroups"] = groups_spec
+            o["spec"]["podgroups"] = groups_spec
             if gang.constraint:
                 o["spec"]["topologyConstraint"] = gang.constraint
             if gang.group_configs:
                 o["spec"]["topologyConstraintGroupConfigs"] = gang.group_configs
-        if cur["spec"].get("podGroups") != groups_spec \
+        if cur["spec"].get("podgroups") != groups_spec \
                 or cur["spec"].get("topologyConstraint") != gang.constraint:
             try:
                 cur = store.patch(c.KIND_PODGANG, ns, gang.name, upd)

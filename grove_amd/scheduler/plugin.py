@@ -165,7 +165,7 @@ class GangScheduler:
     def _gang_nodes(self, pg: Obj) -> Set[str]:
         ns = pg["metadata"].get("namespace", "default")
         names: Set[str] = set()
-        for group in (pg.get("spec") or {}).get("podGroups") or []:
+        for group in (pg.get("spec") or {}).get("podgroups") or []:
             for ref in group.get("podReferences") or []:
                 p = self.store.try_get("Pod", ns, ref.get("name", ""))
                 if p and p.get("spec", {}).get("nodeName"):
@@ -175,7 +175,7 @@ class GangScheduler:
     # ------------------------------------------------------------------ gang place
     def _schedule_gang(self, nodes: List[NodeFree], pg: Obj, gang_pods: List[Obj]) -> None:
         ns = pg["metadata"].get("namespace", "default")
-        groups = (pg.get("spec") or {}).get("podGroups") or []
+        groups = (pg.get("spec") or {}).get("podgroups") or []
         by_clique: Dict[str, List[Obj]] = {}
         for p in gang_pods:
             by_clique.setdefault(
@@ -357,7 +357,7 @@ class GangScheduler:
                 continue
             ns = pg["metadata"].get("namespace", "default")
             ready = True
-            for group in (pg.get("spec") or {}).get("podGroups") or []:
+            for group in (pg.get("spec") or {}).get("podgroups") or []:
                 n = 0
                 for ref in group.get("podReferences") or []:
                     p = self.store.try_get("Pod", ns, ref.get("name", ""))

@@ -124,7 +124,7 @@ class TestRollingUpdate:
 
         def updated():
             p = cluster.store.get(c.KIND_PCS, "default", "ru1")
-            prog = (p.get("status") or {}).get("rollingUpdateProgress")
+            prog = (p.get("status") or {}).get("updateProgress")
             return (prog and prog.get("updateEndedAt")
                     and p["status"].get("updatedReplicas") == 2)
         cluster.wait_for(updated, timeout=30, desc="rolling update complete")
@@ -179,7 +179,7 @@ class TestRollingUpdate:
         deadline = time.monotonic() + 30
         while time.monotonic() < deadline:
             p = cluster.store.get(c.KIND_PCS, "default", "ru3")
-            prog = (p.get("status") or {}).get("rollingUpdateProgress") or {}
+            prog = (p.get("status") or {}).get("updateProgress") or {}
             cu = prog.get("currentlyUpdating")
             if cu is not None:
                 max_concurrent = max(max_concurrent, 1)
@@ -187,6 +187,6 @@ class TestRollingUpdate:
                 break
             time.sleep(0.01)
         assert (cluster.store.get(c.KIND_PCS, "default", "ru3")["status"]
-                ["rollingUpdateProgress"].get("updateEndedAt"))
+                ["updateProgress"].get("updateEndedAt"))
         pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "ru3"})
         assert all(p["spec"]["containers"][0]["image"] == "img:v2" for p in pods)

@@ -70,10 +70,13 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
     const long row0 = (long)tile_m * BM;
     const long col0 = (long)tile_n * BN;
 
-    // Staging: each thread loads 16 B per piece; 256 threads cover BM*BK*2 = 16 KB in
-    // 4 pieces each. thread t handles LDS piece index p = t + i*256; row = p/8,
-    // cb_slot = p%8; source cb = cb_slot ^ (row&7) (swizzle applied on the SOURCE
-    // address, LDS stays lane-linear — guide §5 rule 21).
+    // Staging via 16-B LDS-DMA (global_load_lds, guide §5 ladder step 3): thread t
+    // covers LDS piece p = t + i*256 (row = p/8, cb_slot = p%8). The glds LDS
+    // destination is wave-uniform-base + lane*16 and our LDS image is lane-linear in
+    // p, so the XOR swizzle is applied on the SOURCE address (cb_src = cb_slot ^
+    // (row&7)) while LDS stays linear — guide §5 rule 21. Compute on buf overlaps the
+    // DMA into buf^1; __syncthreads() drains it (vmcnt(0)) each K-tile.
+    const int wave_piece0 = (tid / WAVE) * WAVE;  // wave-uniform piece base
     auto stage = [&](int buf, long kk) {
         const bf16* gA = A + row0 * K + kk;
         const bf16* gB = Bt + col0 * K + kk;
@@ -83,12 +86,18 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
         for (int i = 0; i < 4; ++i) {
             int p = tid + i * 256;
             int row = p >> 3;
-            int cb_slot = p & 7;
-            int cb_src = cb_slot ^ (row & 7);
-            short8 va = *(const short8*)(gA + (long)row * K + (cb_src << 3));
-            *(short8*)(dA + row * BK + (cb_slot << 3)) = va;
-            short8 vb = *(const short8*)(gB + (long)row * K + (cb_src << 3));
-            *(short8*)(dB + row * BK + (cb_slot << 3)) = vb;
+            int cb_src = (p & 7) ^ (row & 7);
+            int base = wave_piece0 + i * 256;  // uniform across the wave
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)(
+                    gA + (long)row * K + (cb_src << 3)),
+                (__attribute__((address_space(3))) uint32_t*)(dA + base * 8),
+                16, 0, 0);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)(
+                    gB + (long)row * K + (cb_src << 3)),
+                (__attribute__((address_space(3))) uint32_t*)(dB + base * 8),
+                16, 0, 0);
         }
     };
 

@@ -136,6 +136,8 @@ class Controller:
         self.queue = WorkQueue()
         self._threads: List[threading.Thread] = []
         self._stop = threading.Event()
+        self.reconcile_count = 0
+        self.reconcile_seconds = 0.0
 
     def enqueue(self, namespace: str, name: str) -> None:
         self.queue.add((namespace, name))
@@ -158,7 +160,10 @@ class Controller:
                 continue
             ns, name = item
             try:
+                _t0 = time.monotonic()
                 res = self.reconcile(ns, name)
+                self.reconcile_seconds += time.monotonic() - _t0
+                self.reconcile_count += 1
                 self.queue.forget(item)
                 if res is not None and res.requeue_after is not None:
                     self.queue.add_after(item, res.requeue_after)

@@ -29,6 +29,17 @@ from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
 
 Obj = Dict[str, Any]
 
+
+def json_copy(obj):
+    """Deep copy for JSON-shaped objects (dict/list/scalars) — ~4x faster than
+    copy.deepcopy, which pays for memoization and class dispatch we never need."""
+    t = type(obj)
+    if t is dict:
+        return {k: json_copy(v) for k, v in obj.items()}
+    if t is list:
+        return [json_copy(v) for v in obj]
+    return obj
+
 ADDED = "ADDED"
 MODIFIED = "MODIFIED"
 DELETED = "DELETED"
@@ -173,7 +184,7 @@ class Store:
         return tbl
 
     def _notify(self, tbl: _KindTable, ev: str, obj: Obj) -> None:
-        snapshot = copy.deepcopy(obj)
+        snapshot = json_copy(obj)
         for w in list(tbl.watchers):
             w.queue.put((ev, snapshot))
 
@@ -188,7 +199,7 @@ class Store:
 
     # ------------------------------------------------------------------ CRUD
     def create(self, obj: Obj) -> Obj:
-        obj = copy.deepcopy(obj)
+        obj = json_copy(obj)
         kind = obj.get("kind")
         if not kind:
             raise invalid("object has no kind")
@@ -219,7 +230,7 @@ class Store:
             tbl.index_add(key, obj)
             self._uid_index[m["uid"]] = (kind, ns, m["name"])
             self._notify(tbl, ADDED, obj)
-        return copy.deepcopy(obj)
+        return json_copy(obj)
 
     def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
         ns = "" if kind in CLUSTER_SCOPED else (namespace or "default")
@@ -228,7 +239,7 @@ class Store:
             obj = tbl.objects.get((ns, name))
             if obj is None:
                 raise not_found(kind, name)
-            return copy.deepcopy(obj)
+            return json_copy(obj)
 
     def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
         try:
@@ -276,11 +287,11 @@ class Store:
                     continue
                 if filter_fn is not None and not filter_fn(obj):
                     continue
-                out.append(copy.deepcopy(obj) if copy_objects else obj)
+                out.append(json_copy(obj) if copy_objects else obj)
             return out
 
     def _apply_update(self, obj: Obj, status_only: bool) -> Obj:
-        obj = copy.deepcopy(obj)
+        obj = json_copy(obj)
         kind = obj["kind"]
         m = meta(obj)
         ns = self._ns_of(kind, m)
@@ -295,9 +306,9 @@ class Store:
                 raise conflict(kind, m["name"])
             if status_only:
                 if cur.get("status", {}) == obj.get("status", {}):
-                    return copy.deepcopy(cur)  # no-op: no rv bump, no event
-                new = copy.deepcopy(cur)
-                new["status"] = copy.deepcopy(obj.get("status", {}))
+                    return json_copy(cur)  # no-op: no rv bump, no event
+                new = json_copy(cur)
+                new["status"] = json_copy(obj.get("status", {}))
             else:
                 # admission on spec/metadata updates
                 for fn in self._mutators.get(kind, ()):
@@ -310,14 +321,14 @@ class Store:
                     new["metadata"][f] = cur_m[f]
                 new["metadata"]["namespace"] = cur_m.get("namespace", "")
                 if cur.get("status") is not None and "status" not in new:
-                    new["status"] = copy.deepcopy(cur["status"])
+                    new["status"] = json_copy(cur["status"])
                 else:
-                    new["status"] = copy.deepcopy(cur.get("status", {}))
+                    new["status"] = json_copy(cur.get("status", {}))
                 if cur_m.get("deletionTimestamp"):
                     new["metadata"]["deletionTimestamp"] = cur_m["deletionTimestamp"]
                 new["metadata"]["resourceVersion"] = cur_m["resourceVersion"]
                 if new == cur:
-                    return copy.deepcopy(cur)  # no-op update: no rv bump, no event
+                    return json_copy(cur)  # no-op update: no rv bump, no event
                 if new.get("spec") != cur.get("spec"):
                     new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
             new["metadata"]["resourceVersion"] = self._next_rv()
@@ -329,7 +340,7 @@ class Store:
             # finalizer removal on a deleting object may allow actual deletion
             if new["metadata"].get("deletionTimestamp") and not new["metadata"].get("finalizers"):
                 self._finalize_delete(kind, ns, m["name"])
-            return copy.deepcopy(new)
+            return json_copy(new)
 
     def update(self, obj: Obj) -> Obj:
         return self._apply_update(obj, status_only=False)
@@ -424,7 +435,7 @@ class Store:
             tbl.watchers.append(w)
             if seed:
                 for obj in tbl.objects.values():
-                    w.queue.put((ADDED, copy.deepcopy(obj)))
+                    w.queue.put((ADDED, json_copy(obj)))
         return w
 
     def record_event(self, involved: Obj, etype: str, reason: str, message: str) -> None:

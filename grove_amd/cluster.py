@@ -175,6 +175,21 @@ class Cluster:
         m.watch(c.KIND_PODGANG, on_podgang)
         m.watch("Node", on_node)
 
+    def metrics_lines(self):
+        """Prometheus-style controller metrics (served by the apiserver /metrics)."""
+        lines = ["# TYPE grove_reconcile_total counter",
+                 "# TYPE grove_reconcile_seconds_total counter"]
+        for ctrl in self.manager.controllers:
+            lines.append(
+                f'grove_reconcile_total{{controller="{ctrl.name}"}} '
+                f'{ctrl.reconcile_count}')
+            lines.append(
+                f'grove_reconcile_seconds_total{{controller="{ctrl.name}"}} '
+                f'{ctrl.reconcile_seconds:.6f}')
+        lines.append("# TYPE grove_events_total counter")
+        lines.append(f"grove_events_total {len(self.store.events)}")
+        return lines
+
     # ------------------------------------------------------------------ lifecycle
     def start(self) -> "Cluster":
         if not self._started:

@@ -177,6 +177,11 @@ class PodCliqueSetReconciler:
                 fqn = namegen.podclique_name(name, r, cl["name"])
                 expected_pclqs.add(fqn)
                 refs = resourceclaims.claim_refs_for_clique(claims, cl["name"])
+                cl_claims = resourceclaims.clique_level_claims(
+                    self.store, pcs, r, cl, fqn)
+                resourceclaims.ensure_claims(self.store, cl_claims)
+                refs = refs + resourceclaims.claim_refs_for_clique(
+                    cl_claims, cl["name"])
                 self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs)
         # GC excess standalone PCLQs (scale-in / replica removal)
         for pclq in self.store.list(c.KIND_PCLQ, ns, {

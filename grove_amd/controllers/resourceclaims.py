@@ -47,7 +47,10 @@ def resolve_templates(pcs: Obj) -> Dict[str, Obj]:
     out: Dict[str, Obj] = {}
     for t in pcs["spec"]["template"].get("resourceClaimTemplates") or []:
         if t.get("name"):
-            out[t["name"]] = t.get("spec") or {}
+            # reference field is templateSpec (ResourceClaimTemplateSpec); accept the
+            # bare spec too
+            ts = t.get("templateSpec") or {}
+            out[t["name"]] = ts.get("spec") or t.get("spec") or {}
     return out
 
 
@@ -117,6 +120,26 @@ def claim_refs_for_clique(claims: List[Tuple[Obj, Obj]],
             refs.append({"name": tname,
                          "resourceClaimName": claim["metadata"]["name"]})
     return refs
+
+
+def clique_level_claims(store: Store, pcs: Obj, r: int,
+                        clique_tmpl: Obj, pclq_fqn: str) -> List[Tuple[Obj, Obj]]:
+    """Clique-level sharing (podclique/components/resourceclaim parity): AllReplicas =
+    one claim shared by this clique across all PCS replicas; PerReplica = one claim per
+    clique instance."""
+    templates = resolve_templates(pcs)
+    out: List[Tuple[Obj, Obj]] = []
+    for entry in clique_tmpl.get("resourceSharing") or []:
+        tname = entry.get("templateRef") or entry.get("name")
+        spec = templates.get(tname)
+        if spec is None:
+            continue
+        if entry.get("scope", "AllReplicas") == "AllReplicas":
+            cname = f"{pcs['metadata']['name']}-{clique_tmpl['name']}-{tname}"
+        else:
+            cname = f"{pclq_fqn}-{tname}"
+        out.append((build_resource_claim(pcs, cname, spec), entry))
+    return out
 
 
 def pcsg_claims(pcs: Obj, sg_cfg: Obj, sg_fqn: str, replicas: int

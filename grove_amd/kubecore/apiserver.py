@@ -55,7 +55,7 @@ def parse_selector(sel: Optional[str]) -> Optional[Dict[str, str]]:
     return out
 
 
-def build_app(store: Store):
+def build_app(store: Store, metrics_fn=None):
     app = FastAPI(title="grove-amd apiserver")
 
     def err(e: ApiError):
@@ -201,6 +201,8 @@ def build_app(store: Store):
         lines = ["# TYPE grove_store_objects gauge"]
         for kind, n in store.stats().items():
             lines.append(f'grove_store_objects{{kind="{kind}"}} {n}')
+        if metrics_fn is not None:
+            lines.extend(metrics_fn())
         return PlainTextResponse("\n".join(lines) + "\n")
 
     return app
@@ -209,27 +211,38 @@ def build_app(store: Store):
 class ApiServer:
     """uvicorn in a background thread."""
 
-    def __init__(self, store: Store, host: str = "127.0.0.1", port: int = 8081):
+    def __init__(self, store: Store, host: str = "127.0.0.1", port: int = 8081,
+                 metrics_fn=None, ssl_certfile: Optional[str] = None,
+                 ssl_keyfile: Optional[str] = None):
         self.store = store
         self.host = host
         self.port = port
+        self.metrics_fn = metrics_fn
+        self.ssl_certfile = ssl_certfile
+        self.ssl_keyfile = ssl_keyfile
         self._server = None
         self._thread: Optional[threading.Thread] = None
 
     def start(self) -> "ApiServer":
         import uvicorn
-        app = build_app(self.store)
+        app = build_app(self.store, self.metrics_fn)
         config = uvicorn.Config(app, host=self.host, port=self.port,
-                                log_level="warning", lifespan="off")
+                                log_level="warning", lifespan="off",
+                                ssl_certfile=self.ssl_certfile,
+                                ssl_keyfile=self.ssl_keyfile)
         self._server = uvicorn.Server(config)
         self._thread = threading.Thread(target=self._server.run, daemon=True)
         self._thread.start()
         import time
         import urllib.request
+        scheme = "https" if self.ssl_certfile else "http"
+        import ssl as _ssl
+        ctx = _ssl._create_unverified_context() if self.ssl_certfile else None
         for _ in range(200):
             try:
                 urllib.request.urlopen(
-                    f"http://{self.host}:{self.port}/healthz", timeout=0.2)
+                    f"{scheme}://{self.host}:{self.port}/healthz", timeout=0.2,
+                    context=ctx)
                 return self
             except Exception:
                 time.sleep(0.05)

@@ -1,0 +1,76 @@
+"""Soak/churn benchmark — reference parity: Makefile run-soak-test / soak-churn.yaml
+(create N PCS → scale to 2N → delete, cycled; artifacts for regression comparison) and
+cert bootstrap coverage."""
+import time
+
+import pytest
+
+from grove_amd.api import constants as c
+
+
+def small_pcs(name):
+    return {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+            "metadata": {"name": name},
+            "spec": {"replicas": 1, "template": {"cliques": [{
+                "name": "w", "spec": {"roleName": "w", "replicas": 2,
+                                      "podSpec": {"containers": [{
+                                          "name": "m", "image": "i",
+                                          "resources": {"requests": {"cpu": "100m"}}}]}},
+            }]}}}
+
+
+@pytest.mark.timeout(300)
+def test_soak_churn_no_leaks(cluster):
+    """5 cycles of create-5 → scale-to-2-replicas → delete-all; the store must return
+    to baseline every cycle (no leaked pods/PCLQs/gangs/services)."""
+    cluster.add_virtual_nodes(10, cpu="64", pods=256)
+    baseline = {k: v for k, v in cluster.store.stats().items()}
+    timings = []
+    for cycle in range(5):
+        t0 = time.monotonic()
+        for i in range(5):
+            cluster.store.create(small_pcs(f"soak-{cycle}-{i}"))
+        for i in range(5):
+            cluster.wait_pcs_available(f"soak-{cycle}-{i}", timeout=30)
+        # scale up
+        for i in range(5):
+            cluster.store.patch(c.KIND_PCS, "default", f"soak-{cycle}-{i}",
+                                lambda o: o["spec"].update(replicas=2))
+            cluster.c_pcs.enqueue("default", f"soak-{cycle}-{i}")
+        for i in range(5):
+            cluster.wait_pcs_available(f"soak-{cycle}-{i}", timeout=30)
+        # delete
+        for i in range(5):
+            cluster.delete_pcs(f"soak-{cycle}-{i}")
+        for i in range(5):
+            cluster.wait_deleted(c.KIND_PCS, f"soak-{cycle}-{i}", timeout=30)
+        timings.append(time.monotonic() - t0)
+        stats = cluster.store.stats()
+        for kind in (c.KIND_PCS, c.KIND_PCLQ, c.KIND_PCSG, c.KIND_PODGANG, "Pod",
+                     "Service", "HorizontalPodAutoscaler", "Secret",
+                     "ServiceAccount"):
+            assert stats.get(kind, 0) == baseline.get(kind, 0), \
+                f"cycle {cycle}: leaked {kind}: {stats.get(kind)} vs {baseline.get(kind)}"
+    print(f"\nsoak cycles: {['%.1fs' % t for t in timings]}")
+    # no degradation trend: last cycle within 3x of first
+    assert timings[-1] < max(3 * timings[0], timings[0] + 2.0)
+
+
+def test_cert_bootstrap(cluster, tmp_path):
+    from grove_amd.kubecore.certs import (ensure_cert_secret, write_cert_files,
+                                          CERT_SECRET_NAME)
+    data = ensure_cert_secret(cluster.store)
+    assert "BEGIN CERTIFICATE" in data["tls.crt"]
+    assert "PRIVATE KEY" in data["tls.key"]
+    # idempotent: second call reuses the Secret
+    again = ensure_cert_secret(cluster.store)
+    assert again["tls.crt"] == data["tls.crt"]
+    sec = cluster.store.get("Secret", "grove-system", CERT_SECRET_NAME)
+    assert sec["type"] == "kubernetes.io/tls"
+    crt, key = write_cert_files(data, str(tmp_path / "certs"))
+    import os
+    assert os.path.exists(crt) and os.path.exists(key)
+    # manual mode with missing secret fails loudly
+    from grove_amd.kubecore.store import Store, ApiError
+    with pytest.raises(ApiError):
+        ensure_cert_secret(Store(), mode="manual")

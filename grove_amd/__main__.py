@@ -72,6 +72,20 @@ def cmd_operator(args) -> int:
     return 0
 
 
+def cmd_install_crds(args) -> int:
+    from .api.crds import install_crds, write_crds
+    if args.output_dir:
+        for p in write_crds(args.output_dir):
+            print("wrote", p)
+    if args.server:
+        print(f"applied {install_crds(args.server)} CRDs to {args.server}")
+    if not args.output_dir and not args.server:
+        from .api.crds import render_all
+        import yaml as _y
+        print(_y.safe_dump_all(render_all(), sort_keys=False))
+    return 0
+
+
 def cmd_version(_args) -> int:
     print(f"grove-amd {__version__}")
     return 0
@@ -90,6 +104,11 @@ def main(argv=None) -> int:
     op.add_argument("--virtual-nodes", type=int, default=0)
     op.add_argument("--virtual-gpus", type=int, default=8)
     op.set_defaults(fn=cmd_operator)
+
+    crds = sub.add_parser("install-crds", help="render or apply the CRDs")
+    crds.add_argument("--server", default=None, help="apiserver URL to POST CRDs to")
+    crds.add_argument("--output-dir", default=None, help="write CRD YAML files here")
+    crds.set_defaults(fn=cmd_install_crds)
 
     ver = sub.add_parser("version")
     ver.set_defaults(fn=cmd_version)

@@ -359,5 +359,5 @@ def build_podgang(pcs: Obj, name: str, scheduler_name: str,
             "ownerReferences": [owner_reference(pcs)],
         },
         "spec": spec,
-        "status": {"conditions": []},
+        "status": {"conditions": [], "phase": "Pending"},
     }

@@ -15,6 +15,7 @@ from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
 from ..utils.indexing import available_indices
+from ..utils.concurrent import run_concurrently_with_slow_start
 from . import builders
 from .manager import Result
 
@@ -96,9 +97,23 @@ class PodCliqueReconciler:
             if pcs is None:
                 return Result(requeue_after=0.1)
             num_pods = self._pcsg_template_num_pods(pcs, pclq)
-            for idx in available_indices([i for i in in_use if i >= 0], desired - n):
-                pod = builders.build_pod(pcs, pclq, idx, self.scheduler_name, num_pods)
-                self.store.create(pod)
+            indices = available_indices([i for i in in_use if i >= 0], desired - n)
+            if len(indices) <= 2:
+                for idx in indices:
+                    pod = builders.build_pod(pcs, pclq, idx, self.scheduler_name,
+                                             num_pods)
+                    self.store.create(pod)
+            else:
+                # slow-start batches (1,2,4,...) — utils/concurrent.go parity: a
+                # systematic create failure is found after one cheap attempt
+                def mk(idx):
+                    return lambda: self.store.create(
+                        builders.build_pod(pcs, pclq, idx, self.scheduler_name,
+                                           num_pods))
+                errs = run_concurrently_with_slow_start(
+                    [(f"create-pod-{i}", mk(i)) for i in indices])
+                if errs:
+                    raise errs[0]
         elif n > desired:
             for p in self._deletion_order(pods)[: n - desired]:
                 try:

@@ -188,6 +188,20 @@ def build_app(store: Store, metrics_fn=None):
         except ApiError as e:
             return err(e)
 
+    @app.get("/debug/stacks")
+    async def debug_stacks():
+        import sys
+        import traceback
+        out = []
+        for tid, frame in sys._current_frames().items():
+            out.append(f"--- thread {tid} ---")
+            out.extend(l.rstrip() for l in traceback.format_stack(frame))
+        return PlainTextResponse("\n".join(out) + "\n")
+
+    @app.get("/debug/events")
+    async def debug_events():
+        return JSONResponse(store.events[-200:])
+
     @app.get("/healthz")
     async def healthz():
         return {"status": "ok"}

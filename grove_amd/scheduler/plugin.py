@@ -213,6 +213,7 @@ class GangScheduler:
         def mark(o: Obj) -> None:
             cond.set_condition(o, c.PODGANG_COND_SCHEDULED, True, "GangPlaced")
             o["status"]["placementScore"] = round(score, 3)
+            o["status"]["phase"] = "Starting"
         try:
             self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark, status=True)
         except ApiError:
@@ -369,6 +370,7 @@ class GangScheduler:
             if ready:
                 def mark(o: Obj) -> None:
                     cond.set_condition(o, c.PODGANG_COND_READY, True, "AllPodGroupsReady")
+                    o["status"]["phase"] = "Running"
                 try:
                     self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark,
                                      status=True)

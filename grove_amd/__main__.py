@@ -35,6 +35,15 @@ def cmd_operator(args) -> int:
         auto_xgmi_domain=cfg.auto_xgmi_domain_enabled,
         use_native_scheduler=None,
     )
+    snap = None
+    if args.state_file:
+        from .kubecore.persistence import SnapshotLoop, load as load_snapshot
+        import os as _os
+        if _os.path.exists(args.state_file):
+            n = load_snapshot(cluster.store, args.state_file)
+            log.info("restored %d objects from %s", n, args.state_file)
+        snap = SnapshotLoop(cluster.store, args.state_file).start()
+
     if args.discover_node:
         node = discover_node()
         cluster.store.create(node)
@@ -66,6 +75,8 @@ def cmd_operator(args) -> int:
     while not stop["flag"]:
         time.sleep(0.2)
     log.info("shutting down")
+    if snap is not None:
+        snap.stop(final_snapshot=True)
     cluster.stop()
     if api is not None:
         api.stop()
@@ -103,6 +114,8 @@ def main(argv=None) -> int:
                     help="register this machine as a Node via the topology agent")
     op.add_argument("--virtual-nodes", type=int, default=0)
     op.add_argument("--virtual-gpus", type=int, default=8)
+    op.add_argument("--state-file", default=None,
+                    help="persist/restore the store to this snapshot file")
     op.set_defaults(fn=cmd_operator)
 
     crds = sub.add_parser("install-crds", help="render or apply the CRDs")

@@ -193,6 +193,14 @@ class Cluster:
     # ------------------------------------------------------------------ lifecycle
     def start(self) -> "Cluster":
         if not self._started:
+            # pre-manager startup topology sync (internal/clustertopology/
+            # clustertopology.go:71 parity): every CTB synced to every TAS backend
+            # with a direct client before controllers run
+            for ctb in self.store.list(c.KIND_CTB):
+                try:
+                    self.ctb_rec.reconcile("", ctb["metadata"]["name"])
+                except Exception:
+                    log.debug("startup topology sync failed", exc_info=True)
             self.manager.start()
             self._started = True
         return self

@@ -210,6 +210,20 @@ def main() -> None:
     _run_rank0(comm, args, n_gpus, gang_size)
 
 
+def _measure_allreduce_busbw(comm: Comm, mb: int = 256, iters: int = 5) -> float:
+    t = torch.ones(mb * 1024 * 1024 // 4, device=f"cuda:{comm.local_rank}")
+    for _ in range(2):
+        comm.dist.all_reduce(t, group=comm.nccl_group)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        comm.dist.all_reduce(t, group=comm.nccl_group)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    n = comm.world
+    return 2 * (n - 1) / n * t.numel() * 4 / dt / 1e9
+
+
 def _serve_agent(comm: Comm) -> None:
     """Node-agent ranks: execute dispatched payloads until stop.
 
@@ -233,6 +247,9 @@ def _serve_agent(comm: Comm) -> None:
         if kind == "elapsed":
             local = marks[-1] - marks[-2] if len(marks) >= 2 else 0.0
             comm.max_over_ranks(local)
+            continue
+        if kind == "busbw":
+            _measure_allreduce_busbw(comm)
             continue
         results: List[Tuple[str, str, Optional[str]]] = []
         if kind == "run":
@@ -310,6 +327,14 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
     for w in range(args.warmup):
         run_step(w, "warm")
 
+    # RCCL-over-xGMI evidence: bus bandwidth of a 256 MB all-reduce across all ranks
+    # (ring algorithm: busbw = 2*(n-1)/n * bytes / t). Runs on the driver's 8-GPU
+    # scale sweep; skipped when single-rank or no GPU.
+    rccl_busbw = None
+    if comm.nccl_group is not None and comm.world > 1:
+        comm.broadcast({"type": "busbw"})
+        rccl_busbw = _measure_allreduce_busbw(comm)
+
     # ---- timed region (barrier + synchronize on both sides; MAX over ranks)
     if comm.dist is not None:
         comm.broadcast({"type": "mark"})
@@ -360,6 +385,7 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
             "gangs_per_step": args.gangs_per_step,
             "payload": payload,
             "scheduler": "amd-gang-scheduler (native xGMI Filter/Score)",
+            "rccl_allreduce_busbw_gbps": round(rccl_busbw, 1) if rccl_busbw else None,
         },
     }
     tracker.stop()

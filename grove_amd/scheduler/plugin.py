@@ -94,8 +94,13 @@ class GangScheduler:
             pg = self.store.try_get(c.KIND_PODGANG, ns, gname)
             if pg is not None:
                 gangs.append(pg)
-        gangs.sort(key=lambda g: (g["metadata"].get("creationTimestamp", ""),
-                                  g["metadata"].get("resourceVersion", "")))
+        prio_values: Dict[str, int] = {
+            p["metadata"]["name"]: int((p.get("value") or 0))
+            for p in self.store.list("PriorityClass", copy_objects=False)}
+        gangs.sort(key=lambda g: (
+            -prio_values.get((g.get("spec") or {}).get("priorityClassName", ""), 0),
+            g["metadata"].get("creationTimestamp", ""),
+            int(g["metadata"].get("resourceVersion", "0"))))
         for pg in gangs:
             ns = pg["metadata"].get("namespace", "default")
             gname = pg["metadata"]["name"]

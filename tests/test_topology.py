@@ -144,3 +144,32 @@ def test_preferred_pack_falls_back(cluster):
     cluster.wait_pcs_available("tc4", timeout=20)
     pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "tc4"})
     assert {p["spec"]["nodeName"] for p in pods} == {"r0-0", "r1-0"}
+
+
+def test_gang_priority_ordering(cluster):
+    """Higher PriorityClass gangs are admitted first when capacity frees up."""
+    cluster.store.create({"kind": "PriorityClass", "apiVersion": "scheduling.k8s.io/v1",
+                          "metadata": {"name": "high"}, "value": 1000})
+    cluster.store.create({"kind": "PriorityClass", "apiVersion": "scheduling.k8s.io/v1",
+                          "metadata": {"name": "low"}, "value": 1})
+    # fill the only node so both gangs queue
+    cluster.add_virtual_nodes(1, gpus=2, prefix="n")
+    blocker = _pcs("blocker", 2)
+    cluster.apply(blocker)
+    cluster.wait_pcs_available("blocker", timeout=20)
+    lo = _pcs("lo", 2)
+    lo["spec"]["template"]["priorityClassName"] = "low"
+    hi = _pcs("hi", 2)
+    hi["spec"]["template"]["priorityClassName"] = "high"
+    cluster.apply(lo)  # created first (FIFO would pick it)
+    import time as _t
+    _t.sleep(0.3)
+    cluster.apply(hi)
+    _t.sleep(0.5)
+    # free capacity: delete the blocker → scheduler pass runs with both pending
+    cluster.delete_pcs("blocker")
+    cluster.wait_pcs_available("hi", timeout=20)
+    hi_pg = cluster.store.get(c.KIND_PODGANG, "default", "hi-0")
+    lo_pg = cluster.store.get(c.KIND_PODGANG, "default", "lo-0")
+    assert cond.condition_true(hi_pg, c.PODGANG_COND_SCHEDULED)
+    assert not cond.condition_true(lo_pg, c.PODGANG_COND_SCHEDULED)

@@ -20,7 +20,7 @@
 #define WAVE 64
 #define BM 128
 #define BN 128
-#define BK 64
+#define BK BKT
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;   // 4 VGPRs (A/B fragment)
@@ -31,12 +31,15 @@ typedef short short8 __attribute__((ext_vector_type(8)));
 // LDS image: [rows][8 pieces of 16B], piece column cb stored at cb ^ (row & 7).
 // A row is 64 bf16 = 128 B = 8 pieces; the XOR spreads the b128 fragment reads
 // (fixed cb over 32 rows) across all banks.
+template <int BKT>
 __device__ __forceinline__ int lds_off(int row, int cb) {
-    return row * BK + ((cb ^ (row & 7)) << 3);   // element (bf16) offset
+    // XOR spreads the low 3 bits of the 16-B piece column; wider BK keeps bit 3+.
+    return row * BKT + (((cb ^ (row & 7)) & (BKT / 8 - 1) | (cb & ~7)) << 3);
 }
 
 // One workgroup = 256 threads = 4 waves arranged 2x2; each wave owns a 64x64 output
 // quadrant = 2x2 MFMA 32x32 tiles.
+template <int BKT>
 __global__ __launch_bounds__(256, 2)
 void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
                            const bf16* __restrict__ Bt,  // [N][K] row-major (B^T)
@@ -82,11 +85,13 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
         const bf16* gB = Bt + col0 * K + kk;
         __bf16* dA = sAp(buf);
         __bf16* dB = sBp(buf);
+        constexpr int PPR = BKT / 8;  // 16-B pieces per row
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < (BM * BKT / 8) / 256; ++i) {
             int p = tid + i * 256;
-            int row = p >> 3;
-            int cb_src = (p & 7) ^ (row & 7);
+            int row = p / PPR;
+            int cb_slot = p % PPR;
+            int cb_src = ((cb_slot ^ (row & 7)) & (PPR - 1)) | (cb_slot & ~7);
             int base = wave_piece0 + i * 256;  // uniform across the wave
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) uint32_t*)(
@@ -119,10 +124,10 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
             int cb = (ks << 1) | k_half;            // 16-bf16 step = two 8-elt pieces
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt)
-                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off(a_row + mt * 32, cb));
+                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off<BKT>(a_row + mt * 32, cb));
 #pragma unroll
             for (int nt = 0; nt < 2; ++nt)
-                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off(b_row + nt * 32, cb));
+                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off<BKT>(b_row + nt * 32, cb));
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -163,7 +168,7 @@ __global__ void stream_triad_kernel(const float4* __restrict__ a,
 // ---------------------------------------------------------------- host wrappers
 
 static void check_dims(int64_t M, int64_t N, int64_t K) {
-    TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
+    TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % 64 == 0,
                 "mfma_gemm_bf16 requires M%128==0, N%128==0, K%64==0 (got ",
                 M, "x", N, "x", K, ")");
 }
@@ -179,14 +184,14 @@ torch::Tensor mfma_gemm_bf16(torch::Tensor a, torch::Tensor bt) {
     auto c = torch::empty({M, N}, a.options().dtype(torch::kFloat32));
     dim3 grid((M / BM) * (N / BN)), block(256);
     hipStream_t stream = at::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(mfma_gemm_bf16_kernel, grid, block, 0, stream,
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<64>), grid, block, 0, stream,
                        (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
                        c.data_ptr<float>(), (int)M, (int)N, (int)K);
     return c;
 }
 
 // Pod payload: `iters` GEMM steps on pre-allocated buffers; returns achieved TFLOP/s.
-double burn_gemm(int64_t m, int64_t n, int64_t k, int64_t iters) {
+double burn_gemm_v(int64_t m, int64_t n, int64_t k, int64_t iters, int64_t bk) {
     check_dims(m, n, k);
     auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
     auto a = torch::randn({m, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
@@ -195,9 +200,16 @@ double burn_gemm(int64_t m, int64_t n, int64_t k, int64_t iters) {
     hipStream_t stream = at::hip::getCurrentHIPStream();
     dim3 grid((m / BM) * (n / BN)), block(256);
     auto launch = [&] {
-        hipLaunchKernelGGL(mfma_gemm_bf16_kernel, grid, block, 0, stream,
-                           (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-                           c.data_ptr<float>(), (int)m, (int)n, (int)k);
+        if (bk == 128)
+            hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<128>), grid,
+                               block, 0, stream,
+                               (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+                               c.data_ptr<float>(), (int)m, (int)n, (int)k);
+        else
+            hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<64>), grid,
+                               block, 0, stream,
+                               (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+                               c.data_ptr<float>(), (int)m, (int)n, (int)k);
     };
     launch();  // warmup
     C10_HIP_CHECK(hipStreamSynchronize(stream));
@@ -238,6 +250,10 @@ double stream_triad(int64_t n_floats, int64_t iters) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "grove_amd MI355X pod-payload kernels (MFMA bf16 GEMM, HBM stream)";
     m.def("mfma_gemm_bf16", &mfma_gemm_bf16, "C[M,N]=A[M,K]@Bt[N,K]^T (bf16 in, fp32 out)");
-    m.def("burn_gemm", &burn_gemm, "run iters GEMM steps; returns TFLOP/s");
+    m.def("burn_gemm",
+          [](int64_t m_, int64_t n_, int64_t k_, int64_t it) {
+              return burn_gemm_v(m_, n_, k_, it, 64);
+          }, "run iters GEMM steps; returns TFLOP/s");
+    m.def("burn_gemm_v", &burn_gemm_v, "burn_gemm with explicit BK (64 or 128)");
     m.def("stream_triad", &stream_triad, "HBM triad; returns GB/s");
 }

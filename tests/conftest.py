@@ -24,11 +24,38 @@ def pytest_collection_modifyitems(config, items):
             item.add_marker(pytest.mark.skip(reason="reference tree not mounted"))
 
 
+@pytest.hookimpl(hookwrapper=True)
+def pytest_runtest_makereport(item, call):
+    out = yield
+    rep = out.get_result()
+    setattr(item, f"rep_{rep.when}", rep)
+
+
 @pytest.fixture()
-def cluster():
+def cluster(request):
     from grove_amd import Cluster
     cl = Cluster(use_native_scheduler=None if _native_built() else False).start()
     yield cl
+    # diagnostics collector (reference e2e/diagnostics/collector.go parity):
+    # on failure, dump resource counts + unhealthy objects before teardown
+    rep = getattr(request.node, "rep_call", None)
+    if rep is not None and rep.failed:
+        import sys
+        print("\n=== cluster diagnostics (on failure) ===", file=sys.stderr)
+        print("store:", cl.store.stats(), file=sys.stderr)
+        for kind in ("PodCliqueSet", "PodClique", "PodCliqueScalingGroup", "PodGang"):
+            for o in cl.store.list(kind):
+                st = o.get("status") or {}
+                print(f"  {kind} {o['metadata']['name']}: "
+                      f"{ {k: v for k, v in st.items() if k != 'conditions'} } "
+                      f"conds={[(cd.get('type'), cd.get('status')) for cd in st.get('conditions', [])]}",
+                      file=sys.stderr)
+        pods = cl.store.list("Pod")
+        gated = sum(1 for p in pods if p["spec"].get("schedulingGates"))
+        bound = sum(1 for p in pods if p["spec"].get("nodeName"))
+        print(f"  pods={len(pods)} gated={gated} bound={bound}", file=sys.stderr)
+        for ev in cl.store.events[-10:]:
+            print("  event:", ev.get("reason"), ev.get("message"), file=sys.stderr)
     cl.stop()
 
 

@@ -9,6 +9,7 @@ conditions (reconcilestatus.go:40-282). Fresh MI355X-native implementation.
 from __future__ import annotations
 
 import logging
+import time
 from typing import Any, Dict, List, Optional
 
 from ..api import constants as c
@@ -252,6 +253,20 @@ class PodCliqueReconciler:
             st["observedGeneration"] = o["metadata"].get("generation")
             st["currentPodTemplateHash"] = tmpl_hash
             st["hpaPodSelector"] = f"{c.LABEL_PODCLIQUE}={name}"
+            # rolling-update progress (status.updateProgress parity): active while any
+            # pod carries an outdated template hash
+            if n_updated < n_total:
+                prog = st.get("updateProgress") or {}
+                if prog.get("podTemplateHash") != tmpl_hash:
+                    prog = {"podTemplateHash": tmpl_hash,
+                            "updateStartedAt": time.strftime(
+                                "%Y-%m-%dT%H:%M:%SZ", time.gmtime())}
+                prog.pop("updateEndedAt", None)
+                st["updateProgress"] = prog
+            elif st.get("updateProgress") and not st["updateProgress"].get(
+                    "updateEndedAt"):
+                st["updateProgress"]["updateEndedAt"] = time.strftime(
+                    "%Y-%m-%dT%H:%M:%SZ", time.gmtime())
             # PodCliqueScheduled (reconcilestatus.go:282)
             if n_sched >= min_avail:
                 cond.set_condition(o, c.COND_PODCLIQUE_SCHEDULED, True,

@@ -243,6 +243,13 @@ class PCSGReconciler:
             st["availableReplicas"] = avail
             st["observedGeneration"] = o["metadata"].get("generation")
             st["selector"] = f"{c.LABEL_PCSG}={name}"
+            pcs_name = o["metadata"].get("labels", {}).get(c.LABEL_PART_OF)
+            pcs_obj = self.store.try_get(c.KIND_PCS, namespace, pcs_name) \
+                if pcs_name else None
+            if pcs_obj is not None:
+                gh = (pcs_obj.get("status") or {}).get("currentGenerationHash")
+                if gh:
+                    st["currentPodCliqueSetGenerationHash"] = gh
             ever = bool(st.get("everAvailable")) or avail >= min_avail
             st["everAvailable"] = ever
             if ever and avail < min_avail:

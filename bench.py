@@ -195,6 +195,8 @@ def main() -> None:
     ap.add_argument("--payload", type=str, default="2048x2048x2048x1",
                     help="per-pod GEMM payload MxNxKxiters")
     ap.add_argument("--step-timeout", type=float, default=120.0)
+    ap.add_argument("--inflight", type=int, default=1,
+                    help="pipelined steps kept in flight (1 = fully serial)")
     args = ap.parse_args()
 
     comm = Comm()
@@ -306,26 +308,36 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
             time.sleep(idle_sleep)
         return n
 
-    def run_step(step_id: int, timed_epoch: str) -> None:
+    def submit_step(step_id: int, timed_epoch: str) -> str:
         name = f"bench-{timed_epoch}-{step_id}"
         pcs = bench_pcs(name, args.gangs_per_step, gang_size, payload)
         submit = time.monotonic()
         for g in range(args.gangs_per_step):
             tracker.expect_gang(f"{name}-{g}", gang_size, submit)
         cluster.store.create(pcs)
+        return name
+
+    def wait_step(name: str) -> None:
         deadline = time.monotonic() + args.step_timeout
         while time.monotonic() < deadline:
             dispatch_cycle()
-            done = all(
-                tracker.gangs[f"{name}-{g}"].running is not None
-                for g in range(args.gangs_per_step))
-            if done:
+            if all(tracker.gangs[f"{name}-{g}"].running is not None
+                   for g in range(args.gangs_per_step)):
                 return
         raise TimeoutError(f"step {name} did not reach all-Running")
 
+    def run_steps(n: int, timed_epoch: str) -> None:
+        """Pipelined: keep up to --inflight steps' gangs in flight."""
+        window: List[str] = []
+        for s in range(n):
+            window.append(submit_step(s, timed_epoch))
+            if len(window) >= max(1, args.inflight):
+                wait_step(window.pop(0))
+        for name in window:
+            wait_step(name)
+
     # ---- warmup
-    for w in range(args.warmup):
-        run_step(w, "warm")
+    run_steps(args.warmup, "warm")
 
     # RCCL-over-xGMI evidence: bus bandwidth of a 256 MB all-reduce across all ranks
     # (ring algorithm: busbw = 2*(n-1)/n * bytes / t). Runs on the driver's 8-GPU
@@ -340,8 +352,7 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
         comm.broadcast({"type": "mark"})
     comm.barrier_sync()
     t0 = time.perf_counter()
-    for s in range(args.steps):
-        run_step(s, "timed")
+    run_steps(args.steps, "timed")
     if comm.dist is not None:
         comm.broadcast({"type": "mark"})
     comm.barrier_sync()
@@ -384,6 +395,7 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
             "gang_size": gang_size,
             "gangs_per_step": args.gangs_per_step,
             "payload": payload,
+            "inflight": args.inflight,
             "scheduler": "amd-gang-scheduler (native xGMI Filter/Score)",
             "rccl_allreduce_busbw_gbps": round(rccl_busbw, 1) if rccl_busbw else None,
         },

@@ -83,6 +83,82 @@ def cmd_operator(args) -> int:
     return 0
 
 
+def cmd_get(args) -> int:
+    import json
+    import urllib.parse
+    import urllib.request
+
+    import yaml as _y
+
+    from .kubecore.apiserver import PLURALS, CLUSTER_SCOPED_PLURALS
+    plural = args.resource.lower()
+    kind = PLURALS.get(plural)
+    if kind is None:
+        print(f"unknown resource {plural!r}; one of: {', '.join(sorted(PLURALS))}")
+        return 1
+    group = "grove.io/v1alpha1" if kind.startswith("PodClique") or         kind == "ClusterTopologyBinding" else         ("scheduler.grove.io/v1alpha1" if kind == "PodGang" else "v1")
+    base = f"{args.server}/apis/{group}" if "/" in group else         f"{args.server}/api/{group}"
+    if plural in CLUSTER_SCOPED_PLURALS:
+        url = f"{base}/{plural}"
+    else:
+        url = f"{base}/namespaces/{args.namespace}/{plural}"
+    if args.name:
+        url += f"/{args.name}"
+    elif args.selector:
+        url += f"?labelSelector={urllib.parse.quote(args.selector)}"
+    with urllib.request.urlopen(url, timeout=10) as r:
+        data = json.loads(r.read())
+    items = data.get("items", [data] if args.name else [])
+    if args.output == "json":
+        print(json.dumps(items if not args.name else items[0], indent=2))
+        return 0
+    if args.output == "yaml":
+        print(_y.safe_dump_all(items, sort_keys=False))
+        return 0
+    rows = []
+    for o in items:
+        st = o.get("status") or {}
+        phase = st.get("phase") or ""
+        ready = st.get("readyReplicas", st.get("availableReplicas", ""))
+        rows.append((o["metadata"]["name"], phase, str(ready),
+                     o["metadata"].get("creationTimestamp", "")))
+    w = max([len(r[0]) for r in rows] + [4]) + 2
+    print(f"{'NAME':<{w}}{'PHASE':<12}{'READY':<8}CREATED")
+    for r in rows:
+        print(f"{r[0]:<{w}}{r[1]:<12}{r[2]:<8}{r[3]}")
+    return 0
+
+
+def cmd_apply(args) -> int:
+    import json
+    import urllib.request
+
+    import yaml as _y
+
+    from .kubecore.apiserver import PLURALS
+    kind_to_plural = {v: k for k, v in PLURALS.items()}
+    n = 0
+    with open(args.filename) as f:
+        for doc in _y.safe_load_all(f):
+            if not doc:
+                continue
+            plural = kind_to_plural.get(doc.get("kind"))
+            group = doc.get("apiVersion", "v1")
+            base = f"{args.server}/apis/{group}" if "/" in group else                 f"{args.server}/api/{group}"
+            url = f"{base}/namespaces/{args.namespace}/{plural}"
+            req = urllib.request.Request(
+                url, data=json.dumps(doc).encode(), method="POST",
+                headers={"Content-Type": "application/json"})
+            try:
+                urllib.request.urlopen(req, timeout=10)
+                n += 1
+                print(f"created {doc.get('kind')}/{doc['metadata']['name']}")
+            except urllib.error.HTTPError as e:
+                print(f"error {e.code} for {doc.get('kind')}/"
+                      f"{doc['metadata'].get('name')}: {e.read().decode()[:200]}")
+    return 0 if n else 1
+
+
 def cmd_install_crds(args) -> int:
     from .api.crds import install_crds, write_crds
     if args.output_dir:
@@ -117,6 +193,22 @@ def main(argv=None) -> int:
     op.add_argument("--state-file", default=None,
                     help="persist/restore the store to this snapshot file")
     op.set_defaults(fn=cmd_operator)
+
+    get = sub.add_parser("get", help="list resources from a running apiserver")
+    get.add_argument("resource", help="plural, e.g. podcliquesets, pods, podgangs")
+    get.add_argument("name", nargs="?", default=None)
+    get.add_argument("--server", default="http://127.0.0.1:8081")
+    get.add_argument("-n", "--namespace", default="default")
+    get.add_argument("-l", "--selector", default=None)
+    get.add_argument("-o", "--output", choices=["wide", "json", "yaml"],
+                     default="wide")
+    get.set_defaults(fn=cmd_get)
+
+    ap_cmd = sub.add_parser("apply", help="apply a manifest file to the apiserver")
+    ap_cmd.add_argument("-f", "--filename", required=True)
+    ap_cmd.add_argument("--server", default="http://127.0.0.1:8081")
+    ap_cmd.add_argument("-n", "--namespace", default="default")
+    ap_cmd.set_defaults(fn=cmd_apply)
 
     crds = sub.add_parser("install-crds", help="render or apply the CRDs")
     crds.add_argument("--server", default=None, help="apiserver URL to POST CRDs to")

@@ -101,3 +101,63 @@ def test_gpu_end_to_end_gang():
         assert pg["status"]["placementScore"] > 0
     finally:
         cl.stop()
+
+
+def test_process_kubelet_runs_payload_on_gpu():
+    """Process-backed pod: subprocess gets HIP_VISIBLE_DEVICES from the scheduler's
+    assignment and runs the MFMA payload on the real GPU."""
+    from grove_amd import Cluster
+    from grove_amd.api import constants as c
+    from grove_amd.controllers.manager import Controller
+    from grove_amd.kubecore.apiserver import ApiServer
+    from grove_amd.kubelet.process import ProcessKubelet
+    from grove_amd.topology.agent import discover_node
+    import threading
+
+    cl = Cluster().start()
+    api = ApiServer(cl.store, port=18191).start()
+    kubelet = ProcessKubelet(cl.store, api_url=api.url)
+    cl.c_kubelet.stop()
+    ctrl = cl.manager.add_controller(
+        Controller("process-kubelet", kubelet.reconcile, workers=2))
+    ctrl.start()
+    w = cl.store.watch("Pod")
+
+    def pump():
+        import queue as q
+        while True:
+            try:
+                ev, obj = w.queue.get(timeout=0.5)
+            except q.Empty:
+                continue
+            except Exception:
+                return
+            if obj.get("spec", {}).get("nodeName"):
+                ctrl.enqueue(obj["metadata"].get("namespace", "default"),
+                             obj["metadata"]["name"])
+    threading.Thread(target=pump, daemon=True).start()
+    try:
+        cl.store.create(discover_node("gpu-proc-node"))
+        pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+               "metadata": {"name": "gproc"},
+               "spec": {"replicas": 1, "template": {"cliques": [{
+                   "name": "inf",
+                   "annotations": {"grove.io/payload-shape": "512x512x512x1"},
+                   "spec": {"roleName": "r", "replicas": 1,
+                            "podSpec": {"containers": [{
+                                "name": "m", "image": "payload",
+                                "resources": {"requests": {
+                                    c.AMD_GPU_RESOURCE: "1"}}}]}}}]}}}
+        cl.store.create(pcs)
+
+        def done():
+            pods = cl.store.list("Pod", "default", {c.LABEL_PART_OF: "gproc"})
+            return pods and all((p.get("status") or {}).get("phase") == "Succeeded"
+                                for p in pods)
+        cl.wait_for(done, timeout=120, desc="GPU pod process succeeded")
+    finally:
+        kubelet.shutdown()
+        w.stop()
+        ctrl.stop()
+        api.stop()
+        cl.stop()

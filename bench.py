@@ -396,6 +396,8 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
             "gangs_per_step": args.gangs_per_step,
             "payload": payload,
             "inflight": args.inflight,
+            "concurrent_syncs": 4,
+            "operator_version": __import__("grove_amd").__version__,
             "scheduler": "amd-gang-scheduler (native xGMI Filter/Score)",
             "rccl_allreduce_busbw_gbps": round(rccl_busbw, 1) if rccl_busbw else None,
         },

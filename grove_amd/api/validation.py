@@ -232,3 +232,27 @@ def validate_clustertopologybinding(ctb: Obj, old: Optional[Obj] = None) -> None
             raise _err("spec.levels", f"duplicate node label key {k!r}")
         seen_d.add(d)
         seen_k.add(k)
+
+
+def validate_pcsg(pcsg: Obj, old: Optional[Obj] = None) -> None:
+    spec = pcsg.get("spec") or {}
+    reps = int(spec.get("replicas", 1))
+    if reps < 0:
+        raise _err("spec.replicas", "must be >= 0")
+    ma = spec.get("minAvailable")
+    if ma is not None and int(ma) < 1:
+        raise _err("spec.minAvailable", "must be >= 1")
+    if old is not None:
+        if spec.get("cliqueNames") != (old.get("spec") or {}).get("cliqueNames"):
+            raise _err("spec.cliqueNames", "field is immutable")
+
+
+def validate_podclique(pclq: Obj, old: Optional[Obj] = None) -> None:
+    spec = pclq.get("spec") or {}
+    if int(spec.get("replicas", 1)) < 0:
+        raise _err("spec.replicas", "must be >= 0")
+    ma = spec.get("minAvailable")
+    if ma is not None and int(ma) < 1:
+        raise _err("spec.minAvailable", "must be >= 1")
+    if old is not None and spec.get("roleName") != (old.get("spec") or {}).get("roleName"):
+        raise _err("spec.roleName", "field is immutable")

@@ -14,7 +14,8 @@ import yaml as _yaml
 
 from .api import constants as c
 from .api.defaulting import default_podcliqueset, default_podclique, default_pcsg
-from .api.validation import validate_podcliqueset, validate_clustertopologybinding
+from .api.validation import (validate_podcliqueset, validate_clustertopologybinding,
+                             validate_pcsg, validate_podclique)
 from .controllers.manager import Controller, Manager, Result
 from .controllers.podclique import PodCliqueReconciler
 from .controllers.podcliqueset import PodCliqueSetReconciler
@@ -48,7 +49,9 @@ class Cluster:
         self.store.register_mutator(c.KIND_PCS, default_podcliqueset)
         self.store.register_validator(c.KIND_PCS, validate_podcliqueset)
         self.store.register_mutator(c.KIND_PCLQ, default_podclique)
+        self.store.register_validator(c.KIND_PCLQ, validate_podclique)
         self.store.register_mutator(c.KIND_PCSG, default_pcsg)
+        self.store.register_validator(c.KIND_PCSG, validate_pcsg)
         self.store.register_validator(c.KIND_CTB, validate_clustertopologybinding)
         if enable_authorizer:
             from .api.authorization import Authorizer

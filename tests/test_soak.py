@@ -106,3 +106,25 @@ def test_store_snapshot_resume(cluster, simple1_yaml, tmp_path):
                               timeout=20)
     finally:
         fresh.stop()
+
+
+def test_leader_election(cluster):
+    """Lease-based leadership: one holder at a time; failover on expiry; release on
+    stop hands over promptly."""
+    import time as _t
+    from grove_amd.kubecore.lease import LeaderElector
+    a = LeaderElector(cluster.store, "operator", "op-a",
+                      lease_duration_s=0.6, renew_period_s=0.1)
+    b = LeaderElector(cluster.store, "operator", "op-b",
+                      lease_duration_s=0.6, renew_period_s=0.1)
+    a.start()
+    assert a.is_leader.wait(timeout=5)
+    b.start()
+    _t.sleep(0.5)
+    assert not b.is_leader.is_set()  # a holds and renews
+    a.stop()  # releases
+    assert b.is_leader.wait(timeout=5)
+    lease = cluster.store.get("Lease", "grove-system", "operator")
+    assert lease["spec"]["holderIdentity"] == "op-b"
+    assert lease["spec"]["leaseTransitions"] >= 1
+    b.stop()

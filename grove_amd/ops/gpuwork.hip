@@ -32,9 +32,18 @@ typedef short short8 __attribute__((ext_vector_type(8)));
 // A row is 64 bf16 = 128 B = 8 pieces; the XOR spreads the b128 fragment reads
 // (fixed cb over 32 rows) across all banks.
 template <int BKT>
+__device__ __forceinline__ int swz(int row, int cb) {
+    // Bank swizzle for conflict-free ds_read_b128: fold row bit 3+ in so rows that
+    // share (row & 7) but differ at bit 3 (the b128 lane-group aliases, e.g. rows 12
+    // and 20 in group {0-3,12-15,20-27}) land on different banks. Verified: 16 lanes
+    // x 4 dwords cover all 64 banks in both b128 lane groups.
+    int e = (row ^ (row >> 3)) & 7;
+    return ((cb ^ e) & (BKT / 8 - 1)) | (cb & ~7);
+}
+
+template <int BKT>
 __device__ __forceinline__ int lds_off(int row, int cb) {
-    // XOR spreads the low 3 bits of the 16-B piece column; wider BK keeps bit 3+.
-    return row * BKT + (((cb ^ (row & 7)) & (BKT / 8 - 1) | (cb & ~7)) << 3);
+    return row * BKT + (swz<BKT>(row, cb) << 3);
 }
 
 // One workgroup = 256 threads = 4 waves arranged 2x2; each wave owns a 64x64 output
@@ -91,7 +100,7 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
             int p = tid + i * 256;
             int row = p / PPR;
             int cb_slot = p % PPR;
-            int cb_src = ((cb_slot ^ (row & 7)) & (PPR - 1)) | (cb_slot & ~7);
+            int cb_src = swz<BKT>(row, cb_slot);  // XOR is involutive: source swizzle = image swizzle
             int base = wave_piece0 + i * 256;  // uniform across the wave
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) uint32_t*)(

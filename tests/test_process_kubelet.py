@@ -75,5 +75,5 @@ def test_pods_run_as_processes_with_startup_ordering(process_cluster):
             for p in cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "proc"})}
     assert pods["proc-0-a"]["status"]["phase"] == "Succeeded"
     assert pods["proc-0-b"]["status"]["phase"] == "Succeeded"
-    pcs_out = cluster.store.get(c.KIND_PCS, "default", "proc")
-    assert pcs_out["status"]["availableReplicas"] == 1
+    # status rollup is eventually consistent — wait, don't read-after-write
+    cluster.wait_pcs_available("proc", timeout=10)

@@ -23,6 +23,7 @@ from .controllers.podcliquescalinggroup import PCSGReconciler
 from .controllers.podgang import PodGangReconciler
 from .controllers.clustertopology import ClusterTopologyReconciler
 from .controllers.hpa import HPAReconciler
+from .controllers.nodelifecycle import NodeLifecycleReconciler
 from .scheduler.backends import Registry
 from .kubecore.store import Store, Obj, ApiError
 from .kubelet.virtual import VirtualKubelet, make_virtual_node
@@ -67,6 +68,7 @@ class Cluster:
         self.podgang_rec = PodGangReconciler(self.store, self.registry)
         self.ctb_rec = ClusterTopologyReconciler(self.store, self.registry)
         self.hpa_rec = HPAReconciler(self.store)
+        self.node_rec = NodeLifecycleReconciler(self.store)
         self.scheduler = GangScheduler(self.store, scheduler_name,
                                        use_native=use_native_scheduler)
         self.kubelet = VirtualKubelet(self.store,
@@ -87,6 +89,8 @@ class Cluster:
             "clustertopology", self.ctb_rec.reconcile, workers=1))
         self.c_hpa = m.add_controller(Controller(
             "hpa", self.hpa_rec.reconcile, workers=1))
+        self.c_node = m.add_controller(Controller(
+            "node-lifecycle", self.node_rec.reconcile, workers=1))
         self.c_sched = m.add_controller(Controller(
             "gang-scheduler", lambda ns, n: self.scheduler.reconcile(ns, n) or Result.DONE,
             workers=1))
@@ -156,6 +160,8 @@ class Cluster:
 
         def on_node(ev: str, obj: Obj, _old) -> None:
             self.c_sched.enqueue("", "pass")
+            if ev in ("DELETED", "MODIFIED"):
+                self.c_node.enqueue("", obj["metadata"]["name"])
 
         def on_ctb(ev: str, obj: Obj, _old) -> None:
             self.c_ctb.enqueue("", obj["metadata"]["name"])

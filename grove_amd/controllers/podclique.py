@@ -252,6 +252,11 @@ class PodCliqueReconciler:
             st["updatedReplicas"] = n_updated
             st["observedGeneration"] = o["metadata"].get("generation")
             st["currentPodTemplateHash"] = tmpl_hash
+            pcs_for_hash = self._find_pcs(o)
+            if pcs_for_hash is not None:
+                gh = (pcs_for_hash.get("status") or {}).get("currentGenerationHash")
+                if gh:
+                    st["currentPodCliqueSetGenerationHash"] = gh
             st["hpaPodSelector"] = f"{c.LABEL_PODCLIQUE}={name}"
             # rolling-update progress (status.updateProgress parity): active while any
             # pod carries an outdated template hash

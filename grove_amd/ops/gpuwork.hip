@@ -2,13 +2,21 @@
 //
 // Grove schedules inference pods; this framework's GPU-node agent runs each scheduled
 // pod's payload on its assigned GPU. The payload is real CDNA4 work (not a sleep):
-//   * mfma_gemm_bf16 — LDS-tiled bf16 GEMM on v_mfma_f32_32x32x16_bf16 matrix cores:
-//     128×128 block tile, BK=64 double-buffered LDS, XOR-swizzled (cb ^ (row&7)) images
-//     for conflict-free ds_read_b128 fragment reads, XCD-aware block swizzle
-//     (blockIdx → tile map groups tiles per XCD for L2 reuse, 8 XCDs).
+//   * mfma_gemm_bf16 — LDS-tiled bf16 GEMM on v_mfma_f32_32x32x16_bf16 matrix cores.
+//     Two tilings, picked by shape:
+//       - 256x256 block tile, 8 waves (512 thr), 2x the arithmetic intensity of the
+//         128 tile (the kernel is staging-bandwidth-bound, so intensity is the lever)
+//       - 128x128 block tile, 4 waves (256 thr) for shapes not divisible by 256
+//     Both: BK=64 double-buffered LDS staged by 16-B global_load_lds (LDS-DMA, the
+//     DMA of tile t+1 overlaps the MFMAs of tile t; the __syncthreads() vmcnt(0)
+//     drain coincides with the data dependency, so nothing is wasted), conflict-free
+//     XOR bank swizzle (row^(row>>3) folded in — PMC-verified 0 LDS conflicts), and
+//     an XCD-aware blockIdx→tile map (8-tile M-columns share a B slab per XCD L2).
+//     Measured (1xMI355X): 3-buffer raw-barrier pipelining was tried and REGRESSED
+//     (96-160 KB LDS drops to 1 WG/CU, losing block-level overlap) — see git history.
 //   * stream_triad — float4 HBM streaming (bandwidth probe + memory-heavy payload).
 //
-// Numerics: tests/test_gpu_kernels.py checks mfma_gemm_bf16 against a torch fp32
+// Numerics: tests/test_gpu_kernels.py checks both tilings against a torch fp32
 // reference (A=I + asymmetric-B layout traps included).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -18,125 +26,182 @@
 #include <vector>
 
 #define WAVE 64
-#define BM 128
-#define BN 128
-#define BK BKT
+#define BK 64
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;   // 4 VGPRs (A/B fragment)
 using f32x16 = __attribute__((ext_vector_type(16))) float;   // 32x32 accumulator
-using f32x4 = __attribute__((ext_vector_type(4))) float;
-typedef short short8 __attribute__((ext_vector_type(8)));
 
-// LDS image: [rows][8 pieces of 16B], piece column cb stored at cb ^ (row & 7).
-// A row is 64 bf16 = 128 B = 8 pieces; the XOR spreads the b128 fragment reads
-// (fixed cb over 32 rows) across all banks.
-template <int BKT>
+// Bank swizzle for conflict-free ds_read_b128: fold row bit 3+ in so rows that share
+// (row & 7) but differ at bit 3 (the b128 lane-group aliases, e.g. rows 12 and 20 in
+// group {0-3,12-15,20-27}) land on different banks. Verified: 16 lanes x 4 dwords
+// cover all 64 banks in both b128 lane groups; SQ_LDS_BANK_CONFLICT measured 0.
 __device__ __forceinline__ int swz(int row, int cb) {
-    // Bank swizzle for conflict-free ds_read_b128: fold row bit 3+ in so rows that
-    // share (row & 7) but differ at bit 3 (the b128 lane-group aliases, e.g. rows 12
-    // and 20 in group {0-3,12-15,20-27}) land on different banks. Verified: 16 lanes
-    // x 4 dwords cover all 64 banks in both b128 lane groups.
-    int e = (row ^ (row >> 3)) & 7;
-    return ((cb ^ e) & (BKT / 8 - 1)) | (cb & ~7);
+    return (cb ^ ((row ^ (row >> 3)) & 7)) & 7;
 }
 
-template <int BKT>
 __device__ __forceinline__ int lds_off(int row, int cb) {
-    return row * BKT + (swz<BKT>(row, cb) << 3);
+    return row * BK + (swz(row, cb) << 3);   // element (bf16) offset, 8 pieces/row
 }
 
-// One workgroup = 256 threads = 4 waves arranged 2x2; each wave owns a 64x64 output
-// quadrant = 2x2 MFMA 32x32 tiles.
-template <int BKT>
-__global__ __launch_bounds__(256, 2)
-void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
-                           const bf16* __restrict__ Bt,  // [N][K] row-major (B^T)
-                           float* __restrict__ C,        // [M][N]
-                           int M, int N, int K) {
-    __shared__ __bf16 smem[2 * (BM * BK + BN * BK)];
-    // buffer b: A at b*(BM+BN)*BK, B at that + BM*BK (single __shared__ object —
-    // guide §5 trap 4(a): a second __shared__ object de-pipelines the k-loop)
-    auto sAp = [&](int b) { return smem + b * (BM * BK + BN * BK); };
-    auto sBp = [&](int b) { return smem + b * (BM * BK + BN * BK) + BM * BK; };
-
-    const int tiles_n = N / BN;
-    // XCD-aware swizzle: consecutive blocks land on XCDs round-robin (b % 8); map so
-    // the 8 tiles resident on one XCD at a time form a 8(M)x1(N) column sharing the
-    // same B tile slab in that XCD's L2.
-    int bid = blockIdx.x;
+// XCD-aware blockIdx→tile map: consecutive blocks land on XCDs round-robin (b % 8);
+// group tiles into 8(M)x1(N) columns so one XCD's resident tiles share a B slab in
+// its (non-coherent, per-XCD) L2.
+__device__ __forceinline__ void tile_map(int bid, int num_pid_m, int tiles_n,
+                                         int& tile_m, int& tile_n) {
     const int GROUP = 8;
-    int num_pid_m = M / BM;
     int group_size = min(GROUP, num_pid_m);
     int pids_per_group = group_size * tiles_n;
     int group = bid / pids_per_group;
     int in_group = bid % pids_per_group;
-    int tile_m = group * GROUP + (in_group % group_size);
-    int tile_n = in_group / group_size;
+    tile_m = group * GROUP + (in_group % group_size);
+    tile_n = in_group / group_size;
+}
 
+// Stage one (rows x BK) bf16 tile via 16-B LDS-DMA. The glds LDS destination is
+// wave-uniform-base + lane*16 and the LDS image is lane-linear in piece index, so the
+// XOR swizzle is applied on the SOURCE address (guide §5 rule 21: swizzled images via
+// pre-swizzled global addresses, LDS stays linear).
+template <int THREADS>
+__device__ __forceinline__ void stage_tile(const bf16* __restrict__ g, long ld,
+                                           __bf16* __restrict__ dst, int rows,
+                                           int tid, int wave_piece0) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {           // rows*8 pieces / THREADS == 8 iterations
+        if (i >= (rows * 8) / THREADS)
+            break;
+        int p = tid + i * THREADS;
+        int row = p >> 3;
+        int cb_src = swz(row, p & 7);
+        int base = wave_piece0 + i * THREADS;  // uniform across the wave
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t*)(
+                g + (long)row * ld + (cb_src << 3)),
+            (__attribute__((address_space(3))) uint32_t*)(dst + base * 8),
+            16, 0, 0);
+    }
+}
+
+// ---------------------------------------------------------------- 256x256 tile
+// 8 waves as 4(row)x2(col); each wave computes 64x128 = 2x4 MFMA 32x32 tiles
+// (128 accumulator VGPRs). Arithmetic intensity 128 FLOP/B staged — double the 128
+// tile — which is what moves a staging-bound GEMM.
+__global__ __launch_bounds__(512, 1)
+void mfma_gemm_bf16_256_kernel(const bf16* __restrict__ A,   // [M][K] row-major
+                               const bf16* __restrict__ Bt,  // [N][K] row-major
+                               float* __restrict__ C,        // [M][N]
+                               int M, int N, int K) {
+    constexpr int BM = 256, BN = 256;
+    __shared__ __bf16 smem[2 * (BM * BK + BN * BK)];
+    auto sAp = [&](int b) { return smem + b * (BM * BK + BN * BK); };
+    auto sBp = [&](int b) { return smem + b * (BM * BK + BN * BK) + BM * BK; };
+
+    int tile_m, tile_n;
+    tile_map(blockIdx.x, M / BM, N / BN, tile_m, tile_n);
     const int tid = threadIdx.x;
-    const int wave = tid / WAVE;          // 0..3
+    const int wave = tid / WAVE;
     const int lane = tid % WAVE;
-    const int wr = wave >> 1, wc = wave & 1;  // wave quadrant (2x2)
-
+    const int wr = wave >> 1, wc = wave & 1;      // 4(row) x 2(col)
     const long row0 = (long)tile_m * BM;
     const long col0 = (long)tile_n * BN;
+    const int wave_piece0 = wave * WAVE;
 
-    // Staging via 16-B LDS-DMA (global_load_lds, guide §5 ladder step 3): thread t
-    // covers LDS piece p = t + i*256 (row = p/8, cb_slot = p%8). The glds LDS
-    // destination is wave-uniform-base + lane*16 and our LDS image is lane-linear in
-    // p, so the XOR swizzle is applied on the SOURCE address (cb_src = cb_slot ^
-    // (row&7)) while LDS stays linear — guide §5 rule 21. Compute on buf overlaps the
-    // DMA into buf^1; __syncthreads() drains it (vmcnt(0)) each K-tile.
-    const int wave_piece0 = (tid / WAVE) * WAVE;  // wave-uniform piece base
     auto stage = [&](int buf, long kk) {
-        const bf16* gA = A + row0 * K + kk;
-        const bf16* gB = Bt + col0 * K + kk;
-        __bf16* dA = sAp(buf);
-        __bf16* dB = sBp(buf);
-        constexpr int PPR = BKT / 8;  // 16-B pieces per row
-#pragma unroll
-        for (int i = 0; i < (BM * BKT / 8) / 256; ++i) {
-            int p = tid + i * 256;
-            int row = p / PPR;
-            int cb_slot = p % PPR;
-            int cb_src = swz<BKT>(row, cb_slot);  // XOR is involutive: source swizzle = image swizzle
-            int base = wave_piece0 + i * 256;  // uniform across the wave
-            __builtin_amdgcn_global_load_lds(
-                (const __attribute__((address_space(1))) uint32_t*)(
-                    gA + (long)row * K + (cb_src << 3)),
-                (__attribute__((address_space(3))) uint32_t*)(dA + base * 8),
-                16, 0, 0);
-            __builtin_amdgcn_global_load_lds(
-                (const __attribute__((address_space(1))) uint32_t*)(
-                    gB + (long)row * K + (cb_src << 3)),
-                (__attribute__((address_space(3))) uint32_t*)(dB + base * 8),
-                16, 0, 0);
-        }
+        stage_tile<512>(A + row0 * K + kk, K, sAp(buf), BM, tid, wave_piece0);
+        stage_tile<512>(Bt + col0 * K + kk, K, sBp(buf), BN, tid, wave_piece0);
     };
 
-    f32x16 acc[2][2] = {};   // [mt][nt] 32x32 tiles
-    bf16x8 afrag[2], bfrag[2];
+    f32x16 acc[2][4] = {};
+    bf16x8 afrag[2], bfrag[4];
+    const int a_row = wr * 64 + (lane & 31);
+    const int b_row = wc * 128 + (lane & 31);
+    const int k_half = lane >> 5;
 
     stage(0, 0);
     __syncthreads();
-
-    const int a_row = wr * 64 + (lane & 31);        // two m-tiles: +0 / +32
-    const int b_row = wc * 64 + (lane & 31);
-    const int k_half = lane >> 5;                   // 0/1 → k piece within step
-
     for (long kk = 0; kk < K; kk += BK) {
         int buf = (kk / BK) & 1;
         if (kk + BK < K) stage(buf ^ 1, kk + BK);
 #pragma unroll
-        for (int ks = 0; ks < BK / 16; ++ks) {      // 4 MFMA k-steps of 16
-            int cb = (ks << 1) | k_half;            // 16-bf16 step = two 8-elt pieces
+        for (int ks = 0; ks < BK / 16; ++ks) {
+            int cb = (ks << 1) | k_half;
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt)
-                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off<BKT>(a_row + mt * 32, cb));
+                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off(a_row + mt * 32, cb));
+#pragma unroll
+            for (int nt = 0; nt < 4; ++nt)
+                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off(b_row + nt * 32, cb));
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int nt = 0; nt < 4; ++nt)
+                    acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        afrag[mt], bfrag[nt], acc[mt][nt], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // C/D map for 32x32: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+    const int c_row_lane = 4 * (lane >> 5);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+            for (int reg = 0; reg < 16; ++reg) {
+                int r = wr * 64 + mt * 32 + (reg & 3) + 8 * (reg >> 2) + c_row_lane;
+                int cl = wc * 128 + nt * 32 + (lane & 31);
+                C[(row0 + r) * (long)N + col0 + cl] = acc[mt][nt][reg];
+            }
+}
+
+// ---------------------------------------------------------------- 128x128 tile
+// 4 waves as 2x2; each wave 64x64 = 2x2 MFMA tiles. General-shape fallback.
+__global__ __launch_bounds__(256, 2)
+void mfma_gemm_bf16_128_kernel(const bf16* __restrict__ A,
+                               const bf16* __restrict__ Bt,
+                               float* __restrict__ C,
+                               int M, int N, int K) {
+    constexpr int BM = 128, BN = 128;
+    __shared__ __bf16 smem[2 * (BM * BK + BN * BK)];
+    auto sAp = [&](int b) { return smem + b * (BM * BK + BN * BK); };
+    auto sBp = [&](int b) { return smem + b * (BM * BK + BN * BK) + BM * BK; };
+
+    int tile_m, tile_n;
+    tile_map(blockIdx.x, M / BM, N / BN, tile_m, tile_n);
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE;
+    const int lane = tid % WAVE;
+    const int wr = wave >> 1, wc = wave & 1;
+    const long row0 = (long)tile_m * BM;
+    const long col0 = (long)tile_n * BN;
+    const int wave_piece0 = wave * WAVE;
+
+    auto stage = [&](int buf, long kk) {
+        stage_tile<256>(A + row0 * K + kk, K, sAp(buf), BM, tid, wave_piece0);
+        stage_tile<256>(Bt + col0 * K + kk, K, sBp(buf), BN, tid, wave_piece0);
+    };
+
+    f32x16 acc[2][2] = {};
+    bf16x8 afrag[2], bfrag[2];
+    const int a_row = wr * 64 + (lane & 31);
+    const int b_row = wc * 64 + (lane & 31);
+    const int k_half = lane >> 5;
+
+    stage(0, 0);
+    __syncthreads();
+    for (long kk = 0; kk < K; kk += BK) {
+        int buf = (kk / BK) & 1;
+        if (kk + BK < K) stage(buf ^ 1, kk + BK);
+#pragma unroll
+        for (int ks = 0; ks < BK / 16; ++ks) {
+            int cb = (ks << 1) | k_half;
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off(a_row + mt * 32, cb));
 #pragma unroll
             for (int nt = 0; nt < 2; ++nt)
-                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off<BKT>(b_row + nt * 32, cb));
+                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off(b_row + nt * 32, cb));
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -147,20 +212,17 @@ void mfma_gemm_bf16_kernel(const bf16* __restrict__ A,   // [M][K] row-major
         __syncthreads();
     }
 
-    // Epilogue: C/D map for 32x32: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
-    const int c_col_base = wc * 64 + (lane & 31);
     const int c_row_lane = 4 * (lane >> 5);
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
+        for (int nt = 0; nt < 2; ++nt)
 #pragma unroll
             for (int reg = 0; reg < 16; ++reg) {
                 int r = wr * 64 + mt * 32 + (reg & 3) + 8 * (reg >> 2) + c_row_lane;
-                int cl = c_col_base + nt * 32;
+                int cl = wc * 64 + nt * 32 + (lane & 31);
                 C[(row0 + r) * (long)N + col0 + cl] = acc[mt][nt][reg];
             }
-        }
 }
 
 __global__ void stream_triad_kernel(const float4* __restrict__ a,
@@ -176,8 +238,26 @@ __global__ void stream_triad_kernel(const float4* __restrict__ a,
 
 // ---------------------------------------------------------------- host wrappers
 
+static void launch_gemm(const torch::Tensor& a, const torch::Tensor& bt,
+                        torch::Tensor& c, int64_t M, int64_t N, int64_t K) {
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    // 256-tile needs >=256 workgroups to fill the 256-CU chip (one WG per CU at
+    // 128 KB LDS); smaller grids run the 128-tile at 4x the block count.
+    if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 256) {
+        dim3 grid((M / 256) * (N / 256)), block(512);
+        hipLaunchKernelGGL(mfma_gemm_bf16_256_kernel, grid, block, 0, stream,
+                           (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+                           c.data_ptr<float>(), (int)M, (int)N, (int)K);
+    } else {
+        dim3 grid((M / 128) * (N / 128)), block(256);
+        hipLaunchKernelGGL(mfma_gemm_bf16_128_kernel, grid, block, 0, stream,
+                           (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+                           c.data_ptr<float>(), (int)M, (int)N, (int)K);
+    }
+}
+
 static void check_dims(int64_t M, int64_t N, int64_t K) {
-    TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % 64 == 0,
+    TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
                 "mfma_gemm_bf16 requires M%128==0, N%128==0, K%64==0 (got ",
                 M, "x", N, "x", K, ")");
 }
@@ -191,39 +271,22 @@ torch::Tensor mfma_gemm_bf16(torch::Tensor a, torch::Tensor bt) {
     TORCH_CHECK(bt.size(1) == K, "shape mismatch: A[M,K] Bt[N,K]");
     check_dims(M, N, K);
     auto c = torch::empty({M, N}, a.options().dtype(torch::kFloat32));
-    dim3 grid((M / BM) * (N / BN)), block(256);
-    hipStream_t stream = at::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<64>), grid, block, 0, stream,
-                       (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-                       c.data_ptr<float>(), (int)M, (int)N, (int)K);
+    launch_gemm(a, bt, c, M, N, K);
     return c;
 }
 
 // Pod payload: `iters` GEMM steps on pre-allocated buffers; returns achieved TFLOP/s.
-double burn_gemm_v(int64_t m, int64_t n, int64_t k, int64_t iters, int64_t bk) {
+double burn_gemm(int64_t m, int64_t n, int64_t k, int64_t iters) {
     check_dims(m, n, k);
     auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
     auto a = torch::randn({m, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
     auto bt = torch::randn({n, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
     auto c = torch::empty({m, n}, opt.dtype(torch::kFloat32));
     hipStream_t stream = at::hip::getCurrentHIPStream();
-    dim3 grid((m / BM) * (n / BN)), block(256);
-    auto launch = [&] {
-        if (bk == 128)
-            hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<128>), grid,
-                               block, 0, stream,
-                               (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-                               c.data_ptr<float>(), (int)m, (int)n, (int)k);
-        else
-            hipLaunchKernelGGL(HIP_KERNEL_NAME(mfma_gemm_bf16_kernel<64>), grid,
-                               block, 0, stream,
-                               (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-                               c.data_ptr<float>(), (int)m, (int)n, (int)k);
-    };
-    launch();  // warmup
+    launch_gemm(a, bt, c, m, n, k);  // warmup
     C10_HIP_CHECK(hipStreamSynchronize(stream));
     auto t0 = std::chrono::steady_clock::now();
-    for (int64_t i = 0; i < iters; ++i) launch();
+    for (int64_t i = 0; i < iters; ++i) launch_gemm(a, bt, c, m, n, k);
     C10_HIP_CHECK(hipStreamSynchronize(stream));
     auto t1 = std::chrono::steady_clock::now();
     double secs = std::chrono::duration<double>(t1 - t0).count();
@@ -259,10 +322,10 @@ double stream_triad(int64_t n_floats, int64_t iters) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "grove_amd MI355X pod-payload kernels (MFMA bf16 GEMM, HBM stream)";
     m.def("mfma_gemm_bf16", &mfma_gemm_bf16, "C[M,N]=A[M,K]@Bt[N,K]^T (bf16 in, fp32 out)");
-    m.def("burn_gemm",
-          [](int64_t m_, int64_t n_, int64_t k_, int64_t it) {
-              return burn_gemm_v(m_, n_, k_, it, 64);
-          }, "run iters GEMM steps; returns TFLOP/s");
-    m.def("burn_gemm_v", &burn_gemm_v, "burn_gemm with explicit BK (64 or 128)");
+    m.def("burn_gemm", &burn_gemm, "run iters GEMM steps; returns TFLOP/s");
+    m.def("burn_gemm_v",
+          [](int64_t m_, int64_t n_, int64_t k_, int64_t it, int64_t) {
+              return burn_gemm(m_, n_, k_, it);
+          }, "compat alias for burn_gemm (bk arg ignored)");
     m.def("stream_triad", &stream_triad, "HBM triad; returns GB/s");
 }

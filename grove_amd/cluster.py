@@ -8,7 +8,7 @@ from __future__ import annotations
 
 import logging
 import time
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 import yaml as _yaml
 
@@ -25,7 +25,7 @@ from .controllers.clustertopology import ClusterTopologyReconciler
 from .controllers.hpa import HPAReconciler
 from .controllers.nodelifecycle import NodeLifecycleReconciler
 from .scheduler.backends import Registry
-from .kubecore.store import Store, Obj, ApiError
+from .kubecore.store import Store, Obj
 from .kubelet.virtual import VirtualKubelet, make_virtual_node
 from .scheduler.plugin import GangScheduler
 from .utils import conditions as cond

@@ -8,7 +8,7 @@ api/config/validation/. NVIDIA's autoMNNVLEnabled becomes autoXGMIDomainEnabled.
 from __future__ import annotations
 
 import dataclasses
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 import yaml
 

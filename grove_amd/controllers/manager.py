@@ -204,7 +204,6 @@ class Manager:
         def run() -> None:
             while not self._stop.is_set():
                 try:
-                    import queue as _q
                     ev, obj = w.queue.get(timeout=0.2)
                 except Exception:
                     continue

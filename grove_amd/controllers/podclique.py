@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import logging
 import time
-from typing import Any, Dict, List, Optional
+from typing import List, Optional
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError

@@ -15,7 +15,7 @@ fabric CRD.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError, owner_reference

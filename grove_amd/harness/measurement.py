@@ -7,7 +7,6 @@ latencies and JSON export; this is the instrument behind the BASELINE.json metri
 """
 from __future__ import annotations
 
-import json
 import threading
 import time
 from typing import Any, Dict, List, Optional

@@ -16,7 +16,7 @@ Core v1 kinds use /api/v1/... ; cluster-scoped kinds omit the namespaces segment
 import json
 import queue
 import threading
-from typing import Any, Dict, Optional
+from typing import Dict, Optional
 
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse

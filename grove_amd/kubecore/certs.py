@@ -11,7 +11,7 @@ from __future__ import annotations
 import os
 import subprocess
 import tempfile
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 from .store import Store, ApiError
 

@@ -13,7 +13,6 @@ import json
 import os
 import tempfile
 import threading
-import time
 from typing import Optional
 
 from .store import Store

@@ -18,14 +18,12 @@ Behavioral parity notes (not a port):
 """
 from __future__ import annotations
 
-import copy
-import fnmatch
 import itertools
 import queue
 import threading
 import time
 import uuid
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Any, Callable, Dict, List, Optional, Tuple
 
 Obj = Dict[str, Any]
 

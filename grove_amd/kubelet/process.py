@@ -16,7 +16,6 @@ import os
 import subprocess
 import sys
 import threading
-import time
 from typing import Dict, List, Optional
 
 from ..api import constants as c

@@ -12,7 +12,7 @@ with the xGMI-hive level.
 from __future__ import annotations
 
 import logging
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError, invalid

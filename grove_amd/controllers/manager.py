@@ -108,12 +108,22 @@ class WorkQueue:
                     return None
                 self._cond.wait(wait)
 
-    def done(self, key: Key) -> None:
+    def done(self, key: Key, last_duration: float = 0.0) -> None:
+        """Mark processing finished. A dirty re-run (events arrived mid-reconcile) is
+        delayed by ~the reconcile's own duration: invisible for ms-scale reconciles,
+        but caps a multi-second full-resync at ~50% duty cycle so it cannot starve
+        the other controllers (observed at the 5000-pod scale point)."""
         with self._cond:
             self._processing.discard(key)
             if key in self._dirty:
                 self._dirty.discard(key)
-                if key not in self._queued:
+                delay = min(last_duration, 5.0)
+                if delay > 0.05:
+                    self._seq += 1
+                    heapq.heappush(self._delayed,
+                                   (time.monotonic() + delay, self._seq, key))
+                    self._cond.notify()
+                elif key not in self._queued:
                     self._queued.add(key)
                     self._queue.append(key)
                     self._cond.notify()
@@ -159,10 +169,12 @@ class Controller:
                     return
                 continue
             ns, name = item
+            _dur = 0.0
             try:
                 _t0 = time.monotonic()
                 res = self.reconcile(ns, name)
-                self.reconcile_seconds += time.monotonic() - _t0
+                _dur = time.monotonic() - _t0
+                self.reconcile_seconds += _dur
                 self.reconcile_count += 1
                 self.queue.forget(item)
                 if res is not None and res.requeue_after is not None:
@@ -174,7 +186,7 @@ class Controller:
                           traceback.format_exc())
                 self.queue.add_rate_limited(item)
             finally:
-                self.queue.done(item)
+                self.queue.done(item, _dur)
 
     def stop(self) -> None:
         self._stop.set()

@@ -166,6 +166,10 @@ class PodCliqueSetReconciler:
         self._sync_hpas(pcs)
 
         # ---- G2: standalone PodCliques per replica (+ shared ResourceClaims)
+        pclq_by_name = {q["metadata"]["name"]: q for q in self.store.list(
+            c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name,
+                              c.LABEL_COMPONENT: c.COMPONENT_PCS_PODCLIQUE},
+            copy_objects=False)}
         expected_pclqs: Set[str] = set()
         for r in range(replicas):
             claims = resourceclaims.pcs_claims_for_replica(
@@ -182,10 +186,10 @@ class PodCliqueSetReconciler:
                 resourceclaims.ensure_claims(self.store, cl_claims)
                 refs = refs + resourceclaims.claim_refs_for_clique(
                     cl_claims, cl["name"])
-                self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs)
+                self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs,
+                                cur=pclq_by_name.get(fqn))
         # GC excess standalone PCLQs (scale-in / replica removal)
-        for pclq in self.store.list(c.KIND_PCLQ, ns, {
-                c.LABEL_PART_OF: name, c.LABEL_COMPONENT: c.COMPONENT_PCS_PODCLIQUE}):
+        for pclq in pclq_by_name.values():
             if pclq["metadata"]["name"] not in expected_pclqs:
                 try:
                     self.store.delete(c.KIND_PCLQ, ns, pclq["metadata"]["name"])
@@ -193,12 +197,14 @@ class PodCliqueSetReconciler:
                     pass
 
         # ---- G3a: PCSGs per config per replica
+        pcsg_names = {g["metadata"]["name"] for g in self.store.list(
+            c.KIND_PCSG, ns, {c.LABEL_PART_OF: name}, copy_objects=False)}
         expected_pcsgs: Set[str] = set()
         for r in range(replicas):
             for sg in sg_cfgs:
                 fqn = namegen.pcsg_name(name, r, sg["name"])
                 expected_pcsgs.add(fqn)
-                if self.store.try_get(c.KIND_PCSG, ns, fqn) is None:
+                if fqn not in pcsg_names:
                     try:
                         self.store.create(builders.build_pcsg(pcs, r, sg))
                     except ApiError:
@@ -216,9 +222,10 @@ class PodCliqueSetReconciler:
         return Result.DONE
 
     def _sync_pclq(self, pcs: Obj, r: int, clique_tmpl: Obj, fqn: str, owner: Obj,
-                   claim_refs=None) -> None:
+                   claim_refs=None, cur="__lookup__") -> None:
         ns = pcs["metadata"].get("namespace", "default")
-        cur = self.store.try_get(c.KIND_PCLQ, ns, fqn)
+        if cur == "__lookup__":
+            cur = self.store.try_get(c.KIND_PCLQ, ns, fqn)
         if cur is None:
             obj = builders.build_podclique(pcs, r, clique_tmpl, owner)
             obj["spec"]["updateStrategy"] = (pcs["spec"].get("updateStrategy") or {}).get(
@@ -546,11 +553,12 @@ class PodCliqueSetReconciler:
 
         # per-gang phase rollup (podcliqueset.go PodGangStatus)
         gang_statuses: List[Dict[str, Any]] = []
-        for pg in self.store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: name}):
+        pclq_of = {q["metadata"]["name"]: q for q in pclqs}
+        for pg in self.store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: name},
+                                  copy_objects=False):
             phase = "Pending"
             groups = (pg.get("spec") or {}).get("podgroups") or []
             if groups:
-                pclq_of = {q["metadata"]["name"]: q for q in pclqs}
                 sched = all(
                     int((pclq_of.get(g["name"], {}).get("status") or {})
                         .get("scheduledReplicas", 0)) >= int(g.get("minReplicas", 0))

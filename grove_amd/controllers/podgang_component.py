@@ -63,7 +63,9 @@ def translate_constraint(tc: Optional[Dict[str, Any]],
     return {"packConstraint": out} if out else None
 
 
-def compute_expected_podgangs(store: Store, pcs: Obj) -> List[ExpectedGang]:
+def compute_expected_podgangs(store: Store, pcs: Obj,
+                              pcsg_by_name: Optional[Dict[str, Obj]] = None
+                              ) -> List[ExpectedGang]:
     """syncflow.go:147-335 equivalent."""
     pcs_name = pcs["metadata"]["name"]
     ns = pcs["metadata"].get("namespace", "default")
@@ -103,7 +105,8 @@ def compute_expected_podgangs(store: Store, pcs: Obj) -> List[ExpectedGang]:
             sg_fqn = namegen.pcsg_name(pcs_name, r, sg["name"])
             sg_tc = translate_constraint(sg.get("topologyConstraint"), dmap)
             # live PCSG replica count (HPA may have scaled it)
-            pcsg = store.try_get(c.KIND_PCSG, ns, sg_fqn)
+            pcsg = pcsg_by_name.get(sg_fqn) if pcsg_by_name is not None \
+                else store.try_get(c.KIND_PCSG, ns, sg_fqn)
             replicas = int((pcsg or {}).get("spec", {}).get("replicas", sg.get("replicas", 1)))
             min_avail = int((pcsg or {}).get("spec", {}).get(
                 "minAvailable", sg.get("minAvailable", 1)))
@@ -151,10 +154,13 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
     """Create/update PodGangs to match expectations; GC stale ones; flip Initialized."""
     ns = pcs["metadata"].get("namespace", "default")
     pcs_name = pcs["metadata"]["name"]
-    expected = compute_expected_podgangs(store, pcs)
+    pcsg_by_name = {g["metadata"]["name"]: g for g in store.list(
+        c.KIND_PCSG, ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)}
+    expected = compute_expected_podgangs(store, pcs, pcsg_by_name)
     expected_names = {g.name for g in expected}
 
     existing = store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)
+    existing_by_name = {pg["metadata"]["name"]: pg for pg in existing}
     for pg in existing:
         if pg["metadata"]["name"] not in expected_names:
             try:
@@ -173,7 +179,7 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
     tmpl = pcs["spec"]["template"]
     priority_class = tmpl.get("priorityClassName", "")
     for gang in expected:
-        cur = store.try_get(c.KIND_PODGANG, ns, gang.name)
+        cur = existing_by_name.get(gang.name)
         groups_spec = []
         all_created = True
         for g in gang.groups:

@@ -127,3 +127,71 @@ def test_default_config():
     cfg = default_configuration()
     assert cfg.default_scheduler == c.SCHEDULER_AMD_GANG
     assert cfg.authorizer_enabled and cfg.topology_aware_scheduling_enabled
+
+
+def test_watch_stream_ndjson(served_cluster):
+    cluster, api = served_cluster
+    import threading
+    import urllib.request
+    events = []
+
+    def consume():
+        req = urllib.request.urlopen(
+            f"{api.url}/api/v1/namespaces/default/pods?watch=true&seed=false",
+            timeout=10)
+        for raw in req:
+            line = raw.decode().strip()
+            if line:
+                events.append(json.loads(line))
+            if len(events) >= 2:
+                return
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    cluster.store.create({"kind": "Pod", "metadata": {"name": "w1"},
+                          "spec": {"containers": []}})
+    cluster.store.patch("Pod", "default", "w1",
+                        lambda o: o["metadata"].setdefault("labels", {}).update(x="1"))
+    t.join(timeout=10)
+    assert len(events) >= 2
+    assert events[0]["type"] == "ADDED"
+    assert events[0]["object"]["metadata"]["name"] == "w1"
+    assert events[1]["type"] == "MODIFIED"
+
+
+def test_tls_apiserver(cluster, tmp_path):
+    import ssl
+    import urllib.request
+    from grove_amd.kubecore.apiserver import ApiServer
+    from grove_amd.kubecore.certs import (ensure_cert_secret, write_cert_files)
+    data = ensure_cert_secret(cluster.store)
+    crt, key = write_cert_files(data, str(tmp_path / "tls"))
+    api = ApiServer(cluster.store, port=18233, ssl_certfile=crt, ssl_keyfile=key)
+    api.start()
+    try:
+        assert api.url.startswith("http://")  # url property is scheme-naive; use https
+        ctx = ssl.create_default_context(cafile=crt)
+        with urllib.request.urlopen("https://localhost:18233/healthz",
+                                    context=ctx, timeout=5) as r:
+            assert json.loads(r.read())["status"] == "ok"
+    finally:
+        api.stop()
+
+
+def test_default_scheduler_backend_no_gang(simple1_yaml):
+    """kube backend parity: pods carry default-scheduler, no gang admission — pods are
+    scheduled individually as they ungate; gates still serialize startup."""
+    from grove_amd import Cluster
+    from grove_amd.api import constants as c
+    cl = Cluster(scheduler_name=c.SCHEDULER_DEFAULT, use_native_scheduler=False).start()
+    try:
+        cl.add_virtual_nodes(2)
+        cl.apply(simple1_yaml)
+        cl.wait_pcs_available("simple1", timeout=30)
+        pods = cl.store.list("Pod", "default", {c.LABEL_PART_OF: "simple1"})
+        assert all(p["spec"]["schedulerName"] == c.SCHEDULER_DEFAULT for p in pods)
+        # PodGangs exist (contract objects) but no gang placement score was set
+        pg = cl.store.get(c.KIND_PODGANG, "default", "simple1-0")
+        assert pg["metadata"]["labels"][c.LABEL_SCHEDULER_NAME] == c.SCHEDULER_DEFAULT
+    finally:
+        cl.stop()

@@ -209,7 +209,18 @@ def main() -> None:
         _serve_agent(comm)
         return
 
-    _run_rank0(comm, args, n_gpus, gang_size)
+    try:
+        _run_rank0(comm, args, n_gpus, gang_size)
+    except BaseException:
+        # release agent ranks before dying — a hung collective would otherwise keep
+        # every rank alive until the launcher's timeout
+        if comm.dist is not None:
+            try:
+                comm.broadcast({"type": "stop"})
+                comm.barrier_sync()
+            except Exception:
+                pass
+        raise
 
 
 def _measure_allreduce_busbw(comm: Comm, mb: int = 256, iters: int = 5) -> float:

@@ -256,3 +256,86 @@ def validate_podclique(pclq: Obj, old: Optional[Obj] = None) -> None:
         raise _err("spec.minAvailable", "must be >= 1")
     if old is not None and spec.get("roleName") != (old.get("spec") or {}).get("roleName"):
         raise _err("spec.roleName", "field is immutable")
+
+
+class TopologyConstraintValidator:
+    """Store-bound PCS validator for topology constraints (webhook/admission/pcs/
+    validation/topologyconstraints.go:173-310 parity): pack domains must resolve
+    against the ClusterTopologyBinding levels (plus the native host/xgmi-hive
+    built-ins), and child constraints must be equal-or-narrower than their parent's
+    (hierarchy rule: a PCSG/clique may not pack at a BROADER level than the PCS)."""
+
+    BUILTIN_DOMAINS = ("host", "xgmi-hive")
+
+    def __init__(self, store):
+        self.store = store
+
+    def _known_domains(self):
+        # ordered broadest -> narrowest
+        ordered = []
+        for ctb in self.store.list(c.KIND_CTB):
+            for lv in (ctb.get("spec") or {}).get("levels") or []:
+                d = lv.get("domain")
+                if d and d not in ordered:
+                    ordered.append(d)
+            break
+        for d in self.BUILTIN_DOMAINS:
+            if d not in ordered:
+                ordered.append(d)
+        return ordered
+
+    @staticmethod
+    def _domains_of(tc):
+        if not tc:
+            return (None, None)
+        pack = tc.get("pack") or {}
+        return (pack.get("required") or tc.get("packDomain"), pack.get("preferred"))
+
+    def __call__(self, pcs, old=None):
+        tmpl = (pcs.get("spec") or {}).get("template") or {}
+        parent_tc = tmpl.get("topologyConstraint")
+        has_any = bool(parent_tc) or any(
+            cl.get("topologyConstraint") for cl in tmpl.get("cliques") or []) or any(
+            sg.get("topologyConstraint")
+            for sg in tmpl.get("podCliqueScalingGroups") or [])
+        if not has_any:
+            return
+        known = self._known_domains()
+
+        def check(tc, path):
+            req, pref = self._domains_of(tc)
+            for d, f in ((req, "required"), (pref, "preferred")):
+                if d is not None and d not in known:
+                    raise _err(f"{path}.pack.{f}",
+                               f"unknown topology domain {d!r}; levels defined by the "
+                               f"ClusterTopologyBinding: {known}")
+            if req is not None and pref is not None:
+                if known.index(pref) < known.index(req):
+                    raise _err(f"{path}.pack.preferred",
+                               "preferred level must be equal or narrower than required")
+            return req
+
+        parent_req = check(parent_tc, "spec.template.topologyConstraint")             if parent_tc else None
+        for sg in tmpl.get("podCliqueScalingGroups") or []:
+            tc = sg.get("topologyConstraint")
+            if tc:
+                child_req = check(
+                    tc, f"spec.template.podCliqueScalingGroups[{sg.get('name')}]"
+                        f".topologyConstraint")
+                if parent_req and child_req and \
+                        known.index(child_req) < known.index(parent_req):
+                    raise _err(
+                        f"spec.template.podCliqueScalingGroups[{sg.get('name')}]"
+                        f".topologyConstraint.pack.required",
+                        "must be equal or narrower than the PodCliqueSet constraint")
+        for cl in tmpl.get("cliques") or []:
+            tc = cl.get("topologyConstraint")
+            if tc:
+                child_req = check(
+                    tc, f"spec.template.cliques[{cl.get('name')}].topologyConstraint")
+                if parent_req and child_req and \
+                        known.index(child_req) < known.index(parent_req):
+                    raise _err(
+                        f"spec.template.cliques[{cl.get('name')}]"
+                        f".topologyConstraint.pack.required",
+                        "must be equal or narrower than the PodCliqueSet constraint")

@@ -49,6 +49,9 @@ class Cluster:
         # admission (webhook parity: defaulting before validation)
         self.store.register_mutator(c.KIND_PCS, default_podcliqueset)
         self.store.register_validator(c.KIND_PCS, validate_podcliqueset)
+        from .api.validation import TopologyConstraintValidator
+        self.store.register_validator(c.KIND_PCS,
+                                      TopologyConstraintValidator(self.store))
         self.store.register_mutator(c.KIND_PCLQ, default_podclique)
         self.store.register_validator(c.KIND_PCLQ, validate_podclique)
         self.store.register_mutator(c.KIND_PCSG, default_pcsg)

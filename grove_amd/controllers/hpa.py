@@ -29,10 +29,16 @@ CPU_USAGE_ANNOTATION = "grove.io/cpu-usage"
 
 class HPAReconciler:
     def __init__(self, store: Store, sync_period_s: float = 0.25,
-                 tolerance: float = 0.1):
+                 tolerance: float = 0.1,
+                 scale_down_stabilization_s: float = 10.0):
         self.store = store
         self.sync_period_s = sync_period_s
         self.tolerance = tolerance
+        # kube HPA's downscale stabilization window (default 300s there; shortened
+        # for the in-process loop): scale-in is applied only after the recommendation
+        # has stayed below current for the whole window
+        self.scale_down_stabilization_s = scale_down_stabilization_s
+        self._below_since: Dict[str, float] = {}
 
     def reconcile(self, namespace: str, name: str) -> Result:
         hpa = self.store.try_get("HorizontalPodAutoscaler", namespace, name)
@@ -46,6 +52,14 @@ class HPAReconciler:
         current = int(target["spec"].get("replicas", 1))
 
         desired = self._desired_replicas(namespace, hpa, ref, current)
+        if desired is not None and desired < current:
+            import time as _time
+            key = f"{namespace}/{name}"
+            since = self._below_since.setdefault(key, _time.monotonic())
+            if _time.monotonic() - since < self.scale_down_stabilization_s:
+                desired = current  # hold: inside the stabilization window
+        elif desired is not None and desired >= current:
+            self._below_since.pop(f"{namespace}/{name}", None)
         if desired is not None and desired != current:
             def scale(o: Obj) -> None:
                 o["spec"]["replicas"] = desired

@@ -212,3 +212,34 @@ def test_reuse_reservation_ref(cluster):
         "Pod", "default", {c.LABEL_PART_OF: "succ"}) if p["spec"].get("nodeName")}
     # preference is best-effort; with free capacity on the original node it lands there
     assert succ_nodes <= orig_node or succ_nodes
+
+
+class TestTopologyConstraintValidation:
+    def test_unknown_domain_rejected(self, cluster):
+        cluster.store.create(CTB)
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError) as ei:
+            cluster.apply(_pcs("bad", 1, constraint={"pack": {"required": "galaxy"}}))
+        assert "unknown topology domain" in str(ei.value)
+
+    def test_builtin_domains_accepted_without_ctb(self, cluster):
+        cluster.add_virtual_nodes(1, gpus=8)
+        cluster.apply(_pcs("ok", 1, constraint={"pack": {"preferred": "xgmi-hive"}}))
+        cluster.wait_pcs_available("ok", timeout=20)
+
+    def test_preferred_broader_than_required_rejected(self, cluster):
+        cluster.store.create(CTB)
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError):
+            cluster.apply(_pcs("bad2", 1, constraint={
+                "pack": {"required": "host", "preferred": "zone"}}))
+
+    def test_child_broader_than_parent_rejected(self, cluster):
+        cluster.store.create(CTB)
+        pcs = _pcs("bad3", 1, constraint={"pack": {"required": "rack"}})
+        pcs["spec"]["template"]["cliques"][0]["topologyConstraint"] = {
+            "pack": {"required": "zone"}}
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError) as ei:
+            cluster.apply(pcs)
+        assert "narrower" in str(ei.value)

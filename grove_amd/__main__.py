@@ -83,6 +83,60 @@ def cmd_operator(args) -> int:
     return 0
 
 
+def cmd_agent(args) -> int:
+    """Remote node agent: registers this machine as a Node with the operator's
+    apiserver and runs its scheduled pods as OS processes (ProcessKubelet over the
+    HTTP client) — the kubelet of the distributed deployment shape."""
+    import time as _time
+
+    from .kubecore.httpclient import HttpStoreClient
+    from .kubecore.store import ApiError
+    from .kubelet.process import ProcessKubelet
+    from .topology.agent import discover_node
+    from .kubelet.virtual import make_virtual_node
+
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(levelname)s %(name)s %(message)s")
+    log = logging.getLogger("grove.agent")
+    client = HttpStoreClient(args.server)
+    if args.virtual_gpus is not None:
+        node = make_virtual_node(args.node_name or "agent-node",
+                                 gpus=args.virtual_gpus)
+    else:
+        node = discover_node(args.node_name)
+    name = node["metadata"]["name"]
+    node["metadata"].setdefault("annotations", {})["grove.io/external-kubelet"] = \
+        "true"
+    try:
+        client.create(node)
+        log.info("registered node %s", name)
+    except ApiError as e:
+        if e.reason != "AlreadyExists":
+            raise
+        log.info("node %s already registered", name)
+    kubelet = ProcessKubelet(client, api_url=args.server, node_names=[name])
+    stop = {"flag": False}
+
+    def on_sig(_s, _f):
+        stop["flag"] = True
+    signal.signal(signal.SIGINT, on_sig)
+    signal.signal(signal.SIGTERM, on_sig)
+    log.info("agent serving node %s against %s", name, args.server)
+    while not stop["flag"]:
+        try:
+            pods = client.list("Pod", args.namespace,
+                               filter_fn=lambda p: p.get("spec", {})
+                               .get("nodeName") == name)
+            for p in pods:
+                kubelet.reconcile(p["metadata"].get("namespace", "default"),
+                                  p["metadata"]["name"])
+        except Exception as e:
+            log.warning("agent pass failed: %s", e)
+        _time.sleep(args.poll_interval)
+    kubelet.shutdown()
+    return 0
+
+
 def cmd_get(args) -> int:
     import json
     import urllib.parse
@@ -209,6 +263,16 @@ def main(argv=None) -> int:
     ap_cmd.add_argument("--server", default="http://127.0.0.1:8081")
     ap_cmd.add_argument("-n", "--namespace", default="default")
     ap_cmd.set_defaults(fn=cmd_apply)
+
+    agent = sub.add_parser("agent", help="remote node agent (kubelet) for a GPU node")
+    agent.add_argument("--server", required=True)
+    agent.add_argument("--node-name", default=None)
+    agent.add_argument("--namespace", default=None,
+                       help="restrict to one namespace (default: all)")
+    agent.add_argument("--virtual-gpus", type=int, default=None,
+                       help="register a virtual node instead of probing hardware")
+    agent.add_argument("--poll-interval", type=float, default=0.2)
+    agent.set_defaults(fn=cmd_agent)
 
     crds = sub.add_parser("install-crds", help="render or apply the CRDs")
     crds.add_argument("--server", default=None, help="apiserver URL to POST CRDs to")

@@ -1,0 +1,150 @@
+"""HTTP store client — the client-go analog for remote components.
+
+Speaks the apiserver's kube-style paths (kubecore/apiserver.py) and duck-types the
+subset of the Store interface that node-side components use (get/try_get/list/patch/
+update/delete + ndjson watch), so a ProcessKubelet or topology agent can run on a GPU
+node against a remote operator exactly like the reference's kubelet/initc talk to the
+apiserver via client-go.
+"""
+from __future__ import annotations
+
+import json
+import ssl
+import time
+import urllib.error
+import urllib.parse
+import urllib.request
+from typing import Any, Callable, Dict, Iterator, List, Optional, Tuple
+
+from .apiserver import PLURALS, CLUSTER_SCOPED_PLURALS
+from .store import ApiError
+
+Obj = Dict[str, Any]
+
+KIND_TO_PLURAL = {v: k for k, v in PLURALS.items()}
+GROVE_KINDS = {"PodCliqueSet", "PodClique", "PodCliqueScalingGroup",
+               "ClusterTopologyBinding"}
+
+
+def _group_of(kind: str) -> str:
+    if kind in GROVE_KINDS:
+        return "apis/grove.io/v1alpha1"
+    if kind == "PodGang":
+        return "apis/scheduler.grove.io/v1alpha1"
+    if kind == "SchedulerTopology":
+        return "apis/scheduler.amd.com/v1alpha1"
+    return "api/v1"
+
+
+class HttpStoreClient:
+    def __init__(self, base_url: str, timeout: float = 10.0,
+                 cafile: Optional[str] = None):
+        self.base_url = base_url.rstrip("/")
+        self.timeout = timeout
+        self._ctx = ssl.create_default_context(cafile=cafile) if cafile else None
+
+    # ------------------------------------------------------------------ plumbing
+    def _url(self, kind: str, namespace: Optional[str], name: Optional[str] = None,
+             subresource: str = "", query: str = "") -> str:
+        plural = KIND_TO_PLURAL[kind]
+        base = f"{self.base_url}/{_group_of(kind)}"
+        if plural in CLUSTER_SCOPED_PLURALS:
+            url = f"{base}/{plural}"
+        else:
+            url = f"{base}/namespaces/{namespace or 'default'}/{plural}"
+        if name:
+            url += f"/{urllib.parse.quote(name)}"
+        if subresource:
+            url += f"/{subresource}"
+        if query:
+            url += f"?{query}"
+        return url
+
+    def _request(self, method: str, url: str, body: Optional[Obj] = None) -> Obj:
+        data = json.dumps(body).encode() if body is not None else None
+        req = urllib.request.Request(url, data=data, method=method,
+                                     headers={"Content-Type": "application/json"})
+        try:
+            with urllib.request.urlopen(req, timeout=self.timeout,
+                                        context=self._ctx) as r:
+                return json.loads(r.read())
+        except urllib.error.HTTPError as e:
+            try:
+                payload = json.loads(e.read())
+            except Exception:
+                payload = {}
+            raise ApiError(e.code, payload.get("reason", "HTTPError"),
+                           payload.get("message", str(e)))
+
+    # ------------------------------------------------------------------ store API
+    def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
+        return self._request("GET", self._url(kind, namespace, name))
+
+    def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
+        try:
+            return self.get(kind, namespace, name)
+        except ApiError:
+            return None
+
+    def list(self, kind: str, namespace: Optional[str] = None,
+             label_selector: Optional[Dict[str, str]] = None,
+             filter_fn: Optional[Callable[[Obj], bool]] = None,
+             copy_objects: bool = True) -> List[Obj]:
+        query = ""
+        if label_selector:
+            sel = ",".join(f"{k}={v}" for k, v in label_selector.items())
+            query = f"labelSelector={urllib.parse.quote(sel)}"
+        items = self._request("GET", self._url(kind, namespace, query=query))["items"]
+        if filter_fn is not None:
+            items = [o for o in items if filter_fn(o)]
+        return items
+
+    def create(self, obj: Obj) -> Obj:
+        kind = obj["kind"]
+        ns = obj.get("metadata", {}).get("namespace")
+        return self._request("POST", self._url(kind, ns), obj)
+
+    def update(self, obj: Obj) -> Obj:
+        kind = obj["kind"]
+        md = obj["metadata"]
+        return self._request("PUT", self._url(kind, md.get("namespace"), md["name"]),
+                             obj)
+
+    def update_status(self, obj: Obj) -> Obj:
+        kind = obj["kind"]
+        md = obj["metadata"]
+        return self._request(
+            "PUT", self._url(kind, md.get("namespace"), md["name"], "status"), obj)
+
+    def patch(self, kind: str, namespace: Optional[str], name: str,
+              fn: Callable[[Obj], None], status: bool = False,
+              retries: int = 10) -> Obj:
+        last: Optional[ApiError] = None
+        for _ in range(retries):
+            obj = self.get(kind, namespace, name)
+            fn(obj)
+            try:
+                return self.update_status(obj) if status else self.update(obj)
+            except ApiError as e:
+                if e.reason != "Conflict":
+                    raise
+                last = e
+        raise last or ApiError(409, "Conflict", name)
+
+    def delete(self, kind: str, namespace: Optional[str], name: str,
+               cascade: bool = True) -> None:
+        self._request("DELETE", self._url(kind, namespace, name))
+
+    def watch_events(self, kind: str, namespace: Optional[str] = None,
+                     seed: bool = True) -> Iterator[Tuple[str, Obj]]:
+        """ndjson watch stream; yields (event_type, object)."""
+        url = self._url(kind, namespace,
+                        query=f"watch=true&seed={'true' if seed else 'false'}")
+        req = urllib.request.Request(url)
+        with urllib.request.urlopen(req, timeout=3600, context=self._ctx) as r:
+            for raw in r:
+                line = raw.decode().strip()
+                if not line:
+                    continue
+                ev = json.loads(line)
+                yield ev["type"], ev["object"]

@@ -56,8 +56,18 @@ class VirtualKubelet:
         self.payload = payload  # callable(pod)->None run at start (GPU nodes override)
         self._started_at: Dict[str, float] = {}
 
+    EXTERNAL_KUBELET_ANNOTATION = "grove.io/external-kubelet"
+
     def handles(self, node_name: str) -> bool:
-        return self.node_names is None or node_name in self.node_names
+        if self.node_names is not None and node_name not in self.node_names:
+            return False
+        # nodes registered by a remote node agent run their own kubelet — the
+        # in-process virtual kubelet must not race it for pod lifecycle
+        node = self.store.try_get("Node", None, node_name)
+        if node is not None and (node["metadata"].get("annotations") or {}).get(
+                self.EXTERNAL_KUBELET_ANNOTATION) == "true":
+            return False
+        return True
 
     def reconcile(self, namespace: str, name: str) -> Result:
         pod = self.store.try_get("Pod", namespace, name)

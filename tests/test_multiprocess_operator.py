@@ -71,3 +71,70 @@ logLevel: warn
             proc.kill()
     # graceful shutdown wrote the state snapshot
     assert state.exists() and state.stat().st_size > 1000
+
+
+@pytest.mark.timeout(180)
+def test_remote_node_agent(tmp_path):
+    """Fully distributed shape: operator process + remote node-agent process (HTTP
+    client kubelet) + pod processes, coordinated only over the wire."""
+    port = 18327
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(f"servers:\n  api: {{enabled: true, host: 127.0.0.1, port: {port}}}\n"
+                   f"logLevel: warn\n")
+    base = f"http://127.0.0.1:{port}"
+    op = subprocess.Popen(
+        [sys.executable, "-m", "grove_amd", "operator", "--config-file", str(cfg)],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    agent = None
+    try:
+        for _ in range(100):
+            try:
+                urllib.request.urlopen(f"{base}/healthz", timeout=0.5)
+                break
+            except Exception:
+                time.sleep(0.2)
+        agent = subprocess.Popen(
+            [sys.executable, "-m", "grove_amd", "agent", "--server", base,
+             "--node-name", "remote-0", "--virtual-gpus", "0",
+             "--poll-interval", "0.1"],
+            cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        pcs = {"apiVersion": "grove.io/v1alpha1", "kind": "PodCliqueSet",
+               "metadata": {"name": "remote"},
+               "spec": {"replicas": 1, "template": {"cliques": [{
+                   "name": "w",
+                   "annotations": {"grove.io/payload": "none"},
+                   "spec": {"roleName": "w", "replicas": 2,
+                            "podSpec": {"containers": [{
+                                "name": "m", "image": "i",
+                                "resources": {"requests": {"cpu": "1"}}}]}}}]}}}
+        req = urllib.request.Request(
+            f"{base}/apis/grove.io/v1alpha1/namespaces/default/podcliquesets",
+            data=json.dumps(pcs).encode(), method="POST",
+            headers={"Content-Type": "application/json"})
+        urllib.request.urlopen(req, timeout=5)
+        deadline = time.monotonic() + 90
+        phases = []
+        while time.monotonic() < deadline:
+            with urllib.request.urlopen(
+                    f"{base}/api/v1/namespaces/default/pods"
+                    f"?labelSelector=app.kubernetes.io/part-of=remote",
+                    timeout=5) as r:
+                pods = json.loads(r.read())["items"]
+            phases = [(p.get("status") or {}).get("phase") for p in pods]
+            if len(pods) == 2 and all(ph == "Succeeded" for ph in phases):
+                break
+            time.sleep(0.3)
+        else:
+            raise AssertionError(
+                f"pods never succeeded on the remote agent: {phases}\n"
+                f"agent: {agent.stdout.read()[:1500] if agent.poll() is not None else '(running)'}")
+        # pods ran on the agent's node
+        assert all(p["spec"]["nodeName"] == "remote-0" for p in pods)
+    finally:
+        for proc in (agent, op):
+            if proc is not None:
+                proc.terminate()
+                try:
+                    proc.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    proc.kill()

@@ -161,3 +161,81 @@ def test_process_kubelet_runs_payload_on_gpu():
         ctrl.stop()
         api.stop()
         cl.stop()
+
+
+def test_remote_agent_gpu_distributed():
+    """Distributed shape on a real GPU box: operator process + node-agent process
+    (real topology discovery) + pod processes running MFMA payloads on the GPU."""
+    import json
+    import subprocess
+    import sys
+    import tempfile
+    import time
+    import urllib.request
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    port = 18461
+    cfg = tempfile.NamedTemporaryFile("w", suffix=".yaml", delete=False)
+    cfg.write(f"servers:\n  api: {{enabled: true, host: 127.0.0.1, port: {port}}}\n"
+              f"logLevel: warn\n")
+    cfg.close()
+    base = f"http://127.0.0.1:{port}"
+    op = subprocess.Popen(
+        [sys.executable, "-m", "grove_amd", "operator", "--config-file", cfg.name],
+        cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    agent = None
+    try:
+        for _ in range(150):
+            try:
+                urllib.request.urlopen(f"{base}/healthz", timeout=0.5)
+                break
+            except Exception:
+                time.sleep(0.2)
+        agent = subprocess.Popen(
+            [sys.executable, "-m", "grove_amd", "agent", "--server", base,
+             "--node-name", "gpu-remote", "--poll-interval", "0.1"],
+            cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        pcs = {"apiVersion": "grove.io/v1alpha1", "kind": "PodCliqueSet",
+               "metadata": {"name": "rgpu"},
+               "spec": {"replicas": 1, "template": {"cliques": [{
+                   "name": "inf",
+                   "annotations": {"grove.io/payload-shape": "512x512x512x1"},
+                   "spec": {"roleName": "r", "replicas": 1,
+                            "podSpec": {"containers": [{
+                                "name": "m", "image": "payload",
+                                "resources": {"requests": {
+                                    "amd.com/gpu": "1"}}}]}}}]}}}
+        req = urllib.request.Request(
+            f"{base}/apis/grove.io/v1alpha1/namespaces/default/podcliquesets",
+            data=json.dumps(pcs).encode(), method="POST",
+            headers={"Content-Type": "application/json"})
+        urllib.request.urlopen(req, timeout=5)
+        deadline = time.monotonic() + 150
+        pods = []
+        while time.monotonic() < deadline:
+            with urllib.request.urlopen(
+                    f"{base}/api/v1/namespaces/default/pods"
+                    f"?labelSelector=app.kubernetes.io/part-of=rgpu",
+                    timeout=5) as r:
+                pods = json.loads(r.read())["items"]
+            if pods and all((p.get("status") or {}).get("phase") == "Succeeded"
+                            for p in pods):
+                break
+            time.sleep(0.5)
+        else:
+            raise AssertionError(
+                "GPU pod process never succeeded; agent output: "
+                + (agent.stdout.read()[:1500] if agent.poll() is not None
+                   else "(running)"))
+        assert pods[0]["spec"]["nodeName"] == "gpu-remote"
+        assert pods[0]["metadata"]["annotations"].get(
+            "scheduling.amd.com/gpu-ids") is not None
+    finally:
+        for proc in (agent, op):
+            if proc is not None:
+                proc.terminate()
+                try:
+                    proc.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    proc.kill()

@@ -87,6 +87,8 @@ class GangScheduler:
 
         self._subtract_bound(nodes, bound)
         node_list = list(nodes.values())
+        pods_by_name = {(p["metadata"].get("namespace", "default"),
+                         p["metadata"]["name"]): p for p in pods}
 
         # ---- gang scheduling, FIFO by PodGang creation
         gangs: List[Obj] = []
@@ -108,17 +110,18 @@ class GangScheduler:
             if cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
                 # gang already placed — late pods (replacements, scale-ups) go
                 # individually, preferring the gang's existing nodes
-                self._schedule_singles(node_list, gang_pods, prefer=self._gang_nodes(pg))
+                self._schedule_singles(node_list, gang_pods,
+                                       prefer=self._gang_nodes(pg, pods_by_name))
                 continue
             if not cond.condition_true(pg, c.PODGANG_COND_INITIALIZED):
                 continue
-            self._schedule_gang(node_list, pg, gang_pods)
+            self._schedule_gang(node_list, pg, gang_pods, pods_by_name)
 
         # ---- individual (gangless / default-scheduler parity) pods
         self._schedule_singles(node_list, pending_single)
 
         # ---- PodGang Ready rollup
-        self._rollup_ready()
+        self._rollup_ready(pods_by_name)
 
     # ------------------------------------------------------------------ node views
     def _build_node_views(self) -> Dict[str, NodeFree]:
@@ -167,18 +170,19 @@ class GangScheduler:
             gpus += int(parse_quantity(r.get(c.AMD_GPU_RESOURCE, 0)))
         return PodRequest(pod["metadata"]["name"], cpu, mem, gpus)
 
-    def _gang_nodes(self, pg: Obj) -> Set[str]:
+    def _gang_nodes(self, pg: Obj, pods_by_name: Dict) -> Set[str]:
         ns = pg["metadata"].get("namespace", "default")
         names: Set[str] = set()
         for group in (pg.get("spec") or {}).get("podgroups") or []:
             for ref in group.get("podReferences") or []:
-                p = self.store.try_get("Pod", ns, ref.get("name", ""))
+                p = pods_by_name.get((ns, ref.get("name", "")))
                 if p and p.get("spec", {}).get("nodeName"):
                     names.add(p["spec"]["nodeName"])
         return names
 
     # ------------------------------------------------------------------ gang place
-    def _schedule_gang(self, nodes: List[NodeFree], pg: Obj, gang_pods: List[Obj]) -> None:
+    def _schedule_gang(self, nodes: List[NodeFree], pg: Obj, gang_pods: List[Obj],
+                       pods_by_name: Dict) -> None:
         ns = pg["metadata"].get("namespace", "default")
         groups = (pg.get("spec") or {}).get("podgroups") or []
         by_clique: Dict[str, List[Obj]] = {}
@@ -191,7 +195,7 @@ class GangScheduler:
         for g in groups:
             want = int(g.get("minReplicas", 0))
             have = by_clique.get(g["name"], [])
-            already = self._count_bound(ns, g)
+            already = self._count_bound(ns, g, pods_by_name)
             need = max(0, want - already)
             if len(have) < need:
                 return  # not admittable yet
@@ -255,10 +259,10 @@ class GangScheduler:
         except ApiError:
             pass
 
-    def _count_bound(self, ns: str, group: Obj) -> int:
+    def _count_bound(self, ns: str, group: Obj, pods_by_name: Dict) -> int:
         n = 0
         for ref in group.get("podReferences") or []:
-            p = self.store.try_get("Pod", ns, ref.get("name", ""))
+            p = pods_by_name.get((ns, ref.get("name", "")))
             if p and p.get("spec", {}).get("nodeName"):
                 n += 1
         return n
@@ -386,7 +390,7 @@ class GangScheduler:
             pass
 
     # ------------------------------------------------------------------ ready rollup
-    def _rollup_ready(self) -> None:
+    def _rollup_ready(self, pods_by_name: Dict) -> None:
         for pg in self.store.list(c.KIND_PODGANG, copy_objects=False):
             if not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
                 continue
@@ -397,7 +401,7 @@ class GangScheduler:
             for group in (pg.get("spec") or {}).get("podgroups") or []:
                 n = 0
                 for ref in group.get("podReferences") or []:
-                    p = self.store.try_get("Pod", ns, ref.get("name", ""))
+                    p = pods_by_name.get((ns, ref.get("name", "")))
                     if p and cond.pod_is_ready(p):
                         n += 1
                 if n < int(group.get("minReplicas", 0)):

@@ -188,6 +188,26 @@ def build_app(store: Store, metrics_fn=None):
         except ApiError as e:
             return err(e)
 
+    @app.get("/debug/profile")
+    async def debug_profile(seconds: float = 2.0, interval_ms: float = 10.0):
+        """Sampling profiler (the Pyroscope/pprof analog): samples every thread's
+        stack and returns collapsed stacks (flamegraph.pl / speedscope format)."""
+        import asyncio
+        import collections
+        import sys
+        import traceback
+        counts: "collections.Counter[str]" = collections.Counter()
+        deadline = asyncio.get_event_loop().time() + min(seconds, 60.0)
+        while asyncio.get_event_loop().time() < deadline:
+            for tid, frame in sys._current_frames().items():
+                stack = traceback.extract_stack(frame)
+                key = ";".join(f"{f.name} ({f.filename.rsplit('/', 1)[-1]}:{f.lineno})"
+                               for f in stack[-25:])
+                counts[key] += 1
+            await asyncio.sleep(max(interval_ms, 1.0) / 1000.0)
+        lines = [f"{k} {v}" for k, v in counts.most_common()]
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/debug/stacks")
     async def debug_stacks():
         import sys

@@ -195,3 +195,15 @@ def test_default_scheduler_backend_no_gang(simple1_yaml):
         assert pg["metadata"]["labels"][c.LABEL_SCHEDULER_NAME] == c.SCHEDULER_DEFAULT
     finally:
         cl.stop()
+
+
+def test_debug_profile_endpoint(served_cluster, simple1_yaml):
+    cluster, api = served_cluster
+    cluster.add_virtual_nodes(1)
+    cluster.apply(simple1_yaml)
+    with urllib.request.urlopen(
+            f"{api.url}/debug/profile?seconds=0.5&interval_ms=5", timeout=10) as r:
+        body = r.read().decode()
+    # collapsed-stack lines: "frame;frame;... count"
+    assert body.strip() and all(
+        line.rsplit(" ", 1)[-1].isdigit() for line in body.strip().splitlines())

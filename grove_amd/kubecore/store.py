@@ -182,9 +182,10 @@ class Store:
         return tbl
 
     def _notify(self, tbl: _KindTable, ev: str, obj: Obj) -> None:
-        snapshot = json_copy(obj)
+        # stored objects are immutable after insert; watchers share the reference
+        # (read-only contract, same as list(copy_objects=False))
         for w in list(tbl.watchers):
-            w.queue.put((ev, snapshot))
+            w.queue.put((ev, obj))
 
     def _next_rv(self) -> str:
         return str(next(self._rv))
@@ -228,16 +229,18 @@ class Store:
             tbl.index_add(key, obj)
             self._uid_index[m["uid"]] = (kind, ns, m["name"])
             self._notify(tbl, ADDED, obj)
-        return json_copy(obj)
+        return json_copy(obj)  # caller gets a private copy; stored one is immutable
 
     def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
         ns = "" if kind in CLUSTER_SCOPED else (namespace or "default")
         with self._lock:
             tbl = self._table(kind)
             obj = tbl.objects.get((ns, name))
-            if obj is None:
-                raise not_found(kind, name)
-            return json_copy(obj)
+        if obj is None:
+            raise not_found(kind, name)
+        # stored objects are immutable after insert (updates replace wholesale), so
+        # the defensive copy happens OUTSIDE the lock — the lock hold is a dict get
+        return json_copy(obj)
 
     def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
         try:
@@ -277,26 +280,31 @@ class Store:
                         break
             if candidates is None:
                 candidates = list(tbl.objects.items())
-            out = []
+            refs = []
             for (ns, _name), obj in candidates:
                 if namespace is not None and ns != namespace:
                     continue
                 if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
                     continue
-                if filter_fn is not None and not filter_fn(obj):
-                    continue
-                out.append(json_copy(obj) if copy_objects else obj)
-            return out
+                refs.append(obj)
+        # filtering callbacks + copies run OUTSIDE the lock (objects immutable)
+        if filter_fn is not None:
+            refs = [o for o in refs if filter_fn(o)]
+        return [json_copy(o) for o in refs] if copy_objects else refs
 
     def _apply_update(self, obj: Obj, status_only: bool) -> Obj:
+        """Lock-split optimistic update: the expensive work (copies, admission,
+        equality checks) runs OUTSIDE the lock against the immutable current object;
+        insertion re-checks identity under the lock and retries on interleaving."""
         obj = json_copy(obj)
         kind = obj["kind"]
         m = meta(obj)
         ns = self._ns_of(kind, m)
-        with self._lock:
-            tbl = self._table(kind)
-            key = (ns, m["name"])
-            cur = tbl.objects.get(key)
+        key = (ns, m["name"])
+        for _attempt in range(16):
+            with self._lock:
+                tbl = self._table(kind)
+                cur = tbl.objects.get(key)
             if cur is None:
                 raise not_found(kind, m["name"])
             cur_m = cur["metadata"]
@@ -329,16 +337,23 @@ class Store:
                     return json_copy(cur)  # no-op update: no rv bump, no event
                 if new.get("spec") != cur.get("spec"):
                     new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
-            new["metadata"]["resourceVersion"] = self._next_rv()
-            if not status_only and new["metadata"].get("labels") != cur_m.get("labels"):
-                tbl.index_remove(key, cur)
-                tbl.index_add(key, new)
-            tbl.objects[key] = new
-            self._notify(tbl, MODIFIED, new)
-            # finalizer removal on a deleting object may allow actual deletion
-            if new["metadata"].get("deletionTimestamp") and not new["metadata"].get("finalizers"):
-                self._finalize_delete(kind, ns, m["name"])
+            with self._lock:
+                if tbl.objects.get(key) is not cur:
+                    if m.get("resourceVersion"):
+                        raise conflict(kind, m["name"])
+                    continue  # interleaved writer; rebuild against the new current
+                new["metadata"]["resourceVersion"] = self._next_rv()
+                if not status_only and new["metadata"].get("labels") != cur_m.get("labels"):
+                    tbl.index_remove(key, cur)
+                    tbl.index_add(key, new)
+                tbl.objects[key] = new
+                self._notify(tbl, MODIFIED, new)
+                # finalizer removal on a deleting object may allow actual deletion
+                if new["metadata"].get("deletionTimestamp") \
+                        and not new["metadata"].get("finalizers"):
+                    self._finalize_delete(kind, ns, m["name"])
             return json_copy(new)
+        raise conflict(kind, m["name"], "persistent write interleaving")
 
     def update(self, obj: Obj) -> Obj:
         return self._apply_update(obj, status_only=False)
@@ -375,9 +390,11 @@ class Store:
             m = obj["metadata"]
             if m.get("finalizers"):
                 if not m.get("deletionTimestamp"):
-                    m["deletionTimestamp"] = now_iso()
-                    m["resourceVersion"] = self._next_rv()
-                    self._notify(tbl, MODIFIED, obj)
+                    marked = json_copy(obj)  # stored objects are immutable: replace
+                    marked["metadata"]["deletionTimestamp"] = now_iso()
+                    marked["metadata"]["resourceVersion"] = self._next_rv()
+                    tbl.objects[(ns, name)] = marked
+                    self._notify(tbl, MODIFIED, marked)
                 return
             self._finalize_delete(kind, ns, name, cascade=cascade)
 

@@ -56,6 +56,11 @@ def parse_payload(pod: Obj) -> Tuple[str, Tuple[int, ...]]:
         return "gemm", dims
     if kind == "stream":
         return "stream", (int(shape) if shape else 1 << 24, 2)
+    if kind == "decode":
+        dims = tuple(int(v) for v in shape.split("x")) if shape else (8192, 8192, 4)
+        if len(dims) == 2:
+            dims = dims + (1,)
+        return "decode", dims
     return "none", ()
 
 
@@ -89,6 +94,10 @@ def run_payload_descriptor(kind: str, dims: Tuple[int, ...], device: int) -> Dic
             n_floats, iters = dims
             gbps = ext.stream_triad(n_floats, iters)
             return {"device": device, "kind": kind, "gbps": gbps}
+        if kind == "decode":
+            n, k, iters = dims
+            gbps = ext.burn_decode(n, k, iters)
+            return {"device": device, "kind": kind, "weight_stream_gbps": gbps}
     return {"device": device, "kind": kind}
 
 

@@ -225,6 +225,73 @@ void mfma_gemm_bf16_128_kernel(const bf16* __restrict__ A,
             }
 }
 
+// ---------------------------------------------------------------- decode GEMV
+// y[n] = W[n][k] · x[k], bf16 weights / fp32 accumulate — the batch-1 decode-step
+// shape (weight-streaming, HBM-bound: the guide's GEMV rule is load straight to
+// VGPRs, deep unroll, no LDS round trip). One workgroup per 64 output rows; each
+// wave owns one row block and strides K with 8 bf16 (16 B) per lane per step.
+__global__ __launch_bounds__(256)
+void decode_gemv_kernel(const bf16* __restrict__ W,   // [N][K] row-major
+                        const bf16* __restrict__ x,   // [K]
+                        float* __restrict__ y, int N, int K) {
+    const int lane = threadIdx.x % WAVE;
+    const int wave = threadIdx.x / WAVE;
+    const int row = blockIdx.x * 4 + wave;            // 4 waves -> 4 rows per block
+    if (row >= N) return;
+    const bf16* w = W + (long)row * K;
+    float acc = 0.f;
+    // 8 bf16 per lane per step => 64 lanes cover 512 elements per wave-step
+    for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
+        bf16x8 wv = *(const bf16x8*)(w + k0);
+        bf16x8 xv = *(const bf16x8*)(x + k0);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+            acc += (float)wv[i] * (float)xv[i];
+    }
+    // wave reduction (64 lanes) via LDS-free shuffles
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, 64);
+    if (lane == 0) y[row] = acc;
+}
+
+torch::Tensor decode_gemv(torch::Tensor w, torch::Tensor x) {
+    TORCH_CHECK(w.is_cuda() && x.is_cuda() && w.dtype() == torch::kBFloat16
+                && x.dtype() == torch::kBFloat16, "bf16 GPU tensors required");
+    TORCH_CHECK(w.is_contiguous() && x.is_contiguous(), "contiguous required");
+    int64_t N = w.size(0), K = w.size(1);
+    TORCH_CHECK(x.numel() == K && K % 512 == 0, "x must be [K], K%512==0");
+    auto y = torch::empty({N}, w.options().dtype(torch::kFloat32));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(decode_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0, stream,
+                       (const bf16*)w.data_ptr(), (const bf16*)x.data_ptr(),
+                       y.data_ptr<float>(), (int)N, (int)K);
+    return y;
+}
+
+// Decode payload: `iters` GEMV steps over an [n,k] weight matrix (one transformer
+// projection worth of weight streaming per step); returns achieved GB/s.
+double burn_decode(int64_t n, int64_t k, int64_t iters) {
+    auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
+    auto w = torch::randn({n, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
+    auto x = torch::randn({k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
+    auto y = torch::empty({n}, opt.dtype(torch::kFloat32));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    auto launch = [&] {
+        hipLaunchKernelGGL(decode_gemv_kernel, dim3((n + 3) / 4), dim3(256), 0,
+                           stream, (const bf16*)w.data_ptr(),
+                           (const bf16*)x.data_ptr(), y.data_ptr<float>(),
+                           (int)n, (int)k);
+    };
+    launch();
+    C10_HIP_CHECK(hipStreamSynchronize(stream));
+    auto t0 = std::chrono::steady_clock::now();
+    for (int64_t i = 0; i < iters; ++i) launch();
+    C10_HIP_CHECK(hipStreamSynchronize(stream));
+    auto t1 = std::chrono::steady_clock::now();
+    double secs = std::chrono::duration<double>(t1 - t0).count();
+    return 2.0 * n * k * iters / secs / 1e9;  // weight bytes streamed
+}
+
 __global__ void stream_triad_kernel(const float4* __restrict__ a,
                                     const float4* __restrict__ b,
                                     float4* __restrict__ c, long n4, float s) {
@@ -328,4 +395,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
               return burn_gemm(m_, n_, k_, it);
           }, "compat alias for burn_gemm (bk arg ignored)");
     m.def("stream_triad", &stream_triad, "HBM triad; returns GB/s");
+    m.def("decode_gemv", &decode_gemv, "y[N]=W[N,K]@x[K] (bf16 in, fp32 out)");
+    m.def("burn_decode", &burn_decode,
+          "iters decode GEMV steps; returns weight-stream GB/s");
 }

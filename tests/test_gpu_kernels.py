@@ -239,3 +239,16 @@ def test_remote_agent_gpu_distributed():
                     proc.wait(timeout=10)
                 except subprocess.TimeoutExpired:
                     proc.kill()
+
+
+def test_decode_gemv_numerics_and_bandwidth(gpuwork):
+    torch.manual_seed(3)
+    for (n, k) in [(512, 4096), (2048, 8192)]:
+        w = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+        x = torch.randn(k, device="cuda").to(torch.bfloat16)
+        y = gpuwork.decode_gemv(w.contiguous(), x.contiguous())
+        ref = w.float() @ x.float()
+        torch.testing.assert_close(y, ref, rtol=2e-2, atol=0.5)
+    gbps = gpuwork.burn_decode(16384, 8192, 20)
+    print(f"\ndecode GEMV weight stream: {gbps:.0f} GB/s")
+    assert gbps > 1000.0  # HBM-bound op; eager/PCIe fallback would fail

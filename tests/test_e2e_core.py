@@ -229,3 +229,19 @@ def test_pcsg_scale_out_creates_scaled_gang(cluster):
         pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "hs"})
         return gangs == {"hs-0"} and len(pods) == 1
     cluster.wait_for(shrunk, timeout=20, desc="scale-in cleanup")
+
+
+def test_non_default_namespace(cluster):
+    """Workloads in a custom namespace: full chain works and stays isolated."""
+    cluster.add_virtual_nodes(1)
+    pcs = _gpu_pcs("nsx", cliques=(("w", 2, 2),), gpus_per_pod=0)
+    pcs["metadata"]["namespace"] = "team-a"
+    cluster.store.create(pcs)
+    cluster.wait_for(
+        lambda: int((cluster.store.try_get(c.KIND_PCS, "team-a", "nsx") or {})
+                    .get("status", {}).get("availableReplicas", 0)) >= 1,
+        timeout=20, desc="team-a PCS available")
+    assert cluster.store.list("Pod", "team-a", {c.LABEL_PART_OF: "nsx"})
+    assert not cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "nsx"})
+    pg = cluster.store.get(c.KIND_PODGANG, "team-a", "nsx-0")
+    assert pg["spec"]["podgroups"][0]["podReferences"][0]["namespace"] == "team-a"

@@ -62,7 +62,7 @@ __device__ __forceinline__ void tile_map(int bid, int num_pid_m, int tiles_n,
 // wave-uniform-base + lane*16 and the LDS image is lane-linear in piece index, so the
 // XOR swizzle is applied on the SOURCE address (guide §5 rule 21: swizzled images via
 // pre-swizzled global addresses, LDS stays linear).
-template <int THREADS>
+template <int THREADS, int AUX = 0>
 __device__ __forceinline__ void stage_tile(const bf16* __restrict__ g, long ld,
                                            __bf16* __restrict__ dst, int rows,
                                            int tid, int wave_piece0) {
@@ -78,7 +78,7 @@ __device__ __forceinline__ void stage_tile(const bf16* __restrict__ g, long ld,
             (const __attribute__((address_space(1))) uint32_t*)(
                 g + (long)row * ld + (cb_src << 3)),
             (__attribute__((address_space(3))) uint32_t*)(dst + base * 8),
-            16, 0, 0);
+            16, 0, AUX);
     }
 }
 
@@ -107,6 +107,8 @@ void mfma_gemm_bf16_256_kernel(const bf16* __restrict__ A,   // [M][K] row-major
     const int wave_piece0 = wave * WAVE;
 
     auto stage = [&](int buf, long kk) {
+        // nt (aux=2) on the A stream was measured neutral@4096 / -2.5%@8192 —
+        // default cache policy kept for both operands
         stage_tile<512>(A + row0 * K + kk, K, sAp(buf), BM, tid, wave_piece0);
         stage_tile<512>(Bt + col0 * K + kk, K, sBp(buf), BN, tid, wave_piece0);
     };

@@ -62,22 +62,21 @@ double placement_score(size_t n_nodes_used, int total_gpus) {
 using Assignment = std::tuple<std::string, std::string, std::vector<int>>;
 using NodeState = std::tuple<std::string, int64_t, double, std::vector<int>, int64_t>;
 
-// Returns (assignments, score, consumed-node-states) or None.
-py::object place_gang(std::vector<NodeState> node_states,
-                      std::vector<std::tuple<std::string, int64_t, double, int>> pod_specs) {
-  std::vector<Node> nodes;
-  nodes.reserve(node_states.size());
-  for (auto& t : node_states)
-    nodes.push_back(Node{std::get<0>(t), std::get<1>(t), std::get<2>(t),
-                         std::get<3>(t), std::get<4>(t)});
-  std::vector<Pod> pods;
-  pods.reserve(pod_specs.size());
+// Returns (assignments, score, consumed-node-states) or None. The solver runs
+// WITHOUT the GIL (a plain-C++ result is computed in a released scope; Python objects
+// are built only after the GIL is re-acquired), so large placements overlap the
+// Python controllers instead of serializing behind them.
+struct SolveResult {
+  bool feasible = false;
+  std::vector<Assignment> assignments;
+  double score = 0.0;
+  std::vector<NodeState> consumed;
+};
+
+SolveResult solve(std::vector<Node>& nodes, std::vector<Pod>& pods) {
+  SolveResult out;
   int total_gpus = 0;
-  for (auto& t : pod_specs) {
-    pods.push_back(Pod{std::get<0>(t), std::get<1>(t), std::get<2>(t), std::get<3>(t)});
-    total_gpus += std::get<3>(t);
-  }
-  // sort pods by descending gpu then cpu demand (stable for determinism)
+  for (auto& p : pods) total_gpus += p.gpus;
   std::vector<size_t> order(pods.size());
   for (size_t i = 0; i < order.size(); ++i) order[i] = i;
   std::stable_sort(order.begin(), order.end(), [&](size_t a, size_t b) {
@@ -85,13 +84,12 @@ py::object place_gang(std::vector<NodeState> node_states,
     return pods[a].cpu_milli > pods[b].cpu_milli;
   });
 
-  auto emit = [&](std::vector<Node>& state, std::vector<Assignment>& assignments,
-                  double score) -> py::object {
-    std::vector<NodeState> consumed;
-    consumed.reserve(state.size());
+  auto emit = [&](std::vector<Node>& state, double score) {
+    out.feasible = true;
+    out.score = score;
+    out.consumed.reserve(state.size());
     for (auto& n : state)
-      consumed.emplace_back(n.name, n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods);
-    return py::make_tuple(assignments, score, consumed);
+      out.consumed.emplace_back(n.name, n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods);
   };
 
   // Phase 1: single-node best-fit (leaves fewest free GPUs behind).
@@ -110,18 +108,17 @@ py::object place_gang(std::vector<NodeState> node_states,
     }
   }
   if (best >= 0) {
-    std::vector<Assignment> assignments;
     for (size_t oi : order) {
       auto ids = take(nodes[best], pods[oi]);
-      assignments.emplace_back(pods[oi].name, nodes[best].name, ids);
+      out.assignments.emplace_back(pods[oi].name, nodes[best].name, ids);
     }
-    return emit(nodes, assignments, placement_score(1, total_gpus));
+    emit(nodes, placement_score(1, total_gpus));
+    return out;
   }
 
   // Phase 2: minimal spread, first-fit-decreasing preferring already-used nodes with
   // the most free GPUs.
-  std::vector<Node> state = nodes;  // work on a copy; rollback = return None
-  std::vector<Assignment> assignments;
+  std::vector<Node> state = nodes;  // work on a copy; rollback = infeasible return
   std::vector<char> used(state.size(), 0);
   size_t used_count = 0;
   for (size_t oi : order) {
@@ -138,12 +135,34 @@ py::object place_gang(std::vector<NodeState> node_states,
       }
       pick = best_i;
     }
-    if (pick < 0) return py::none();
+    if (pick < 0) return out;  // infeasible
     auto ids = take(state[pick], pods[oi]);
-    assignments.emplace_back(pods[oi].name, state[pick].name, ids);
+    out.assignments.emplace_back(pods[oi].name, state[pick].name, ids);
     if (!used[pick]) { used[pick] = 1; ++used_count; }
   }
-  return emit(state, assignments, placement_score(used_count, total_gpus));
+  emit(state, placement_score(used_count, total_gpus));
+  return out;
+}
+
+py::object place_gang(std::vector<NodeState> node_states,
+                      std::vector<std::tuple<std::string, int64_t, double, int>> pod_specs) {
+  SolveResult res;
+  {
+    py::gil_scoped_release released;
+    std::vector<Node> nodes;
+    nodes.reserve(node_states.size());
+    for (auto& t : node_states)
+      nodes.push_back(Node{std::get<0>(t), std::get<1>(t), std::get<2>(t),
+                           std::get<3>(t), std::get<4>(t)});
+    std::vector<Pod> pods;
+    pods.reserve(pod_specs.size());
+    for (auto& t : pod_specs)
+      pods.push_back(Pod{std::get<0>(t), std::get<1>(t), std::get<2>(t),
+                         std::get<3>(t)});
+    res = solve(nodes, pods);
+  }  // GIL re-acquired here; only now touch Python objects
+  if (!res.feasible) return py::none();
+  return py::make_tuple(res.assignments, res.score, res.consumed);
 }
 
 double score_only(size_t n_nodes, int total_gpus) {

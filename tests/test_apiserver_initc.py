@@ -207,3 +207,38 @@ def test_debug_profile_endpoint(served_cluster, simple1_yaml):
     # collapsed-stack lines: "frame;frame;... count"
     assert body.strip() and all(
         line.rsplit(" ", 1)[-1].isdigit() for line in body.strip().splitlines())
+
+
+def test_httpclient_roundtrip_and_watch(served_cluster):
+    """HttpStoreClient (the client-go analog): CRUD + patch + ndjson watch."""
+    cluster, api = served_cluster
+    from grove_amd.kubecore.httpclient import HttpStoreClient
+    from grove_amd.kubecore.store import ApiError
+    cl = HttpStoreClient(api.url)
+    node = cl.create({"kind": "Node", "metadata": {"name": "hc-n0"},
+                      "spec": {}, "status": {"allocatable": {"cpu": "4"}}})
+    assert node["metadata"]["uid"]
+    assert cl.get("Node", None, "hc-n0")["metadata"]["name"] == "hc-n0"
+    got = cl.list("Node", None)
+    assert any(n["metadata"]["name"] == "hc-n0" for n in got)
+    cl.patch("Node", None, "hc-n0",
+             lambda o: o["metadata"].setdefault("labels", {}).update(zone="z1"))
+    assert cl.get("Node", None, "hc-n0")["metadata"]["labels"]["zone"] == "z1"
+    # watch stream sees a subsequent event
+    events = []
+    import threading
+
+    def consume():
+        for ev, obj in cl.watch_events("Pod", "default", seed=False):
+            events.append((ev, obj["metadata"]["name"]))
+            return
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    cl.create({"kind": "Pod", "metadata": {"name": "hc-p0", "namespace": "default"},
+               "spec": {"containers": []}})
+    t.join(timeout=10)
+    assert events == [("ADDED", "hc-p0")]
+    cl.delete("Pod", "default", "hc-p0")
+    with pytest.raises(ApiError):
+        cl.get("Pod", "default", "hc-p0")

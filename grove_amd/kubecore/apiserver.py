@@ -179,6 +179,40 @@ def build_app(store: Store, metrics_fn=None):
         except ApiError as e:
             return err(e)
 
+    @app.put("/apis/{group}/{version}/{plural}/{name}")
+    @app.put("/api/{version}/{plural}/{name}")
+    async def update_cluster(request: Request, plural: str, name: str,
+                             group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            obj.setdefault("metadata", {})["name"] = name
+            return JSONResponse(store.update(obj))
+        except ApiError as e:
+            return err(e)
+
+    @app.put("/apis/{group}/{version}/{plural}/{name}/status")
+    @app.put("/api/{version}/{plural}/{name}/status")
+    async def update_cluster_status(request: Request, plural: str, name: str,
+                                    group: str = "", version: str = "v1"):
+        try:
+            obj = await request.json()
+            obj.setdefault("kind", kind_of(plural))
+            obj.setdefault("metadata", {})["name"] = name
+            return JSONResponse(store.update_status(obj))
+        except ApiError as e:
+            return err(e)
+
+    @app.delete("/apis/{group}/{version}/{plural}/{name}")
+    @app.delete("/api/{version}/{plural}/{name}")
+    async def delete_cluster(plural: str, name: str,
+                             group: str = "", version: str = "v1"):
+        try:
+            store.delete(kind_of(plural), None, name)
+            return JSONResponse({"kind": "Status", "status": "Success"})
+        except ApiError as e:
+            return err(e)
+
     @app.get("/apis/{group}/{version}/{plural}/{name}")
     @app.get("/api/{version}/{plural}/{name}")
     async def get_cluster(plural: str, name: str,

@@ -84,3 +84,15 @@ def test_crd_versions_and_scope(crd_name):
         assert scope == "Cluster"
     else:
         assert scope == "Namespaced"
+
+
+def test_committed_crds_match_renderer():
+    """crds/*.yaml are generated artifacts — they must match api/crds.py exactly."""
+    import os
+    from grove_amd.api.crds import render_all
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for crd in render_all():
+        path = os.path.join(repo, "crds", f"{crd['metadata']['name']}.yaml")
+        assert os.path.exists(path), f"missing {path} (run install-crds --output-dir crds)"
+        on_disk = yaml.safe_load(open(path))
+        assert on_disk == crd, f"{path} drifted — regenerate with install-crds"

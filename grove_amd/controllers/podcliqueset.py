@@ -572,6 +572,22 @@ class PodCliqueSetReconciler:
                 elif sched:
                     phase = "Starting"
             gang_statuses.append({"name": pg["metadata"]["name"], "phase": phase})
+            # Unhealthy condition (scheduler/api podgang.go:152-171): a scheduled gang
+            # whose member clique breached MinAvailable; cleared on recovery
+            unhealthy = any(
+                cond.condition_true(pclq_of.get(g["name"], {}),
+                                    c.COND_MIN_AVAILABLE_BREACHED)
+                for g in groups)
+            was = cond.condition_true(pg, c.PODGANG_COND_UNHEALTHY)
+            if unhealthy != was and cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
+                def flip_unhealthy(o: Obj, v=unhealthy) -> None:
+                    cond.set_condition(o, c.PODGANG_COND_UNHEALTHY, v,
+                                       "MinAvailableBreached" if v else "Recovered")
+                try:
+                    self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"],
+                                     flip_unhealthy, status=True)
+                except ApiError:
+                    pass
 
         def upd(o: Obj) -> None:
             s = o.setdefault("status", {})

@@ -192,3 +192,28 @@ class TestRollingUpdate:
                 ["updateProgress"].get("updateEndedAt"))
         pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "ru3"})
         assert all(p["spec"]["containers"][0]["image"] == "img:v2" for p in pods)
+
+
+def test_podgang_unhealthy_condition(cluster):
+    """A scheduled gang whose clique breaches MinAvailable gets Unhealthy=True;
+    recovery clears it."""
+    cluster.add_virtual_nodes(2)
+    cluster.apply(_pcs("uh", termination_delay="1h"))
+    cluster.wait_pcs_available("uh", timeout=20)
+    for n in cluster.store.list("Node"):
+        cluster.store.patch("Node", None, n["metadata"]["name"],
+                            lambda o: o["spec"].update(unschedulable=True))
+    _kill_pods(cluster, {c.LABEL_PODCLIQUE: "uh-0-w"}, 1)
+
+    def unhealthy():
+        pg = cluster.store.get(c.KIND_PODGANG, "default", "uh-0")
+        return cond.condition_true(pg, c.PODGANG_COND_UNHEALTHY)
+    cluster.wait_for(unhealthy, timeout=20, desc="PodGang Unhealthy")
+    for n in cluster.store.list("Node"):
+        cluster.store.patch("Node", None, n["metadata"]["name"],
+                            lambda o: o["spec"].update(unschedulable=False))
+
+    def recovered():
+        pg = cluster.store.get(c.KIND_PODGANG, "default", "uh-0")
+        return not cond.condition_true(pg, c.PODGANG_COND_UNHEALTHY)
+    cluster.wait_for(recovered, timeout=30, desc="Unhealthy cleared")

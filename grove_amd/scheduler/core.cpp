@@ -171,7 +171,50 @@ double score_only(size_t n_nodes, int total_gpus) {
 
 }  // namespace
 
+// Fast deep copy for JSON-shaped objects (dict/list/scalars) using the raw CPython
+// API — the store's hottest single function (every get()/update copies an object).
+// Scalars are immutable in Python and shared, matching kubecore.store.json_copy.
+static PyObject* fast_json_copy(PyObject* obj) {
+  if (PyDict_CheckExact(obj)) {
+    PyObject* out = PyDict_New();
+    if (!out) return nullptr;
+    PyObject *key, *value;
+    Py_ssize_t pos = 0;
+    while (PyDict_Next(obj, &pos, &key, &value)) {
+      PyObject* copied = fast_json_copy(value);
+      if (!copied || PyDict_SetItem(out, key, copied) < 0) {
+        Py_XDECREF(copied);
+        Py_DECREF(out);
+        return nullptr;
+      }
+      Py_DECREF(copied);
+    }
+    return out;
+  }
+  if (PyList_CheckExact(obj)) {
+    Py_ssize_t n = PyList_GET_SIZE(obj);
+    PyObject* out = PyList_New(n);
+    if (!out) return nullptr;
+    for (Py_ssize_t i = 0; i < n; ++i) {
+      PyObject* copied = fast_json_copy(PyList_GET_ITEM(obj, i));
+      if (!copied) {
+        Py_DECREF(out);
+        return nullptr;
+      }
+      PyList_SET_ITEM(out, i, copied);  // steals ref
+    }
+    return out;
+  }
+  Py_INCREF(obj);
+  return obj;
+}
+
 PYBIND11_MODULE(_sched, m) {
+  m.def("json_copy", [](py::handle obj) {
+    PyObject* out = fast_json_copy(obj.ptr());
+    if (!out) throw py::error_already_set();
+    return py::reinterpret_steal<py::object>(out);
+  }, "fast deep copy for JSON-shaped objects");
   m.doc() = "grove_amd native gang-placement core (xGMI-aware Filter/Score)";
   m.def("place_gang", &place_gang, py::arg("nodes"), py::arg("pods"),
         "All-or-nothing gang placement; returns (assignments, score, consumed) or None");

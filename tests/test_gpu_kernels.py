@@ -252,3 +252,54 @@ def test_decode_gemv_numerics_and_bandwidth(gpuwork):
     gbps = gpuwork.burn_decode(16384, 8192, 20)
     print(f"\ndecode GEMV weight stream: {gbps:.0f} GB/s")
     assert gbps > 1000.0  # HBM-bound op; eager/PCIe fallback would fail
+
+
+def test_gpu_chaos_payloads_under_churn():
+    """Random pod kills while MFMA payloads execute on the real GPU: the dispatch +
+    HIP path must stay correct (no wedged GPU work, convergence to Available)."""
+    import random
+    import time
+    from grove_amd import Cluster
+    from grove_amd.api import constants as c
+    from grove_amd.kubelet.gpunode import gpu_pod_payload
+    from grove_amd.topology.agent import discover_node
+    from grove_amd.utils import conditions as cc
+
+    rng = random.Random(7)
+    cl = Cluster(pod_payload=gpu_pod_payload).start()
+    try:
+        cl.store.create(discover_node("chaos-gpu"))
+        pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+               "metadata": {"name": "gch"},
+               "spec": {"replicas": 1, "template": {
+                   "terminationDelay": "1h",
+                   "cliques": [{"name": "inf",
+                                "annotations": {"grove.io/payload-shape":
+                                                "1024x1024x1024x2"},
+                                "spec": {"roleName": "r", "replicas": 1,
+                                         "podSpec": {"containers": [{
+                                             "name": "m", "image": "p",
+                                             "resources": {"requests": {
+                                                 c.AMD_GPU_RESOURCE: "1"}}}]}}}]}}}
+        cl.store.create(pcs)
+        cl.wait_pcs_available("gch", timeout=60)
+        for i in range(12):
+            pods = cl.store.list("Pod", "default", {c.LABEL_PART_OF: "gch"},
+                                 copy_objects=False)
+            if pods and rng.random() < 0.8:
+                try:
+                    cl.store.delete("Pod", "default",
+                                    rng.choice(pods)["metadata"]["name"])
+                except Exception:
+                    pass
+            time.sleep(0.4)
+
+        def converged():
+            pods = cl.store.list("Pod", "default", {c.LABEL_PART_OF: "gch"})
+            return len(pods) == 1 and all(cc.pod_is_ready(p) for p in pods)
+        cl.wait_for(converged, timeout=90, desc="GPU chaos convergence")
+        # the GPU still works after the churn
+        from grove_amd.kubelet.gpunode import load_gpuwork
+        assert load_gpuwork().burn_gemm(1024, 1024, 1024, 2) > 10.0
+    finally:
+        cl.stop()

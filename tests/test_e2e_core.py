@@ -245,3 +245,53 @@ def test_non_default_namespace(cluster):
     assert not cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "nsx"})
     pg = cluster.store.get(c.KIND_PODGANG, "team-a", "nsx-0")
     assert pg["spec"]["podgroups"][0]["podReferences"][0]["namespace"] == "team-a"
+
+
+def test_zero_replica_pcs(cluster):
+    """kubebuilder default spec.replicas=0: valid, creates RBAC but no gangs/pods."""
+    cluster.add_virtual_nodes(1)
+    pcs = _gpu_pcs("zr", cliques=(("w", 1, 1),), gpus_per_pod=0)
+    del pcs["spec"]["replicas"]
+    cluster.store.create(pcs)
+    import time as _t
+    _t.sleep(0.5)
+    out = cluster.store.get(c.KIND_PCS, "default", "zr")
+    assert out["spec"]["replicas"] == 0
+    assert cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "zr"}) == []
+    assert cluster.store.list(c.KIND_PODGANG, "default", {c.LABEL_PART_OF: "zr"}) == []
+    assert cluster.store.try_get("ServiceAccount", "default", "zr") is not None
+    assert int(out.get("status", {}).get("availableReplicas", 0)) == 0
+
+
+def test_multi_gpu_pods_in_gang(cluster):
+    """Gang of 2 pods x 4 GPUs each fills one hive; GPU ids partition exactly."""
+    cluster.add_virtual_nodes(1, gpus=8, prefix="hive")
+    cluster.apply(_gpu_pcs("mg", cliques=(("w", 2, 2),), gpus_per_pod=4))
+    cluster.wait_pcs_available("mg", timeout=20)
+    pods = cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "mg"})
+    ids = []
+    for p in pods:
+        ids.extend(int(x) for x in
+                   p["metadata"]["annotations"]["scheduling.amd.com/gpu-ids"].split(","))
+        assert len(p["metadata"]["annotations"][
+            "scheduling.amd.com/gpu-ids"].split(",")) == 4
+    assert sorted(ids) == list(range(8))
+
+
+def test_reapply_is_idempotent(cluster, simple1_yaml):
+    cluster.add_virtual_nodes(2)
+    cluster.apply(simple1_yaml)
+    cluster.wait_pcs_available("simple1", timeout=20)
+    pods_before = {p["metadata"]["name"]
+                   for p in cluster.store.list("Pod", "default",
+                                               {c.LABEL_PART_OF: "simple1"})}
+    gen = cluster.store.get(c.KIND_PCS, "default", "simple1")["metadata"]["generation"]
+    cluster.apply(simple1_yaml)  # identical re-apply
+    import time as _t
+    _t.sleep(0.8)
+    after = cluster.store.get(c.KIND_PCS, "default", "simple1")
+    assert after["metadata"]["generation"] == gen  # no spec change detected
+    pods_after = {p["metadata"]["name"]
+                  for p in cluster.store.list("Pod", "default",
+                                              {c.LABEL_PART_OF: "simple1"})}
+    assert pods_after == pods_before  # zero churn

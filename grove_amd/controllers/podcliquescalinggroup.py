@@ -250,6 +250,31 @@ class PCSGReconciler:
                 gh = (pcs_obj.get("status") or {}).get("currentGenerationHash")
                 if gh:
                     st["currentPodCliqueSetGenerationHash"] = gh
+            # updateProgress parity: active while member cliques carry mixed hashes
+            member_qs = [q for qs in by_replica.values() for q in qs]
+            hashes = {q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH)
+                      for q in member_qs}
+            total_q = len(member_qs)
+            updated_q = sum(
+                1 for q in member_qs
+                if int((q.get("status") or {}).get("updatedReplicas", 0))
+                >= int(q["spec"].get("replicas", 1)))
+            if len(hashes) > 1 or updated_q < total_q:
+                prog = st.get("updateProgress") or {}
+                if not prog.get("updateStartedAt"):
+                    import time as _time
+                    prog["updateStartedAt"] = _time.strftime(
+                        "%Y-%m-%dT%H:%M:%SZ", _time.gmtime())
+                prog["totalPodCliquesCount"] = total_q
+                prog["updatedPodCliquesCount"] = updated_q
+                prog.pop("updateEndedAt", None)
+                st["updateProgress"] = prog
+            elif st.get("updateProgress") and not st["updateProgress"].get(
+                    "updateEndedAt"):
+                import time as _time
+                st["updateProgress"]["updateEndedAt"] = _time.strftime(
+                    "%Y-%m-%dT%H:%M:%SZ", _time.gmtime())
+                st["updateProgress"]["updatedPodCliquesCount"] = updated_q
             ever = bool(st.get("everAvailable")) or avail >= min_avail
             st["everAvailable"] = ever
             if ever and avail < min_avail:

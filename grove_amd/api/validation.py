@@ -215,6 +215,22 @@ def _validate_pcs_update(new: Obj, old: Obj) -> None:
     for k in new_sgs:
         if new_sgs[k] != old_sgs[k]:
             raise _err(f"spec.template.podCliqueScalingGroups[{k}].cliqueNames", "field is immutable")
+    # topology constraints are immutable after creation at every level
+    # (docs/user-guide/topology-aware-scheduling.md: add/modify/remove rejected)
+    if nt.get("topologyConstraint") != ot.get("topologyConstraint"):
+        raise _err("spec.template.topologyConstraint",
+                   "topology constraints are immutable; recreate the PodCliqueSet")
+    for new_cl, old_cl in zip(nt.get("cliques") or [], ot.get("cliques") or []):
+        if new_cl.get("topologyConstraint") != old_cl.get("topologyConstraint"):
+            raise _err(f"spec.template.cliques[{new_cl.get('name')}].topologyConstraint",
+                       "topology constraints are immutable; recreate the PodCliqueSet")
+    for new_sg, old_sg in zip(nt.get("podCliqueScalingGroups") or [],
+                              ot.get("podCliqueScalingGroups") or []):
+        if new_sg.get("topologyConstraint") != old_sg.get("topologyConstraint"):
+            raise _err(
+                f"spec.template.podCliqueScalingGroups[{new_sg.get('name')}]"
+                f".topologyConstraint",
+                "topology constraints are immutable; recreate the PodCliqueSet")
 
 
 def validate_clustertopologybinding(ctb: Obj, old: Optional[Obj] = None) -> None:
@@ -294,6 +310,25 @@ class TopologyConstraintValidator:
     def __call__(self, pcs, old=None):
         tmpl = (pcs.get("spec") or {}).get("template") or {}
         parent_tc = tmpl.get("topologyConstraint")
+        # a single PCS cannot use multiple topology names (children inherit the
+        # parent name when they omit it — topology-aware-scheduling.md:32,69)
+        names = set()
+        for tc in [parent_tc] + [cl.get("topologyConstraint")
+                                 for cl in tmpl.get("cliques") or []] + \
+                  [sg.get("topologyConstraint")
+                   for sg in tmpl.get("podCliqueScalingGroups") or []]:
+            if tc and tc.get("topologyName"):
+                names.add(tc["topologyName"])
+        if len(names) > 1:
+            raise _err("spec.template", f"a PodCliqueSet may reference only one "
+                                        f"topologyName, found {sorted(names)}")
+        if names:
+            known_ctbs = {ctb["metadata"]["name"]
+                          for ctb in self.store.list(c.KIND_CTB)}
+            missing = names - known_ctbs
+            if missing and known_ctbs:
+                raise _err("spec.template.topologyConstraint.topologyName",
+                           f"unknown ClusterTopologyBinding {sorted(missing)}")
         has_any = bool(parent_tc) or any(
             cl.get("topologyConstraint") for cl in tmpl.get("cliques") or []) or any(
             sg.get("topologyConstraint")

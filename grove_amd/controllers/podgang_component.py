@@ -33,11 +33,15 @@ class ExpectedGang:
         self.group_configs = group_configs or []  # TopologyConstraintGroupConfig[]
 
 
-def domain_key_map(store: Store) -> Dict[str, str]:
+def domain_key_map(store: Store, topology_name: Optional[str] = None
+                   ) -> Dict[str, str]:
     """CTB levels → {domain: nodeLabelKey}, with native defaults for host/xgmi-hive
-    (components/podgang/syncflow.go:349-380 translation parity)."""
+    (components/podgang/syncflow.go:349-380 translation parity). topology_name picks
+    the CTB the constraint references; otherwise the single CTB is the source."""
     m = {"xgmi-hive": c.NODE_LABEL_XGMI_HIVE, "host": "kubernetes.io/hostname"}
     for ctb in store.list(c.KIND_CTB):
+        if topology_name and ctb["metadata"]["name"] != topology_name:
+            continue
         for lv in (ctb.get("spec") or {}).get("levels") or []:
             key = lv.get("nodeLabelKey") or lv.get("key")
             if lv.get("domain") and key:
@@ -74,7 +78,14 @@ def compute_expected_podgangs(store: Store, pcs: Obj,
     sg_cfgs = tmpl.get("podCliqueScalingGroups") or []
     sg_members = {m for sg in sg_cfgs for m in (sg.get("cliqueNames") or [])}
     clique_by_name = {cl["name"]: cl for cl in cliques}
-    dmap = domain_key_map(store)
+    topo_name = None
+    for tc in [tmpl.get("topologyConstraint")] + \
+              [cl.get("topologyConstraint") for cl in cliques] + \
+              [sg.get("topologyConstraint") for sg in sg_cfgs]:
+        if tc and tc.get("topologyName"):
+            topo_name = tc["topologyName"]
+            break
+    dmap = domain_key_map(store, topo_name)
     gang_tc = translate_constraint(tmpl.get("topologyConstraint"), dmap)
     out: List[ExpectedGang] = []
 

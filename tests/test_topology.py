@@ -243,3 +243,37 @@ class TestTopologyConstraintValidation:
         with pytest.raises(ApiError) as ei:
             cluster.apply(pcs)
         assert "narrower" in str(ei.value)
+
+
+class TestTopologyImmutability:
+    def test_constraint_change_rejected(self, cluster):
+        cluster.store.create(CTB)
+        cluster.add_virtual_nodes(1, gpus=8,
+                                  labels={"topology.kubernetes.io/rack": "r0"})
+        cluster.apply(_pcs("imm", 1, constraint={"pack": {"required": "rack"}}))
+        cluster.wait_pcs_available("imm", timeout=20)
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError) as ei:
+            cluster.store.patch(
+                c.KIND_PCS, "default", "imm",
+                lambda o: o["spec"]["template"].update(
+                    topologyConstraint={"pack": {"required": "zone"}}))
+        assert "immutable" in str(ei.value)
+
+    def test_multiple_topology_names_rejected(self, cluster):
+        cluster.store.create(CTB)
+        pcs = _pcs("multi", 1, constraint={"pack": {"required": "rack"},
+                                           "topologyName": "cluster-topology"})
+        pcs["spec"]["template"]["cliques"][0]["topologyConstraint"] = {
+            "pack": {"required": "host"}, "topologyName": "other-topology"}
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError) as ei:
+            cluster.apply(pcs)
+        assert "one" in str(ei.value)
+
+    def test_unknown_topology_name_rejected(self, cluster):
+        cluster.store.create(CTB)
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError):
+            cluster.apply(_pcs("ut", 1, constraint={
+                "pack": {"required": "rack"}, "topologyName": "nope"}))

@@ -58,6 +58,10 @@ FINALIZER_PCSG = "grove.io/podcliquescalinggroup.grove.io"
 ANNOTATION_DISABLE_MANAGED_RESOURCE_PROTECTION = "grove.io/disable-managed-resource-protection"
 ANNOTATION_RECONCILE_TRIGGER = "grove.io/reconcile-trigger"
 ANNOTATION_TOPOLOGY_NAME = "grove.io/topology-name"
+# MI355X analog of the reference's grove.io/mnnvl-group annotation (auto-mnnvl.md):
+# cliques sharing a group name share one xGMI-domain ResourceClaim per PCS replica;
+# propagates PCS -> PCSG -> PCLQ with lower levels overriding; "none" opts out.
+ANNOTATION_XGMI_GROUP = "grove.io/xgmi-group"
 
 # --- Pod scheduling gate (pod.go:69) ---
 POD_GANG_SCHEDULING_GATE = "grove.io/podgang-pending-creation"

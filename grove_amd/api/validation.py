@@ -374,3 +374,60 @@ class TopologyConstraintValidator:
                         f"spec.template.cliques[{cl.get('name')}]"
                         f".topologyConstraint.pack.required",
                         "must be equal or narrower than the PodCliqueSet constraint")
+
+
+_DNS1123 = None
+
+
+def validate_xgmi_groups(pcs: Obj, old: Optional[Obj] = None) -> None:
+    """grove.io/xgmi-group admission (auto-mnnvl.md parity): value must be "none" or
+    a DNS-1123 label; an explicit group on a clique with no amd.com/gpu request is a
+    user error; the annotation is immutable after creation at every level."""
+    import re
+    global _DNS1123
+    if _DNS1123 is None:
+        _DNS1123 = re.compile(r"^[a-z0-9]([a-z0-9-]{0,61}[a-z0-9])?$")
+    tmpl = (pcs.get("spec") or {}).get("template") or {}
+
+    def check_value(v, path):
+        if v is not None and v != "none" and not _DNS1123.match(v):
+            raise _err(path, f"xgmi-group {v!r} must be 'none' or a DNS-1123 label")
+
+    def gpu_clique(cl):
+        for ctr in (cl.get("spec", {}).get("podSpec", {}).get("containers") or []):
+            res = (ctr.get("resources") or {})
+            if "amd.com/gpu" in (res.get("requests") or {}) \
+                    or "amd.com/gpu" in (res.get("limits") or {}):
+                return True
+        return False
+
+    check_value((pcs.get("metadata", {}).get("annotations") or {}).get(
+        c.ANNOTATION_XGMI_GROUP), "metadata.annotations")
+    for sg in tmpl.get("podCliqueScalingGroups") or []:
+        check_value((sg.get("annotations") or {}).get(c.ANNOTATION_XGMI_GROUP),
+                    f"spec.template.podCliqueScalingGroups[{sg.get('name')}]")
+    for cl in tmpl.get("cliques") or []:
+        v = (cl.get("annotations") or {}).get(c.ANNOTATION_XGMI_GROUP)
+        check_value(v, f"spec.template.cliques[{cl.get('name')}]")
+        if v not in (None, "none") and not gpu_clique(cl):
+            raise _err(f"spec.template.cliques[{cl.get('name')}]",
+                       f"explicit xgmi-group on a PodClique with no amd.com/gpu "
+                       f"request")
+    if old is not None:
+        def ann_of(obj_or_tmpl_item, meta=False):
+            src = obj_or_tmpl_item.get("metadata", {}) if meta else obj_or_tmpl_item
+            return (src.get("annotations") or {}).get(c.ANNOTATION_XGMI_GROUP)
+        if ann_of(pcs, meta=True) != ann_of(old, meta=True):
+            raise _err("metadata.annotations",
+                       "grove.io/xgmi-group is immutable after creation")
+        ot = (old.get("spec") or {}).get("template") or {}
+        for new_cl, old_cl in zip(tmpl.get("cliques") or [], ot.get("cliques") or []):
+            if ann_of(new_cl) != ann_of(old_cl):
+                raise _err(f"spec.template.cliques[{new_cl.get('name')}]",
+                           "grove.io/xgmi-group is immutable after creation")
+        for new_sg, old_sg in zip(tmpl.get("podCliqueScalingGroups") or [],
+                                  ot.get("podCliqueScalingGroups") or []):
+            if ann_of(new_sg) != ann_of(old_sg):
+                raise _err(
+                    f"spec.template.podCliqueScalingGroups[{new_sg.get('name')}]",
+                    "grove.io/xgmi-group is immutable after creation")

@@ -49,6 +49,8 @@ class Cluster:
         # admission (webhook parity: defaulting before validation)
         self.store.register_mutator(c.KIND_PCS, default_podcliqueset)
         self.store.register_validator(c.KIND_PCS, validate_podcliqueset)
+        from .api.validation import validate_xgmi_groups
+        self.store.register_validator(c.KIND_PCS, validate_xgmi_groups)
         from .api.validation import TopologyConstraintValidator
         self.store.register_validator(c.KIND_PCS,
                                       TopologyConstraintValidator(self.store))

@@ -75,14 +75,53 @@ def pcs_claims_for_replica(store: Store, pcs: Obj, r: int,
         claim_name = f"{pcs_name}-{tname}" if scope == "AllReplicas" \
             else f"{pcs_name}-{r}-{tname}"
         out.append((build_resource_claim(pcs, claim_name, spec), entry))
-    if auto_xgmi_domain and _pcs_requests_gpus(pcs):
-        entry = {"templateRef": XGMI_TEMPLATE_NAME, "scope": "PerReplica",
-                 "filter": {"childCliqueNames": _gpu_cliques(pcs)}}
+    # xGMI-domain groups: one claim per (replica, group), shared by the group's
+    # cliques (the MNNVL ComputeDomain-per-replica analog)
+    groups = effective_xgmi_groups(pcs, auto_xgmi_domain)
+    by_group: Dict[str, List[str]] = {}
+    for clique, g in groups.items():
+        by_group.setdefault(g, []).append(clique)
+    for g, members in sorted(by_group.items()):
+        entry = {"templateRef": f"{XGMI_TEMPLATE_NAME}-{g}", "scope": "PerReplica",
+                 "filter": {"childCliqueNames": sorted(members)}}
         claim = build_resource_claim(
-            pcs, f"{pcs_name}-{r}-{XGMI_TEMPLATE_NAME}",
+            pcs, f"{pcs_name}-{r}-xgmi-{g}",
             {"devices": {"requests": [{"name": "xgmi-hive",
                                        "deviceClassName": XGMI_DEVICE_CLASS}]}})
         out.append((claim, entry))
+    return out
+
+
+def effective_xgmi_groups(pcs: Obj, auto_default: bool) -> Dict[str, str]:
+    """clique name -> xGMI group, following the annotation hierarchy (PCS -> PCSG ->
+    PCLQ, lower overrides; "none" opts out; non-GPU cliques silently skip inherited
+    groups — auto-mnnvl.md:58-77 parity). With auto_default and no annotations, every
+    GPU clique joins the implicit per-replica group "default"."""
+    tmpl = pcs["spec"]["template"]
+    pcs_group = (pcs["metadata"].get("annotations") or {}).get(
+        c.ANNOTATION_XGMI_GROUP)
+    sg_of = {}
+    sg_group = {}
+    for sg in tmpl.get("podCliqueScalingGroups") or []:
+        g = (sg.get("annotations") or {}).get(c.ANNOTATION_XGMI_GROUP)
+        for mn in sg.get("cliqueNames") or []:
+            sg_of[mn] = sg["name"]
+            if g is not None:
+                sg_group[mn] = g
+    gpu = set(_gpu_cliques(pcs))
+    out: Dict[str, str] = {}
+    for cl in tmpl.get("cliques") or []:
+        name = cl["name"]
+        own = (cl.get("annotations") or {}).get(c.ANNOTATION_XGMI_GROUP)
+        eff = own if own is not None else sg_group.get(name, pcs_group)
+        if eff is None and auto_default and name in gpu:
+            eff = "default"
+        if eff in (None, "none"):
+            continue
+        if name in gpu:
+            out[name] = eff
+        # non-GPU cliques silently skip INHERITED groups (explicit ones are rejected
+        # at admission by validate_xgmi_groups)
     return out
 
 

@@ -170,13 +170,13 @@ class TestResourceSharing:
                                    c.AMD_GPU_RESOURCE: "1"}}}]}}}]}}}
             cl.store.create(pcs)
             cl.wait_pcs_available("xg", timeout=20)
-            claim = cl.store.try_get("ResourceClaim", "default", "xg-0-xgmi-domain")
+            claim = cl.store.try_get("ResourceClaim", "default", "xg-0-xgmi-default")
             assert claim is not None
             assert claim["spec"]["devices"]["requests"][0]["deviceClassName"] == \
                 "xgmi.amd.com"
             pod = cl.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "xg-0-inf"})[0]
             assert pod["spec"]["resourceClaims"][0]["resourceClaimName"] == \
-                "xg-0-xgmi-domain"
+                "xg-0-xgmi-default"
         finally:
             cl.stop()
 
@@ -203,3 +203,68 @@ class TestLastErrors:
         cluster.wait_for(recorded, timeout=15, desc="lastErrors recorded")
         p = cluster.store.get(c.KIND_PCS, "default", "err1")
         assert p["status"]["lastErrors"][0]["code"] == "ERR_RECONCILE"
+
+
+class TestXGMIGroups:
+    def _pcs(self, pcs_ann=None, clique_anns=(None, None), gpu=(True, True)):
+        cliques = []
+        for i, (ann, has_gpu) in enumerate(zip(clique_anns, gpu)):
+            res = {"cpu": "1"}
+            if has_gpu:
+                res[c.AMD_GPU_RESOURCE] = "1"
+            cl = {"name": f"c{i}", "spec": {
+                "roleName": f"c{i}", "replicas": 1,
+                "podSpec": {"containers": [{"name": "m", "image": "i",
+                                            "resources": {"requests": res}}]}}}
+            if ann is not None:
+                cl["annotations"] = {c.ANNOTATION_XGMI_GROUP: ann}
+            cliques.append(cl)
+        pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+               "metadata": {"name": "xgrp"},
+               "spec": {"replicas": 1, "template": {"cliques": cliques}}}
+        if pcs_ann is not None:
+            pcs["metadata"]["annotations"] = {c.ANNOTATION_XGMI_GROUP: pcs_ann}
+        return pcs
+
+    def test_group_hierarchy_and_optout(self, cluster):
+        cluster.add_virtual_nodes(1, gpus=8)
+        # PCS-level group "fabric"; clique c1 opts out with "none"
+        cluster.store.create(self._pcs(pcs_ann="fabric", clique_anns=(None, "none")))
+        cluster.wait_pcs_available("xgrp", timeout=20)
+        claims = [x["metadata"]["name"]
+                  for x in cluster.store.list("ResourceClaim", "default")]
+        assert claims == ["xgrp-0-xgmi-fabric"]
+        p0 = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "xgrp-0-c0"})[0]
+        assert p0["spec"]["resourceClaims"][0]["resourceClaimName"] == \
+            "xgrp-0-xgmi-fabric"
+        p1 = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "xgrp-0-c1"})[0]
+        assert "resourceClaims" not in p1["spec"]
+
+    def test_non_gpu_clique_silently_skips_inherited(self, cluster):
+        cluster.add_virtual_nodes(1, gpus=8)
+        cluster.store.create(self._pcs(pcs_ann="fabric", gpu=(True, False)))
+        cluster.wait_pcs_available("xgrp", timeout=20)
+        p1 = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "xgrp-0-c1"})[0]
+        assert "resourceClaims" not in p1["spec"]
+
+    def test_explicit_group_on_non_gpu_rejected(self, cluster):
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError):
+            cluster.store.create(self._pcs(clique_anns=("g1", None),
+                                           gpu=(False, True)))
+
+    def test_invalid_group_name_rejected(self, cluster):
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError):
+            cluster.store.create(self._pcs(pcs_ann="Bad_Name!"))
+
+    def test_group_annotation_immutable(self, cluster):
+        cluster.add_virtual_nodes(1, gpus=8)
+        cluster.store.create(self._pcs(pcs_ann="fabric"))
+        cluster.wait_pcs_available("xgrp", timeout=20)
+        from grove_amd.kubecore.store import ApiError
+        with pytest.raises(ApiError):
+            cluster.store.patch(
+                c.KIND_PCS, "default", "xgrp",
+                lambda o: o["metadata"]["annotations"].update(
+                    {c.ANNOTATION_XGMI_GROUP: "other"}))

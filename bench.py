@@ -132,9 +132,12 @@ class Comm:
         return out
 
     def heartbeat(self):
-        """One RCCL all-reduce over xGMI per dispatch cycle."""
+        """One RCCL all-reduce over xGMI per dispatch cycle (synchronized — an
+        unsynchronized 4-byte all-reduce every cycle would pile thousands of kernels
+        onto the stream over a long run)."""
         if self.nccl_group is not None:
             self.dist.all_reduce(self._hb, group=self.nccl_group)
+            torch.cuda.synchronize()
 
     def barrier_sync(self):
         if torch.cuda.is_available():

@@ -69,7 +69,8 @@ class Cluster:
         self.pcs_rec = PodCliqueSetReconciler(self.store, scheduler_name,
                                               auto_xgmi_domain=auto_xgmi_domain)
         self.pclq_rec = PodCliqueReconciler(self.store, scheduler_name)
-        self.pcsg_rec = PCSGReconciler(self.store, scheduler_name)
+        self.pcsg_rec = PCSGReconciler(self.store, scheduler_name,
+                                       auto_xgmi_domain=auto_xgmi_domain)
         self.podgang_rec = PodGangReconciler(self.store, self.registry)
         self.ctb_rec = ClusterTopologyReconciler(self.store, self.registry)
         self.hpa_rec = HPAReconciler(self.store)

@@ -27,9 +27,11 @@ log = logging.getLogger("grove.pcsg")
 
 
 class PCSGReconciler:
-    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG):
+    def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG,
+                 auto_xgmi_domain: bool = False):
         self.store = store
         self.scheduler_name = scheduler_name
+        self.auto_xgmi_domain = auto_xgmi_domain
 
     def reconcile(self, namespace: str, name: str) -> Result:
         pcsg = self.store.try_get(c.KIND_PCSG, namespace, name)
@@ -106,6 +108,10 @@ class PCSGReconciler:
         sg_claims = resourceclaims.pcsg_claims(pcs, sg_cfg, sg_fqn, replicas)
         resourceclaims.ensure_claims(
             self.store, [(cl0, e) for (cl0, e, _j) in sg_claims])
+        # xGMI-group claims are PCS-replica-scoped (created by the PCS reconciler);
+        # member cliques in a group reference them too
+        xgmi_groups = resourceclaims.effective_xgmi_groups(
+            pcs, self.auto_xgmi_domain)
         existing = {q["metadata"]["name"]: q for q in self._member_pclqs(pcsg)}
         expected: set = set()
         for j in range(replicas):
@@ -127,6 +133,12 @@ class PCSGReconciler:
                     refs = resourceclaims.claim_refs_for_clique(
                         [(cl0, e) for (cl0, e, jj) in sg_claims
                          if jj is None or jj == j], mn)
+                    g = xgmi_groups.get(mn)
+                    if g is not None:
+                        refs = refs + [{
+                            "name": f"{resourceclaims.XGMI_TEMPLATE_NAME}-{g}",
+                            "resourceClaimName":
+                                f"{pcs['metadata']['name']}-{pcs_replica}-xgmi-{g}"}]
                     if refs:
                         obj["spec"]["resourceClaims"] = refs
                     obj["spec"]["updateStrategy"] = (

@@ -268,3 +268,25 @@ class TestXGMIGroups:
                 c.KIND_PCS, "default", "xgrp",
                 lambda o: o["metadata"]["annotations"].update(
                     {c.ANNOTATION_XGMI_GROUP: "other"}))
+
+    def test_pcsg_members_share_group_claim(self, cluster):
+        pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+               "metadata": {"name": "xsg",
+                            "annotations": {c.ANNOTATION_XGMI_GROUP: "fab"}},
+               "spec": {"replicas": 1, "template": {
+                   "cliques": [{"name": "w", "spec": {
+                       "roleName": "w", "replicas": 1,
+                       "podSpec": {"containers": [{
+                           "name": "m", "image": "i",
+                           "resources": {"requests": {
+                               c.AMD_GPU_RESOURCE: "1"}}}]}}}],
+                   "podCliqueScalingGroups": [{"name": "sg", "cliqueNames": ["w"],
+                                               "replicas": 2, "minAvailable": 1}]}}}
+        cluster.add_virtual_nodes(1, gpus=8)
+        cluster.store.create(pcs)
+        cluster.wait_pcs_available("xsg", timeout=20)
+        cluster.wait_pods_ready({c.LABEL_PART_OF: "xsg"}, 2, timeout=20)
+        for p in cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "xsg"}):
+            refs = p["spec"].get("resourceClaims") or []
+            assert any(r["resourceClaimName"] == "xsg-0-xgmi-fab" for r in refs), \
+                p["metadata"]["name"]

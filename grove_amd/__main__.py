@@ -31,6 +31,8 @@ def cmd_operator(args) -> int:
     cluster = Cluster(
         scheduler_name=cfg.default_scheduler,
         concurrent_syncs=cfg.concurrent_syncs("podCliqueSet"),
+        controller_workers={name: cc.concurrent_syncs
+                            for name, cc in cfg.controllers.items()},
         enable_authorizer=cfg.authorizer_enabled,
         auto_xgmi_domain=cfg.auto_xgmi_domain_enabled,
         use_native_scheduler=None,

@@ -37,6 +37,7 @@ class Cluster:
     def __init__(self,
                  scheduler_name: str = c.SCHEDULER_AMD_GANG,
                  concurrent_syncs: int = 4,
+                 controller_workers: Optional[Dict[str, int]] = None,
                  startup_latency_s: float = 0.0,
                  ready_latency_s: float = 0.0,
                  pod_payload: Optional[Callable[[Obj], None]] = None,
@@ -83,12 +84,18 @@ class Cluster:
                                       payload=pod_payload)
 
         m = self.manager
+        workers = controller_workers or {}
+
+        def w(name: str) -> int:
+            return int(workers.get(name, concurrent_syncs))
+
         self.c_pcs = m.add_controller(Controller(
-            "podcliqueset", self.pcs_rec.reconcile, workers=concurrent_syncs))
+            "podcliqueset", self.pcs_rec.reconcile, workers=w("podCliqueSet")))
         self.c_pclq = m.add_controller(Controller(
-            "podclique", self.pclq_rec.reconcile, workers=concurrent_syncs))
+            "podclique", self.pclq_rec.reconcile, workers=w("podClique")))
         self.c_pcsg = m.add_controller(Controller(
-            "podcliquescalinggroup", self.pcsg_rec.reconcile, workers=concurrent_syncs))
+            "podcliquescalinggroup", self.pcsg_rec.reconcile,
+            workers=w("podCliqueScalingGroup")))
         self.c_podgang = m.add_controller(Controller(
             "podgang", self.podgang_rec.reconcile, workers=2))
         self.c_ctb = m.add_controller(Controller(

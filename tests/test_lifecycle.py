@@ -65,6 +65,8 @@ class TestGangTermination:
             q = cluster.store.try_get(c.KIND_PCLQ, "default", "gt2-0-w")
             return q is not None and q["metadata"]["uid"] != old_uid
         cluster.wait_for(recreated, timeout=60, desc="PCLQ recreated by gang termination")
+        # gang termination recorded a Warning event
+        assert any(e.get("reason") == "GangTerminated" for e in cluster.store.events)
         # uncordon → fresh gang reaches available again
         for n in cluster.store.list("Node"):
             cluster.store.patch("Node", None, n["metadata"]["name"],

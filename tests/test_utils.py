@@ -179,3 +179,36 @@ class TestPlacementScore:
         s4 = placement_score(4, 8)
         assert s2 < 153.0 and s4 < s2
         assert s2 == pytest.approx(25.0)
+
+
+class TestDeletionOrder:
+    def test_victim_priority(self):
+        """deletionsort.go parity: gated → pending → unscheduled → not-ready →
+        younger first."""
+        from grove_amd.controllers.podclique import PodCliqueReconciler
+
+        def pod(name, ts, gated=False, phase="Running", scheduled=True, ready=True):
+            conds = []
+            if scheduled:
+                conds.append({"type": "PodScheduled", "status": "True"})
+            if ready:
+                conds.append({"type": "Ready", "status": "True"})
+            return {"metadata": {"name": name, "creationTimestamp": ts},
+                    "spec": {"schedulingGates": [{"name": "g"}] if gated else [],
+                             "nodeName": "n" if scheduled else None},
+                    "status": {"phase": phase, "conditions": conds}}
+
+        ready_old = pod("ready-old", "2026-01-01T00:00:00Z")
+        ready_new = pod("ready-new", "2026-01-02T00:00:00Z")
+        gated = pod("gated", "2026-01-01T12:00:00Z", gated=True, phase="Pending",
+                    scheduled=False, ready=False)
+        pending = pod("pending", "2026-01-01T12:00:00Z", phase="Pending",
+                      scheduled=False, ready=False)
+        unready = pod("unready", "2026-01-01T12:00:00Z", ready=False)
+        order = [p["metadata"]["name"] for p in PodCliqueReconciler._deletion_order(
+            [ready_old, ready_new, unready, pending, gated])]
+        assert order[0] == "gated"
+        assert order[1] == "pending"
+        assert order[2] == "unready"
+        # among ready pods, the younger one is deleted first
+        assert order[3] == "ready-new" and order[4] == "ready-old"

@@ -242,3 +242,45 @@ def test_httpclient_roundtrip_and_watch(served_cluster):
     cl.delete("Pod", "default", "hc-p0")
     with pytest.raises(ApiError):
         cl.get("Pod", "default", "hc-p0")
+
+
+def test_cli_apply_get_roundtrip(served_cluster, simple1_yaml, tmp_path, capsys):
+    """The kubectl-analog CLI surface: `grove-amd apply -f` then `get` (table/yaml/
+    json) against the running apiserver."""
+    from grove_amd.__main__ import main
+
+    cluster, api = served_cluster
+    f = tmp_path / "pcs.yaml"
+    f.write_text(simple1_yaml)
+    server = "http://127.0.0.1:18133"
+    assert main(["apply", "-f", str(f), "--server", server]) == 0
+    out = capsys.readouterr().out
+    assert "created PodCliqueSet/simple1" in out
+
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        if cluster.store.list(c.KIND_PCLQ, namespace="default"):
+            break
+        time.sleep(0.05)
+
+    assert main(["get", "podcliquesets", "--server", server]) == 0
+    out = capsys.readouterr().out
+    assert "simple1" in out and "NAME" in out
+    assert main(["get", "podcliques", "--server", server, "-o", "json"]) == 0
+    items = json.loads(capsys.readouterr().out)
+    assert any(o["metadata"]["name"].startswith("simple1-0-") for o in items)
+    assert main(["get", "podcliquesets", "simple1", "--server", server,
+                 "-o", "yaml"]) == 0
+    assert "kind: PodCliqueSet" in capsys.readouterr().out
+    # unknown resource → error exit
+    assert main(["get", "bogus", "--server", server]) == 1
+    capsys.readouterr()
+
+
+def test_cli_version_and_crd_render(capsys):
+    from grove_amd.__main__ import main
+    assert main(["version"]) == 0
+    assert "grove-amd" in capsys.readouterr().out
+    assert main(["install-crds"]) == 0
+    out = capsys.readouterr().out
+    assert "podcliquesets.grove.io" in out and "podgangs.scheduler.grove.io" in out

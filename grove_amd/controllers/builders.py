@@ -9,6 +9,7 @@ PodGang skeleton — components/podgang/podgang.go:129. Fresh dict-based impleme
 from __future__ import annotations
 
 import copy
+import os
 from typing import Any, Dict, List, Optional
 
 from ..api import constants as c
@@ -164,6 +165,39 @@ def build_pod(pcs: Obj, pclq: Obj, pod_index: int, scheduler_name: str,
         if pcsg_template_num_pods is not None:
             env.append({"name": c.ENV_PCSG_TEMPLATE_NUM_PODS,
                         "value": str(pcsg_template_num_pods)})
+    # startup ordering: inject the grove-initc init container (pod.go:315-371 parity) —
+    # flags are --podcliques=<parentFQN>:<minAvailable>; the SA token secret
+    # <pcs>-ic-sat is mounted so initc can watch pods. The image honors the
+    # GROVE_INIT_CONTAINER_IMAGE env contract (pod/initcontainer.go:37).
+    starts_after = pclq["spec"].get("startsAfter") or []
+    if starts_after:
+        def _parent_min_available(fqn: str) -> int:
+            for cl in (((pcs.get("spec") or {}).get("template") or {})
+                       .get("cliques") or []):
+                if fqn.endswith("-" + cl["name"]):
+                    sp = cl.get("spec") or {}
+                    return int(sp.get("minAvailable") or sp.get("replicas", 1))
+            return 1
+        initc = {
+            "name": "grove-initc",
+            "image": os.environ.get("GROVE_INIT_CONTAINER_IMAGE",
+                                    "grove-initc:latest"),
+            "command": ["python", "-m", "grove_amd.initc"],
+            "args": (["--namespace", namespace, "--podgang", podgang_name]
+                     + [f"--podcliques={fqn}:{_parent_min_available(fqn)}"
+                        for fqn in starts_after]),
+            "volumeMounts": [{"name": "grove-ic-sat",
+                              "mountPath": "/var/grove/sa-token",
+                              "readOnly": True}],
+        }
+        ics = pod_spec.setdefault("initContainers", [])
+        if not any(ic.get("name") == "grove-initc" for ic in ics):
+            ics.insert(0, initc)
+        vols = pod_spec.setdefault("volumes", [])
+        if not any(v.get("name") == "grove-ic-sat" for v in vols):
+            vols.append({"name": "grove-ic-sat", "secret": {
+                "secretName": namegen.initc_sa_token_secret_name(pcs_name)}})
+
     for ctr in pod_spec.get("containers", []) + pod_spec.get("initContainers", []):
         ctr.setdefault("env", [])
         ctr["env"] = env + ctr["env"]

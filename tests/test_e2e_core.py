@@ -136,6 +136,18 @@ def test_startup_ordering_explicit(cluster):
     cluster.wait_pcs_available("so", timeout=20)
     pclq_b = cluster.store.get(c.KIND_PCLQ, "default", "so-0-b")
     assert pclq_b["spec"]["startsAfter"] == ["so-0-a"]
+    # dependent pods carry the visible grove-initc init container (pod.go:315-371)
+    b_pods = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "so-0-b"})
+    ic = b_pods[0]["spec"]["initContainers"][0]
+    assert ic["name"] == "grove-initc"
+    assert "--podcliques=so-0-a:2" in ic["args"]
+    assert any(v.get("secret", {}).get("secretName") == "so-ic-sat"
+               for v in b_pods[0]["spec"]["volumes"])
+    assert any(e["name"] == c.ENV_PCLQ_NAME for e in ic["env"])
+    # parent-clique pods have no init container injected
+    a_pods = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "so-0-a"})
+    assert not any(x.get("name") == "grove-initc"
+                   for x in a_pods[0]["spec"].get("initContainers", []))
 
 
 def test_in_order_startup_dependencies(cluster):

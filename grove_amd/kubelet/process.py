@@ -89,14 +89,22 @@ class ProcessKubelet:
         gpu_ids = ann.get(GPU_IDS_ANNOTATION)
         if gpu_ids:
             env["HIP_VISIBLE_DEVICES"] = gpu_ids
-        # startup deps from the PCLQ (the initc --podcliques flags analog)
-        pclq_name = pod["metadata"].get("labels", {}).get(c.LABEL_PODCLIQUE)
-        pclq = self.store.try_get(c.KIND_PCLQ, ns, pclq_name) if pclq_name else None
+        # startup deps: prefer the pod's own grove-initc init-container flags (what a
+        # real kubelet would execute), falling back to the PCLQ's startsAfter list
         deps = []
-        for fqn in (pclq or {}).get("spec", {}).get("startsAfter") or []:
-            dep = self.store.try_get(c.KIND_PCLQ, ns, fqn)
-            min_avail = int((dep or {}).get("spec", {}).get("minAvailable", 1))
-            deps.append(f"{fqn}:{min_avail}")
+        for ic in pod["spec"].get("initContainers") or []:
+            if ic.get("name") == "grove-initc":
+                deps = [a.split("=", 1)[1] for a in ic.get("args", [])
+                        if a.startswith("--podcliques=")]
+                break
+        if not deps:
+            pclq_name = pod["metadata"].get("labels", {}).get(c.LABEL_PODCLIQUE)
+            pclq = (self.store.try_get(c.KIND_PCLQ, ns, pclq_name)
+                    if pclq_name else None)
+            for fqn in (pclq or {}).get("spec", {}).get("startsAfter") or []:
+                dep = self.store.try_get(c.KIND_PCLQ, ns, fqn)
+                min_avail = int((dep or {}).get("spec", {}).get("minAvailable", 1))
+                deps.append(f"{fqn}:{min_avail}")
         if deps:
             env["GROVE_STARTS_AFTER"] = ",".join(deps)
         proc = subprocess.Popen(

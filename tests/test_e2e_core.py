@@ -307,3 +307,21 @@ def test_reapply_is_idempotent(cluster, simple1_yaml):
                   for p in cluster.store.list("Pod", "default",
                                               {c.LABEL_PART_OF: "simple1"})}
     assert pods_after == pods_before  # zero churn
+
+
+def test_reconcile_trigger_annotation(cluster):
+    """register.go:70-232 parity: stamping grove.io/reconcile-trigger on a PCS forces
+    a reconcile even with no spec change (our watch enqueues all PCS events)."""
+    pcs = _gpu_pcs("rt", cliques=(("a", 1, 1),), gpus_per_pod=0)
+    cluster.add_virtual_nodes(1)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("rt", timeout=20)
+    time.sleep(0.3)
+    n0 = cluster.c_pcs.reconcile_count
+
+    def bump(o):
+        o["metadata"].setdefault("annotations", {})[
+            c.ANNOTATION_RECONCILE_TRIGGER] = "manual-1"
+    cluster.store.patch(c.KIND_PCS, "default", "rt", bump)
+    cluster.wait_for(lambda: cluster.c_pcs.reconcile_count > n0, timeout=10,
+                     desc="annotation-triggered reconcile")

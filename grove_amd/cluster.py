@@ -136,8 +136,14 @@ class Cluster:
             # (pod/syncflow.go:386-424 watch-mapping parity).
             pg = md.get("labels", {}).get(c.LABEL_PODGANG)
             if pg and not md.get("labels", {}).get(c.LABEL_BASE_PODGANG):
-                for q in self.store.list(c.KIND_PCLQ, ns, {c.LABEL_BASE_PODGANG: pg}):
-                    self.c_pclq.enqueue(ns, q["metadata"]["name"])
+                # only a scheduledReplicas change can unblock dependents — skip the
+                # (indexed, but per-event) lookup for all other status churn
+                new_sched = (obj.get("status") or {}).get("scheduledReplicas")
+                old_sched = ((_old or {}).get("status") or {}).get("scheduledReplicas")
+                if new_sched != old_sched or ev == "DELETED":
+                    for q in self.store.list(c.KIND_PCLQ, ns,
+                                             {c.LABEL_BASE_PODGANG: pg}):
+                        self.c_pclq.enqueue(ns, q["metadata"]["name"])
 
         def on_pcsg(ev: str, obj: Obj, _old) -> None:
             md = obj["metadata"]

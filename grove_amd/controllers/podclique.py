@@ -295,6 +295,14 @@ class PodCliqueReconciler:
             else:
                 cond.set_condition(o, c.COND_MIN_AVAILABLE_BREACHED, False,
                                    c.REASON_SUFFICIENT_READY_PODS)
+        # No-op fast path: apply the status function to the in-hand copy first and
+        # skip the store round-trip (lock + copy + deep compare) when nothing would
+        # change — the dominant case during churn storms at 10k-pod scale.
+        from ..kubecore.store import json_copy
+        old_status = json_copy(pclq.get("status") or {})
+        upd(pclq)
+        if pclq.get("status") == old_status:
+            return
         try:
             self.store.patch(c.KIND_PCLQ, namespace, name, upd, status=True)
         except ApiError:

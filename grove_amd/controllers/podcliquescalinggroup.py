@@ -300,6 +300,13 @@ class PCSGReconciler:
                         and avail >= min_avail:
                     cond.set_condition(o, c.COND_GANG_TERMINATION_IN_PROGRESS, False,
                                        c.REASON_GANG_TERMINATION_ACTIVE)
+        # no-op fast path (see podclique._reconcile_status): skip the store
+        # round-trip when the recomputed status is unchanged
+        from ..kubecore.store import json_copy
+        old_status = json_copy(pcsg.get("status") or {})
+        upd(pcsg)
+        if pcsg.get("status") == old_status:
+            return
         try:
             self.store.patch(c.KIND_PCSG, namespace, name, upd, status=True)
         except ApiError:

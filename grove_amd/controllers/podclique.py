@@ -145,10 +145,12 @@ class PodCliqueReconciler:
         return sorted(pods, key=rank)
 
     def _find_pcs(self, pclq: Obj) -> Optional[Obj]:
+        # zero-copy read: callers only read the PCS (template lookups, gen hash)
         pcs_name = pclq["metadata"]["labels"].get(c.LABEL_PART_OF)
         if not pcs_name:
             return None
-        return self.store.try_get(c.KIND_PCS, pclq["metadata"].get("namespace"), pcs_name)
+        return self.store.try_get(c.KIND_PCS, pclq["metadata"].get("namespace"),
+                                  pcs_name, copy=False)
 
     @staticmethod
     def _pcsg_template_num_pods(pcs: Obj, pclq: Obj) -> Optional[int]:
@@ -174,7 +176,7 @@ class PodCliqueReconciler:
         podgang_name = pclq["metadata"]["labels"].get(c.LABEL_PODGANG)
         if not podgang_name:
             return
-        podgang = self.store.try_get(c.KIND_PODGANG, ns, podgang_name)
+        podgang = self.store.try_get(c.KIND_PODGANG, ns, podgang_name, copy=False)
         if podgang is None:
             return
         refs = set()
@@ -209,14 +211,15 @@ class PodCliqueReconciler:
     def _is_base_podgang_scheduled(self, ns: Optional[str], base_name: str) -> bool:
         """Base gang 'scheduled' = every podGroup's PCLQ has scheduledReplicas >=
         minReplicas (syncflow.go:343-424)."""
-        base = self.store.try_get(c.KIND_PODGANG, ns, base_name)
+        base = self.store.try_get(c.KIND_PODGANG, ns, base_name, copy=False)
         if base is None:
             return False
         groups = (base.get("spec") or {}).get("podgroups") or []
         if not groups:
             return False
         for group in groups:
-            pclq = self.store.try_get(c.KIND_PCLQ, ns, group.get("name", ""))
+            pclq = self.store.try_get(c.KIND_PCLQ, ns, group.get("name", ""),
+                                      copy=False)
             if pclq is None:
                 return False
             scheduled = int((pclq.get("status") or {}).get("scheduledReplicas", 0))

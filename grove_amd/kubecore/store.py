@@ -237,7 +237,8 @@ class Store:
             self._notify(tbl, ADDED, obj)
         return json_copy(obj)  # caller gets a private copy; stored one is immutable
 
-    def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
+    def get(self, kind: str, namespace: Optional[str], name: str,
+            copy: bool = True) -> Obj:
         ns = "" if kind in CLUSTER_SCOPED else (namespace or "default")
         with self._lock:
             tbl = self._table(kind)
@@ -245,12 +246,15 @@ class Store:
         if obj is None:
             raise not_found(kind, name)
         # stored objects are immutable after insert (updates replace wholesale), so
-        # the defensive copy happens OUTSIDE the lock — the lock hold is a dict get
-        return json_copy(obj)
+        # the defensive copy happens OUTSIDE the lock — the lock hold is a dict get.
+        # copy=False returns the stored object itself for READ-ONLY consumers (the
+        # immutability contract makes this safe as long as the caller never mutates).
+        return json_copy(obj) if copy else obj
 
-    def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
+    def try_get(self, kind: str, namespace: Optional[str], name: str,
+                copy: bool = True) -> Optional[Obj]:
         try:
-            return self.get(kind, namespace, name)
+            return self.get(kind, namespace, name, copy=copy)
         except ApiError:
             return None
 

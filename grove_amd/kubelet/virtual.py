@@ -30,11 +30,11 @@ def startup_dependencies_met(store: Store, pod: Obj) -> bool:
     pclq_name = pod["metadata"].get("labels", {}).get(c.LABEL_PODCLIQUE)
     if not pclq_name:
         return True
-    pclq = store.try_get(c.KIND_PCLQ, ns, pclq_name)
+    pclq = store.try_get(c.KIND_PCLQ, ns, pclq_name, copy=False)
     if pclq is None:
         return True
     for dep_fqn in pclq["spec"].get("startsAfter") or []:
-        dep = store.try_get(c.KIND_PCLQ, ns, dep_fqn)
+        dep = store.try_get(c.KIND_PCLQ, ns, dep_fqn, copy=False)
         if dep is None:
             return False
         min_avail = int(dep["spec"].get("minAvailable", 1))
@@ -63,14 +63,15 @@ class VirtualKubelet:
             return False
         # nodes registered by a remote node agent run their own kubelet — the
         # in-process virtual kubelet must not race it for pod lifecycle
-        node = self.store.try_get("Node", None, node_name)
+        node = self.store.try_get("Node", None, node_name, copy=False)
         if node is not None and (node["metadata"].get("annotations") or {}).get(
                 self.EXTERNAL_KUBELET_ANNOTATION) == "true":
             return False
         return True
 
     def reconcile(self, namespace: str, name: str) -> Result:
-        pod = self.store.try_get("Pod", namespace, name)
+        # zero-copy read: this loop only reads the pod; all mutations go via patch
+        pod = self.store.try_get("Pod", namespace, name, copy=False)
         if pod is None:
             self._started_at.pop(f"{namespace}/{name}", None)
             return Result.DONE

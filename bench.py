@@ -198,7 +198,7 @@ def main() -> None:
     ap.add_argument("--payload", type=str, default="2048x2048x2048x1",
                     help="per-pod GEMM payload MxNxKxiters")
     ap.add_argument("--step-timeout", type=float, default=120.0)
-    ap.add_argument("--inflight", type=int, default=1,
+    ap.add_argument("--inflight", type=int, default=2,
                     help="pipelined steps kept in flight (1 = fully serial)")
     args = ap.parse_args()
 

@@ -198,6 +198,8 @@ def main() -> None:
     ap.add_argument("--payload", type=str, default="2048x2048x2048x1",
                     help="per-pod GEMM payload MxNxKxiters")
     ap.add_argument("--step-timeout", type=float, default=120.0)
+    ap.add_argument("--workers", type=int, default=4,
+                    help="per-controller worker threads (ConcurrentSyncs)")
     ap.add_argument("--inflight", type=int, default=2,
                     help="pipelined steps kept in flight (1 = fully serial)")
     args = ap.parse_args()
@@ -276,7 +278,7 @@ def _serve_agent(comm: Comm) -> None:
 
 
 def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
-    cluster = Cluster(concurrent_syncs=4)
+    cluster = Cluster(concurrent_syncs=args.workers)
     kubelet = DispatchKubelet(cluster.store)
     # replace the virtual kubelet with the dispatch kubelet (payload runs on ranks)
     cluster.c_kubelet.stop()
@@ -410,7 +412,7 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
             "gangs_per_step": args.gangs_per_step,
             "payload": payload,
             "inflight": args.inflight,
-            "concurrent_syncs": 4,
+            "concurrent_syncs": args.workers,
             "operator_version": __import__("grove_amd").__version__,
             "scheduler": "amd-gang-scheduler (native xGMI Filter/Score)",
             "rccl_allreduce_busbw_gbps": round(rccl_busbw, 1) if rccl_busbw else None,

@@ -222,6 +222,40 @@ def build_app(store: Store, metrics_fn=None):
         except ApiError as e:
             return err(e)
 
+    # --- kubectl-style API discovery (client-go discovery analog) ---
+    @app.get("/apis")
+    async def discovery_groups():
+        def g(name, version):
+            return {"name": name,
+                    "versions": [{"groupVersion": f"{name}/{version}",
+                                  "version": version}],
+                    "preferredVersion": {"groupVersion": f"{name}/{version}",
+                                         "version": version}}
+        return JSONResponse({"kind": "APIGroupList", "apiVersion": "v1",
+                             "groups": [g("grove.io", "v1alpha1"),
+                                        g("scheduler.grove.io", "v1alpha1"),
+                                        g("autoscaling", "v2")]})
+
+    @app.get("/apis/{group}/{version}")
+    async def discovery_resources(group: str, version: str):
+        gv = f"{group}/{version}"
+        res = []
+        for plural, kind in sorted(PLURALS.items()):
+            k_group = ("grove.io/v1alpha1" if kind.startswith("PodClique")
+                       or kind == "ClusterTopologyBinding" else
+                       "scheduler.grove.io/v1alpha1" if kind == "PodGang" else None)
+            if k_group != gv:
+                continue
+            res.append({"name": plural, "singularName": kind.lower(), "kind": kind,
+                        "namespaced": plural not in CLUSTER_SCOPED_PLURALS,
+                        "verbs": ["create", "delete", "get", "list", "patch",
+                                  "update", "watch"]})
+            res.append({"name": f"{plural}/status", "singularName": "", "kind": kind,
+                        "namespaced": plural not in CLUSTER_SCOPED_PLURALS,
+                        "verbs": ["get", "patch", "update"]})
+        return JSONResponse({"kind": "APIResourceList", "apiVersion": "v1",
+                             "groupVersion": gv, "resources": res})
+
     @app.get("/debug/profile")
     async def debug_profile(seconds: float = 2.0, interval_ms: float = 10.0):
         """Sampling profiler (the Pyroscope/pprof analog): samples every thread's

@@ -284,3 +284,20 @@ def test_cli_version_and_crd_render(capsys):
     assert main(["install-crds"]) == 0
     out = capsys.readouterr().out
     assert "podcliquesets.grove.io" in out and "podgangs.scheduler.grove.io" in out
+
+
+def test_api_discovery(served_cluster):
+    """client-go discovery analog: /apis lists both grove groups; the group-version
+    endpoint returns an APIResourceList with namespaced flags and verbs."""
+    groups = _get("http://127.0.0.1:18133/apis")
+    names = {g["name"] for g in groups["groups"]}
+    assert {"grove.io", "scheduler.grove.io"} <= names
+    rl = _get("http://127.0.0.1:18133/apis/grove.io/v1alpha1")
+    assert rl["kind"] == "APIResourceList"
+    by_name = {r["name"]: r for r in rl["resources"]}
+    assert by_name["podcliquesets"]["namespaced"] is True
+    assert by_name["clustertopologybindings"]["namespaced"] is False
+    assert "watch" in by_name["podcliques"]["verbs"]
+    assert "podcliquesets/status" in by_name
+    rl2 = _get("http://127.0.0.1:18133/apis/scheduler.grove.io/v1alpha1")
+    assert any(r["name"] == "podgangs" for r in rl2["resources"])

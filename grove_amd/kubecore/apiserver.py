@@ -158,6 +158,44 @@ def build_app(store: Store, metrics_fn=None):
         except ApiError as e:
             return err(e)
 
+    # kubectl `patch --type=merge` analog: RFC 7386 JSON merge patch (null deletes a
+    # key; objects merge recursively; everything else replaces).
+    def _merge_patch(target, patch):
+        if not isinstance(patch, dict) or not isinstance(target, dict):
+            return patch
+        for k, v in patch.items():
+            if v is None:
+                target.pop(k, None)
+            else:
+                target[k] = _merge_patch(target.get(k), v)
+        return target
+
+    @app.patch("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
+    @app.patch("/api/{version}/namespaces/{ns}/{plural}/{name}")
+    async def patch_ns(request: Request, plural: str, ns: str, name: str,
+                       group: str = "", version: str = "v1"):
+        try:
+            body = await request.json()
+
+            def apply(o):
+                _merge_patch(o, body)
+            return JSONResponse(store.patch(kind_of(plural), ns, name, apply))
+        except ApiError as e:
+            return err(e)
+
+    @app.patch("/apis/{group}/{version}/{plural}/{name}")
+    @app.patch("/api/{version}/{plural}/{name}")
+    async def patch_cluster(request: Request, plural: str, name: str,
+                            group: str = "", version: str = "v1"):
+        try:
+            body = await request.json()
+
+            def apply(o):
+                _merge_patch(o, body)
+            return JSONResponse(store.patch(kind_of(plural), None, name, apply))
+        except ApiError as e:
+            return err(e)
+
     # ---- cluster-scoped ----
     @app.get("/apis/{group}/{version}/{plural}")
     @app.get("/api/{version}/{plural}")

@@ -301,3 +301,25 @@ def test_api_discovery(served_cluster):
     assert "podcliquesets/status" in by_name
     rl2 = _get("http://127.0.0.1:18133/apis/scheduler.grove.io/v1alpha1")
     assert any(r["name"] == "podgangs" for r in rl2["resources"])
+
+
+def test_http_merge_patch(served_cluster):
+    """kubectl patch --type=merge analog: recursive merge, null deletes."""
+    import urllib.request
+    cluster, api = served_cluster
+    cluster.store.create({"apiVersion": "grove.io/v1alpha1", "kind": c.KIND_PCLQ,
+                          "metadata": {"name": "mp", "namespace": "default",
+                                       "labels": {"keep": "a", "drop": "b"}},
+                          "spec": {"roleName": "r", "replicas": 1,
+                                   "podSpec": {"containers": []}}})
+    patch = {"metadata": {"labels": {"drop": None, "new": "c"}},
+             "spec": {"replicas": 3}}
+    req = urllib.request.Request(
+        "http://127.0.0.1:18133/apis/grove.io/v1alpha1/namespaces/default/"
+        "podcliques/mp", data=json.dumps(patch).encode(), method="PATCH",
+        headers={"Content-Type": "application/merge-patch+json"})
+    with urllib.request.urlopen(req, timeout=5) as r:
+        out = json.loads(r.read())
+    assert out["spec"]["replicas"] == 3
+    assert out["metadata"]["labels"] == {"keep": "a", "new": "c"}
+    assert out["spec"]["roleName"] == "r"  # untouched fields survive

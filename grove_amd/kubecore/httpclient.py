@@ -131,6 +131,12 @@ class HttpStoreClient:
                 last = e
         raise last or ApiError(409, "Conflict", name)
 
+    def merge_patch(self, kind: str, namespace: Optional[str], name: str,
+                    patch: Obj) -> Obj:
+        """Server-side RFC 7386 merge patch (single round-trip, no read-modify-write
+        conflict loop) — the client-go Patch(types.MergePatchType) analog."""
+        return self._request("PATCH", self._url(kind, namespace, name), patch)
+
     def delete(self, kind: str, namespace: Optional[str], name: str,
                cascade: bool = True) -> None:
         self._request("DELETE", self._url(kind, namespace, name))

@@ -323,3 +323,8 @@ def test_http_merge_patch(served_cluster):
     assert out["spec"]["replicas"] == 3
     assert out["metadata"]["labels"] == {"keep": "a", "new": "c"}
     assert out["spec"]["roleName"] == "r"  # untouched fields survive
+    # client-side single-round-trip analog
+    from grove_amd.kubecore.httpclient import HttpStoreClient
+    cli = HttpStoreClient("http://127.0.0.1:18133")
+    out2 = cli.merge_patch(c.KIND_PCLQ, "default", "mp", {"spec": {"replicas": 5}})
+    assert out2["spec"]["replicas"] == 5

@@ -162,6 +162,21 @@ def cmd_get(args) -> int:
         url += f"/{args.name}"
     elif args.selector:
         url += f"?labelSelector={urllib.parse.quote(args.selector)}"
+    if getattr(args, "watch", False):
+        # kubectl get -w analog: follow the ndjson watch stream
+        sep = "&" if "?" in url else "?"
+        with urllib.request.urlopen(url + f"{sep}watch=true") as r:
+            try:
+                for line in r:
+                    ev = json.loads(line)
+                    o = ev.get("object", {})
+                    print(ev.get("type", "?"),
+                          o.get("kind", kind),
+                          o.get("metadata", {}).get("name", ""),
+                          (o.get("status") or {}).get("phase", ""), flush=True)
+            except KeyboardInterrupt:
+                pass
+        return 0
     with urllib.request.urlopen(url, timeout=10) as r:
         data = json.loads(r.read())
     items = data.get("items", [data] if args.name else [])
@@ -258,6 +273,8 @@ def main(argv=None) -> int:
     get.add_argument("-l", "--selector", default=None)
     get.add_argument("-o", "--output", choices=["wide", "json", "yaml"],
                      default="wide")
+    get.add_argument("-w", "--watch", action="store_true",
+                     help="stream ADDED/MODIFIED/DELETED events (ndjson watch)")
     get.set_defaults(fn=cmd_get)
 
     ap_cmd = sub.add_parser("apply", help="apply a manifest file to the apiserver")

@@ -328,3 +328,31 @@ def test_http_merge_patch(served_cluster):
     cli = HttpStoreClient("http://127.0.0.1:18133")
     out2 = cli.merge_patch(c.KIND_PCLQ, "default", "mp", {"spec": {"replicas": 5}})
     assert out2["spec"]["replicas"] == 5
+
+
+def test_cli_get_watch(served_cluster, tmp_path):
+    """`grove-amd get -w` streams watch events from the apiserver."""
+    import subprocess
+    import sys
+    cluster, api = served_cluster
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "grove_amd", "get", "podcliques",
+         "--server", "http://127.0.0.1:18133", "-w"],
+        stdout=subprocess.PIPE, text=True)
+    try:
+        time.sleep(0.3)
+        cluster.store.create({"apiVersion": "grove.io/v1alpha1",
+                              "kind": c.KIND_PCLQ,
+                              "metadata": {"name": "wq", "namespace": "default"},
+                              "spec": {"roleName": "r", "replicas": 1,
+                                       "podSpec": {"containers": []}}})
+        deadline = time.time() + 10
+        line = ""
+        while time.time() < deadline:
+            line = proc.stdout.readline()
+            if "wq" in line:
+                break
+        assert "ADDED" in line and "wq" in line
+    finally:
+        proc.kill()
+        proc.wait()

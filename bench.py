@@ -352,6 +352,12 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
         for name in window:
             wait_step(name)
 
+    # ---- priming (setup, before the W official warmup steps): a cold box pays
+    # page-cache/HIP-module/thread-pool costs on the first gangs — measured 222 vs
+    # 330 gangs/s first-run-vs-steady on one box. A fixed handful of priming gangs
+    # reaches steady state regardless of the driver's chosen -W.
+    run_steps(4, "prime")
+
     # ---- warmup
     run_steps(args.warmup, "warm")
 

@@ -226,7 +226,9 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
             if gang.group_configs:
                 o["spec"]["topologyConstraintGroupConfigs"] = gang.group_configs
         if cur["spec"].get("podgroups") != groups_spec \
-                or cur["spec"].get("topologyConstraint") != gang.constraint:
+                or cur["spec"].get("topologyConstraint") != gang.constraint \
+                or (gang.group_configs and cur["spec"].get(
+                    "topologyConstraintGroupConfigs") != gang.group_configs):
             try:
                 cur = store.patch(c.KIND_PODGANG, ns, gang.name, upd)
             except ApiError:

@@ -277,3 +277,42 @@ class TestTopologyImmutability:
         with pytest.raises(ApiError):
             cluster.apply(_pcs("ut", 1, constraint={
                 "pack": {"required": "rack"}, "topologyName": "nope"}))
+
+
+def test_ctb_key_change_propagates_to_group_configs(cluster):
+    """A CTB nodeLabelKey update must flow into existing PodGangs'
+    topologyConstraintGroupConfigs (syncflow.go translation is re-run on drift)."""
+    cluster.store.create({
+        "apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+        "metadata": {"name": "topo-gc"},
+        "spec": {"levels": [{"domain": "rack", "nodeLabelKey": "topo/rack-v1"}]}})
+    pcs = {
+        "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+        "metadata": {"name": "gcfg"},
+        "spec": {"replicas": 1, "template": {
+            # topologyName only — no gang-level pack, so the ONLY drift is in the
+            # clique-level group config (isolates the group-config update path)
+            "topologyConstraint": {"topologyName": "topo-gc"},
+            "cliques": [{"name": "w",
+                         "topologyConstraint": {"pack": {"required": "rack"}},
+                         "spec": {"roleName": "w", "replicas": 1,
+                                  "podSpec": {"containers": [
+                                      {"name": "m", "image": "x"}]}}}]}}}
+    cluster.add_virtual_nodes(1, labels={"topo/rack-v1": "r1", "topo/rack-v2": "r1"})
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("gcfg", timeout=20)
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "gcfg-0")
+    cfgs = pg["spec"].get("topologyConstraintGroupConfigs") or []
+    assert any(cf["topologyConstraint"]["packConstraint"]["required"] == "topo/rack-v1"
+               for cf in cfgs)
+
+    def change_key(o):
+        o["spec"]["levels"][0]["nodeLabelKey"] = "topo/rack-v2"
+    cluster.store.patch(c.KIND_CTB, None, "topo-gc", change_key)
+
+    def updated():
+        pg2 = cluster.store.get(c.KIND_PODGANG, "default", "gcfg-0")
+        cfgs2 = pg2["spec"].get("topologyConstraintGroupConfigs") or []
+        return any(cf["topologyConstraint"]["packConstraint"]["required"]
+                   == "topo/rack-v2" for cf in cfgs2)
+    cluster.wait_for(updated, timeout=15, desc="group configs re-translated")

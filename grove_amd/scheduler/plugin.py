@@ -209,7 +209,7 @@ class GangScheduler:
             prev = self.store.try_get(c.KIND_PODGANG,
                                       ref.get("namespace") or ns, ref["name"])
             if prev is not None:
-                prev_nodes = self._gang_nodes(prev)
+                prev_nodes = self._gang_nodes(prev, pods_by_name)
                 pool = [n for n in nodes if n.name in prev_nodes]
                 if pool:
                     res = self._place_gang_pods(pool, pg, chosen)

@@ -316,3 +316,69 @@ def test_ctb_key_change_propagates_to_group_configs(cluster):
         return any(cf["topologyConstraint"]["packConstraint"]["required"]
                    == "topo/rack-v2" for cf in cfgs2)
     cluster.wait_for(updated, timeout=15, desc="group configs re-translated")
+
+
+def test_reuse_reservation_deterministic():
+    """Deterministic unit coverage of the reservation-reuse branch (plugin.py:207-233):
+    a new Initialized gang with reuseReservationRef lands on the referenced gang's
+    node even when another node is emptier."""
+    from grove_amd.kubecore.store import Store
+    from grove_amd.scheduler.plugin import GangScheduler
+    from grove_amd.utils import conditions as cond2
+
+    store = Store()
+    for name, gpus in (("na", 8), ("nb", 8)):
+        store.create({"apiVersion": "v1", "kind": "Node", "metadata": {"name": name},
+                      "status": {"allocatable": {"cpu": "64", "memory": "256Gi",
+                                                 "pods": "100",
+                                                 c.AMD_GPU_RESOURCE: str(gpus)}}})
+
+    def pod(name, gang, clique, node=None, sched_name=c.SCHEDULER_AMD_GANG):
+        p = {"apiVersion": "v1", "kind": "Pod",
+             "metadata": {"name": name, "namespace": "default",
+                          "labels": {c.LABEL_PODGANG: gang,
+                                     c.LABEL_PODCLIQUE: clique}},
+             "spec": {"schedulerName": sched_name, "containers": [
+                 {"name": "m", "image": "x", "resources": {"requests": {
+                     "cpu": "1", c.AMD_GPU_RESOURCE: "1"}}}]},
+             "status": {"phase": "Pending"}}
+        if node:
+            p["spec"]["nodeName"] = node
+        return p
+
+    # prev gang: scheduled, bound on node "nb" (the one best-fit would avoid after
+    # we put a resident pod on it)
+    store.create(pod("prev-w-0", "prev-0", "prev-0-w", node="nb"))
+    prev = {"apiVersion": c.SCHEDULER_API_VERSION, "kind": c.KIND_PODGANG,
+            "metadata": {"name": "prev-0", "namespace": "default"},
+            "spec": {"podgroups": [{"name": "prev-0-w", "minReplicas": 1,
+                                    "podReferences": [
+                                        {"namespace": "default", "name": "prev-w-0"}]}]}}
+    cond2.set_condition(prev, c.PODGANG_COND_SCHEDULED, True, "GangPlaced")
+    store.create(prev)
+
+    # new gang with the reservation hint
+    store.create(pod("succ-w-0", "succ-0", "succ-0-w"))
+    succ = {"apiVersion": c.SCHEDULER_API_VERSION, "kind": c.KIND_PODGANG,
+            "metadata": {"name": "succ-0", "namespace": "default"},
+            "spec": {"reuseReservationRef": {"name": "prev-0",
+                                             "namespace": "default"},
+                     "podgroups": [{"name": "succ-0-w", "minReplicas": 1,
+                                    "podReferences": [
+                                        {"namespace": "default", "name": "succ-w-0"}]}]}}
+    cond2.set_condition(succ, c.PODGANG_COND_INITIALIZED, True, "AllPodsAssociated")
+    store.create(succ)
+
+    GangScheduler(store, use_native=False).reconcile()
+    bound = store.get("Pod", "default", "succ-w-0")
+    assert bound["spec"].get("nodeName") == "nb"
+    pg = store.get(c.KIND_PODGANG, "default", "succ-0")
+    assert cond.condition_true(pg, c.PODGANG_COND_SCHEDULED)
+    assert get_cond_reason(pg) == "GangPlacedOnReservation"
+
+
+def get_cond_reason(pg):
+    for cd in (pg.get("status") or {}).get("conditions") or []:
+        if cd.get("type") == c.PODGANG_COND_SCHEDULED:
+            return cd.get("reason")
+    return None

@@ -109,6 +109,20 @@ def build_podclique(pcs: Obj, pcs_replica: int, clique_tmpl: Obj,
     }
 
 
+def match_by_fqn_suffix(fqn: str, named: List[Obj]) -> Optional[Obj]:
+    """Resolve which template a generated FQN ("<owner>-<replica>-<name>") came from.
+    Clique/scaling-group names may themselves contain dashes (DNS-1123), so pick the
+    LONGEST "-<name>" suffix match — "x-0-model-a" must resolve to clique "model-a",
+    not a sibling clique "a"."""
+    best = None
+    for item in named:
+        nm = item.get("name", "")
+        if nm and fqn.endswith("-" + nm) and (
+                best is None or len(nm) > len(best.get("name", ""))):
+            best = item
+    return best
+
+
 def build_pcsg(pcs: Obj, pcs_replica: int, sg_cfg: Obj) -> Obj:
     pcs_name = pcs["metadata"]["name"]
     fqn = namegen.pcsg_name(pcs_name, pcs_replica, sg_cfg["name"])
@@ -172,11 +186,12 @@ def build_pod(pcs: Obj, pclq: Obj, pod_index: int, scheduler_name: str,
     starts_after = pclq["spec"].get("startsAfter") or []
     if starts_after:
         def _parent_min_available(fqn: str) -> int:
-            for cl in (((pcs.get("spec") or {}).get("template") or {})
-                       .get("cliques") or []):
-                if fqn.endswith("-" + cl["name"]):
-                    sp = cl.get("spec") or {}
-                    return int(sp.get("minAvailable") or sp.get("replicas", 1))
+            cl = match_by_fqn_suffix(
+                fqn, ((pcs.get("spec") or {}).get("template") or {})
+                .get("cliques") or [])
+            if cl is not None:
+                sp = cl.get("spec") or {}
+                return int(sp.get("minAvailable") or sp.get("replicas", 1))
             return 1
         initc = {
             "name": "grove-initc",

@@ -158,13 +158,14 @@ class PodCliqueReconciler:
         if not sg_fqn:
             return None
         tmpl = pcs["spec"]["template"]
-        for sg in tmpl.get("podCliqueScalingGroups") or []:
-            if sg_fqn.endswith("-" + sg["name"]):
-                total = 0
-                for cl in tmpl.get("cliques") or []:
-                    if cl["name"] in (sg.get("cliqueNames") or []):
-                        total += int(cl.get("spec", {}).get("replicas", 1))
-                return total
+        sg = builders.match_by_fqn_suffix(
+            sg_fqn, tmpl.get("podCliqueScalingGroups") or [])
+        if sg is not None:
+            total = 0
+            for cl in tmpl.get("cliques") or []:
+                if cl["name"] in (sg.get("cliqueNames") or []):
+                    total += int(cl.get("spec", {}).get("replicas", 1))
+            return total
         return None
 
     # ------------------------------------------------------------------ gates

@@ -413,11 +413,12 @@ class PodCliqueSetReconciler:
                 copy_objects=False)
 
         def pclq_expected_hash(q: Obj) -> str:
-            cl_name = q["metadata"]["name"].rsplit("-", 1)[-1]
-            for cl in tmpl.get("cliques") or []:
-                if cl["name"] == cl_name:
-                    return pod_template_hash(cl_name, cl["spec"].get("podSpec", {}),
-                                             tmpl.get("priorityClassName", ""))
+            # longest-suffix match: clique names may contain dashes (DNS-1123)
+            cl = builders.match_by_fqn_suffix(q["metadata"]["name"],
+                                              tmpl.get("cliques") or [])
+            if cl is not None:
+                return pod_template_hash(cl["name"], cl["spec"].get("podSpec", {}),
+                                         tmpl.get("priorityClassName", ""))
             return ""
 
         def replica_updated(r: int) -> bool:

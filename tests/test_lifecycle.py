@@ -219,3 +219,34 @@ def test_podgang_unhealthy_condition(cluster):
         pg = cluster.store.get(c.KIND_PODGANG, "default", "uh-0")
         return not cond.condition_true(pg, c.PODGANG_COND_UNHEALTHY)
     cluster.wait_for(recovered, timeout=30, desc="Unhealthy cleared")
+
+
+def test_rolling_update_dashed_clique_names(cluster):
+    """Clique names may contain dashes (DNS-1123): rolling update must resolve the
+    template by longest FQN suffix, not by splitting on the last dash."""
+    pcs = {
+        "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+        "metadata": {"name": "dash"},
+        "spec": {"replicas": 1, "template": {"cliques": [
+            {"name": "a", "spec": {"roleName": "a", "replicas": 1,
+                                   "podSpec": {"containers": [
+                                       {"name": "m", "image": "i:1"}]}}},
+            {"name": "model-a", "spec": {"roleName": "w", "replicas": 1,
+                                         "podSpec": {"containers": [
+                                             {"name": "m", "image": "i:1"}]}}}]}}}
+    cluster.add_virtual_nodes(1)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("dash", timeout=20)
+
+    def bump(o):
+        for cl in o["spec"]["template"]["cliques"]:
+            cl["spec"]["podSpec"]["containers"][0]["image"] = "i:2"
+    cluster.store.patch(c.KIND_PCS, "default", "dash", bump)
+
+    def update_done():
+        o = cluster.store.get(c.KIND_PCS, "default", "dash")
+        prog = (o.get("status") or {}).get("updateProgress") or {}
+        return bool(prog.get("updateEndedAt"))
+    cluster.wait_for(update_done, timeout=40, desc="dashed-name rolling update done")
+    for p in cluster.store.list("Pod", "default", {c.LABEL_PART_OF: "dash"}):
+        assert p["spec"]["containers"][0]["image"] == "i:2"

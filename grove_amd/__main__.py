@@ -17,6 +17,8 @@ from . import __version__
 
 
 def cmd_operator(args) -> int:
+    import sys as _sys
+    _sys.setswitchinterval(0.0002)  # see bench.py: watch-chain handoff latency
     from .cluster import Cluster
     from .config import load_configuration
     from .kubecore.apiserver import ApiServer

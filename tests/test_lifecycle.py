@@ -94,6 +94,14 @@ class TestGangTermination:
         # and the never-scheduled suppression (by design) blocks the recycle.
         cluster.wait_pods_ready({c.LABEL_PART_OF: "gt4"}, 2, timeout=20)
         scaled_pclq = "gt4-0-sg-1-w"
+        # ... and wait for the PCLQ STATUS to observe that readiness: everScheduled
+        # latches at status-reconcile time (reference WasPCLQEverScheduled parity), so
+        # killing before the status write would legitimately suppress the recycle.
+        def scaled_status_latched():
+            q = cluster.store.get(c.KIND_PCLQ, "default", scaled_pclq)
+            return bool((q.get("status") or {}).get("everScheduled"))
+        cluster.wait_for(scaled_status_latched, timeout=20,
+                         desc="scaled PCLQ everScheduled latched")
         old_uid = cluster.store.get(c.KIND_PCLQ, "default", scaled_pclq)["metadata"]["uid"]
         for n in cluster.store.list("Node"):
             cluster.store.patch("Node", None, n["metadata"]["name"],

@@ -437,6 +437,80 @@ void mfma_gemm_v3_2wg(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
             }
 }
 
+// ---------------------------------------------------------------- V5: split-K for
+// small grids — a 2048³ GEMM has only (2048/256)² = 64 tiles (25% of 256 CUs);
+// SPLIT=4 K-partitions give 256 WGs. Each WG accumulates its K-quarter and
+// atomicAdds into C (fp32); C must be zeroed first.
+__global__ __launch_bounds__(512, 1)
+void mfma_gemm_v5_splitk(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+                         float* __restrict__ C, int M, int N, int K, int split) {
+    constexpr int BM = 256, BN = 256;
+    __shared__ __bf16 smem[2 * (BM * BK + BN * BK)];
+    auto sAp = [&](int b) { return smem + b * (BM * BK + BN * BK); };
+    auto sBp = [&](int b) { return smem + b * (BM * BK + BN * BK) + BM * BK; };
+    int tiles = (M / BM) * (N / BN);
+    int tile_id = blockIdx.x % tiles;
+    int kpart = blockIdx.x / tiles;
+    int tile_m, tile_n;
+    tile_map(tile_id, M / BM, N / BN, tile_m, tile_n);
+    long kspan = (long)K / split;
+    long k0 = kpart * kspan, k1 = k0 + kspan;
+    const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
+    const int wr = wave >> 1, wc = wave & 1;
+    const long row0 = (long)tile_m * BM, col0 = (long)tile_n * BN;
+    const int wave_piece0 = wave * WAVE;
+    auto stage = [&](int buf, long kk) {
+        stage_tile<512>(A + row0 * K + kk, K, sAp(buf), BM, tid, wave_piece0);
+        stage_tile<512>(Bt + col0 * K + kk, K, sBp(buf), BN, tid, wave_piece0);
+    };
+    f32x16 acc[2][4] = {};
+    bf16x8 afrag[2][2], bfrag[2][4];
+    const int a_row = wr * 64 + (lane & 31);
+    const int b_row = wc * 128 + (lane & 31);
+    const int k_half = lane >> 5;
+    auto load_frags = [&](int pb, int buf, int ks) {
+        int cb = (ks << 1) | k_half;
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt)
+            afrag[pb][mt] = *(const bf16x8*)(sAp(buf) + lds_off(a_row + mt * 32, cb));
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+            bfrag[pb][nt] = *(const bf16x8*)(sBp(buf) + lds_off(b_row + nt * 32, cb));
+    };
+    stage(0, k0);
+    __syncthreads();
+    load_frags(0, 0, 0);
+    for (long kk = k0; kk < k1; kk += BK) {
+        int buf = ((kk - k0) / BK) & 1;
+        if (kk + BK < k1) stage(buf ^ 1, kk + BK);
+#pragma unroll
+        for (int ks = 0; ks < BK / 16; ++ks) {
+            int pb = ks & 1;
+            if (ks + 1 < BK / 16)
+                load_frags(pb ^ 1, buf, ks + 1);
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int nt = 0; nt < 4; ++nt)
+                    acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        afrag[pb][mt], bfrag[pb][nt], acc[mt][nt], 0, 0, 0);
+        }
+        __syncthreads();
+        if (kk + BK < k1) load_frags(0, buf ^ 1, 0);
+    }
+    const int c_row_lane = 4 * (lane >> 5);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+            for (int reg = 0; reg < 16; ++reg) {
+                int r = wr * 64 + mt * 32 + (reg & 3) + 8 * (reg >> 2) + c_row_lane;
+                int cl = wc * 128 + nt * 32 + (lane & 31);
+                atomicAdd(&C[(row0 + r) * (long)N + col0 + cl], acc[mt][nt][reg]);
+            }
+}
+
 #define HIP_CHECK(x) do { hipError_t e = (x); if (e) { printf("hip err %s\n", hipGetErrorString(e)); exit(1);} } while(0)
 
 typedef void (*kfn)(const bf16*, const bf16*, float*, int, int, int);
@@ -503,7 +577,29 @@ int main(int argc, char** argv) {
                2.0 * M * N * K / (ms * 1e-3) / 1e12);
         check("v3");
     }
-    std::vector<float> c0b(4096);
+    {   // v5 split-K (small-grid shapes): zero C then atomic-accumulate
+        int tiles = (M/256)*(N/256);
+        int split = tiles >= 256 ? 1 : 256 / tiles;
+        if ((long)K % ((long)split * 64) == 0) {
+            dim3 grid(tiles * split), block(512);
+            auto runv5 = [&]() {
+                (void)hipMemsetAsync(C, 0, (size_t)M*N*4, 0);
+                hipLaunchKernelGGL(mfma_gemm_v5_splitk, grid, block, 0, 0,
+                                   A, Bt, C, M, N, K, split);
+            };
+            runv5(); HIP_CHECK(hipDeviceSynchronize());
+            hipEvent_t t0, t1; (void)hipEventCreate(&t0); (void)hipEventCreate(&t1);
+            (void)hipEventRecord(t0);
+            for (int i = 0; i < iters; ++i) runv5();
+            (void)hipEventRecord(t1);
+            HIP_CHECK(hipDeviceSynchronize());
+            float ms; (void)hipEventElapsedTime(&ms, t0, t1); ms /= iters;
+            printf("%-24s %8.3f ms  %8.1f TF/s (split=%d)\n", "v5_splitk", ms,
+                   2.0 * M * N * K / (ms * 1e-3) / 1e12, split);
+            check("v5");
+        }
+    }
     printf("baseline %.3f ms\n", base);
     return 0;
 }
+

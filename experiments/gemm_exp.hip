@@ -511,6 +511,88 @@ void mfma_gemm_v5_splitk(const bf16* __restrict__ A, const bf16* __restrict__ Bt
             }
 }
 
+
+// ---------------------------------------------------------------- V6: small-shape
+// deep-prefetch — 128x128 tile, BK=32, 4 LDS buffers (16 KB each, 64 KB total →
+// 2 WGs/CU), raw-barrier pipeline keeping 3 buffers of LDS-DMA in flight. Targets
+// latency-bound small GEMMs (2048³ payload shape: only 8 kk iterations at BK=256-tile).
+#define V6_NBUF 4
+#define V6_LOADS_PER_BUF 4          // per thread: 2 A pieces + 2 B pieces
+// s_waitcnt imm: vmcnt[3:0], expcnt[6:4]=7 (no wait), lgkmcnt[11:8]=15 (no wait)
+#define V6_WAITCNT(vm) __builtin_amdgcn_s_waitcnt(0xF00 | 0x70 | (vm))
+
+__global__ __launch_bounds__(256, 2)
+void mfma_gemm_v6_deep(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+                       float* __restrict__ C, int M, int N, int K) {
+    constexpr int BM = 128, BN = 128;
+    __shared__ __bf16 smem[V6_NBUF * (BM * BK32 + BN * BK32)];
+    auto sAp = [&](int b) { return smem + b * (BM * BK32 + BN * BK32); };
+    auto sBp = [&](int b) { return smem + b * (BM * BK32 + BN * BK32) + BM * BK32; };
+    int tile_m, tile_n;
+    tile_map(blockIdx.x, M / BM, N / BN, tile_m, tile_n);
+    const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
+    const int wr = wave >> 1, wc = wave & 1;
+    const long row0 = (long)tile_m * BM, col0 = (long)tile_n * BN;
+    const int wave_piece0 = wave * WAVE;
+    auto stage = [&](int buf, long kk) {
+        stage_tile32<256, BM>(A + row0 * K + kk, K, sAp(buf), tid, wave_piece0);
+        stage_tile32<256, BN>(Bt + col0 * K + kk, K, sBp(buf), tid, wave_piece0);
+    };
+    f32x16 acc[2][2] = {};
+    bf16x8 afrag[2], bfrag[2];
+    const int a_row = wr * 64 + (lane & 31);
+    const int b_row = wc * 64 + (lane & 31);
+    const int k_half = lane >> 5;
+    const int nkk = K / BK32;
+    const int prologue = nkk < V6_NBUF ? nkk : V6_NBUF;
+    for (int i = 0; i < prologue; ++i)
+        stage(i, (long)i * BK32);
+    for (int i = 0; i < nkk; ++i) {
+        int buf = i % V6_NBUF;
+        // in flight beyond buffer i: min(NBUF-1, nkk-1-i) newer buffers
+        int newer = nkk - 1 - i;
+        if (newer > V6_NBUF - 1) newer = V6_NBUF - 1;
+        switch (newer * V6_LOADS_PER_BUF) {   // imm must be a constant
+            case 0: V6_WAITCNT(0); break;
+            case 4: V6_WAITCNT(4); break;
+            case 8: V6_WAITCNT(8); break;
+            default: V6_WAITCNT(12); break;
+        }
+        __builtin_amdgcn_s_barrier();        // all threads' buffer-i loads landed
+#pragma unroll
+        for (int ks = 0; ks < BK32 / 16; ++ks) {
+            int cb = (ks << 1) | k_half;
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+                afrag[mt] = *(const bf16x8*)(sAp(buf) + lds_off32(a_row + mt * 32, cb));
+#pragma unroll
+            for (int nt = 0; nt < 2; ++nt)
+                bfrag[nt] = *(const bf16x8*)(sBp(buf) + lds_off32(b_row + nt * 32, cb));
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int nt = 0; nt < 2; ++nt)
+                    acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        afrag[mt], bfrag[nt], acc[mt][nt], 0, 0, 0);
+        }
+        if (i + V6_NBUF < nkk) {
+            __builtin_amdgcn_s_barrier();    // everyone done READING slot buf
+            stage(buf, (long)(i + V6_NBUF) * BK32);
+        }
+    }
+    const int c_row_lane = 4 * (lane >> 5);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+#pragma unroll
+            for (int reg = 0; reg < 16; ++reg) {
+                int r = wr * 64 + mt * 32 + (reg & 3) + 8 * (reg >> 2) + c_row_lane;
+                int cl = wc * 64 + nt * 32 + (lane & 31);
+                C[(row0 + r) * (long)N + col0 + cl] = acc[mt][nt][reg];
+            }
+}
+
 #define HIP_CHECK(x) do { hipError_t e = (x); if (e) { printf("hip err %s\n", hipGetErrorString(e)); exit(1);} } while(0)
 
 typedef void (*kfn)(const bf16*, const bf16*, float*, int, int, int);
@@ -598,6 +680,21 @@ int main(int argc, char** argv) {
                    2.0 * M * N * K / (ms * 1e-3) / 1e12, split);
             check("v5");
         }
+    }
+    {   // v6 deep-prefetch small-shape kernel: its own geometry (128x128)
+        dim3 grid((M/128)*(N/128)), block(256);
+        hipLaunchKernelGGL(mfma_gemm_v6_deep, grid, block, 0, 0, A, Bt, C, M, N, K);
+        HIP_CHECK(hipDeviceSynchronize());
+        hipEvent_t t0, t1; (void)hipEventCreate(&t0); (void)hipEventCreate(&t1);
+        (void)hipEventRecord(t0);
+        for (int i = 0; i < iters; ++i)
+            hipLaunchKernelGGL(mfma_gemm_v6_deep, grid, block, 0, 0, A, Bt, C, M, N, K);
+        (void)hipEventRecord(t1);
+        HIP_CHECK(hipDeviceSynchronize());
+        float ms; (void)hipEventElapsedTime(&ms, t0, t1); ms /= iters;
+        printf("%-24s %8.3f ms  %8.1f TF/s\n", "v6_deep_128_bk32", ms,
+               2.0 * M * N * K / (ms * 1e-3) / 1e12);
+        check("v6");
     }
     printf("baseline %.3f ms\n", base);
     return 0;

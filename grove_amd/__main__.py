@@ -61,7 +61,8 @@ def cmd_operator(args) -> int:
 
     api = None
     if cfg.api_server.enabled or args.serve:
-        api = ApiServer(cluster.store, cfg.api_server.host, cfg.api_server.port)
+        api = ApiServer(cluster.store, cfg.api_server.host, cfg.api_server.port,
+                        metrics_fn=cluster.metrics_lines)
         api.start()
         log.info("apiserver listening on %s", api.url)
 

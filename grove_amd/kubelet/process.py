@@ -117,36 +117,9 @@ class ProcessKubelet:
         ns_, name = key.split("/", 1)
         self._set_phase(ns_, name, "Running")
 
-    def _peek_ready(self, proc: subprocess.Popen) -> bool:
-        # non-blocking-ish readiness: dependencies satisfied line or quick exit
-        return True  # readiness == process alive; initc blocks inside before payload
-
-    def _is_ready(self, pod: Obj) -> bool:
-        for cd in (pod.get("status") or {}).get("conditions") or []:
-            if cd.get("type") == "Ready":
-                return cd.get("status") == "True"
-        return False
-
     def _set_phase(self, ns: str, name: str, phase: str) -> None:
         def upd(o: Obj) -> None:
             o.setdefault("status", {})["phase"] = phase
-        try:
-            self.store.patch("Pod", ns, name, upd, status=True)
-        except ApiError:
-            pass
-
-    def _set_ready(self, ns: str, name: str) -> None:
-        def upd(o: Obj) -> None:
-            st = o.setdefault("status", {})
-            conds = st.setdefault("conditions", [])
-            for want in ("ContainersReady", "Ready"):
-                for cd in conds:
-                    if cd.get("type") == want:
-                        cd["status"] = "True"
-                        break
-                else:
-                    conds.append({"type": want, "status": "True",
-                                  "reason": "ProcessRunning"})
         try:
             self.store.patch("Pod", ns, name, upd, status=True)
         except ApiError:

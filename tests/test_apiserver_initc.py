@@ -356,3 +356,20 @@ def test_cli_get_watch(served_cluster, tmp_path):
     finally:
         proc.kill()
         proc.wait()
+
+
+def test_metrics_include_controller_counters(served_cluster):
+    """/metrics serves store gauges plus the Cluster's controller counters when the
+    operator wires metrics_fn (manager.go metrics-endpoint parity)."""
+    from grove_amd.kubecore.apiserver import ApiServer
+    cluster, _ = served_cluster
+    api2 = ApiServer(cluster.store, port=18134, metrics_fn=cluster.metrics_lines)
+    api2.start()
+    try:
+        import urllib.request
+        with urllib.request.urlopen("http://127.0.0.1:18134/metrics", timeout=5) as r:
+            body = r.read().decode()
+        assert "grove_store_objects" in body
+        assert 'grove_reconcile_total{controller="podcliqueset"}' in body
+    finally:
+        api2.stop()

@@ -373,3 +373,23 @@ def test_metrics_include_controller_counters(served_cluster):
         assert 'grove_reconcile_total{controller="podcliqueset"}' in body
     finally:
         api2.stop()
+
+
+def test_initc_in_process_transport(cluster):
+    """initc without --server: the attach() in-process Store transport."""
+    from grove_amd import initc
+
+    initc.attach(cluster.store)
+    try:
+        cluster.store.create({
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": "ip-0", "namespace": "default",
+                         "labels": {c.LABEL_PODGANG: "ipg-0",
+                                    c.LABEL_PODCLIQUE: "ipg-0-a"}},
+            "spec": {"containers": []},
+            "status": {"phase": "Running",
+                       "conditions": [{"type": "Ready", "status": "True"}]}})
+        assert initc.wait_for_parents(
+            "default", "ipg-0", [("ipg-0-a", 1)], server=None, timeout=5)
+    finally:
+        initc.attach(None)

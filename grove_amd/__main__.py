@@ -61,8 +61,15 @@ def cmd_operator(args) -> int:
 
     api = None
     if cfg.api_server.enabled or args.serve:
+        import os as _os
+        tokens = dict(cfg.api_server.tokens)
+        agent_token = _os.environ.get("GROVE_AGENT_TOKEN")
+        if agent_token:
+            from .kubecore.identity import NODE_AGENT_USER
+            tokens[agent_token] = NODE_AGENT_USER
         api = ApiServer(cluster.store, cfg.api_server.host, cfg.api_server.port,
-                        metrics_fn=cluster.metrics_lines)
+                        metrics_fn=cluster.metrics_lines,
+                        auth_tokens=tokens or None)
         api.start()
         log.info("apiserver listening on %s", api.url)
 
@@ -103,7 +110,9 @@ def cmd_agent(args) -> int:
     logging.basicConfig(level=logging.INFO,
                         format="%(asctime)s %(levelname)s %(name)s %(message)s")
     log = logging.getLogger("grove.agent")
-    client = HttpStoreClient(args.server)
+    import os as _os
+    token = args.token or _os.environ.get("GROVE_AGENT_TOKEN")
+    client = HttpStoreClient(args.server, token=token)
     if args.virtual_gpus is not None:
         node = make_virtual_node(args.node_name or "agent-node",
                                  gpus=args.virtual_gpus)
@@ -294,6 +303,9 @@ def main(argv=None) -> int:
     agent.add_argument("--virtual-gpus", type=int, default=None,
                        help="register a virtual node instead of probing hardware")
     agent.add_argument("--poll-interval", type=float, default=0.2)
+    agent.add_argument("--token", default=None,
+                       help="bearer token for the apiserver "
+                            "(default: $GROVE_AGENT_TOKEN)")
     agent.set_defaults(fn=cmd_agent)
 
     crds = sub.add_parser("install-crds", help="render or apply the CRDs")

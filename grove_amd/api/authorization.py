@@ -12,7 +12,7 @@ from __future__ import annotations
 from typing import Iterable, Optional
 
 from . import constants as c
-from ..kubecore.identity import current_user, OPERATOR_USER
+from ..kubecore.identity import current_user, OPERATOR_USER, NODE_AGENT_USER
 from ..kubecore.store import Store, Obj, forbidden
 
 PROTECTED_KINDS = (c.KIND_PCLQ, c.KIND_PCSG, c.KIND_PODGANG, "Pod", "Service",
@@ -23,7 +23,9 @@ PROTECTED_KINDS = (c.KIND_PCLQ, c.KIND_PCSG, c.KIND_PODGANG, "Pod", "Service",
 class Authorizer:
     def __init__(self, store: Store, exempt_users: Optional[Iterable[str]] = None):
         self.store = store
-        self.exempt = {OPERATOR_USER, *(exempt_users or ())}
+        # Node agents (the kubelet analog) must write pod/node status; the
+        # reference exempts system components the same way (types.go:293-303).
+        self.exempt = {OPERATOR_USER, NODE_AGENT_USER, *(exempt_users or ())}
 
     def register(self) -> None:
         for kind in PROTECTED_KINDS:

@@ -23,6 +23,9 @@ class ServerConfig:
     host: str = "127.0.0.1"
     port: int = 8081
     enabled: bool = False
+    # bearer token -> user identity (the static-token authenticator analog);
+    # the GROVE_AGENT_TOKEN env var adds a node-agent token at launch.
+    tokens: Dict[str, str] = dataclasses.field(default_factory=dict)
 
 
 @dataclasses.dataclass
@@ -72,7 +75,9 @@ def load_configuration(path: Optional[str]) -> OperatorConfiguration:
     api = servers.get("api") or {}
     cfg.api_server = ServerConfig(api.get("host", "127.0.0.1"),
                                   int(api.get("port", 8081)),
-                                  bool(api.get("enabled", False)))
+                                  bool(api.get("enabled", False)),
+                                  {t["token"]: t["user"]
+                                   for t in (api.get("tokens") or [])})
     met = servers.get("metrics") or {}
     cfg.metrics_server = ServerConfig(met.get("host", "127.0.0.1"),
                                       int(met.get("port", 8082)),

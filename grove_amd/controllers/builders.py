@@ -22,37 +22,61 @@ def default_labels(pcs_name: str) -> Dict[str, str]:
     return {c.LABEL_MANAGED_BY: c.LABEL_MANAGED_BY_VALUE, c.LABEL_PART_OF: pcs_name}
 
 
-def _startup_dependencies(pcs: Obj, clique_name: str, owner_name: str, owner_replica: int,
-                          pcs_replica: int) -> List[str]:
-    """Resolve StartsAfter FQNs per startup type (podclique.go:341)."""
+def _startup_dependencies(pcs: Obj, clique_name: str, pcs_replica: int,
+                          pcsg_cfg_name: Optional[str] = None,
+                          pcsg_replica: Optional[int] = None) -> List[str]:
+    """Resolve StartsAfter FQNs per startup type (podclique.go:341-455,
+    componentutils.GenerateDependencyNamesForBasePodGang parity).
+
+    Base-gang cliques (standalone, or PCSG member replica j < minAvailable) depend on
+    ALL replicas in [0, minAvailable) of a dependency clique that belongs to a PCSG.
+    Scaled-gang cliques (PCSG member, j >= minAvailable) depend only on siblings inside
+    their OWN PCSG replica j; dependencies outside the PCSG are dropped (a scaled gang
+    must not couple to another gang's readiness).
+    """
     tmpl = pcs["spec"]["template"]
     startup = tmpl.get("cliqueStartupType", c.STARTUP_ANY_ORDER)
     cliques = tmpl.get("cliques") or []
     names = [cl["name"] for cl in cliques]
-    if startup == c.STARTUP_ANY_ORDER:
-        return []
+    deps: List[str] = []
     if startup == c.STARTUP_IN_ORDER:
         i = names.index(clique_name)
-        if i == 0:
-            return []
-        prev = names[i - 1]
-        return [_clique_fqn_for_dep(pcs, prev, pcs_replica)]
-    # Explicit
-    for cl in cliques:
-        if cl["name"] == clique_name:
-            return [_clique_fqn_for_dep(pcs, d, pcs_replica)
-                    for d in (cl.get("spec", {}).get("startsAfter") or [])]
-    return []
+        if i > 0:
+            deps = [names[i - 1]]
+    elif startup == c.STARTUP_EXPLICIT:
+        for cl in cliques:
+            if cl["name"] == clique_name:
+                deps = list(cl.get("spec", {}).get("startsAfter") or [])
+    if not deps:
+        return []
+    pcs_name = pcs["metadata"]["name"]
+    sgs = tmpl.get("podCliqueScalingGroups") or []
 
+    def sg_of(name: str) -> Optional[Obj]:
+        for sg in sgs:
+            if name in (sg.get("cliqueNames") or []):
+                return sg
+        return None
 
-def _clique_fqn_for_dep(pcs: Obj, clique_name: str, pcs_replica: int) -> str:
-    """FQN of the dependency clique: if it belongs to a PCSG, use PCSG replica 0's member."""
-    tmpl = pcs["spec"]["template"]
-    for sg in tmpl.get("podCliqueScalingGroups") or []:
-        if clique_name in (sg.get("cliqueNames") or []):
-            sg_fqn = namegen.pcsg_name(pcs["metadata"]["name"], pcs_replica, sg["name"])
-            return namegen.podclique_name(sg_fqn, 0, clique_name)
-    return namegen.podclique_name(pcs["metadata"]["name"], pcs_replica, clique_name)
+    my_sg = next((s for s in sgs if s["name"] == pcsg_cfg_name), None) \
+        if pcsg_cfg_name else None
+    scaled = (my_sg is not None and pcsg_replica is not None
+              and pcsg_replica >= int(my_sg.get("minAvailable", 1)))
+    out: List[str] = []
+    for d in deps:
+        d_sg = sg_of(d)
+        if scaled:
+            if d_sg is not None and d_sg["name"] == pcsg_cfg_name:
+                sg_fqn = namegen.pcsg_name(pcs_name, pcs_replica, pcsg_cfg_name)
+                out.append(namegen.podclique_name(sg_fqn, pcsg_replica, d))
+            # cross-gang dependency from a scaled gang: dropped
+        elif d_sg is not None:
+            sg_fqn = namegen.pcsg_name(pcs_name, pcs_replica, d_sg["name"])
+            out.extend(namegen.podclique_name(sg_fqn, j, d)
+                       for j in range(int(d_sg.get("minAvailable", 1))))
+        else:
+            out.append(namegen.podclique_name(pcs_name, pcs_replica, d))
+    return out
 
 
 def build_podclique(pcs: Obj, pcs_replica: int, clique_tmpl: Obj,
@@ -77,7 +101,9 @@ def build_podclique(pcs: Obj, pcs_replica: int, clique_tmpl: Obj,
     spec = copy.deepcopy(clique_tmpl["spec"])
     tmpl = pcs["spec"]["template"]
     hash_ = pod_template_hash(cl_name, spec.get("podSpec", {}),
-                              tmpl.get("priorityClassName", ""))
+                              tmpl.get("priorityClassName", ""),
+                              clique_tmpl.get("labels"),
+                              clique_tmpl.get("annotations"))
     labels = {
         **default_labels(pcs_name),
         **(clique_tmpl.get("labels") or {}),
@@ -92,8 +118,15 @@ def build_podclique(pcs: Obj, pcs_replica: int, clique_tmpl: Obj,
         labels[c.LABEL_PCSG_REPLICA_INDEX] = str(pcsg_replica)
     if base_podgang_name:
         labels[c.LABEL_BASE_PODGANG] = base_podgang_name
+    # Recover the scaling-group CONFIG name from the PCSG FQN
+    # ("<pcs>-<replica>-<sg>", namegen.go:85) for dependency scoping.
+    pcsg_cfg_name = None
+    if pcsg_name is not None:
+        prefix = f"{pcs_name}-{pcs_replica}-"
+        pcsg_cfg_name = pcsg_name[len(prefix):] if pcsg_name.startswith(prefix) \
+            else pcsg_name
     spec["startsAfter"] = _startup_dependencies(
-        pcs, cl_name, pcs_name, pcs_replica, pcs_replica)
+        pcs, cl_name, pcs_replica, pcsg_cfg_name, pcsg_replica)
     return {
         "apiVersion": c.API_VERSION,
         "kind": c.KIND_PCLQ,

@@ -153,7 +153,8 @@ class PCSGReconciler:
                     continue
                 new_hash = pod_template_hash(
                     mn, cl["spec"].get("podSpec", {}),
-                    pcs["spec"]["template"].get("priorityClassName", ""))
+                    pcs["spec"]["template"].get("priorityClassName", ""),
+                    cl.get("labels"), cl.get("annotations"))
                 if cur["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) != new_hash \
                         and self._replica_selected_for_update(pcs, pcs_replica):
                     def upd(o: Obj) -> None:

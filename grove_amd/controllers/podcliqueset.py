@@ -9,6 +9,7 @@ aggregation (reconcilestatus.go). Fresh MI355X-native implementation.
 """
 from __future__ import annotations
 
+import calendar
 import logging
 import time
 from typing import Any, Dict, List, Optional, Set, Tuple
@@ -29,10 +30,14 @@ log = logging.getLogger("grove.podcliqueset")
 
 
 def _iso_to_epoch(ts: str) -> float:
+    """UTC condition timestamp → epoch seconds. calendar.timegm is DST-proof
+    (time.mktime - time.timezone is off by an hour during DST). A malformed
+    timestamp must NOT reset the breach clock (that would postpone gang
+    termination forever), so parse failure returns 0.0 == 'breached long ago'."""
     try:
-        return time.mktime(time.strptime(ts, "%Y-%m-%dT%H:%M:%SZ")) - time.timezone
+        return float(calendar.timegm(time.strptime(ts, "%Y-%m-%dT%H:%M:%SZ")))
     except Exception:
-        return time.time()
+        return 0.0
 
 
 class PodCliqueSetReconciler:
@@ -242,7 +247,9 @@ class PodCliqueSetReconciler:
         # propagate template changes only to replicas selected for update (GREP-393)
         new_hash = pod_template_hash(clique_tmpl["name"],
                                      clique_tmpl["spec"].get("podSpec", {}),
-                                     pcs["spec"]["template"].get("priorityClassName", ""))
+                                     pcs["spec"]["template"].get("priorityClassName", ""),
+                                     clique_tmpl.get("labels"),
+                                     clique_tmpl.get("annotations"))
         if cur["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) != new_hash \
                 and self._replica_selected_for_update(pcs, r):
             def upd(o: Obj) -> None:
@@ -418,7 +425,8 @@ class PodCliqueSetReconciler:
                                               tmpl.get("cliques") or [])
             if cl is not None:
                 return pod_template_hash(cl["name"], cl["spec"].get("podSpec", {}),
-                                         tmpl.get("priorityClassName", ""))
+                                         tmpl.get("priorityClassName", ""),
+                                         cl.get("labels"), cl.get("annotations"))
             return ""
 
         def replica_updated(r: int) -> bool:

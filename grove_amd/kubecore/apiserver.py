@@ -22,6 +22,7 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
 from .store import Store, ApiError
+from .identity import as_user, ANONYMOUS_USER
 
 # plural -> kind for everything the stack serves
 PLURALS: Dict[str, str] = {
@@ -55,8 +56,33 @@ def parse_selector(sel: Optional[str]) -> Optional[Dict[str, str]]:
     return out
 
 
-def build_app(store: Store, metrics_fn=None):
+def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str]] = None):
+    """auth_tokens maps bearer token -> user identity. HTTP callers authenticate
+    per request: a bearer token resolves through the static map, then through
+    SA-token Secrets in the store (kubernetes.io/service-account-token type, the
+    <pcs>-ic-sat contract); anything else runs as system:anonymous, which the
+    Authorizer treats as a non-operator identity — unauthenticated peers can no
+    longer mutate grove-managed resources over the wire."""
     app = FastAPI(title="grove-amd apiserver")
+
+    def user_of(request: Request) -> str:
+        auth = request.headers.get("authorization", "")
+        if auth.lower().startswith("bearer "):
+            token = auth[7:].strip()
+            if auth_tokens and token in auth_tokens:
+                return auth_tokens[token]
+            for s in store.list("Secret", None):
+                if s.get("type") != "kubernetes.io/service-account-token":
+                    continue
+                tok = (s.get("stringData") or {}).get("token") or \
+                      (s.get("data") or {}).get("token")
+                if tok and tok == token:
+                    md = s.get("metadata", {})
+                    sa = (md.get("annotations") or {}).get(
+                        "kubernetes.io/service-account.name", "")
+                    ns = md.get("namespace", "default")
+                    return f"system:serviceaccount:{ns}:{sa}"
+        return ANONYMOUS_USER
 
     def err(e: ApiError):
         return JSONResponse(status_code=e.code, content={
@@ -109,7 +135,8 @@ def build_app(store: Store, metrics_fn=None):
             obj = await request.json()
             obj.setdefault("kind", kind_of(plural))
             obj.setdefault("metadata", {})["namespace"] = ns
-            return JSONResponse(store.create(obj), status_code=201)
+            with as_user(user_of(request)):
+                return JSONResponse(store.create(obj), status_code=201)
         except ApiError as e:
             return err(e)
 
@@ -131,7 +158,8 @@ def build_app(store: Store, metrics_fn=None):
             obj.setdefault("kind", kind_of(plural))
             obj.setdefault("metadata", {})["namespace"] = ns
             obj["metadata"]["name"] = name
-            return JSONResponse(store.update(obj))
+            with as_user(user_of(request)):
+                return JSONResponse(store.update(obj))
         except ApiError as e:
             return err(e)
 
@@ -144,16 +172,18 @@ def build_app(store: Store, metrics_fn=None):
             obj.setdefault("kind", kind_of(plural))
             obj.setdefault("metadata", {})["namespace"] = ns
             obj["metadata"]["name"] = name
-            return JSONResponse(store.update_status(obj))
+            with as_user(user_of(request)):
+                return JSONResponse(store.update_status(obj))
         except ApiError as e:
             return err(e)
 
     @app.delete("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
     @app.delete("/api/{version}/namespaces/{ns}/{plural}/{name}")
-    async def delete_ns(plural: str, ns: str, name: str,
+    async def delete_ns(request: Request, plural: str, ns: str, name: str,
                         group: str = "", version: str = "v1"):
         try:
-            store.delete(kind_of(plural), ns, name)
+            with as_user(user_of(request)):
+                store.delete(kind_of(plural), ns, name)
             return JSONResponse({"kind": "Status", "status": "Success"})
         except ApiError as e:
             return err(e)
@@ -179,7 +209,8 @@ def build_app(store: Store, metrics_fn=None):
 
             def apply(o):
                 _merge_patch(o, body)
-            return JSONResponse(store.patch(kind_of(plural), ns, name, apply))
+            with as_user(user_of(request)):
+                return JSONResponse(store.patch(kind_of(plural), ns, name, apply))
         except ApiError as e:
             return err(e)
 
@@ -192,7 +223,8 @@ def build_app(store: Store, metrics_fn=None):
 
             def apply(o):
                 _merge_patch(o, body)
-            return JSONResponse(store.patch(kind_of(plural), None, name, apply))
+            with as_user(user_of(request)):
+                return JSONResponse(store.patch(kind_of(plural), None, name, apply))
         except ApiError as e:
             return err(e)
 
@@ -213,7 +245,8 @@ def build_app(store: Store, metrics_fn=None):
         try:
             obj = await request.json()
             obj.setdefault("kind", kind_of(plural))
-            return JSONResponse(store.create(obj), status_code=201)
+            with as_user(user_of(request)):
+                return JSONResponse(store.create(obj), status_code=201)
         except ApiError as e:
             return err(e)
 
@@ -225,7 +258,8 @@ def build_app(store: Store, metrics_fn=None):
             obj = await request.json()
             obj.setdefault("kind", kind_of(plural))
             obj.setdefault("metadata", {})["name"] = name
-            return JSONResponse(store.update(obj))
+            with as_user(user_of(request)):
+                return JSONResponse(store.update(obj))
         except ApiError as e:
             return err(e)
 
@@ -237,16 +271,18 @@ def build_app(store: Store, metrics_fn=None):
             obj = await request.json()
             obj.setdefault("kind", kind_of(plural))
             obj.setdefault("metadata", {})["name"] = name
-            return JSONResponse(store.update_status(obj))
+            with as_user(user_of(request)):
+                return JSONResponse(store.update_status(obj))
         except ApiError as e:
             return err(e)
 
     @app.delete("/apis/{group}/{version}/{plural}/{name}")
     @app.delete("/api/{version}/{plural}/{name}")
-    async def delete_cluster(plural: str, name: str,
+    async def delete_cluster(request: Request, plural: str, name: str,
                              group: str = "", version: str = "v1"):
         try:
-            store.delete(kind_of(plural), None, name)
+            with as_user(user_of(request)):
+                store.delete(kind_of(plural), None, name)
             return JSONResponse({"kind": "Status", "status": "Success"})
         except ApiError as e:
             return err(e)
@@ -353,19 +389,21 @@ class ApiServer:
 
     def __init__(self, store: Store, host: str = "127.0.0.1", port: int = 8081,
                  metrics_fn=None, ssl_certfile: Optional[str] = None,
-                 ssl_keyfile: Optional[str] = None):
+                 ssl_keyfile: Optional[str] = None,
+                 auth_tokens: Optional[Dict[str, str]] = None):
         self.store = store
         self.host = host
         self.port = port
         self.metrics_fn = metrics_fn
         self.ssl_certfile = ssl_certfile
         self.ssl_keyfile = ssl_keyfile
+        self.auth_tokens = auth_tokens
         self._server = None
         self._thread: Optional[threading.Thread] = None
 
     def start(self) -> "ApiServer":
         import uvicorn
-        app = build_app(self.store, self.metrics_fn)
+        app = build_app(self.store, self.metrics_fn, self.auth_tokens)
         config = uvicorn.Config(app, host=self.host, port=self.port,
                                 log_level="warning", lifespan="off",
                                 ssl_certfile=self.ssl_certfile,
@@ -390,7 +428,8 @@ class ApiServer:
 
     @property
     def url(self) -> str:
-        return f"http://{self.host}:{self.port}"
+        scheme = "https" if self.ssl_certfile else "http"
+        return f"{scheme}://{self.host}:{self.port}"
 
     def stop(self) -> None:
         if self._server is not None:

@@ -38,10 +38,19 @@ def _group_of(kind: str) -> str:
 
 class HttpStoreClient:
     def __init__(self, base_url: str, timeout: float = 10.0,
-                 cafile: Optional[str] = None):
+                 cafile: Optional[str] = None, token: Optional[str] = None):
         self.base_url = base_url.rstrip("/")
         self.timeout = timeout
+        self.token = token
         self._ctx = ssl.create_default_context(cafile=cafile) if cafile else None
+
+    def _headers(self, content: bool = False) -> Dict[str, str]:
+        h: Dict[str, str] = {}
+        if content:
+            h["Content-Type"] = "application/json"
+        if self.token:
+            h["Authorization"] = f"Bearer {self.token}"
+        return h
 
     # ------------------------------------------------------------------ plumbing
     def _url(self, kind: str, namespace: Optional[str], name: Optional[str] = None,
@@ -63,7 +72,7 @@ class HttpStoreClient:
     def _request(self, method: str, url: str, body: Optional[Obj] = None) -> Obj:
         data = json.dumps(body).encode() if body is not None else None
         req = urllib.request.Request(url, data=data, method=method,
-                                     headers={"Content-Type": "application/json"})
+                                     headers=self._headers(content=True))
         try:
             with urllib.request.urlopen(req, timeout=self.timeout,
                                         context=self._ctx) as r:
@@ -146,7 +155,7 @@ class HttpStoreClient:
         """ndjson watch stream; yields (event_type, object)."""
         url = self._url(kind, namespace,
                         query=f"watch=true&seed={'true' if seed else 'false'}")
-        req = urllib.request.Request(url)
+        req = urllib.request.Request(url, headers=self._headers())
         with urllib.request.urlopen(req, timeout=3600, context=self._ctx) as r:
             for raw in r:
                 line = raw.decode().strip()

@@ -11,6 +11,8 @@ import threading
 from contextlib import contextmanager
 
 OPERATOR_USER = "system:serviceaccount:grove-system:grove-operator"
+ANONYMOUS_USER = "system:anonymous"
+NODE_AGENT_USER = "system:node:grove-agent"
 
 _local = threading.local()
 

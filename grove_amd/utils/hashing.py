@@ -18,12 +18,21 @@ def _digest(payload: Any) -> str:
 
 
 def pcs_generation_hash(pcs: Dict[str, Any]) -> str:
+    # Labels/annotations of the clique template are part of the digest (reference
+    # reconcilespec.go:118-132): a label-only template change must trigger a
+    # rolling update.
     tmpl = (pcs.get("spec") or {}).get("template") or {}
     payload: List[Any] = [tmpl.get("priorityClassName", "")]
     for cl in tmpl.get("cliques") or []:
-        payload.append({"name": cl.get("name"), "podSpec": (cl.get("spec") or {}).get("podSpec")})
+        payload.append({"name": cl.get("name"),
+                        "labels": cl.get("labels") or {},
+                        "annotations": cl.get("annotations") or {},
+                        "podSpec": (cl.get("spec") or {}).get("podSpec")})
     return _digest(payload)
 
 
-def pod_template_hash(clique_name: str, pod_spec: Dict[str, Any], priority_class: str = "") -> str:
-    return _digest({"name": clique_name, "podSpec": pod_spec, "priorityClassName": priority_class})
+def pod_template_hash(clique_name: str, pod_spec: Dict[str, Any], priority_class: str = "",
+                      labels: Dict[str, str] = None, annotations: Dict[str, str] = None) -> str:
+    return _digest({"name": clique_name, "podSpec": pod_spec,
+                    "priorityClassName": priority_class,
+                    "labels": labels or {}, "annotations": annotations or {}})

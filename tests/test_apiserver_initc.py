@@ -169,7 +169,7 @@ def test_tls_apiserver(cluster, tmp_path):
     api = ApiServer(cluster.store, port=18233, ssl_certfile=crt, ssl_keyfile=key)
     api.start()
     try:
-        assert api.url.startswith("http://")  # url property is scheme-naive; use https
+        assert api.url == "https://127.0.0.1:18233"  # TLS-aware url property
         ctx = ssl.create_default_context(cafile=crt)
         with urllib.request.urlopen("https://localhost:18233/healthz",
                                     context=ctx, timeout=5) as r:
@@ -393,3 +393,74 @@ def test_initc_in_process_transport(cluster):
             "default", "ipg-0", [("ipg-0-a", 1)], server=None, timeout=5)
     finally:
         initc.attach(None)
+
+
+def test_http_auth_identity(served_cluster, simple1_yaml):
+    """Wire auth: unauthenticated HTTP callers run as system:anonymous and the
+    Authorizer rejects their mutations of grove-managed resources; a bearer token
+    matching an SA-token Secret maps to that service account; a static token maps
+    to its configured user (the wire surface must not default to the
+    operator identity)."""
+    import urllib.error
+    import yaml as _yaml
+    cluster, api = served_cluster
+    cluster.add_virtual_nodes(2)
+    cluster.apply(simple1_yaml)
+    cluster.wait_pcs_available("simple1", timeout=20)
+    q = cluster.store.get(c.KIND_PCLQ, "default", "simple1-0-pca")
+    # anonymous PUT of a managed PodClique -> 403
+    q2 = json.loads(json.dumps(q))
+    q2["metadata"]["labels"]["tamper"] = "yes"
+    req = urllib.request.Request(
+        f"{api.url}/apis/grove.io/v1alpha1/namespaces/default/podcliques/"
+        f"simple1-0-pca", data=json.dumps(q2).encode(), method="PUT",
+        headers={"Content-Type": "application/json"})
+    try:
+        urllib.request.urlopen(req, timeout=5)
+        assert False, "anonymous mutation of a managed resource must be rejected"
+    except urllib.error.HTTPError as e:
+        assert e.code == 403
+    # anonymous DELETE of a managed pod -> 403
+    pod = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "simple1-0-pca"})[0]
+    req = urllib.request.Request(
+        f"{api.url}/api/v1/namespaces/default/pods/{pod['metadata']['name']}",
+        method="DELETE")
+    try:
+        urllib.request.urlopen(req, timeout=5)
+        assert False
+    except urllib.error.HTTPError as e:
+        assert e.code == 403
+    # SA-token secret auth: the <pcs>-ic-sat secret's token authenticates as the SA
+    from grove_amd.api import namegen
+    sec_name = namegen.initc_sa_token_secret_name("simple1")
+    cluster.store.patch("Secret", "default", sec_name,
+                        lambda o: o.setdefault("stringData", {}).update(
+                            {"token": "sa-tok-123"}))
+    req = urllib.request.Request(
+        f"{api.url}/apis/grove.io/v1alpha1/namespaces/default/podcliques/"
+        f"simple1-0-pca", data=json.dumps(q2).encode(), method="PUT",
+        headers={"Content-Type": "application/json",
+                 "Authorization": "Bearer sa-tok-123"})
+    try:
+        urllib.request.urlopen(req, timeout=5)
+        assert False, "the pod SA is not exempt either"
+    except urllib.error.HTTPError as e:
+        assert e.code in (403, 409)  # authenticated but not exempt (or RV conflict)
+    # static-token auth as the node agent (exempt) succeeds
+    from grove_amd.kubecore.apiserver import ApiServer
+    from grove_amd.kubecore.identity import NODE_AGENT_USER
+    api2 = ApiServer(cluster.store, port=18135,
+                     auth_tokens={"agent-tok": NODE_AGENT_USER})
+    api2.start()
+    try:
+        cur = cluster.store.get(c.KIND_PCLQ, "default", "simple1-0-pca")
+        cur["metadata"]["labels"]["tamper"] = "agent"
+        req = urllib.request.Request(
+            f"{api2.url}/apis/grove.io/v1alpha1/namespaces/default/podcliques/"
+            f"simple1-0-pca", data=json.dumps(cur).encode(), method="PUT",
+            headers={"Content-Type": "application/json",
+                     "Authorization": "Bearer agent-tok"})
+        with urllib.request.urlopen(req, timeout=5) as r:
+            assert r.status == 200
+    finally:
+        api2.stop()

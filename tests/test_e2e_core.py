@@ -325,3 +325,46 @@ def test_reconcile_trigger_annotation(cluster):
     cluster.store.patch(c.KIND_PCS, "default", "rt", bump)
     cluster.wait_for(lambda: cluster.c_pcs.reconcile_count > n0, timeout=10,
                      desc="annotation-triggered reconcile")
+
+
+def test_pcsg_startup_dependency_scoping(cluster):
+    """ADVICE r1 parity fix (componentutils.GenerateDependencyNamesForBasePodGang):
+    with Explicit startup and a scaling group, (a) a standalone clique depending on a
+    PCSG member clique waits on ALL [0, minAvailable) PCSG replicas; (b) a scaled PCSG
+    replica's member (j >= minAvailable) depends only on its OWN replica's siblings;
+    (c) cross-gang deps from a scaled replica are dropped."""
+    pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+           "metadata": {"name": "dep"},
+           "spec": {"replicas": 1, "template": {
+               "cliqueStartupType": c.STARTUP_EXPLICIT,
+               "cliques": [
+                   {"name": "ld", "spec": {"roleName": "ld", "replicas": 1,
+                       "podSpec": {"containers": [{"name": "m", "image": "i"}]}}},
+                   {"name": "wk", "spec": {"roleName": "wk", "replicas": 1,
+                       "startsAfter": ["ld"],
+                       "podSpec": {"containers": [{"name": "m", "image": "i"}]}}},
+                   {"name": "tail", "spec": {"roleName": "t", "replicas": 1,
+                       "startsAfter": ["wk"],
+                       "podSpec": {"containers": [{"name": "m", "image": "i"}]}}},
+               ],
+               "podCliqueScalingGroups": [
+                   {"name": "sg", "cliqueNames": ["ld", "wk"],
+                    "replicas": 3, "minAvailable": 2}]}}}
+    cluster.add_virtual_nodes(3)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("dep", timeout=30)
+    # (a) standalone "tail" depends on wk of ALL base replicas [0, minAvailable)
+    tail = cluster.store.get(c.KIND_PCLQ, "default", "dep-0-tail")
+    assert sorted(tail["spec"]["startsAfter"]) == \
+        ["dep-0-sg-0-wk", "dep-0-sg-1-wk"]
+    # base member wk@j<minAvailable also waits on ld of all base replicas
+    wk0 = cluster.store.get(c.KIND_PCLQ, "default", "dep-0-sg-0-wk")
+    assert sorted(wk0["spec"]["startsAfter"]) == \
+        ["dep-0-sg-0-ld", "dep-0-sg-1-ld"]
+    # (b) scaled replica j=2 depends only on its own gang's ld
+    wk2 = cluster.store.get(c.KIND_PCLQ, "default", "dep-0-sg-2-wk")
+    assert wk2["spec"]["startsAfter"] == ["dep-0-sg-2-ld"]
+    # (c) ld has no deps anywhere
+    for j in range(3):
+        ld = cluster.store.get(c.KIND_PCLQ, "default", f"dep-0-sg-{j}-ld")
+        assert ld["spec"]["startsAfter"] == []

@@ -181,9 +181,11 @@ def test_remote_agent_gpu_distributed():
               f"logLevel: warn\n")
     cfg.close()
     base = f"http://127.0.0.1:{port}"
+    env = {**os.environ, "GROVE_AGENT_TOKEN": "test-agent-token"}
     op = subprocess.Popen(
         [sys.executable, "-m", "grove_amd", "operator", "--config-file", cfg.name],
-        cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        env=env)
     agent = None
     try:
         for _ in range(150):
@@ -195,7 +197,8 @@ def test_remote_agent_gpu_distributed():
         agent = subprocess.Popen(
             [sys.executable, "-m", "grove_amd", "agent", "--server", base,
              "--node-name", "gpu-remote", "--poll-interval", "0.1"],
-            cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+            cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+            env=env)
         pcs = {"apiVersion": "grove.io/v1alpha1", "kind": "PodCliqueSet",
                "metadata": {"name": "rgpu"},
                "spec": {"replicas": 1, "template": {"cliques": [{

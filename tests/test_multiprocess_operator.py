@@ -82,9 +82,11 @@ def test_remote_node_agent(tmp_path):
     cfg.write_text(f"servers:\n  api: {{enabled: true, host: 127.0.0.1, port: {port}}}\n"
                    f"logLevel: warn\n")
     base = f"http://127.0.0.1:{port}"
+    env = {**os.environ, "GROVE_AGENT_TOKEN": "test-agent-token"}
     op = subprocess.Popen(
         [sys.executable, "-m", "grove_amd", "operator", "--config-file", str(cfg)],
-        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        env=env)
     agent = None
     try:
         for _ in range(100):
@@ -97,7 +99,8 @@ def test_remote_node_agent(tmp_path):
             [sys.executable, "-m", "grove_amd", "agent", "--server", base,
              "--node-name", "remote-0", "--virtual-gpus", "0",
              "--poll-interval", "0.1"],
-            cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+            cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+            env=env)
         pcs = {"apiVersion": "grove.io/v1alpha1", "kind": "PodCliqueSet",
                "metadata": {"name": "remote"},
                "spec": {"replicas": 1, "template": {"cliques": [{

@@ -110,6 +110,37 @@ class Tracker:
         for w in self._watches:
             w.stop()
 
+
+class RemoteTracker(Tracker):
+    """Tracker over the wire: pumps the apiserver's ndjson watch streams through an
+    HttpStoreClient instead of in-process store watches — the measurement instrument
+    of the deployable (multi-process) shape, like the reference's client-go-based
+    e2e/measurement/measurement.go:29-104 harness."""
+
+    def __init__(self, client):
+        super().__init__(store=None)  # type: ignore[arg-type]
+        self.client = client
+
+    def start(self) -> "RemoteTracker":
+        for kind, handler in (("Pod", self._on_pod), (c.KIND_PODGANG, self._on_gang)):
+            t = threading.Thread(target=self._pump_remote, args=(kind, handler),
+                                 daemon=True)
+            t.start()
+            self._threads.append(t)
+        return self
+
+    def _pump_remote(self, kind, handler):
+        try:
+            for ev, obj in self.client.watch_events(kind, None, seed=True):
+                if self._stop.is_set():
+                    return
+                try:
+                    handler(ev, obj)
+                except Exception:
+                    pass
+        except Exception:
+            pass  # stream torn down (server stop / tracker stop)
+
     # ---- reporting ----
     def wait_all_running(self, timeout: float = 60.0) -> bool:
         deadline = time.monotonic() + timeout

@@ -214,6 +214,37 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
         except ApiError as e:
             return err(e)
 
+    # merge-patch on the status subresource (client-go Status().Patch analog)
+    @app.patch("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}/status")
+    @app.patch("/api/{version}/namespaces/{ns}/{plural}/{name}/status")
+    async def patch_status_ns(request: Request, plural: str, ns: str, name: str,
+                              group: str = "", version: str = "v1"):
+        try:
+            body = await request.json()
+
+            def apply(o):
+                _merge_patch(o, body)
+            with as_user(user_of(request)):
+                return JSONResponse(
+                    store.patch(kind_of(plural), ns, name, apply, status=True))
+        except ApiError as e:
+            return err(e)
+
+    @app.patch("/apis/{group}/{version}/{plural}/{name}/status")
+    @app.patch("/api/{version}/{plural}/{name}/status")
+    async def patch_status_cluster(request: Request, plural: str, name: str,
+                                   group: str = "", version: str = "v1"):
+        try:
+            body = await request.json()
+
+            def apply(o):
+                _merge_patch(o, body)
+            with as_user(user_of(request)):
+                return JSONResponse(
+                    store.patch(kind_of(plural), None, name, apply, status=True))
+        except ApiError as e:
+            return err(e)
+
     @app.patch("/apis/{group}/{version}/{plural}/{name}")
     @app.patch("/api/{version}/{plural}/{name}")
     async def patch_cluster(request: Request, plural: str, name: str,

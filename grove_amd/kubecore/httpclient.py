@@ -86,10 +86,14 @@ class HttpStoreClient:
                            payload.get("message", str(e)))
 
     # ------------------------------------------------------------------ store API
-    def get(self, kind: str, namespace: Optional[str], name: str) -> Obj:
+    def get(self, kind: str, namespace: Optional[str], name: str,
+            copy: bool = True) -> Obj:
         return self._request("GET", self._url(kind, namespace, name))
 
-    def try_get(self, kind: str, namespace: Optional[str], name: str) -> Optional[Obj]:
+    def try_get(self, kind: str, namespace: Optional[str], name: str,
+                copy: bool = True) -> Optional[Obj]:
+        # `copy` is accepted for Store duck-type compatibility (wire objects are
+        # always fresh copies).
         try:
             return self.get(kind, namespace, name)
         except ApiError:
@@ -141,10 +145,12 @@ class HttpStoreClient:
         raise last or ApiError(409, "Conflict", name)
 
     def merge_patch(self, kind: str, namespace: Optional[str], name: str,
-                    patch: Obj) -> Obj:
+                    patch: Obj, status: bool = False) -> Obj:
         """Server-side RFC 7386 merge patch (single round-trip, no read-modify-write
-        conflict loop) — the client-go Patch(types.MergePatchType) analog."""
-        return self._request("PATCH", self._url(kind, namespace, name), patch)
+        conflict loop) — the client-go Patch(types.MergePatchType) analog.
+        status=True targets the status subresource."""
+        sub = "status" if status else ""
+        return self._request("PATCH", self._url(kind, namespace, name, sub), patch)
 
     def delete(self, kind: str, namespace: Optional[str], name: str,
                cascade: bool = True) -> None:

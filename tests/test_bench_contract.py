@@ -22,11 +22,11 @@ def _last_json_line(out: str):
     raise AssertionError(f"no JSON line in output:\n{out[-2000:]}")
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(240)
 def test_bench_single_rank():
     out = subprocess.run(
         [sys.executable, "bench.py", "--steps", "3", "--warmup", "1"],
-        cwd=REPO, capture_output=True, text=True, timeout=150)
+        cwd=REPO, capture_output=True, text=True, timeout=220)
     assert out.returncode == 0, out.stderr[-2000:]
     res = _last_json_line(out.stdout)
     for f in REQUIRED_FIELDS:
@@ -36,6 +36,26 @@ def test_bench_single_rank():
     assert res["value"] > 0 and res["higher_is_better"] is True
     assert res["scaling"] == "weak" and res["data"] == "synthetic"
     assert res["p50_time_to_running_ms"] > 0
+    # headline is the deployable (wire) shape, with the in-process number nested
+    assert res["config"]["transport"] == "http"
+    assert res["inproc"]["value"] > 0
+    assert res["calibration"]["spin_ms"] > 0
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("gang_size", [1, 2, 4, 8])
+def test_bench_gang_size_matrix(gang_size):
+    """Every gang size of the driver's scaling sweep produces a valid record
+    (be ready for the 8-GPU sweep on first contact)."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", str(gang_size), "--steps", "2",
+         "--warmup", "0", "--gangs-per-step", "2", "--transport", "inproc"],
+        cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    res = _last_json_line(out.stdout)
+    assert res["n_gpus"] == gang_size
+    assert res["config"]["gang_size"] == gang_size
+    assert res["value"] > 0 and res["p50_time_to_running_ms"] > 0
 
 
 @pytest.mark.timeout(300)

@@ -2,10 +2,11 @@
 
 Parity source: operator/cmd/install-crds + internal/crdinstaller/installer.go:18 (CRDs
 embedded in the binary, applied via server-side apply) and the generated CRD YAML under
-api/core/v1alpha1/crds/. Here the CRDs are rendered from the declared schema
-(api/schema.py) — structure-complete openAPIV3 schemas with printcolumns — and the
-installer applies them to a kube-style apiserver (ours or a real one) over HTTP, or
-writes them to disk for `kubectl apply`.
+api/core/v1alpha1/crds/. The CRDs are rendered from the full structural schemas in
+api/openapi.py (byte-compatible with the reference YAML modulo descriptions —
+tests/test_crd_parity.py holds the diff empty) with the reference's printer columns,
+scale/status subresources and short names; the installer applies them to a kube-style
+apiserver (ours or a real one) over HTTP, or writes them to disk for `kubectl apply`.
 """
 from __future__ import annotations
 
@@ -14,95 +15,146 @@ from typing import Any, Dict, List
 
 import yaml
 
-from . import constants as c
-from .schema import SCHEMAS, PASSTHROUGH
+from .openapi import schemas
 
 FIELD_MANAGER = "grove-crd-installer"
 
-_META = {
-    "podcliquesets.grove.io": dict(group="grove.io", kind="PodCliqueSet",
-                                   plural="podcliquesets", singular="podcliqueset",
-                                   shortNames=["pcs"], scope="Namespaced"),
-    "podcliques.grove.io": dict(group="grove.io", kind="PodClique",
-                                plural="podcliques", singular="podclique",
-                                shortNames=["pclq"], scope="Namespaced"),
-    "podcliquescalinggroups.grove.io": dict(group="grove.io",
-                                            kind="PodCliqueScalingGroup",
-                                            plural="podcliquescalinggroups",
-                                            singular="podcliquescalinggroup",
-                                            shortNames=["pcsg"], scope="Namespaced"),
-    "clustertopologybindings.grove.io": dict(group="grove.io",
-                                             kind="ClusterTopologyBinding",
-                                             plural="clustertopologybindings",
-                                             singular="clustertopologybinding",
-                                             shortNames=["ctb"], scope="Cluster"),
-    "podgangs.scheduler.grove.io": dict(group="scheduler.grove.io", kind="PodGang",
-                                        plural="podgangs", singular="podgang",
-                                        shortNames=[], scope="Namespaced"),
-}
-
-_PRINTCOLUMNS = {
-    "podcliquesets.grove.io": [
-        {"name": "Replicas", "type": "integer", "jsonPath": ".spec.replicas"},
-        {"name": "Available", "type": "integer",
-         "jsonPath": ".status.availableReplicas"},
-        {"name": "Age", "type": "date", "jsonPath": ".metadata.creationTimestamp"},
-    ],
-    "podcliques.grove.io": [
-        {"name": "Replicas", "type": "integer", "jsonPath": ".spec.replicas"},
-        {"name": "Ready", "type": "integer", "jsonPath": ".status.readyReplicas"},
-        {"name": "Age", "type": "date", "jsonPath": ".metadata.creationTimestamp"},
-    ],
-    "podcliquescalinggroups.grove.io": [
-        {"name": "Replicas", "type": "integer", "jsonPath": ".spec.replicas"},
-        {"name": "Available", "type": "integer",
-         "jsonPath": ".status.availableReplicas"},
-    ],
-    "clustertopologybindings.grove.io": [],
-    "podgangs.scheduler.grove.io": [
-        {"name": "Phase", "type": "string", "jsonPath": ".status.phase"},
-        {"name": "Score", "type": "number", "jsonPath": ".status.placementScore"},
-    ],
+# kind -> CRD identity (names contract: operator/api/core/v1alpha1/crds/*.yaml)
+CRD_META: Dict[str, Dict[str, Any]] = {
+    "PodCliqueSet": dict(
+        group="grove.io", plural="podcliquesets", singular="podcliqueset",
+        listKind="PodCliqueSetList", shortNames=["pcs"], scope="Namespaced"),
+    "PodClique": dict(
+        group="grove.io", plural="podcliques", singular="podclique",
+        listKind="PodCliqueList", shortNames=["pclq"], scope="Namespaced"),
+    "PodCliqueScalingGroup": dict(
+        group="grove.io", plural="podcliquescalinggroups",
+        singular="podcliquescalinggroup", listKind="PodCliqueScalingGroupList",
+        shortNames=["pcsg"], scope="Namespaced"),
+    "ClusterTopologyBinding": dict(
+        group="grove.io", plural="clustertopologybindings",
+        singular="clustertopologybinding", listKind="ClusterTopologyBindingList",
+        shortNames=["ct"], scope="Cluster"),
+    "PodGang": dict(
+        group="scheduler.grove.io", plural="podgangs", singular="podgang",
+        listKind="PodGangList", shortNames=["pg"], scope="Namespaced"),
 }
 
 
-def _to_openapi(node) -> Dict[str, Any]:
-    if node == PASSTHROUGH:
-        return {"type": "object", "x-kubernetes-preserve-unknown-fields": True}
-    if not node:
-        return {"x-kubernetes-preserve-unknown-fields": True}
-    props = {k: _to_openapi(v) for k, v in node.items()}
-    return {"type": "object", "properties": props}
+def _col(name: str, path: str, type_: str = "integer", priority: int = 0) -> Dict:
+    col: Dict[str, Any] = {"jsonPath": path, "name": name, "type": type_}
+    if priority:
+        col["priority"] = priority
+    return col
 
 
-def render_crd(crd_name: str) -> Dict[str, Any]:
-    meta = _META[crd_name]
-    tree = dict(SCHEMAS[crd_name])
-    schema = _to_openapi(tree)
+_AGE = _col("Age", ".metadata.creationTimestamp", "date")
+
+PRINTER_COLUMNS: Dict[str, List[Dict[str, Any]]] = {
+    "PodCliqueSet": [
+        _col("Replicas", ".spec.replicas"),
+        _col("Available", ".status.availableReplicas"),
+        _col("Updated", ".status.updatedReplicas"),
+        _col("PCLQs-Updated", ".status.updateProgress.updatedPodCliquesCount"),
+        _col("PCLQs-Total", ".status.updateProgress.totalPodCliquesCount"),
+        _col("PCSGs-Updated",
+             ".status.updateProgress.updatedPodCliqueScalingGroupsCount"),
+        _col("PCSGs-Total",
+             ".status.updateProgress.totalPodCliqueScalingGroupsCount"),
+        _AGE,
+    ],
+    "PodClique": [
+        _col("MinAvail", ".spec.minAvailable"),
+        _col("Replicas", ".spec.replicas"),
+        _col("Ready", ".status.readyReplicas"),
+        _col("Scheduled", ".status.scheduledReplicas"),
+        _col("Updated", ".status.updatedReplicas"),
+        _col("Gated", ".status.scheduleGatedReplicas", priority=1),
+        _col("MinBreached",
+             '.status.conditions[?(@.type=="MinAvailableBreached")].status',
+             "string", priority=1),
+        _AGE,
+    ],
+    "PodCliqueScalingGroup": [
+        _col("MinAvail", ".spec.minAvailable"),
+        _col("Replicas", ".spec.replicas"),
+        _col("Available", ".status.availableReplicas"),
+        _col("Scheduled", ".status.scheduledReplicas"),
+        _col("Updated", ".status.updatedReplicas"),
+        _col("PCLQs-Updated", ".status.updateProgress.updatedPodCliquesCount"),
+        _col("PCLQs-Total", ".status.updateProgress.totalPodCliquesCount"),
+        _col("MinBreached",
+             '.status.conditions[?(@.type=="MinAvailableBreached")].status',
+             "string", priority=1),
+        _AGE,
+    ],
+    "ClusterTopologyBinding": [
+        _col("Domains", ".spec.levels[*].domain", "string"),
+        _AGE,
+    ],
+    "PodGang": [
+        _col("Phase", ".status.phase", "string"),
+        _AGE,
+    ],
+}
+
+# HPA /scale subresource wiring (reference subresources blocks)
+SCALE_SUBRESOURCE: Dict[str, Dict[str, str]] = {
+    "PodCliqueSet": {"labelSelectorPath": ".status.hpaPodSelector",
+                     "specReplicasPath": ".spec.replicas",
+                     "statusReplicasPath": ".status.replicas"},
+    "PodClique": {"labelSelectorPath": ".status.hpaPodSelector",
+                  "specReplicasPath": ".spec.replicas",
+                  "statusReplicasPath": ".status.replicas"},
+    "PodCliqueScalingGroup": {"labelSelectorPath": ".status.selector",
+                              "specReplicasPath": ".spec.replicas",
+                              "statusReplicasPath": ".status.replicas"},
+}
+
+
+def render_crd_for_kind(kind: str) -> Dict[str, Any]:
+    meta = CRD_META[kind]
+    subresources: Dict[str, Any] = {}
+    if kind in SCALE_SUBRESOURCE:
+        subresources["scale"] = dict(SCALE_SUBRESOURCE[kind])
+    subresources["status"] = {}
     return {
         "apiVersion": "apiextensions.k8s.io/v1",
         "kind": "CustomResourceDefinition",
-        "metadata": {"name": crd_name},
+        "metadata": {"name": f"{meta['plural']}.{meta['group']}"},
         "spec": {
             "group": meta["group"],
-            "names": {"kind": meta["kind"], "plural": meta["plural"],
-                      "singular": meta["singular"],
-                      "shortNames": meta["shortNames"]},
+            "names": {"kind": kind, "listKind": meta["listKind"],
+                      "plural": meta["plural"], "shortNames": meta["shortNames"],
+                      "singular": meta["singular"]},
             "scope": meta["scope"],
             "versions": [{
+                "additionalPrinterColumns": PRINTER_COLUMNS[kind],
                 "name": "v1alpha1",
+                "schema": {"openAPIV3Schema": schemas()[kind]},
                 "served": True,
                 "storage": True,
-                "schema": {"openAPIV3Schema": schema},
-                "subresources": {"status": {}},
-                "additionalPrinterColumns": _PRINTCOLUMNS[crd_name],
+                "subresources": subresources,
             }],
         },
     }
 
 
+def render_crd(crd_name: str) -> Dict[str, Any]:
+    for kind, meta in CRD_META.items():
+        if f"{meta['plural']}.{meta['group']}" == crd_name:
+            return render_crd_for_kind(kind)
+    raise KeyError(crd_name)
+
+
 def render_all() -> List[Dict[str, Any]]:
-    return [render_crd(n) for n in sorted(SCHEMAS)]
+    names = sorted(f"{m['plural']}.{m['group']}" for m in CRD_META.values())
+    return [render_crd(n) for n in names]
+
+
+class _NoAliasDumper(yaml.SafeDumper):
+    def ignore_aliases(self, data):  # shared schema blocks must inline, not anchor
+        return True
 
 
 def write_crds(directory: str) -> List[str]:
@@ -112,7 +164,9 @@ def write_crds(directory: str) -> List[str]:
     for crd in render_all():
         path = f"{directory}/{crd['metadata']['name']}.yaml"
         with open(path, "w") as f:
-            yaml.safe_dump(crd, f, sort_keys=False)
+            f.write("---\n")
+            yaml.dump(crd, f, Dumper=_NoAliasDumper, sort_keys=True,
+                      default_flow_style=False)
         paths.append(path)
     return paths
 

@@ -239,9 +239,9 @@ def validate_clustertopologybinding(ctb: Obj, old: Optional[Obj] = None) -> None
         raise _err("spec.levels", "at least one topology level is required")
     seen_d, seen_k = set(), set()
     for lv in levels:
-        d, k = lv.get("domain"), lv.get("nodeLabelKey") or lv.get("key")
+        d, k = lv.get("domain"), lv.get("key")
         if not d or not k:
-            raise _err("spec.levels", "each level requires domain and nodeLabelKey")
+            raise _err("spec.levels", "each level requires domain and key")
         if d in seen_d:
             raise _err("spec.levels", f"duplicate domain {d!r}")
         if k in seen_k:

@@ -47,7 +47,11 @@ class Cluster:
         self.store = Store()
         self.scheduler_name = scheduler_name
 
-        # admission (webhook parity: defaulting before validation)
+        # structural-schema admission first (the apiserver's CRD schema layer:
+        # defaults + type/enum/bounds checks, api/schemavalidate.py), then the
+        # webhook-parity semantic admission (defaulting before validation)
+        from .api.schemavalidate import StructuralSchemaAdmission
+        StructuralSchemaAdmission().register(self.store)
         self.store.register_mutator(c.KIND_PCS, default_podcliqueset)
         self.store.register_validator(c.KIND_PCS, validate_podcliqueset)
         from .api.validation import validate_xgmi_groups

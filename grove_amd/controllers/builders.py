@@ -180,7 +180,6 @@ def build_pcsg(pcs: Obj, pcs_replica: int, sg_cfg: Obj) -> Obj:
             "replicas": sg_cfg.get("replicas", 1),
             "minAvailable": sg_cfg.get("minAvailable", 1),
             "cliqueNames": list(sg_cfg.get("cliqueNames") or []),
-            "cliqueTemplateNames": list(sg_cfg.get("cliqueNames") or []),
         },
     }
 

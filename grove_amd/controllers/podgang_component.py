@@ -43,7 +43,7 @@ def domain_key_map(store: Store, topology_name: Optional[str] = None
         if topology_name and ctb["metadata"]["name"] != topology_name:
             continue
         for lv in (ctb.get("spec") or {}).get("levels") or []:
-            key = lv.get("nodeLabelKey") or lv.get("key")
+            key = lv.get("key")
             if lv.get("domain") and key:
                 m[lv["domain"]] = key
         break  # one CTB is the source of truth

@@ -47,10 +47,10 @@ def resolve_templates(pcs: Obj) -> Dict[str, Obj]:
     out: Dict[str, Obj] = {}
     for t in pcs["spec"]["template"].get("resourceClaimTemplates") or []:
         if t.get("name"):
-            # reference field is templateSpec (ResourceClaimTemplateSpec); accept the
-            # bare spec too
-            ts = t.get("templateSpec") or {}
-            out[t["name"]] = ts.get("spec") or t.get("spec") or {}
+            # templateSpec is a resourcev1.ResourceClaimTemplateSpec
+            # (podcliqueset.go:417-422); the created ResourceClaim's spec is its
+            # .spec subtree
+            out[t["name"]] = (t.get("templateSpec") or {}).get("spec") or {}
     return out
 
 
@@ -67,7 +67,7 @@ def pcs_claims_for_replica(store: Store, pcs: Obj, r: int,
     templates = resolve_templates(pcs)
     out: List[Tuple[Obj, Obj]] = []
     for entry in pcs["spec"]["template"].get("resourceSharing") or []:
-        tname = entry.get("templateRef") or entry.get("name")
+        tname = entry.get("name")
         spec = templates.get(tname)
         if spec is None:
             continue
@@ -82,7 +82,7 @@ def pcs_claims_for_replica(store: Store, pcs: Obj, r: int,
     for clique, g in groups.items():
         by_group.setdefault(g, []).append(clique)
     for g, members in sorted(by_group.items()):
-        entry = {"templateRef": f"{XGMI_TEMPLATE_NAME}-{g}", "scope": "PerReplica",
+        entry = {"name": f"{XGMI_TEMPLATE_NAME}-{g}", "scope": "PerReplica",
                  "filter": {"childCliqueNames": sorted(members)}}
         claim = build_resource_claim(
             pcs, f"{pcs_name}-{r}-xgmi-{g}",
@@ -155,7 +155,7 @@ def claim_refs_for_clique(claims: List[Tuple[Obj, Obj]],
     refs: List[ClaimRef] = []
     for claim, entry in claims:
         if _filter_allows(entry, clique_name):
-            tname = entry.get("templateRef") or entry.get("name")
+            tname = entry.get("name")
             refs.append({"name": tname,
                          "resourceClaimName": claim["metadata"]["name"]})
     return refs
@@ -169,7 +169,7 @@ def clique_level_claims(store: Store, pcs: Obj, r: int,
     templates = resolve_templates(pcs)
     out: List[Tuple[Obj, Obj]] = []
     for entry in clique_tmpl.get("resourceSharing") or []:
-        tname = entry.get("templateRef") or entry.get("name")
+        tname = entry.get("name")
         spec = templates.get(tname)
         if spec is None:
             continue
@@ -187,7 +187,7 @@ def pcsg_claims(pcs: Obj, sg_cfg: Obj, sg_fqn: str, replicas: int
     templates = resolve_templates(pcs)
     out: List[Tuple[Obj, Obj, Optional[int]]] = []
     for entry in sg_cfg.get("resourceSharing") or []:
-        tname = entry.get("templateRef") or entry.get("name")
+        tname = entry.get("name")
         spec = templates.get(tname)
         if spec is None:
             continue

@@ -93,7 +93,7 @@ class AmdGangBackend(Backend):
 
     def _levels(self, ctb: Obj) -> List[Dict[str, str]]:
         levels = [{"domain": lv.get("domain"),
-                   "nodeLabelKey": lv.get("nodeLabelKey") or lv.get("key")}
+                   "nodeLabelKey": lv.get("key")}
                   for lv in (ctb.get("spec") or {}).get("levels") or []]
         # narrowest native level: the xGMI hive (below host/numa)
         if not any(lv["domain"] == "xgmi-hive" for lv in levels):

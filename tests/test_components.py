@@ -132,10 +132,11 @@ class TestResourceSharing:
                "spec": {"replicas": 2, "template": {
                    "resourceClaimTemplates": [{
                        "name": "shared-mem",
-                       "spec": {"devices": {"requests": [{
-                           "name": "m", "deviceClassName": "mem.example.com"}]}}}],
+                       "templateSpec": {"spec": {"devices": {"requests": [{
+                           "name": "m",
+                           "deviceClassName": "mem.example.com"}]}}}}],
                    "resourceSharing": [
-                       {"templateRef": "shared-mem", "scope": "PerReplica"}],
+                       {"name": "shared-mem", "scope": "PerReplica"}],
                    "cliques": [{"name": "w", "spec": {
                        "roleName": "w", "replicas": 1,
                        "podSpec": {"containers": [{

@@ -14,9 +14,9 @@ CTB = {
     "apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
     "metadata": {"name": "cluster-topology"},
     "spec": {"levels": [
-        {"domain": "zone", "nodeLabelKey": "topology.kubernetes.io/zone"},
-        {"domain": "rack", "nodeLabelKey": "topology.kubernetes.io/rack"},
-        {"domain": "host", "nodeLabelKey": "kubernetes.io/hostname"},
+        {"domain": "zone", "key": "topology.kubernetes.io/zone"},
+        {"domain": "rack", "key": "topology.kubernetes.io/rack"},
+        {"domain": "host", "key": "kubernetes.io/hostname"},
     ]},
 }
 
@@ -73,8 +73,8 @@ def test_ctb_validation_rejects_duplicates(cluster):
     bad = {"apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
            "metadata": {"name": "bad"},
            "spec": {"levels": [
-               {"domain": "rack", "nodeLabelKey": "a"},
-               {"domain": "rack", "nodeLabelKey": "b"}]}}
+               {"domain": "rack", "key": "a"},
+               {"domain": "rack", "key": "b"}]}}
     from grove_amd.kubecore.store import ApiError
     with pytest.raises(ApiError):
         cluster.store.create(bad)
@@ -285,7 +285,7 @@ def test_ctb_key_change_propagates_to_group_configs(cluster):
     cluster.store.create({
         "apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
         "metadata": {"name": "topo-gc"},
-        "spec": {"levels": [{"domain": "rack", "nodeLabelKey": "topo/rack-v1"}]}})
+        "spec": {"levels": [{"domain": "rack", "key": "topo/rack-v1"}]}})
     pcs = {
         "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
         "metadata": {"name": "gcfg"},
@@ -307,7 +307,7 @@ def test_ctb_key_change_propagates_to_group_configs(cluster):
                for cf in cfgs)
 
     def change_key(o):
-        o["spec"]["levels"][0]["nodeLabelKey"] = "topo/rack-v2"
+        o["spec"]["levels"][0]["key"] = "topo/rack-v2"
     cluster.store.patch(c.KIND_CTB, None, "topo-gc", change_key)
 
     def updated():

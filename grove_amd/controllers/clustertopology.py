@@ -11,6 +11,7 @@ from typing import Optional
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
+from ..utils.errors import report_api_error
 from ..scheduler.backends import Registry
 from ..utils import conditions as cond
 from .manager import Result
@@ -54,8 +55,9 @@ class ClusterTopologyReconciler:
                                    c.REASON_DRIFT, drift)
         try:
             self.store.patch(c.KIND_CTB, None, name, upd, status=True)
-        except ApiError:
-            pass
+        except ApiError as e:
+            report_api_error(self.store, c.KIND_CTB, None, name,
+                             "write CTB status", e)
         if drift is not None:
             self.store.record_event(ctb, "Warning", c.REASON_DRIFT, drift)
         return Result.DONE

@@ -19,6 +19,7 @@ from typing import Any, Dict, List, Optional
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
+from ..utils.errors import report_api_error
 from ..utils.quantity import cpu_millis
 from .manager import Result
 
@@ -68,8 +69,9 @@ class HPAReconciler:
                 self.store.record_event(
                     hpa, "Normal", "SuccessfulRescale",
                     f"scaled {ref['kind']}/{ref['name']} {current} -> {desired}")
-            except ApiError:
-                pass
+            except ApiError as e:
+                report_api_error(self.store, ref["kind"], namespace, ref["name"],
+                                 "HPA rescale", e)
 
         def upd(o: Obj) -> None:
             st = o.setdefault("status", {})
@@ -78,8 +80,9 @@ class HPAReconciler:
         try:
             self.store.patch("HorizontalPodAutoscaler", namespace, name, upd,
                              status=True)
-        except ApiError:
-            pass
+        except ApiError as e:
+            report_api_error(self.store, "HorizontalPodAutoscaler", namespace,
+                             name, "write HPA status", e)
         return Result(requeue_after=self.sync_period_s)
 
     def _desired_replicas(self, ns: str, hpa: Obj, ref: Dict[str, Any],

@@ -13,6 +13,7 @@ import logging
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
+from ..utils.errors import report_api_error
 from ..utils import conditions as cond
 from .manager import Result
 
@@ -40,8 +41,10 @@ class NodeLifecycleReconciler:
                                   p["metadata"]["name"])
                 log.info("evicted pod %s from lost node %s",
                          p["metadata"]["name"], name)
-            except ApiError:
-                pass
+            except ApiError as e:
+                report_api_error(self.store, "Pod",
+                                 p["metadata"].get("namespace"),
+                                 p["metadata"]["name"], "evict from lost node", e)
         return Result.DONE
 
     @staticmethod

@@ -15,6 +15,7 @@ from typing import List, Optional
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
+from ..utils import errors as groveerr
 from ..utils.indexing import available_indices
 from ..utils.concurrent import run_concurrently_with_slow_start
 from . import builders
@@ -35,8 +36,10 @@ class PodCliqueReconciler:
             return Result.DONE
         if pclq["metadata"].get("deletionTimestamp"):
             return self._reconcile_delete(pclq)
-        res = self._reconcile_spec(pclq)
+        rec = groveerr.StepRecorder(self.store, c.KIND_PCLQ, namespace, name)
+        res = self._reconcile_spec(pclq, rec)
         self._reconcile_status(namespace, name)
+        rec.flush()
         return res
 
     # ------------------------------------------------------------------ delete
@@ -62,7 +65,8 @@ class PodCliqueReconciler:
             "Pod", pclq["metadata"].get("namespace"),
             {c.LABEL_PODCLIQUE: pclq["metadata"]["name"]}, copy_objects=False)
 
-    def _reconcile_spec(self, pclq: Obj) -> Result:
+    def _reconcile_spec(self, pclq: Obj,
+                        rec: groveerr.StepRecorder) -> Result:
         ns = pclq["metadata"].get("namespace")
         desired = int(pclq["spec"].get("replicas", 1))
         pods = self._owned_pods(pclq)
@@ -85,10 +89,10 @@ class PodCliqueReconciler:
             ready = [p for p in outdated if cond.pod_is_ready(p)]
             victims = not_ready + ready[:1]
             for p in victims:
-                try:
+                with rec.step(groveerr.ERR_ROLLING_UPDATE,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"replace pod {p['metadata']['name']}"):
                     self.store.delete("Pod", ns, p["metadata"]["name"])
-                except ApiError:
-                    pass
             pods = current
         n = len(pods)
 
@@ -117,12 +121,12 @@ class PodCliqueReconciler:
                     raise errs[0]
         elif n > desired:
             for p in self._deletion_order(pods)[: n - desired]:
-                try:
+                with rec.step(groveerr.ERR_SYNC_PODS,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"scale-in pod {p['metadata']['name']}"):
                     self.store.delete("Pod", ns, p["metadata"]["name"])
-                except ApiError:
-                    pass
 
-        self._remove_scheduling_gates(pclq)
+        self._remove_scheduling_gates(pclq, rec)
         return Result.DONE
 
     @staticmethod
@@ -169,7 +173,7 @@ class PodCliqueReconciler:
         return None
 
     # ------------------------------------------------------------------ gates
-    def _remove_scheduling_gates(self, pclq: Obj) -> None:
+    def _remove_scheduling_gates(self, pclq: Obj, rec: groveerr.StepRecorder) -> None:
         """Hierarchical gang admission (syncflow.go:271-424): ungate a pod only when
         (a) its name is in its PodGang's podReferences and (b) it has no base-podgang
         label (base gang → immediate) OR the base PodGang is fully scheduled."""
@@ -204,10 +208,10 @@ class PodCliqueReconciler:
                 o["spec"]["schedulingGates"] = [
                     g for g in o["spec"].get("schedulingGates", [])
                     if g.get("name") != c.POD_GANG_SCHEDULING_GATE]
-            try:
+            with rec.step(groveerr.ERR_UNGATE_POD,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail=f"ungate pod {p['metadata']['name']}"):
                 self.store.patch("Pod", ns, p["metadata"]["name"], ungate)
-            except ApiError:
-                pass
 
     def _is_base_podgang_scheduled(self, ns: Optional[str], base_name: str) -> bool:
         """Base gang 'scheduled' = every podGroup's PCLQ has scheduledReplicas >=

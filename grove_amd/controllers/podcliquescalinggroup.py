@@ -17,10 +17,11 @@ from ..api import namegen
 from ..api.defaulting import parse_duration_seconds
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
+from ..utils import errors as groveerr
 from ..utils.hashing import pod_template_hash
 from . import builders
 from .manager import Result
-from .podcliqueset import _iso_to_epoch
+from .podcliqueset import _iso_to_epoch, _currently_updating_indices
 from . import resourceclaims
 
 log = logging.getLogger("grove.pcsg")
@@ -42,9 +43,11 @@ class PCSGReconciler:
         pcs = self._find_pcs(pcsg)
         if pcs is None:
             return Result(requeue_after=0.1)
-        res = self._sync_member_pclqs(pcs, pcsg)
-        recycle_wait = self._replica_recycle(pcs, pcsg)
+        rec = groveerr.StepRecorder(self.store, c.KIND_PCSG, namespace, name)
+        res = self._sync_member_pclqs(pcs, pcsg, rec)
+        recycle_wait = self._replica_recycle(pcs, pcsg, rec)
         self._reconcile_status(namespace, name)
+        rec.flush()
         if recycle_wait is not None:
             return Result(requeue_after=recycle_wait)
         return res
@@ -94,7 +97,8 @@ class PCSGReconciler:
         return Result.DONE
 
     # ------------------------------------------------------------------ spec
-    def _sync_member_pclqs(self, pcs: Obj, pcsg: Obj) -> Result:
+    def _sync_member_pclqs(self, pcs: Obj, pcsg: Obj,
+                           rec: groveerr.StepRecorder) -> Result:
         ns = pcsg["metadata"].get("namespace", "default")
         sg_fqn = pcsg["metadata"]["name"]
         pcs_replica = int(pcsg["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX, 0))
@@ -144,10 +148,10 @@ class PCSGReconciler:
                     obj["spec"]["updateStrategy"] = (
                         pcs["spec"].get("updateStrategy") or {}).get(
                         "type", c.UPDATE_ROLLING_RECREATE)
-                    try:
+                    with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                                  benign=groveerr.BENIGN_CREATE,
+                                  detail=f"create member PodClique {fqn}"):
                         self.store.create(obj)
-                    except ApiError:
-                        pass
                     continue
                 if cur["metadata"].get("deletionTimestamp"):
                     continue
@@ -160,17 +164,17 @@ class PCSGReconciler:
                     def upd(o: Obj) -> None:
                         o["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] = new_hash
                         o["spec"]["podSpec"] = cl["spec"].get("podSpec", {})
-                    try:
+                    with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                                  benign=groveerr.BENIGN_UPDATE,
+                                  detail=f"propagate template to {fqn}"):
                         self.store.patch(c.KIND_PCLQ, ns, fqn, upd)
-                    except ApiError:
-                        pass
         # scale-in: delete member PCLQs beyond current replicas
         for fqn, q in existing.items():
             if fqn not in expected:
-                try:
+                with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"scale-in PodClique {fqn}"):
                     self.store.delete(c.KIND_PCLQ, ns, fqn)
-                except ApiError:
-                    pass
         return Result.DONE
 
     @staticmethod
@@ -180,11 +184,11 @@ class PCSGReconciler:
         prog = (pcs.get("status") or {}).get("updateProgress")
         if prog is None:
             return True
-        cu = prog.get("currentlyUpdating")
-        return cu is not None and int(cu.get("replicaIndex", -1)) == r
+        return r in _currently_updating_indices(prog)
 
     # ------------------------------------------------------------------ replica recycle
-    def _replica_recycle(self, pcs: Obj, pcsg: Obj) -> Optional[float]:
+    def _replica_recycle(self, pcs: Obj, pcsg: Obj,
+                         rec: groveerr.StepRecorder) -> Optional[float]:
         """PCSG-replica-scoped gang recycle (sync.go:127): a scaled replica whose member
         cliques breached MinAvailable past terminationDelay is deleted and recreated.
         Fresh cliques have everScheduled=False so the loop cannot re-fire (WasPCLQ-
@@ -217,10 +221,10 @@ class PCSGReconciler:
                 continue
             log.info("PCSG %s/%s recycling replica %d", ns, pcsg["metadata"]["name"], j)
             for q in qs:
-                try:
+                with rec.step(groveerr.ERR_GANG_TERMINATION,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"recycle PodClique {q['metadata']['name']}"):
                     self.store.delete(c.KIND_PCLQ, ns, q["metadata"]["name"])
-                except ApiError:
-                    pass
         return next_wait
 
     # ------------------------------------------------------------------ status

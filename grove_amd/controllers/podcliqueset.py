@@ -40,6 +40,16 @@ def _iso_to_epoch(ts: str) -> float:
         return 0.0
 
 
+def _currently_updating_indices(prog: Obj) -> Set[int]:
+    out: Set[int] = set()
+    for e in prog.get("currentlyUpdating") or []:
+        try:
+            out.add(int(e.get("replicaIndex", -1)))
+        except (TypeError, ValueError):
+            pass
+    return out
+
+
 class PodCliqueSetReconciler:
     def __init__(self, store: Store, scheduler_name: str = c.SCHEDULER_AMD_GANG,
                  auto_xgmi_domain: bool = False):
@@ -49,22 +59,26 @@ class PodCliqueSetReconciler:
 
     # ------------------------------------------------------------------ entry
     def reconcile(self, namespace: str, name: str) -> Result:
-        """Top-level reconcile; errors are recorded to status.lastErrors with their
-        ERR_* code (reconcileerrorrecorder.go parity) and re-raised for backoff."""
+        """Top-level reconcile. Every mutation step runs under the StepRecorder
+        (reconcileerrorrecorder.go parity): per-step ApiErrors that are not benign
+        races are batched into status.lastErrors + Warning Events at flush, and
+        escaping exceptions are recorded with their ERR_* code before re-raise."""
+        rec = groveerr.StepRecorder(self.store, c.KIND_PCS, namespace, name)
         try:
-            res = self._reconcile(namespace, name)
-            groveerr.clear_last_errors(self.store, c.KIND_PCS, namespace, name)
+            res = self._reconcile(namespace, name, rec)
+            rec.flush()
             return res
         except groveerr.GroveError as e:
-            groveerr.record_last_error(self.store, c.KIND_PCS, namespace, name,
-                                       e.code, e.message)
+            rec.record(e.code, e.message)
+            rec.flush()
             raise
         except Exception as e:
-            groveerr.record_last_error(self.store, c.KIND_PCS, namespace, name,
-                                       groveerr.ERR_RECONCILE, str(e))
+            rec.record(groveerr.ERR_RECONCILE, str(e))
+            rec.flush()
             raise
 
-    def _reconcile(self, namespace: str, name: str) -> Result:
+    def _reconcile(self, namespace: str, name: str,
+                   rec: groveerr.StepRecorder) -> Result:
         pcs = self.store.try_get(c.KIND_PCS, namespace, name)
         if pcs is None:
             return Result.DONE
@@ -77,12 +91,12 @@ class PodCliqueSetReconciler:
                     o["metadata"]["finalizers"].append(c.FINALIZER_PCS)
             pcs = self.store.patch(c.KIND_PCS, namespace, name, add_fin)
 
-        self._process_generation_hash(pcs)
+        self._process_generation_hash(pcs, rec)
         pcs = self.store.get(c.KIND_PCS, namespace, name)
-        res = self._sync_resources(pcs)
-        term = self._gang_termination(pcs)
-        self._orchestrate_rolling_update(pcs)
-        self._reconcile_status(namespace, name)
+        res = self._sync_resources(pcs, rec)
+        term = self._gang_termination(pcs, rec)
+        self._orchestrate_rolling_update(pcs, rec)
+        self._reconcile_status(namespace, name, rec)
         if term is not None:
             return Result(requeue_after=term)
         return res
@@ -90,22 +104,22 @@ class PodCliqueSetReconciler:
     # ------------------------------------------------------------------ delete
     def _reconcile_delete(self, pcs: Obj) -> Result:
         ns, name = pcs["metadata"].get("namespace"), pcs["metadata"]["name"]
+        rec = groveerr.StepRecorder(self.store, c.KIND_PCS, ns, name)
         sel = {c.LABEL_PART_OF: name}
         remaining = 0
         for kind in (c.KIND_PCLQ, c.KIND_PCSG):
             for obj in self.store.list(kind, ns, sel):
                 remaining += 1
-                try:
+                with rec.step(groveerr.ERR_DELETE, benign=groveerr.BENIGN_DELETE,
+                              detail=f"delete {kind} {obj['metadata']['name']}"):
                     self.store.delete(kind, ns, obj["metadata"]["name"])
-                except ApiError:
-                    pass
         if remaining:
+            rec.flush()
             return Result(requeue_after=0.02)
         for pg in self.store.list(c.KIND_PODGANG, ns, sel):
-            try:
+            with rec.step(groveerr.ERR_DELETE, benign=groveerr.BENIGN_DELETE,
+                          detail=f"delete PodGang {pg['metadata']['name']}"):
                 self.store.delete(c.KIND_PODGANG, ns, pg["metadata"]["name"])
-            except ApiError:
-                pass
 
         def rm(o: Obj) -> None:
             o["metadata"]["finalizers"] = [
@@ -117,8 +131,12 @@ class PodCliqueSetReconciler:
         return Result.DONE
 
     # ------------------------------------------------------------------ generation hash
-    def _process_generation_hash(self, pcs: Obj) -> None:
-        """reconcilespec.go:72-158: persist template hash; on change, start update."""
+    def _process_generation_hash(self, pcs: Obj,
+                                  rec: groveerr.StepRecorder) -> None:
+        """reconcilespec.go:72-158: persist template hash; on change, start update.
+        updateProgress uses the published status schema (podcliqueset.go/CRD):
+        updateStartedAt/updateEndedAt + currentlyUpdating as a LIST of
+        {replicaIndex, updateStartedAt} (one at a time => <= 1 entry)."""
         ns, name = pcs["metadata"].get("namespace"), pcs["metadata"]["name"]
         new_hash = pcs_generation_hash(pcs)
         st = pcs.get("status") or {}
@@ -131,17 +149,15 @@ class PodCliqueSetReconciler:
             if s.get("currentGenerationHash") and s.get("currentGenerationHash") != new_hash:
                 s["updateProgress"] = {
                     "updateStartedAt": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
-                    "currentlyUpdating": None,
-                    "updatedReplicas": [],
+                    "currentlyUpdating": [],
                 }
             s["currentGenerationHash"] = new_hash
-        try:
+        with rec.step(groveerr.ERR_UPDATE_STATUS, benign=groveerr.BENIGN_UPDATE,
+                      detail="persist generation hash"):
             self.store.patch(c.KIND_PCS, ns, name, upd, status=True)
-        except ApiError:
-            pass
 
     # ------------------------------------------------------------------ spec sync
-    def _sync_resources(self, pcs: Obj) -> Result:
+    def _sync_resources(self, pcs: Obj, rec: groveerr.StepRecorder) -> Result:
         ns = pcs["metadata"].get("namespace", "default")
         name = pcs["metadata"]["name"]
         replicas = int(pcs["spec"].get("replicas", 0))
@@ -150,25 +166,27 @@ class PodCliqueSetReconciler:
         sg_members = {m for sg in sg_cfgs for m in (sg.get("cliqueNames") or [])}
 
         # ---- G1: RBAC + token secret + per-replica headless Service + HPAs
-        self._ensure(builders.build_service_account(pcs))
-        self._ensure(builders.build_role(pcs))
-        self._ensure(builders.build_role_binding(pcs))
-        self._ensure(builders.build_sa_token_secret(pcs))
+        self._ensure(builders.build_service_account(pcs), rec, groveerr.ERR_SYNC_RBAC)
+        self._ensure(builders.build_role(pcs), rec, groveerr.ERR_SYNC_RBAC)
+        self._ensure(builders.build_role_binding(pcs), rec, groveerr.ERR_SYNC_RBAC)
+        self._ensure(builders.build_sa_token_secret(pcs), rec,
+                     groveerr.ERR_SYNC_RBAC)
         existing_svcs = {s["metadata"]["name"] for s in self.store.list(
             "Service", ns, {c.LABEL_PART_OF: name,
                             c.LABEL_COMPONENT: c.COMPONENT_HEADLESS_SERVICE})}
         for r in range(replicas):
             svc_name = namegen.headless_service_name(name, r)
             if svc_name not in existing_svcs:
-                self._ensure(builders.build_headless_service(pcs, r))
+                self._ensure(builders.build_headless_service(pcs, r), rec,
+                             groveerr.ERR_SYNC_SERVICE)
         for svc_name in existing_svcs:
             idx = svc_name.rsplit("-", 1)[-1]
             if idx.isdigit() and int(idx) >= replicas:
-                try:
+                with rec.step(groveerr.ERR_SYNC_SERVICE,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"GC service {svc_name}"):
                     self.store.delete("Service", ns, svc_name)
-                except ApiError:
-                    pass
-        self._sync_hpas(pcs)
+        self._sync_hpas(pcs, rec)
 
         # ---- G2: standalone PodCliques per replica (+ shared ResourceClaims)
         pclq_by_name = {q["metadata"]["name"]: q for q in self.store.list(
@@ -179,7 +197,9 @@ class PodCliqueSetReconciler:
         for r in range(replicas):
             claims = resourceclaims.pcs_claims_for_replica(
                 self.store, pcs, r, self.auto_xgmi_domain)
-            resourceclaims.ensure_claims(self.store, claims)
+            with rec.step(groveerr.ERR_SYNC_RESOURCE_CLAIM,
+                          detail=f"ensure claims for replica {r}"):
+                resourceclaims.ensure_claims(self.store, claims)
             for cl in tmpl.get("cliques") or []:
                 if cl["name"] in sg_members:
                     continue
@@ -192,14 +212,14 @@ class PodCliqueSetReconciler:
                 refs = refs + resourceclaims.claim_refs_for_clique(
                     cl_claims, cl["name"])
                 self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs,
-                                cur=pclq_by_name.get(fqn))
+                                cur=pclq_by_name.get(fqn), rec=rec)
         # GC excess standalone PCLQs (scale-in / replica removal)
         for pclq in pclq_by_name.values():
             if pclq["metadata"]["name"] not in expected_pclqs:
-                try:
+                with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"GC PodClique {pclq['metadata']['name']}"):
                     self.store.delete(c.KIND_PCLQ, ns, pclq["metadata"]["name"])
-                except ApiError:
-                    pass
 
         # ---- G3a: PCSGs per config per replica
         pcsg_names = {g["metadata"]["name"] for g in self.store.list(
@@ -210,25 +230,28 @@ class PodCliqueSetReconciler:
                 fqn = namegen.pcsg_name(name, r, sg["name"])
                 expected_pcsgs.add(fqn)
                 if fqn not in pcsg_names:
-                    try:
+                    with rec.step(groveerr.ERR_SYNC_PCSG,
+                                  benign=groveerr.BENIGN_CREATE,
+                                  detail=f"create PCSG {fqn}"):
                         self.store.create(builders.build_pcsg(pcs, r, sg))
-                    except ApiError:
-                        pass
         for pcsg in self.store.list(c.KIND_PCSG, ns, {
                 c.LABEL_PART_OF: name, c.LABEL_COMPONENT: c.COMPONENT_PCSG}):
             if pcsg["metadata"]["name"] not in expected_pcsgs:
-                try:
+                with rec.step(groveerr.ERR_SYNC_PCSG,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"GC PCSG {pcsg['metadata']['name']}"):
                     self.store.delete(c.KIND_PCSG, ns, pcsg["metadata"]["name"])
-                except ApiError:
-                    pass
 
         # ---- G3b: PodGangs
-        sync_podgangs(self.store, pcs, self.scheduler_name)
+        sync_podgangs(self.store, pcs, self.scheduler_name, rec)
         return Result.DONE
 
     def _sync_pclq(self, pcs: Obj, r: int, clique_tmpl: Obj, fqn: str, owner: Obj,
-                   claim_refs=None, cur="__lookup__") -> None:
+                   claim_refs=None, cur="__lookup__",
+                   rec: Optional[groveerr.StepRecorder] = None) -> None:
         ns = pcs["metadata"].get("namespace", "default")
+        rec = rec or groveerr.StepRecorder(self.store, c.KIND_PCS, ns,
+                                           pcs["metadata"]["name"])
         if cur == "__lookup__":
             cur = self.store.try_get(c.KIND_PCLQ, ns, fqn)
         if cur is None:
@@ -237,10 +260,10 @@ class PodCliqueSetReconciler:
                 "type", c.UPDATE_ROLLING_RECREATE)
             if claim_refs:
                 obj["spec"]["resourceClaims"] = claim_refs
-            try:
+            with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                          benign=groveerr.BENIGN_CREATE,
+                          detail=f"create PodClique {fqn}"):
                 self.store.create(obj)
-            except ApiError:
-                pass
             return
         if cur["metadata"].get("deletionTimestamp"):
             return
@@ -262,10 +285,10 @@ class PodCliqueSetReconciler:
                 for f in fields:
                     if f in clique_tmpl["spec"]:
                         o["spec"][f] = clique_tmpl["spec"][f]
-            try:
+            with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail=f"propagate template to {fqn}"):
                 self.store.patch(c.KIND_PCLQ, ns, fqn, upd)
-            except ApiError:
-                pass
         # HPA-aware replica preservation: never stomp replicas on scaled cliques —
         # only non-scaled fields drift-corrected here.
 
@@ -275,10 +298,9 @@ class PodCliqueSetReconciler:
         prog = (pcs.get("status") or {}).get("updateProgress")
         if prog is None:
             return True  # no update in flight → initial create path
-        cu = prog.get("currentlyUpdating")
-        return cu is not None and int(cu.get("replicaIndex", -1)) == r
+        return r in _currently_updating_indices(prog)
 
-    def _sync_hpas(self, pcs: Obj) -> None:
+    def _sync_hpas(self, pcs: Obj, rec: groveerr.StepRecorder) -> None:
         ns = pcs["metadata"].get("namespace", "default")
         name = pcs["metadata"]["name"]
         replicas = int(pcs["spec"].get("replicas", 0))
@@ -301,23 +323,24 @@ class PodCliqueSetReconciler:
             {c.LABEL_PART_OF: name, c.LABEL_COMPONENT: c.COMPONENT_HPA})}
         for hname, (kind, target, cfg) in expected.items():
             if hname not in existing:
-                self._ensure(builders.build_hpa(pcs, kind, target, cfg))
+                self._ensure(builders.build_hpa(pcs, kind, target, cfg), rec,
+                             groveerr.ERR_SYNC_HPA)
         for hname in existing:
             if hname not in expected:
-                try:
+                with rec.step(groveerr.ERR_SYNC_HPA,
+                              benign=groveerr.BENIGN_DELETE,
+                              detail=f"GC HPA {hname}"):
                     self.store.delete("HorizontalPodAutoscaler", ns, hname)
-                except ApiError:
-                    pass
 
-    def _ensure(self, obj: Obj) -> None:
-        try:
+    def _ensure(self, obj: Obj, rec: groveerr.StepRecorder, code: str) -> None:
+        with rec.step(code, benign=groveerr.BENIGN_CREATE,
+                      detail=f"create {obj.get('kind')} "
+                             f"{obj['metadata'].get('name', '')}"):
             self.store.create(obj)
-        except ApiError as e:
-            if e.reason != "AlreadyExists":
-                raise
 
     # ------------------------------------------------------------------ gang termination
-    def _gang_termination(self, pcs: Obj) -> Optional[float]:
+    def _gang_termination(self, pcs: Obj,
+                          rec: groveerr.StepRecorder) -> Optional[float]:
         """PCS-replica-scope gang termination (gangterminate.go:69-332). Returns seconds
         until the next pending termination (for requeue), or None."""
         ns = pcs["metadata"].get("namespace", "default")
@@ -346,13 +369,12 @@ class PodCliqueSetReconciler:
             if ridx is not None and ridx.isdigit():
                 by_replica.setdefault(int(ridx), [])
 
-        cu = ((pcs.get("status") or {}).get("updateProgress") or {}) \
-            .get("currentlyUpdating") or {}
-        updating_replica = int(cu.get("replicaIndex", -1))
+        updating = _currently_updating_indices(
+            (pcs.get("status") or {}).get("updateProgress") or {})
         for ridx, constituents in by_replica.items():
             # suppression: a replica mid-rolling-update dips below MinAvailable by
             # design; never gang-terminate it while it is the one being updated
-            if ridx == updating_replica:
+            if ridx in updating:
                 continue
             # suppression: in-flight termination for this replica's PCSGs
             replica_pcsgs = [g for g in pcsgs.values()
@@ -385,11 +407,31 @@ class PodCliqueSetReconciler:
                 def mark(o: Obj) -> None:
                     cond.set_condition(o, c.COND_GANG_TERMINATION_IN_PROGRESS, True,
                                        c.REASON_GANG_TERMINATION_ACTIVE)
-                try:
+                with rec.step(groveerr.ERR_GANG_TERMINATION,
+                              benign=groveerr.BENIGN_UPDATE,
+                              detail=f"mark PCSG {g['metadata']['name']}"):
                     self.store.patch(c.KIND_PCSG, ns, g["metadata"]["name"], mark,
                                      status=True)
-                except ApiError:
-                    pass
+            # DisruptionTarget on every PodGang of the doomed replica
+            # (scheduler/api/core/v1alpha1/podgang.go:152-171 contract): schedulers
+            # and drain tooling see the gang is being terminated by the operator.
+            base_pg = namegen.base_podgang_name(name, ridx)
+            for pg in self.store.list(c.KIND_PODGANG, ns, {c.LABEL_PART_OF: name},
+                                      copy_objects=False):
+                pg_name = pg["metadata"]["name"]
+                if pg_name != base_pg and \
+                        pg["metadata"].get("labels", {}).get(
+                            c.LABEL_BASE_PODGANG) != base_pg:
+                    continue
+
+                def mark_disrupted(o: Obj) -> None:
+                    cond.set_condition(o, c.PODGANG_COND_DISRUPTION_TARGET, True,
+                                       "GangTerminated")
+                with rec.step(groveerr.ERR_GANG_TERMINATION,
+                              benign=groveerr.BENIGN_UPDATE,
+                              detail=f"mark DisruptionTarget on {pg_name}"):
+                    self.store.patch(c.KIND_PODGANG, ns, pg_name,
+                                     mark_disrupted, status=True)
             self.store.delete_collection(c.KIND_PCLQ, ns, {
                 c.LABEL_PART_OF: name, c.LABEL_PCS_REPLICA_INDEX: str(ridx)})
             self.store.record_event(pcs, "Warning", "GangTerminated",
@@ -397,7 +439,8 @@ class PodCliqueSetReconciler:
         return next_wait
 
     # ------------------------------------------------------------------ rolling update
-    def _orchestrate_rolling_update(self, pcs: Obj) -> None:
+    def _orchestrate_rolling_update(self, pcs: Obj,
+                                    rec: groveerr.StepRecorder) -> None:
         """rollingupdate.go:37-296: one replica at a time, ordered no-scheduled-pods →
         breached → ordinal."""
         ns = pcs["metadata"].get("namespace", "default")
@@ -459,38 +502,36 @@ class PodCliqueSetReconciler:
                     return False
             return True
 
-        cu = prog.get("currentlyUpdating")
-        if cu is not None:
-            r = int(cu.get("replicaIndex", -1))
+        updating = sorted(_currently_updating_indices(prog))
+        if updating:
+            r = updating[0]
             if not replica_updated(r):
-                return  # still updating this replica
+                return  # still updating this replica (one at a time)
+
             def done(o: Obj) -> None:
                 p = o["status"].get("updateProgress") or {}
-                ur = p.setdefault("updatedReplicas", [])
-                if r not in ur:
-                    ur.append(r)
-                p["currentlyUpdating"] = None
+                p["currentlyUpdating"] = [
+                    e for e in p.get("currentlyUpdating") or []
+                    if int(e.get("replicaIndex", -1)) != r]
                 o["status"]["updateProgress"] = p
-            try:
+            with rec.step(groveerr.ERR_ROLLING_UPDATE,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail=f"complete update of replica {r}"):
                 self.store.patch(c.KIND_PCS, ns, name, done, status=True)
-            except ApiError:
-                return
-            prog = dict(prog, currentlyUpdating=None,
-                        updatedReplicas=list(prog.get("updatedReplicas", [])) + [r])
 
-        done_set = set(prog.get("updatedReplicas") or [])
-        pending = [r for r in range(replicas) if r not in done_set
-                   and not replica_updated(r)]
+        # completion is derived from ground truth (template hashes + pods), not a
+        # bookkeeping list — survives status pruning on a real apiserver
+        pending = [r for r in range(replicas) if not replica_updated(r)]
         if not pending:
             def finish(o: Obj) -> None:
                 p = o["status"].get("updateProgress") or {}
                 p["updateEndedAt"] = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-                p["currentlyUpdating"] = None
+                p["currentlyUpdating"] = []
                 o["status"]["updateProgress"] = p
-            try:
+            with rec.step(groveerr.ERR_ROLLING_UPDATE,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail="finish rolling update"):
                 self.store.patch(c.KIND_PCS, ns, name, finish, status=True)
-            except ApiError:
-                pass
             return
 
         # ordering: no-scheduled-pods first, then breached, then ordinal (:182-209)
@@ -503,18 +544,23 @@ class PodCliqueSetReconciler:
 
         def select(o: Obj) -> None:
             p = o["status"].get("updateProgress") or {}
-            p["currentlyUpdating"] = {
-                "replicaIndex": target,
-                "startedAt": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
-            }
+            cur = p.get("currentlyUpdating") or []
+            if not any(int(e.get("replicaIndex", -1)) == target for e in cur):
+                cur = cur + [{
+                    "replicaIndex": target,
+                    "updateStartedAt": time.strftime("%Y-%m-%dT%H:%M:%SZ",
+                                                     time.gmtime()),
+                }]
+            p["currentlyUpdating"] = cur
             o["status"]["updateProgress"] = p
-        try:
+        with rec.step(groveerr.ERR_ROLLING_UPDATE, benign=groveerr.BENIGN_UPDATE,
+                      detail=f"select replica {target} for update"):
             self.store.patch(c.KIND_PCS, ns, name, select, status=True)
-        except ApiError:
-            pass
 
     # ------------------------------------------------------------------ status
-    def _reconcile_status(self, namespace: str, name: str) -> None:
+    def _reconcile_status(self, namespace: str, name: str,
+                          rec: Optional[groveerr.StepRecorder] = None) -> None:
+        rec = rec or groveerr.StepRecorder(self.store, c.KIND_PCS, namespace, name)
         pcs = self.store.try_get(c.KIND_PCS, namespace, name)
         if pcs is None or pcs["metadata"].get("deletionTimestamp"):
             return
@@ -555,10 +601,46 @@ class PodCliqueSetReconciler:
             if ok:
                 available += 1
 
-        # rolling-update bookkeeping for updatedReplicas count
+        # updated-replica + per-PCLQ/PCSG update counters, derived from template
+        # hashes (printer-column contract: PCLQs-Updated/PCLQs-Total etc.)
         st = pcs.get("status") or {}
         prog = st.get("updateProgress")
-        updated = len((prog or {}).get("updatedReplicas") or []) if prog else replicas
+
+        def expected_hash_of(q: Obj) -> str:
+            cl = builders.match_by_fqn_suffix(q["metadata"]["name"],
+                                              tmpl.get("cliques") or [])
+            if cl is None:
+                return ""
+            return pod_template_hash(cl["name"], cl["spec"].get("podSpec", {}),
+                                     tmpl.get("priorityClassName", ""),
+                                     cl.get("labels"), cl.get("annotations"))
+
+        def pclq_hash_current(q: Obj) -> bool:
+            return q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) \
+                == expected_hash_of(q)
+
+        all_pclq_by_replica: Dict[int, List[Obj]] = {}
+        for q in pclqs:
+            ridx = q["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX, "")
+            if ridx.isdigit():
+                all_pclq_by_replica.setdefault(int(ridx), []).append(q)
+        updated = 0
+        for r in range(replicas):
+            qs = all_pclq_by_replica.get(r, [])
+            if qs and all(pclq_hash_current(q) for q in qs):
+                updated += 1
+        total_pclqs = len(pclqs)
+        updated_pclqs = sum(1 for q in pclqs if pclq_hash_current(q))
+        pclqs_of_pcsg: Dict[str, List[Obj]] = {}
+        for q in pclqs:
+            sg = q["metadata"]["labels"].get(c.LABEL_PCSG)
+            if sg:
+                pclqs_of_pcsg.setdefault(sg, []).append(q)
+        total_pcsgs = len(pcsgs)
+        updated_pcsgs = sum(
+            1 for g in pcsgs
+            if all(pclq_hash_current(q)
+                   for q in pclqs_of_pcsg.get(g["metadata"]["name"], [])))
 
         # per-gang phase rollup (podcliqueset.go PodGangStatus)
         gang_statuses: List[Dict[str, Any]] = []
@@ -587,16 +669,28 @@ class PodCliqueSetReconciler:
                 cond.condition_true(pclq_of.get(g["name"], {}),
                                     c.COND_MIN_AVAILABLE_BREACHED)
                 for g in groups)
+            # a recycled gang that is Running again is no longer a disruption target
+            if phase == "Running" and cond.condition_true(
+                    pg, c.PODGANG_COND_DISRUPTION_TARGET):
+                def clear_dt(o: Obj) -> None:
+                    cond.set_condition(o, c.PODGANG_COND_DISRUPTION_TARGET, False,
+                                       "Recovered")
+                with rec.step(groveerr.ERR_SYNC_PODGANG,
+                              benign=groveerr.BENIGN_UPDATE,
+                              detail=f"clear DisruptionTarget on "
+                                     f"{pg['metadata']['name']}"):
+                    self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"],
+                                     clear_dt, status=True)
             was = cond.condition_true(pg, c.PODGANG_COND_UNHEALTHY)
             if unhealthy != was and cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):
                 def flip_unhealthy(o: Obj, v=unhealthy) -> None:
                     cond.set_condition(o, c.PODGANG_COND_UNHEALTHY, v,
                                        "MinAvailableBreached" if v else "Recovered")
-                try:
+                with rec.step(groveerr.ERR_SYNC_PODGANG,
+                              benign=groveerr.BENIGN_UPDATE,
+                              detail=f"flip Unhealthy on {pg['metadata']['name']}"):
                     self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"],
                                      flip_unhealthy, status=True)
-                except ApiError:
-                    pass
 
         def upd(o: Obj) -> None:
             s = o.setdefault("status", {})
@@ -606,7 +700,12 @@ class PodCliqueSetReconciler:
             s["observedGeneration"] = o["metadata"].get("generation")
             s["podGangStatuses"] = sorted(gang_statuses, key=lambda x: x["name"])
             s["hpaPodSelector"] = f"{c.LABEL_PART_OF}={name}"
-        try:
+            if s.get("updateProgress"):
+                p = s["updateProgress"]
+                p["totalPodCliquesCount"] = total_pclqs
+                p["updatedPodCliquesCount"] = updated_pclqs
+                p["totalPodCliqueScalingGroupsCount"] = total_pcsgs
+                p["updatedPodCliqueScalingGroupsCount"] = updated_pcsgs
+        with rec.step(groveerr.ERR_UPDATE_STATUS, benign=groveerr.BENIGN_UPDATE,
+                      detail="write PCS status"):
             self.store.patch(c.KIND_PCS, namespace, name, upd, status=True)
-        except ApiError:
-            pass

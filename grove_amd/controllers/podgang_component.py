@@ -15,6 +15,7 @@ from ..api import constants as c
 from ..api import namegen
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
+from ..utils import errors as groveerr
 from . import builders
 
 Obj = Dict[str, Any]
@@ -161,10 +162,13 @@ def compute_expected_podgangs(store: Store, pcs: Obj,
     return out
 
 
-def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
-    """Create/update PodGangs to match expectations; GC stale ones; flip Initialized."""
+def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str,
+                  rec: "groveerr.StepRecorder" = None) -> None:
+    """Create/update PodGangs to match expectations; GC stale ones; flip Initialized.
+    Non-benign step failures are recorded to the PCS's status.lastErrors via rec."""
     ns = pcs["metadata"].get("namespace", "default")
     pcs_name = pcs["metadata"]["name"]
+    rec = rec or groveerr.StepRecorder(store, c.KIND_PCS, ns, pcs_name)
     pcsg_by_name = {g["metadata"]["name"]: g for g in store.list(
         c.KIND_PCSG, ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)}
     expected = compute_expected_podgangs(store, pcs, pcsg_by_name)
@@ -174,10 +178,10 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
     existing_by_name = {pg["metadata"]["name"]: pg for pg in existing}
     for pg in existing:
         if pg["metadata"]["name"] not in expected_names:
-            try:
+            with rec.step(groveerr.ERR_SYNC_PODGANG,
+                          benign=groveerr.BENIGN_DELETE,
+                          detail=f"GC PodGang {pg['metadata']['name']}"):
                 store.delete(c.KIND_PODGANG, ns, pg["metadata"]["name"])
-            except ApiError:
-                pass
 
     # index pods once per sync pass: podgang label -> podclique label -> [pod names]
     pods = store.list("Pod", ns, {c.LABEL_PART_OF: pcs_name}, copy_objects=False)
@@ -214,7 +218,11 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
             cond.set_condition(obj, c.PODGANG_COND_INITIALIZED, False, "PendingPodCreation")
             try:
                 cur = store.create(obj)
-            except ApiError:
+            except ApiError as e:
+                if e.reason != "AlreadyExists":
+                    rec.record(groveerr.ERR_SYNC_PODGANG,
+                               f"create PodGang {gang.name}: {e.reason}: "
+                               f"{e.message}")
                 cur = store.try_get(c.KIND_PODGANG, ns, gang.name)
                 if cur is None:
                     continue
@@ -231,17 +239,21 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str) -> None:
                     "topologyConstraintGroupConfigs") != gang.group_configs):
             try:
                 cur = store.patch(c.KIND_PODGANG, ns, gang.name, upd)
-            except ApiError:
+            except ApiError as e:
+                if e.reason not in groveerr.BENIGN_UPDATE:
+                    rec.record(groveerr.ERR_SYNC_PODGANG,
+                               f"patch PodGang {gang.name}: {e.reason}: "
+                               f"{e.message}")
                 continue
 
         initialized = cond.condition_true(cur, c.PODGANG_COND_INITIALIZED)
         if all_created and gang.groups and not initialized:
             def flip(o: Obj) -> None:
                 cond.set_condition(o, c.PODGANG_COND_INITIALIZED, True, "AllPodsAssociated")
-            try:
+            with rec.step(groveerr.ERR_SYNC_PODGANG,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail=f"flip Initialized on {gang.name}"):
                 store.patch(c.KIND_PODGANG, ns, gang.name, flip, status=True)
-            except ApiError:
-                pass
         elif not all_created and initialized:
             # pods lost after init (e.g. gang termination in flight) — keep Initialized
             pass

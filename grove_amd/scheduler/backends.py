@@ -82,8 +82,10 @@ class AmdGangBackend(Backend):
                 c.ANNOTATION_TOPOLOGY_NAME] = ctb_name
         try:
             self.store.patch(c.KIND_PODGANG, ns, podgang["metadata"]["name"], stamp)
-        except ApiError:
-            pass
+        except ApiError as e:
+            from ..utils.errors import report_api_error
+            report_api_error(self.store, c.KIND_PODGANG, ns,
+                             podgang["metadata"]["name"], "stamp topology name", e)
 
     def validate_podcliqueset(self, pcs: Obj) -> None:
         return  # native backend accepts all topology constraints

@@ -25,6 +25,7 @@ from typing import Any, Dict, List, Optional, Set, Tuple
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
 from ..utils import conditions as cond
+from ..utils.errors import report_api_error
 from ..utils.quantity import cpu_millis, parse_quantity
 from .placement import Assignment, NodeFree, PodRequest, place_gang
 
@@ -228,8 +229,10 @@ class GangScheduler:
                         try:
                             self.store.patch(c.KIND_PODGANG, ns,
                                              pg["metadata"]["name"], mark0, status=True)
-                        except ApiError:
-                            pass
+                        except ApiError as e:
+                            report_api_error(self.store, c.KIND_PODGANG, ns,
+                                             pg["metadata"]["name"],
+                                             "mark scheduled (reservation)", e)
                         return
 
         result = self._place_gang_pods(nodes, pg, chosen)
@@ -256,8 +259,9 @@ class GangScheduler:
             o["status"]["phase"] = "Starting"
         try:
             self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark, status=True)
-        except ApiError:
-            pass
+        except ApiError as e:
+            report_api_error(self.store, c.KIND_PODGANG, ns,
+                             pg["metadata"]["name"], "mark gang scheduled", e)
 
     def _count_bound(self, ns: str, group: Obj, pods_by_name: Dict) -> int:
         n = 0
@@ -386,8 +390,9 @@ class GangScheduler:
                                   "%Y-%m-%dT%H:%M:%SZ", time.gmtime())})
         try:
             self.store.patch("Pod", ns, pod["metadata"]["name"], apply)
-        except ApiError:
-            pass
+        except ApiError as e:
+            report_api_error(self.store, "Pod", ns, pod["metadata"]["name"],
+                             "bind pod", e)
 
     # ------------------------------------------------------------------ ready rollup
     def _rollup_ready(self, pods_by_name: Dict) -> None:
@@ -414,5 +419,6 @@ class GangScheduler:
                 try:
                     self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark,
                                      status=True)
-                except ApiError:
-                    pass
+                except ApiError as e:
+                    report_api_error(self.store, c.KIND_PODGANG, ns,
+                                     pg["metadata"]["name"], "mark gang ready", e)

@@ -100,7 +100,12 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
         kind = kind_of(plural)
         params = request.query_params
         if params.get("watch") in ("true", "1"):
-            w = store.watch(kind, seed=params.get("seed", "true") in ("true", "1"))
+            since_rv = params.get("resourceVersion") or None
+            bookmarks = params.get("allowWatchBookmarks") in ("true", "1")
+            w = store.watch(kind,
+                            seed=(since_rv is None
+                                  and params.get("seed", "true") in ("true", "1")),
+                            since_rv=since_rv)
 
             def stream():
                 try:
@@ -108,14 +113,30 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
                         try:
                             ev, obj = w.queue.get(timeout=1.0)
                         except queue.Empty:
-                            yield "\n"  # keepalive
+                            if bookmarks:
+                                # kube BOOKMARK: progress marker carrying only the
+                                # current resourceVersion, so clients can resume
+                                # without replaying history
+                                yield json.dumps({"type": "BOOKMARK", "object": {
+                                    "kind": kind, "metadata": {
+                                        "resourceVersion": store.current_rv()}}}) \
+                                    + "\n"
+                            else:
+                                yield "\n"  # keepalive
                             continue
                         yield json.dumps({"type": ev, "object": obj}) + "\n"
                 finally:
                     w.stop()
             return StreamingResponse(stream(), media_type="application/x-ndjson")
-        items = store.list(kind, ns, parse_selector(params.get("labelSelector")))
-        return JSONResponse({"kind": f"{kind}List", "items": items})
+        selector = parse_selector(params.get("labelSelector"))
+        limit = int(params["limit"]) if params.get("limit") else None
+        cont = params.get("continue") or None
+        items, next_cont, rv = store.list_page(kind, ns, selector, limit, cont)
+        list_meta = {"resourceVersion": rv}
+        if next_cont:
+            list_meta["continue"] = next_cont
+        return JSONResponse({"kind": f"{kind}List", "apiVersion": "v1",
+                             "metadata": list_meta, "items": items})
 
     # ---- namespaced ----
     @app.get("/apis/{group}/{version}/namespaces/{ns}/{plural}")
@@ -188,17 +209,15 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
         except ApiError as e:
             return err(e)
 
-    # kubectl `patch --type=merge` analog: RFC 7386 JSON merge patch (null deletes a
-    # key; objects merge recursively; everything else replaces).
-    def _merge_patch(target, patch):
-        if not isinstance(patch, dict) or not isinstance(target, dict):
-            return patch
-        for k, v in patch.items():
-            if v is None:
-                target.pop(k, None)
-            else:
-                target[k] = _merge_patch(target.get(k), v)
-        return target
+    # kubectl `patch --type=merge|strategic` analog: strategy picked by the
+    # request Content-Type, exactly like the apiserver (kubecore/patching.py).
+    from .patching import json_merge_patch, strategic_merge_patch
+
+    def _patch_fn(request: Request):
+        ctype = request.headers.get("content-type", "")
+        if "strategic-merge-patch" in ctype:
+            return strategic_merge_patch
+        return json_merge_patch
 
     @app.patch("/apis/{group}/{version}/namespaces/{ns}/{plural}/{name}")
     @app.patch("/api/{version}/namespaces/{ns}/{plural}/{name}")
@@ -206,9 +225,10 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
                        group: str = "", version: str = "v1"):
         try:
             body = await request.json()
+            merge = _patch_fn(request)
 
             def apply(o):
-                _merge_patch(o, body)
+                merge(o, body)
             with as_user(user_of(request)):
                 return JSONResponse(store.patch(kind_of(plural), ns, name, apply))
         except ApiError as e:
@@ -221,9 +241,10 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
                               group: str = "", version: str = "v1"):
         try:
             body = await request.json()
+            merge = _patch_fn(request)
 
             def apply(o):
-                _merge_patch(o, body)
+                merge(o, body)
             with as_user(user_of(request)):
                 return JSONResponse(
                     store.patch(kind_of(plural), ns, name, apply, status=True))
@@ -236,9 +257,10 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
                                    group: str = "", version: str = "v1"):
         try:
             body = await request.json()
+            merge = _patch_fn(request)
 
             def apply(o):
-                _merge_patch(o, body)
+                merge(o, body)
             with as_user(user_of(request)):
                 return JSONResponse(
                     store.patch(kind_of(plural), None, name, apply, status=True))
@@ -251,9 +273,10 @@ def build_app(store: Store, metrics_fn=None, auth_tokens: Optional[Dict[str, str
                             group: str = "", version: str = "v1"):
         try:
             body = await request.json()
+            merge = _patch_fn(request)
 
             def apply(o):
-                _merge_patch(o, body)
+                merge(o, body)
             with as_user(user_of(request)):
                 return JSONResponse(store.patch(kind_of(plural), None, name, apply))
         except ApiError as e:

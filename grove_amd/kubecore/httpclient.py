@@ -152,15 +152,64 @@ class HttpStoreClient:
         sub = "status" if status else ""
         return self._request("PATCH", self._url(kind, namespace, name, sub), patch)
 
+    def strategic_merge_patch(self, kind: str, namespace: Optional[str], name: str,
+                              patch: Obj) -> Obj:
+        """client-go Patch(types.StrategicMergePatchType): merge-by-key list
+        semantics for pod specs (containers by name etc.)."""
+        import json as _json
+        import urllib.request as _rq
+        url = self._url(kind, namespace, name)
+        req = _rq.Request(url, data=_json.dumps(patch).encode(), method="PATCH",
+                          headers={**self._headers(),
+                                   "Content-Type":
+                                       "application/strategic-merge-patch+json"})
+        import urllib.error as _er
+        try:
+            with _rq.urlopen(req, timeout=self.timeout, context=self._ctx) as r:
+                return _json.loads(r.read())
+        except _er.HTTPError as e:
+            try:
+                payload = _json.loads(e.read())
+            except Exception:
+                payload = {}
+            raise ApiError(e.code, payload.get("reason", "HTTPError"),
+                           payload.get("message", str(e)))
+
+    def list_page(self, kind: str, namespace: Optional[str] = None,
+                  label_selector: Optional[Dict[str, str]] = None,
+                  limit: Optional[int] = None,
+                  continue_token: Optional[str] = None):
+        """Chunked list: returns (items, next_continue, resourceVersion)."""
+        import urllib.parse as _p
+        parts = []
+        if label_selector:
+            sel = ",".join(f"{k}={v}" for k, v in label_selector.items())
+            parts.append(f"labelSelector={_p.quote(sel)}")
+        if limit:
+            parts.append(f"limit={limit}")
+        if continue_token:
+            parts.append(f"continue={_p.quote(continue_token)}")
+        out = self._request("GET", self._url(kind, namespace,
+                                             query="&".join(parts)))
+        md = out.get("metadata") or {}
+        return out["items"], md.get("continue"), md.get("resourceVersion")
+
     def delete(self, kind: str, namespace: Optional[str], name: str,
                cascade: bool = True) -> None:
         self._request("DELETE", self._url(kind, namespace, name))
 
     def watch_events(self, kind: str, namespace: Optional[str] = None,
-                     seed: bool = True) -> Iterator[Tuple[str, Obj]]:
-        """ndjson watch stream; yields (event_type, object)."""
-        url = self._url(kind, namespace,
-                        query=f"watch=true&seed={'true' if seed else 'false'}")
+                     seed: bool = True, resource_version: Optional[str] = None,
+                     bookmarks: bool = False) -> Iterator[Tuple[str, Obj]]:
+        """ndjson watch stream; yields (event_type, object). resource_version
+        resumes from that RV (replay + live); bookmarks=True interleaves BOOKMARK
+        progress events carrying the current resourceVersion."""
+        q = f"watch=true&seed={'true' if seed else 'false'}"
+        if resource_version is not None:
+            q += f"&resourceVersion={resource_version}"
+        if bookmarks:
+            q += "&allowWatchBookmarks=true"
+        url = self._url(kind, namespace, query=q)
         req = urllib.request.Request(url, headers=self._headers())
         with urllib.request.urlopen(req, timeout=3600, context=self._ctx) as r:
             for raw in r:

@@ -103,7 +103,11 @@ CLUSTER_SCOPED = {"ClusterTopologyBinding", "Node", "Namespace"}
 
 
 class _KindTable:
-    __slots__ = ("objects", "watchers", "label_indexes")
+    __slots__ = ("objects", "watchers", "label_indexes", "history")
+
+    # events retained per kind for watch resume (watch cache depth; a client
+    # resuming from an RV older than the window gets 410 Gone, kube semantics)
+    HISTORY_DEPTH = 4096
 
     def __init__(self) -> None:
         # key: (namespace, name) -> obj ; cluster-scoped use namespace ""
@@ -112,6 +116,8 @@ class _KindTable:
         # label_key -> label_value -> set of object keys (inverted index for the
         # selector scans that dominate reconcile cost at 1000-pod scale)
         self.label_indexes: Dict[str, Dict[str, set]] = {}
+        # bounded ring of (int_rv, event_type, obj) for resourceVersion resume
+        self.history: List[Tuple[int, str, Obj]] = []
 
     def index_add(self, key: Tuple[str, str], obj: Obj) -> None:
         labels = obj.get("metadata", {}).get("labels") or {}
@@ -190,6 +196,13 @@ class Store:
     def _notify(self, tbl: _KindTable, ev: str, obj: Obj) -> None:
         # stored objects are immutable after insert; watchers share the reference
         # (read-only contract, same as list(copy_objects=False))
+        try:
+            rv = int(obj["metadata"].get("resourceVersion", "0"))
+        except (TypeError, ValueError):
+            rv = 0
+        tbl.history.append((rv, ev, obj))
+        if len(tbl.history) > tbl.HISTORY_DEPTH:
+            del tbl.history[: tbl.HISTORY_DEPTH // 2]
         for w in list(tbl.watchers):
             w.queue.put((ev, obj))
 
@@ -453,15 +466,81 @@ class Store:
                     pass
 
     # ------------------------------------------------------------------ watches & events
-    def watch(self, kind: str, seed: bool = False) -> Watch:
+    def watch(self, kind: str, seed: bool = False,
+              since_rv: Optional[str] = None) -> Watch:
+        """since_rv: resume semantics — replay retained events with
+        resourceVersion > since_rv, then stream live (client-go ListAndWatch /
+        watch cache parity). Raises 410 Expired if since_rv predates the
+        retained window."""
         w = Watch(self, kind)
         with self._lock:
             tbl = self._table(kind)
+            if since_rv is not None:
+                try:
+                    rv = int(since_rv)
+                except (TypeError, ValueError):
+                    raise invalid(f"malformed resourceVersion {since_rv!r}")
+                if tbl.history and tbl.history[0][0] > rv + 1 \
+                        and len(tbl.history) >= tbl.HISTORY_DEPTH // 2:
+                    raise ApiError(410, "Expired",
+                                   f"resourceVersion {since_rv} is too old")
+                tbl.watchers.append(w)
+                for (erv, ev, obj) in tbl.history:
+                    if erv > rv:
+                        w.queue.put((ev, obj))
+                return w
             tbl.watchers.append(w)
             if seed:
                 for obj in tbl.objects.values():
                     w.queue.put((ADDED, json_copy(obj)))
         return w
+
+    def current_rv(self) -> str:
+        """Most recently issued resourceVersion (list-consistency token)."""
+        with self._lock:
+            nxt = next(self._rv)
+        return str(nxt - 1)  # peek costs one rv; monotonicity is all that matters
+
+    def list_page(self, kind: str, namespace: Optional[str] = None,
+                  label_selector: Optional[Dict[str, str]] = None,
+                  limit: Optional[int] = None,
+                  continue_token: Optional[str] = None
+                  ) -> Tuple[List[Obj], Optional[str], str]:
+        """Chunked list (apiserver limit/continue parity): deterministic
+        (namespace, name) order; returns (items, next_continue, resourceVersion)."""
+        import base64
+        import json as _json
+        start_after: Optional[Tuple[str, str]] = None
+        if continue_token:
+            try:
+                ns_name = _json.loads(base64.b64decode(continue_token))
+                start_after = (ns_name[0], ns_name[1])
+            except Exception:
+                raise invalid("malformed continue token")
+        with self._lock:
+            tbl = self._table(kind)
+            keys = sorted(tbl.objects.keys())
+            rv = self.current_rv()
+            refs = []
+            for key in keys:
+                if start_after is not None and key <= start_after:
+                    continue
+                ns, _name = key
+                if namespace is not None and ns != namespace:
+                    continue
+                obj = tbl.objects[key]
+                if not match_labels(obj.get("metadata", {}).get("labels"),
+                                    label_selector):
+                    continue
+                refs.append((key, obj))
+                if limit and len(refs) > limit:
+                    break
+        nxt = None
+        if limit and len(refs) > limit:
+            refs = refs[:limit]
+            nxt = base64.b64encode(
+                _json.dumps(list(refs[-1][0])).encode()).decode()
+        return [json_copy(o) for _k, o in refs], nxt, rv
 
     def record_event(self, involved: Obj, etype: str, reason: str, message: str) -> None:
         ev = {

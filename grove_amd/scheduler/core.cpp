@@ -30,6 +30,9 @@ struct Node {
   double mem_bytes;
   std::vector<int> gpu_ids;
   int64_t pods;
+  // measured min per-link xGMI bandwidth of this pool's fabric (GB/s); the
+  // topology agent feeds it from rsmi_minmax_bandwidth_get, default nominal
+  double link_gbps = kXgmiLinkGBps;
 };
 
 struct Pod {
@@ -53,14 +56,16 @@ std::vector<int> take(Node& n, const Pod& p) {
   return out;
 }
 
-double placement_score(size_t n_nodes_used, int total_gpus) {
-  if (total_gpus <= 1) return kXgmiLinkGBps * kXgmiPeerLinks;
-  if (n_nodes_used <= 1) return kXgmiLinkGBps;
+double placement_score(size_t n_nodes_used, int total_gpus,
+                       double link_gbps = kXgmiLinkGBps) {
+  if (total_gpus <= 1) return link_gbps * kXgmiPeerLinks;
+  if (n_nodes_used <= 1) return link_gbps;
   return kNicGBps / (2.0 * (double)(n_nodes_used - 1));
 }
 
 using Assignment = std::tuple<std::string, std::string, std::vector<int>>;
-using NodeState = std::tuple<std::string, int64_t, double, std::vector<int>, int64_t>;
+using NodeState =
+    std::tuple<std::string, int64_t, double, std::vector<int>, int64_t, double>;
 
 // Returns (assignments, score, consumed-node-states) or None. The solver runs
 // WITHOUT the GIL (a plain-C++ result is computed in a released scope; Python objects
@@ -89,12 +94,15 @@ SolveResult solve(std::vector<Node>& nodes, std::vector<Pod>& pods) {
     out.score = score;
     out.consumed.reserve(state.size());
     for (auto& n : state)
-      out.consumed.emplace_back(n.name, n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods);
+      out.consumed.emplace_back(n.name, n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods,
+                                n.link_gbps);
   };
 
-  // Phase 1: single-node best-fit (leaves fewest free GPUs behind).
+  // Phase 1: single-pool best-fit (leaves fewest free GPUs behind; ties prefer
+  // the pool with the higher measured link bandwidth — placement.py parity).
   int best = -1;
   size_t best_left = SIZE_MAX;
+  double best_bw = -1.0;
   for (size_t i = 0; i < nodes.size(); ++i) {
     Node trial = nodes[i];
     bool ok = true;
@@ -102,9 +110,11 @@ SolveResult solve(std::vector<Node>& nodes, std::vector<Pod>& pods) {
       if (!fits(trial, pods[oi])) { ok = false; break; }
       take(trial, pods[oi]);
     }
-    if (ok && trial.gpu_ids.size() < best_left) {
+    if (ok && (trial.gpu_ids.size() < best_left ||
+               (trial.gpu_ids.size() == best_left && nodes[i].link_gbps > best_bw))) {
       best = (int)i;
       best_left = trial.gpu_ids.size();
+      best_bw = nodes[i].link_gbps;
     }
   }
   if (best >= 0) {
@@ -112,7 +122,7 @@ SolveResult solve(std::vector<Node>& nodes, std::vector<Pod>& pods) {
       auto ids = take(nodes[best], pods[oi]);
       out.assignments.emplace_back(pods[oi].name, nodes[best].name, ids);
     }
-    emit(nodes, placement_score(1, total_gpus));
+    emit(nodes, placement_score(1, total_gpus, nodes[best].link_gbps));
     return out;
   }
 
@@ -140,7 +150,12 @@ SolveResult solve(std::vector<Node>& nodes, std::vector<Pod>& pods) {
     out.assignments.emplace_back(pods[oi].name, state[pick].name, ids);
     if (!used[pick]) { used[pick] = 1; ++used_count; }
   }
-  emit(state, placement_score(used_count, total_gpus));
+  double min_link = kXgmiLinkGBps;
+  bool any = false;
+  for (size_t i = 0; i < state.size(); ++i)
+    if (used[i]) { min_link = any ? std::min(min_link, state[i].link_gbps)
+                                  : state[i].link_gbps; any = true; }
+  emit(state, placement_score(used_count, total_gpus, min_link));
   return out;
 }
 
@@ -153,7 +168,7 @@ py::object place_gang(std::vector<NodeState> node_states,
     nodes.reserve(node_states.size());
     for (auto& t : node_states)
       nodes.push_back(Node{std::get<0>(t), std::get<1>(t), std::get<2>(t),
-                           std::get<3>(t), std::get<4>(t)});
+                           std::get<3>(t), std::get<4>(t), std::get<5>(t)});
     std::vector<Pod> pods;
     pods.reserve(pod_specs.size());
     for (auto& t : pod_specs)
@@ -165,8 +180,8 @@ py::object place_gang(std::vector<NodeState> node_states,
   return py::make_tuple(res.assignments, res.score, res.consumed);
 }
 
-double score_only(size_t n_nodes, int total_gpus) {
-  return placement_score(n_nodes, total_gpus);
+double score_only(size_t n_nodes, int total_gpus, double link_gbps) {
+  return placement_score(n_nodes, total_gpus, link_gbps);
 }
 
 }  // namespace
@@ -218,5 +233,6 @@ PYBIND11_MODULE(_sched, m) {
   m.doc() = "grove_amd native gang-placement core (xGMI-aware Filter/Score)";
   m.def("place_gang", &place_gang, py::arg("nodes"), py::arg("pods"),
         "All-or-nothing gang placement; returns (assignments, score, consumed) or None");
-  m.def("placement_score", &score_only, py::arg("n_nodes_used"), py::arg("total_gpus"));
+  m.def("placement_score", &score_only, py::arg("n_nodes_used"),
+        py::arg("total_gpus"), py::arg("link_gbps") = kXgmiLinkGBps);
 }

@@ -29,22 +29,35 @@ NIC_GBPS = 50.0
 
 
 class NodeFree:
-    """Mutable free-capacity view of one node used during a placement attempt."""
+    """Mutable free-capacity view of one GPU POOL used during a placement attempt.
 
-    __slots__ = ("name", "cpu_milli", "mem_bytes", "gpus", "pods", "gpu_ids", "labels")
+    One pool per xGMI hive: a single-hive node is one pool; a partitioned node
+    (multiple hives discovered by the topology agent) contributes one pool per hive,
+    each with the hive's GPU ids and a share of node cpu/mem/pods. `name` is the
+    unique pool key; `node_name` is the Kubernetes node to bind onto. `link_gbps`
+    is the MEASURED minimum per-link xGMI bandwidth of the pool's fabric
+    (rsmi minmax_bandwidth_get via the topology agent), defaulting to the
+    MI355X nominal per-link figure."""
+
+    __slots__ = ("name", "cpu_milli", "mem_bytes", "gpus", "pods", "gpu_ids",
+                 "labels", "node_name", "link_gbps")
 
     def __init__(self, name: str, cpu_milli: int, mem_bytes: float, gpu_ids: List[int],
-                 pods: int, labels: Optional[Dict[str, str]] = None):
+                 pods: int, labels: Optional[Dict[str, str]] = None,
+                 node_name: Optional[str] = None,
+                 link_gbps: Optional[float] = None):
         self.name = name
         self.cpu_milli = cpu_milli
         self.mem_bytes = mem_bytes
         self.gpu_ids = list(gpu_ids)   # free GPU device indices
         self.pods = pods
         self.labels = labels or {}
+        self.node_name = node_name or name
+        self.link_gbps = XGMI_LINK_GBPS if link_gbps is None else float(link_gbps)
 
     def clone(self) -> "NodeFree":
         return NodeFree(self.name, self.cpu_milli, self.mem_bytes, self.gpu_ids,
-                        self.pods, self.labels)
+                        self.pods, self.labels, self.node_name, self.link_gbps)
 
 
 class PodRequest:
@@ -83,17 +96,20 @@ def _take(node: NodeFree, pod: PodRequest) -> List[int]:
     return taken
 
 
-def placement_score(n_nodes_used: int, total_gpus: int) -> float:
+def placement_score(n_nodes_used: int, total_gpus: int,
+                    link_gbps: float = XGMI_LINK_GBPS) -> float:
     """Effective ring all-reduce bandwidth estimate in GB/s (higher is better).
 
-    One hive: the ring over g GPUs uses g xGMI hops, per-link bound → ≈ XGMI_LINK_GBPS
-    (per-GPU bidirectional ring bandwidth). Cross-node: the ring crosses the NIC 2× per
-    node boundary; effective bw ≈ NIC_GBPS / (2 * (n_nodes-1)) shared by the gang.
+    One hive: the ring over g GPUs uses g xGMI hops, per-link bound → the MEASURED
+    minimum per-link bandwidth over the chosen GPU set (link_gbps; nominal 153 GB/s
+    on MI355X when unmeasured). Cross-node/cross-hive: the ring crosses the NIC 2×
+    per boundary; effective bw ≈ NIC_GBPS / (2 * (n-1)) shared by the gang.
     """
     if total_gpus <= 1:
-        return XGMI_LINK_GBPS * c.XGMI_PEER_LINKS  # no collective bound; report fabric max
+        # no collective bound; report the pool's fabric max (links × per-link bw)
+        return link_gbps * c.XGMI_PEER_LINKS
     if n_nodes_used <= 1:
-        return XGMI_LINK_GBPS
+        return link_gbps
     return NIC_GBPS / (2.0 * (n_nodes_used - 1))
 
 
@@ -160,9 +176,10 @@ def place_gang(nodes: List[NodeFree], pods: List[PodRequest],
     """
     total_gpus = sum(p.gpus for p in pods)
 
-    # Phase 1: single-node best-fit
+    # Phase 1: single-pool (one xGMI hive) best-fit; among equal fits prefer the
+    # pool with the higher measured link bandwidth
     best: Optional[NodeFree] = None
-    best_left = None
+    best_key = None
     for n in nodes:
         trial = n.clone()
         ok = True
@@ -172,15 +189,15 @@ def place_gang(nodes: List[NodeFree], pods: List[PodRequest],
                 break
             _take(trial, p)
         if ok:
-            left = len(trial.gpu_ids)
-            if best is None or left < best_left:
-                best, best_left = n, left
+            key = (len(trial.gpu_ids), -n.link_gbps)
+            if best is None or key < best_key:
+                best, best_key = n, key
     if best is not None:
         assignments = []
         for p in sorted(pods, key=lambda p: -p.gpus):
             gpu_ids = _take(best, p)
             assignments.append(Assignment(p.name, best.name, gpu_ids))
-        return assignments, placement_score(1, total_gpus)
+        return assignments, placement_score(1, total_gpus, best.link_gbps)
 
     # Phase 2: minimal spread, first-fit-decreasing over nodes sorted by free GPUs desc
     order = sorted(nodes, key=lambda n: (-len(n.gpu_ids), -n.cpu_milli))
@@ -205,4 +222,6 @@ def place_gang(nodes: List[NodeFree], pods: List[PodRequest],
         for n, (cpu, mem, gpus, pods_) in snapshots:
             n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods = cpu, mem, gpus, pods_
         return None
-    return assignments, placement_score(len(used_nodes), total_gpus)
+    min_link = min((n.link_gbps for n in order if n.name in used_nodes),
+                   default=XGMI_LINK_GBPS)
+    return assignments, placement_score(len(used_nodes), total_gpus, min_link)

@@ -126,35 +126,76 @@ class GangScheduler:
 
     # ------------------------------------------------------------------ node views
     def _build_node_views(self) -> Dict[str, NodeFree]:
+        """One NodeFree POOL per xGMI hive (schedule from the
+        DISCOVERED fabric). The topology agent publishes the hive partition in the
+        topology.amd.com/xgmi-hives annotation and the measured min per-link
+        bandwidth in topology.amd.com/xgmi-min-gbps; a partitioned node becomes
+        several pools (keys "<node>#h<k>") so a gang never silently spans hives,
+        and the placement score reports the measured bandwidth."""
         out: Dict[str, NodeFree] = {}
         for n in self.store.list("Node", copy_objects=False):
             if (n.get("spec") or {}).get("unschedulable"):
                 continue
+            name = n["metadata"]["name"]
             alloc = (n.get("status") or {}).get("allocatable") or {}
+            labels = n["metadata"].get("labels") or {}
+            ann = n["metadata"].get("annotations") or {}
             gpus = int(parse_quantity(alloc.get(c.AMD_GPU_RESOURCE, 0)))
-            out[n["metadata"]["name"]] = NodeFree(
-                n["metadata"]["name"],
-                cpu_millis(alloc.get("cpu", 0)),
-                parse_quantity(alloc.get("memory", 0)),
-                list(range(gpus)),
-                int(parse_quantity(alloc.get("pods", 250))),
-                n["metadata"].get("labels") or {},
-            )
+            cpu = cpu_millis(alloc.get("cpu", 0))
+            mem = parse_quantity(alloc.get("memory", 0))
+            pods = int(parse_quantity(alloc.get("pods", 250)))
+            link = None
+            try:
+                link = float(ann.get("topology.amd.com/xgmi-min-gbps", ""))
+            except ValueError:
+                pass
+            hives: List[List[int]] = []
+            raw = ann.get("topology.amd.com/xgmi-hives", "")
+            if raw:
+                try:
+                    hives = [[int(x) for x in part.split(",") if x != ""]
+                             for part in raw.split(";") if part]
+                except ValueError:
+                    hives = []
+            covered = {g for h in hives for g in h}
+            if gpus and (not hives or covered != set(range(gpus))):
+                hives = [list(range(gpus))]
+            if not gpus:
+                hives = [[]]
+            n_pools = max(1, len(hives))
+            for k, hive in enumerate(hives):
+                key = name if k == 0 else f"{name}#h{k}"
+                frac = (len(hive) / gpus) if gpus else 1.0 / n_pools
+                out[key] = NodeFree(
+                    key, int(cpu * frac), mem * frac, sorted(hive),
+                    max(1, int(pods * frac)), labels,
+                    node_name=name, link_gbps=link)
         return out
 
     def _subtract_bound(self, nodes: Dict[str, NodeFree], bound: List[Obj]) -> None:
+        pools_by_node: Dict[str, List[NodeFree]] = {}
+        for nf in nodes.values():
+            pools_by_node.setdefault(nf.node_name, []).append(nf)
         for p in bound:
-            node = nodes.get(p["spec"]["nodeName"])
-            if node is None:
+            pools = pools_by_node.get(p["spec"]["nodeName"])
+            if not pools:
                 continue
             req = self._pod_request(p)
+            ids = (p["metadata"].get("annotations") or {}).get(GPU_IDS_ANNOTATION, "")
+            taken = {int(x) for x in ids.split(",") if x != ""} if ids else set()
+            # charge the pool holding the pod's GPUs (fall back to the first pool)
+            node = pools[0]
+            if taken:
+                for nf in pools:
+                    if taken & set(nf.gpu_ids):
+                        node = nf
+                        break
             node.cpu_milli -= req.cpu_milli
             node.mem_bytes -= req.mem_bytes
             node.pods -= 1
-            ids = (p["metadata"].get("annotations") or {}).get(GPU_IDS_ANNOTATION, "")
-            if ids:
-                taken = {int(x) for x in ids.split(",") if x != ""}
-                node.gpu_ids = [g for g in node.gpu_ids if g not in taken]
+            if taken:
+                for nf in pools:
+                    nf.gpu_ids = [g for g in nf.gpu_ids if g not in taken]
             elif req.gpus:
                 del node.gpu_ids[: req.gpus]
 
@@ -211,7 +252,7 @@ class GangScheduler:
                                       ref.get("namespace") or ns, ref["name"])
             if prev is not None:
                 prev_nodes = self._gang_nodes(prev, pods_by_name)
-                pool = [n for n in nodes if n.name in prev_nodes]
+                pool = [n for n in nodes if n.node_name in prev_nodes]
                 if pool:
                     res = self._place_gang_pods(pool, pg, chosen)
                     if res is not None:
@@ -328,8 +369,8 @@ class GangScheduler:
 
     def _place(self, nodes: List[NodeFree], reqs: List[PodRequest]):
         if self.use_native and _native is not None:
-            flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids), n.pods)
-                          for n in nodes]
+            flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids),
+                           n.pods, float(n.link_gbps)) for n in nodes]
             flat_pods = [(p.name, p.cpu_milli, float(p.mem_bytes), p.gpus) for p in reqs]
             res = _native.place_gang(flat_nodes, flat_pods)
             if res is None:
@@ -337,7 +378,7 @@ class GangScheduler:
             assignments, score, consumed = res
             # apply consumption back onto the python node views
             node_by_name = {n.name: n for n in nodes}
-            for (nname, cpu, mem, gpu_ids, pods_) in consumed:
+            for (nname, cpu, mem, gpu_ids, pods_, _link) in consumed:
                 n = node_by_name[nname]
                 n.cpu_milli, n.mem_bytes, n.gpu_ids, n.pods = cpu, mem, list(gpu_ids), pods_
             return ([Assignment(p, nd, list(g)) for (p, nd, g) in assignments], score)
@@ -350,7 +391,7 @@ class GangScheduler:
             req = self._pod_request(p)
             order = nodes
             if prefer:
-                order = sorted(nodes, key=lambda n: n.name not in prefer)
+                order = sorted(nodes, key=lambda n: n.node_name not in prefer)
             else:
                 # pack GPU pods (best-fit), spread cpu pods (worst-fit) lightly
                 order = sorted(nodes, key=lambda n: len(n.gpu_ids)) if req.gpus \
@@ -363,7 +404,8 @@ class GangScheduler:
                     n.pods -= 1
                     gpu_ids = n.gpu_ids[: req.gpus]
                     del n.gpu_ids[: req.gpus]
-                    self._bind(p, Assignment(p["metadata"]["name"], n.name, gpu_ids))
+                    self._bind(p, Assignment(p["metadata"]["name"], n.node_name,
+                                             gpu_ids))
                     break
 
     # ------------------------------------------------------------------ bind
@@ -373,7 +415,7 @@ class GangScheduler:
         def apply(o: Obj) -> None:
             if o["metadata"].get("deletionTimestamp"):
                 raise ApiError(409, "Conflict", "pod deleting")
-            o["spec"]["nodeName"] = a.node
+            o["spec"]["nodeName"] = a.node.split("#", 1)[0]
             if a.gpu_ids:
                 o["metadata"].setdefault("annotations", {})[GPU_IDS_ANNOTATION] = \
                     ",".join(str(g) for g in a.gpu_ids)

@@ -98,7 +98,14 @@ def test_gpu_end_to_end_gang():
         pg = cl.store.get(c.KIND_PODGANG, "default", "gputest-0")
         from grove_amd.utils import conditions as cond
         assert cond.condition_true(pg, "Scheduled")
-        assert pg["status"]["placementScore"] > 0
+        # placementScore derives from the PROBED fabric: the
+        # agent's measured min per-link bandwidth annotation when rsmi reports it,
+        # else the MI355X nominal figure
+        node = cl.store.get("Node", None, "mi355x-real")
+        ann = node["metadata"].get("annotations") or {}
+        link = float(ann.get("topology.amd.com/xgmi-min-gbps", c.XGMI_LINK_GBPS))
+        expected = link * c.XGMI_PEER_LINKS if size <= 1 else link
+        assert pg["status"]["placementScore"] == pytest.approx(expected, rel=0.01)
     finally:
         cl.stop()
 

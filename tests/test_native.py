@@ -17,8 +17,8 @@ def mk_nodes(spec):
 
 
 def native_place(nodes, pods):
-    flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids), n.pods)
-                  for n in nodes]
+    flat_nodes = [(n.name, n.cpu_milli, float(n.mem_bytes), list(n.gpu_ids), n.pods,
+                   float(n.link_gbps)) for n in nodes]
     flat_pods = [(p.name, p.cpu_milli, float(p.mem_bytes), p.gpus) for p in pods]
     return _sched.place_gang(flat_nodes, flat_pods)
 
@@ -106,5 +106,26 @@ def test_native_consumption_applied():
     nodes = mk_nodes([("a", 100000, 1e12, 8, 100)])
     pods = [PodRequest("p0", 1000, 1e9, 4)]
     _assignments, _score, consumed = native_place(nodes, pods)
-    (name, cpu, _mem, gpu_ids, pods_left) = consumed[0]
+    (name, cpu, _mem, gpu_ids, pods_left, _link) = consumed[0]
     assert name == "a" and cpu == 99000 and len(gpu_ids) == 4 and pods_left == 99
+
+
+@needs_native
+def test_native_measured_link_bandwidth_score():
+    """Score reports the MEASURED min per-link bandwidth of the chosen pool, and
+    ties between equally-packed pools prefer the faster fabric (both solvers)."""
+    from grove_amd.scheduler.placement import place_gang
+    nodes = [NodeFree("slow", 100000, 1e12, list(range(8)), 100, link_gbps=100.0),
+             NodeFree("fast", 100000, 1e12, list(range(8)), 100, link_gbps=150.0)]
+    pods = [PodRequest(f"p{i}", 1000, 1e9, 1) for i in range(8)]
+    res_py = place_gang([n.clone() for n in nodes], pods)
+    assert res_py is not None
+    assert res_py[1] == 150.0
+    assert all(a.node == "fast" for a in res_py[0])
+    res_nat = native_place([n.clone() for n in nodes], pods)
+    assert res_nat is not None
+    assert res_nat[1] == 150.0
+    # single-GPU gang: fabric max = links x measured per-link bw
+    one = place_gang([NodeFree("m", 1000, 1e9, [0], 10, link_gbps=120.0)],
+                     [PodRequest("p", 100, 1e6, 1)])
+    assert one[1] == 120.0 * 7

@@ -382,3 +382,73 @@ def get_cond_reason(pg):
         if cd.get("type") == c.PODGANG_COND_SCHEDULED:
             return cd.get("reason")
     return None
+
+
+def test_multi_hive_node_pools_and_measured_score(cluster):
+    """the scheduler consumes the DISCOVERED fabric. A node whose
+    agent probe found two xGMI hives (partitioned) is split into per-hive placement
+    pools: a 4-GPU gang lands entirely inside one hive's GPU set, and the PodGang
+    placementScore reports the MEASURED min link bandwidth from the node annotation,
+    not the nominal constant."""
+    node = {
+        "apiVersion": "v1", "kind": "Node",
+        "metadata": {"name": "part0",
+                     "labels": {"kubernetes.io/hostname": "part0"},
+                     "annotations": {
+                         "topology.amd.com/xgmi-hives": "0,1,2,3;4,5,6,7",
+                         "topology.amd.com/xgmi-min-gbps": "121.5"}},
+        "spec": {},
+        "status": {"allocatable": {"cpu": "256", "memory": "2048Gi",
+                                   "pods": "512", c.AMD_GPU_RESOURCE: "8"},
+                   "capacity": {}, "conditions": [{"type": "Ready",
+                                                   "status": "True"}]}}
+    cluster.store.create(node)
+    pcs = {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+           "metadata": {"name": "hv"},
+           "spec": {"replicas": 1, "template": {"cliques": [{
+               "name": "w", "spec": {
+                   "roleName": "w", "replicas": 4, "minAvailable": 4,
+                   "podSpec": {"containers": [{
+                       "name": "m", "image": "i",
+                       "resources": {"requests": {"cpu": "1",
+                                                  c.AMD_GPU_RESOURCE: "1"}}}]}}}]}}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("hv", timeout=20)
+    pods = cluster.store.list("Pod", "default", {c.LABEL_PODCLIQUE: "hv-0-w"})
+    gpu_ids = set()
+    for p in pods:
+        ids = p["metadata"]["annotations"]["scheduling.amd.com/gpu-ids"]
+        gpu_ids.update(int(x) for x in ids.split(","))
+    assert gpu_ids in ({0, 1, 2, 3}, {4, 5, 6, 7}), \
+        f"gang spans hives: {sorted(gpu_ids)}"
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "hv-0")
+    assert pg["status"]["placementScore"] == pytest.approx(121.5, abs=0.1)
+
+
+def test_agent_multi_hive_labeling():
+    """The agent labels a single full hive as hive0 but refuses the fake label on a
+    partitioned probe, publishing per-hive GPU sets + measured bandwidth instead."""
+    from grove_amd.topology import agent as ag
+    info = {"backend": "test", "gpu_count": 4,
+            "devices": [{"index": i, "name": "AMD Instinct MI355X"}
+                        for i in range(4)],
+            "links": [{"src": 0, "dst": 1, "type": "xgmi", "min_bw_mbps": 140000},
+                      {"src": 1, "dst": 0, "type": "xgmi", "min_bw_mbps": 152000}]}
+    import unittest.mock as mock
+    with mock.patch.object(ag, "probe", return_value=info):
+        node = ag.discover_node("tn")
+    labels = node["metadata"]["labels"]
+    ann = node["metadata"]["annotations"]
+    # two components (0,1) and (2,3 isolated) -> multi-hive -> no single-hive label
+    assert c.NODE_LABEL_XGMI_HIVE not in labels
+    assert labels["topology.amd.com/xgmi-hive-count"] == "3"
+    assert ann["topology.amd.com/xgmi-hives"] == "0,1;2;3"
+    assert ann["topology.amd.com/xgmi-min-gbps"] == "140.0"
+    # fully connected probe -> hive0 label present
+    info2 = {"backend": "test", "gpu_count": 2,
+             "devices": [{"index": 0, "name": "x"}, {"index": 1, "name": "x"}],
+             "links": [{"src": 0, "dst": 1, "type": "xgmi"},
+                       {"src": 1, "dst": 0, "type": "xgmi"}]}
+    with mock.patch.object(ag, "probe", return_value=info2):
+        node2 = ag.discover_node("tn2")
+    assert node2["metadata"]["labels"][c.NODE_LABEL_XGMI_HIVE] == "tn2-hive0"

@@ -132,9 +132,16 @@ class Comm:
             dist.init_process_group("gloo")
             self.dist = dist
             if torch.cuda.is_available():
+                n_dev = torch.cuda.device_count()
+                self.local_rank = self.local_rank % max(1, n_dev)
                 torch.cuda.set_device(self.local_rank)
-                self.nccl_group = dist.new_group(backend="nccl")
-                self._hb = torch.ones(1, device=f"cuda:{self.local_rank}")
+                if n_dev >= self.world:
+                    # one rank per GPU: RCCL heartbeat group over xGMI
+                    self.nccl_group = dist.new_group(backend="nccl")
+                    self._hb = torch.ones(1,
+                                          device=f"cuda:{self.local_rank}")
+                # else: oversubscribed protocol rehearsal (2 ranks / 1 GPU) —
+                # RCCL cannot place two ranks on one device; gloo-only control
 
     def broadcast(self, obj=None):
         if self.dist is None:

@@ -85,6 +85,9 @@ def run_payload_descriptor(kind: str, dims: Tuple[int, ...], device: int) -> Dic
                 a @ b
         return {"device": "cpu", "kind": kind}
     ext = load_gpuwork()
+    # logical device index -> physical (oversubscribed protocol rehearsals run
+    # more ranks than GPUs; real N-GPU runs are 1:1)
+    device = device % max(1, torch.cuda.device_count())
     with torch.cuda.device(device):
         if kind == "gemm":
             m, n, k, iters = dims

@@ -315,11 +315,15 @@ class Store:
             refs = [o for o in refs if filter_fn(o)]
         return [json_copy(o) for o in refs] if copy_objects else refs
 
-    def _apply_update(self, obj: Obj, status_only: bool) -> Obj:
+    def _apply_update(self, obj: Obj, status_only: bool,
+                      owned: bool = False) -> Obj:
         """Lock-split optimistic update: the expensive work (copies, admission,
         equality checks) runs OUTSIDE the lock against the immutable current object;
-        insertion re-checks identity under the lock and retries on interleaving."""
-        obj = json_copy(obj)
+        insertion re-checks identity under the lock and retries on interleaving.
+        owned=True promises obj is a private copy the store may keep (patch()
+        fast-path — skips one full deep copy per write)."""
+        if not owned:
+            obj = json_copy(obj)
         kind = obj["kind"]
         m = meta(obj)
         ns = self._ns_of(kind, m)
@@ -386,13 +390,19 @@ class Store:
 
     def patch(self, kind: str, namespace: Optional[str], name: str,
               fn: Callable[[Obj], None], status: bool = False, retries: int = 10) -> Obj:
-        """Optimistic-concurrency retry loop: get → fn(obj) → update."""
+        """Optimistic-concurrency retry loop: get → fn(obj) → update.
+
+        The working copy is handed to _apply_update as `owned` (it was copied from
+        the immutable stored object here, and fn sees only the copy), saving one
+        full deep copy per patch — the store's hottest write path at 1000-pod
+        scale."""
         last: Optional[ApiError] = None
         for _ in range(retries):
-            obj = self.get(kind, namespace, name)
+            cur = self.get(kind, namespace, name, copy=False)
+            obj = json_copy(cur)
             fn(obj)
             try:
-                return self.update_status(obj) if status else self.update(obj)
+                return self._apply_update(obj, status_only=status, owned=True)
             except ApiError as e:
                 if e.reason != "Conflict":
                     raise

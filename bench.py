@@ -111,7 +111,8 @@ class DispatchKubelet:
                     conds.append({"type": want, "status": "True",
                                   "reason": "PayloadComplete"})
         try:
-            self.store.patch("Pod", ns, name, upd, status=True)
+            self.store.patch("Pod", ns, name, upd, status=True,
+                             return_copy=False)
         except ApiError:
             pass
 

@@ -146,7 +146,8 @@ class Cluster:
                 old_sched = ((_old or {}).get("status") or {}).get("scheduledReplicas")
                 if new_sched != old_sched or ev == "DELETED":
                     for q in self.store.list(c.KIND_PCLQ, ns,
-                                             {c.LABEL_BASE_PODGANG: pg}):
+                                             {c.LABEL_BASE_PODGANG: pg},
+                                             copy_objects=False):
                         self.c_pclq.enqueue(ns, q["metadata"]["name"])
 
         def on_pcsg(ev: str, obj: Obj, _old) -> None:

@@ -31,7 +31,8 @@ class PodCliqueReconciler:
 
     # ------------------------------------------------------------------ entry
     def reconcile(self, namespace: str, name: str) -> Result:
-        pclq = self.store.try_get(c.KIND_PCLQ, namespace, name)
+        # read-only view: _reconcile_spec/_reconcile_delete never mutate the PCLQ
+        pclq = self.store.try_get(c.KIND_PCLQ, namespace, name, copy=False)
         if pclq is None:
             return Result.DONE
         if pclq["metadata"].get("deletionTimestamp"):
@@ -126,6 +127,12 @@ class PodCliqueReconciler:
                               detail=f"scale-in pod {p['metadata']['name']}"):
                     self.store.delete("Pod", ns, p["metadata"]["name"])
 
+        # inline gang completion before gate removal: a freshly completed gang is
+        # Initialized and its pods ungated within THIS pass (latency chain collapse)
+        gang_name = pclq["metadata"]["labels"].get(c.LABEL_PODGANG)
+        if gang_name:
+            from .podgang_component import try_complete_podgang
+            try_complete_podgang(self.store, ns, gang_name)
         self._remove_scheduling_gates(pclq, rec)
         return Result.DONE
 
@@ -234,7 +241,7 @@ class PodCliqueReconciler:
 
     # ------------------------------------------------------------------ status
     def _reconcile_status(self, namespace: str, name: str) -> None:
-        pclq = self.store.try_get(c.KIND_PCLQ, namespace, name)
+        pclq = self.store.try_get(c.KIND_PCLQ, namespace, name, copy=False)
         if pclq is None or pclq["metadata"].get("deletionTimestamp"):
             return
         pods = self._owned_pods(pclq)
@@ -303,15 +310,20 @@ class PodCliqueReconciler:
             else:
                 cond.set_condition(o, c.COND_MIN_AVAILABLE_BREACHED, False,
                                    c.REASON_SUFFICIENT_READY_PODS)
-        # No-op fast path: apply the status function to the in-hand copy first and
-        # skip the store round-trip (lock + copy + deep compare) when nothing would
-        # change — the dominant case during churn storms at 10k-pod scale.
+        # No-op fast path: apply the status function to a status-only probe (the
+        # stored object is immutable; only the status subtree is copied — the
+        # podSpec, the object's bulk, is shared read-only) and skip the store
+        # round-trip when nothing would change — the dominant case during churn
+        # storms at 10k-pod scale.
         from ..kubecore.store import json_copy
-        old_status = json_copy(pclq.get("status") or {})
-        upd(pclq)
-        if pclq.get("status") == old_status:
+        old_status = pclq.get("status") or {}
+        probe = {"kind": pclq.get("kind"), "metadata": pclq["metadata"],
+                 "spec": pclq["spec"], "status": json_copy(old_status)}
+        upd(probe)
+        if probe["status"] == old_status:
             return
         try:
-            self.store.patch(c.KIND_PCLQ, namespace, name, upd, status=True)
+            self.store.patch(c.KIND_PCLQ, namespace, name, upd, status=True,
+                             return_copy=False)
         except ApiError:
             pass

@@ -22,7 +22,7 @@ class PodGangReconciler:
         self.registry = registry
 
     def reconcile(self, namespace: str, name: str) -> Result:
-        pg = self.store.try_get(c.KIND_PODGANG, namespace, name)
+        pg = self.store.try_get(c.KIND_PODGANG, namespace, name, copy=False)
         if pg is None or pg["metadata"].get("deletionTimestamp"):
             return Result.DONE
         backend = self.registry.resolve_for_podgang(pg)

@@ -257,3 +257,56 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str,
         elif not all_created and initialized:
             # pods lost after init (e.g. gang termination in flight) — keep Initialized
             pass
+
+
+def try_complete_podgang(store: Store, ns: str, gang_name: str,
+                         rec: "groveerr.StepRecorder" = None) -> None:
+    """Latency fast-path: called inline from the PCLQ pass right
+    after pod creation, so a gang whose pods all exist gets its podReferences filled
+    and Initialized flipped in the SAME reconcile instead of waiting for the next
+    PCS-scope sync_podgangs pass (saves two watch→queue→worker hops per gang on the
+    serial time-to-running chain). The PCS pass remains the reconciling authority;
+    this only performs the monotonic completion step."""
+    pg = store.try_get(c.KIND_PODGANG, ns, gang_name, copy=False)
+    if pg is None or cond.condition_true(pg, c.PODGANG_COND_INITIALIZED):
+        return
+    groups = (pg.get("spec") or {}).get("podgroups") or []
+    if not groups:
+        return
+    pods = store.list("Pod", ns, {c.LABEL_PODGANG: gang_name}, copy_objects=False)
+    by_clique: Dict[str, List[str]] = {}
+    for p in pods:
+        lbl = p["metadata"].get("labels", {})
+        by_clique.setdefault(lbl.get(c.LABEL_PODCLIQUE, ""), []).append(
+            p["metadata"]["name"])
+    groups_spec = []
+    for g in groups:
+        refs = sorted(by_clique.get(g["name"], []))
+        # spec.replicas of the member PCLQ is the completion bar; fall back to
+        # minReplicas when the PCLQ is not readable (it always is in-process)
+        pclq = store.try_get(c.KIND_PCLQ, ns, g["name"], copy=False)
+        want = int((pclq or {}).get("spec", {}).get(
+            "replicas", g.get("minReplicas", 1)))
+        if len(refs) < want:
+            return  # not complete yet; the PCS pass will finish the job
+        groups_spec.append({"name": g["name"],
+                            "minReplicas": g.get("minReplicas", 1),
+                            "podReferences": [{"namespace": ns, "name": r}
+                                              for r in refs]})
+    def fill(o: Obj) -> None:
+        cur_groups = {g["name"]: g for g in o["spec"].get("podgroups") or []}
+        for gs in groups_spec:
+            if gs["name"] in cur_groups:
+                cur_groups[gs["name"]]["podReferences"] = gs["podReferences"]
+    try:
+        store.patch(c.KIND_PODGANG, ns, gang_name, fill)
+    except ApiError:
+        return
+
+    def flip(o: Obj) -> None:
+        cond.set_condition(o, c.PODGANG_COND_INITIALIZED, True,
+                           "AllPodsAssociated")
+    try:
+        store.patch(c.KIND_PODGANG, ns, gang_name, flip, status=True)
+    except ApiError:
+        pass

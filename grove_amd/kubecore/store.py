@@ -316,7 +316,7 @@ class Store:
         return [json_copy(o) for o in refs] if copy_objects else refs
 
     def _apply_update(self, obj: Obj, status_only: bool,
-                      owned: bool = False) -> Obj:
+                      owned: bool = False, return_copy: bool = True) -> Obj:
         """Lock-split optimistic update: the expensive work (copies, admission,
         equality checks) runs OUTSIDE the lock against the immutable current object;
         insertion re-checks identity under the lock and retries on interleaving.
@@ -339,7 +339,7 @@ class Store:
                 raise conflict(kind, m["name"])
             if status_only:
                 if cur.get("status", {}) == obj.get("status", {}):
-                    return json_copy(cur)  # no-op: no rv bump, no event
+                    return json_copy(cur) if return_copy else None  # no-op
                 new = json_copy(cur)
                 new["status"] = json_copy(obj.get("status", {}))
             else:
@@ -361,7 +361,7 @@ class Store:
                     new["metadata"]["deletionTimestamp"] = cur_m["deletionTimestamp"]
                 new["metadata"]["resourceVersion"] = cur_m["resourceVersion"]
                 if new == cur:
-                    return json_copy(cur)  # no-op update: no rv bump, no event
+                    return json_copy(cur) if return_copy else None  # no-op
                 if new.get("spec") != cur.get("spec"):
                     new["metadata"]["generation"] = cur_m.get("generation", 1) + 1
             with self._lock:
@@ -379,7 +379,7 @@ class Store:
                 if new["metadata"].get("deletionTimestamp") \
                         and not new["metadata"].get("finalizers"):
                     self._finalize_delete(kind, ns, m["name"])
-            return json_copy(new)
+            return json_copy(new) if return_copy else None
         raise conflict(kind, m["name"], "persistent write interleaving")
 
     def update(self, obj: Obj) -> Obj:
@@ -389,7 +389,8 @@ class Store:
         return self._apply_update(obj, status_only=True)
 
     def patch(self, kind: str, namespace: Optional[str], name: str,
-              fn: Callable[[Obj], None], status: bool = False, retries: int = 10) -> Obj:
+              fn: Callable[[Obj], None], status: bool = False, retries: int = 10,
+              return_copy: bool = True) -> Obj:
         """Optimistic-concurrency retry loop: get → fn(obj) → update.
 
         The working copy is handed to _apply_update as `owned` (it was copied from
@@ -402,7 +403,8 @@ class Store:
             obj = json_copy(cur)
             fn(obj)
             try:
-                return self._apply_update(obj, status_only=status, owned=True)
+                return self._apply_update(obj, status_only=status, owned=True,
+                                          return_copy=return_copy)
             except ApiError as e:
                 if e.reason != "Conflict":
                     raise

@@ -97,11 +97,15 @@ class VirtualKubelet:
                     log.exception("pod payload failed for %s", key)
                     self._fail(namespace, name, "PayloadFailed")
                     return Result.DONE
+            if self.ready_latency_s <= 0:
+                # zero-latency fast path: Running + Ready in ONE status patch
+                # (halves the per-pod patch count on the serial latency chain)
+                self._mark_ready(namespace, name, set_start=True)
+                self._started_at.pop(key, None)
+                return Result.DONE
             self._transition_running(namespace, name)
             self._started_at[key] = time.monotonic()
-            if self.ready_latency_s > 0:
-                return Result(requeue_after=self.ready_latency_s)
-            phase = "Running"
+            return Result(requeue_after=self.ready_latency_s)
 
         if phase == "Running":
             t0 = self._started_at.get(key, time.monotonic())
@@ -118,14 +122,18 @@ class VirtualKubelet:
             st["phase"] = "Running"
             st["startTime"] = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
         try:
-            self.store.patch("Pod", ns, name, upd, status=True)
+            self.store.patch("Pod", ns, name, upd, status=True,
+                             return_copy=False)
         except ApiError:
             pass
 
-    def _mark_ready(self, ns: str, name: str) -> None:
+    def _mark_ready(self, ns: str, name: str, set_start: bool = False) -> None:
         def upd(o: Obj) -> None:
             st = o.setdefault("status", {})
             st["phase"] = "Running"
+            if set_start and not st.get("startTime"):
+                st["startTime"] = time.strftime("%Y-%m-%dT%H:%M:%SZ",
+                                                time.gmtime())
             conds = st.setdefault("conditions", [])
             for want in ("ContainersReady", "Ready"):
                 for cd in conds:
@@ -138,7 +146,8 @@ class VirtualKubelet:
                                   "lastTransitionTime": time.strftime(
                                       "%Y-%m-%dT%H:%M:%SZ", time.gmtime())})
         try:
-            self.store.patch("Pod", ns, name, upd, status=True)
+            self.store.patch("Pod", ns, name, upd, status=True,
+                             return_copy=False)
         except ApiError:
             pass
 
@@ -147,7 +156,8 @@ class VirtualKubelet:
             o.setdefault("status", {})["phase"] = "Failed"
             o["status"]["reason"] = reason
         try:
-            self.store.patch("Pod", ns, name, upd, status=True)
+            self.store.patch("Pod", ns, name, upd, status=True,
+                             return_copy=False)
         except ApiError:
             pass
 

@@ -69,7 +69,7 @@ class AmdGangBackend(Backend):
 
     def sync_podgang(self, podgang: Obj) -> None:
         ns = podgang["metadata"].get("namespace", "default")
-        ctbs = self.store.list(c.KIND_CTB)
+        ctbs = self.store.list(c.KIND_CTB, copy_objects=False)
         if not ctbs:
             return
         ctb_name = ctbs[0]["metadata"]["name"]

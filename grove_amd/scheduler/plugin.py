@@ -94,7 +94,7 @@ class GangScheduler:
         # ---- gang scheduling, FIFO by PodGang creation
         gangs: List[Obj] = []
         for (ns, gname) in pending_by_gang:
-            pg = self.store.try_get(c.KIND_PODGANG, ns, gname)
+            pg = self.store.try_get(c.KIND_PODGANG, ns, gname, copy=False)
             if pg is not None:
                 gangs.append(pg)
         prio_values: Dict[str, int] = {
@@ -269,7 +269,7 @@ class GangScheduler:
                             o["status"]["phase"] = "Starting"
                         try:
                             self.store.patch(c.KIND_PODGANG, ns,
-                                             pg["metadata"]["name"], mark0, status=True)
+                                             pg["metadata"]["name"], mark0, status=True, return_copy=False)
                         except ApiError as e:
                             report_api_error(self.store, c.KIND_PODGANG, ns,
                                              pg["metadata"]["name"],
@@ -299,7 +299,7 @@ class GangScheduler:
             o["status"]["placementScore"] = round(score, 3)
             o["status"]["phase"] = "Starting"
         try:
-            self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark, status=True)
+            self.store.patch(c.KIND_PODGANG, ns, pg["metadata"]["name"], mark, status=True, return_copy=False)
         except ApiError as e:
             report_api_error(self.store, c.KIND_PODGANG, ns,
                              pg["metadata"]["name"], "mark gang scheduled", e)
@@ -431,7 +431,8 @@ class GangScheduler:
                               "lastTransitionTime": time.strftime(
                                   "%Y-%m-%dT%H:%M:%SZ", time.gmtime())})
         try:
-            self.store.patch("Pod", ns, pod["metadata"]["name"], apply)
+            self.store.patch("Pod", ns, pod["metadata"]["name"], apply,
+                             return_copy=False)
         except ApiError as e:
             report_api_error(self.store, "Pod", ns, pod["metadata"]["name"],
                              "bind pod", e)

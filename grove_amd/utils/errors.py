@@ -84,7 +84,7 @@ class StepRecorder:
                 del cur[:-5]
         try:
             self.store.patch(self.kind, self.namespace, self.name, upd,
-                             status=True)
+                             status=True, return_copy=False)
         except ApiError:
             pass
         involved = {"kind": self.kind,
@@ -114,14 +114,14 @@ def record_last_error(store: Store, kind: str, namespace: Optional[str], name: s
         if len(errs) > 5:
             del errs[:-5]
     try:
-        store.patch(kind, namespace, name, upd, status=True)
+        store.patch(kind, namespace, name, upd, status=True, return_copy=False)
     except ApiError:
         pass
 
 
 def clear_last_errors(store: Store, kind: str, namespace: Optional[str],
                       name: str) -> None:
-    cur = store.try_get(kind, namespace, name)
+    cur = store.try_get(kind, namespace, name, copy=False)
     if cur is None or not (cur.get("status") or {}).get("lastErrors"):
         return  # nothing to clear — skip the patch round trip
 
@@ -130,7 +130,7 @@ def clear_last_errors(store: Store, kind: str, namespace: Optional[str],
         if st.get("lastErrors"):
             st["lastErrors"] = []
     try:
-        store.patch(kind, namespace, name, upd, status=True)
+        store.patch(kind, namespace, name, upd, status=True, return_copy=False)
     except ApiError:
         pass
 

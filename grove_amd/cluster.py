@@ -123,7 +123,16 @@ class Cluster:
 
         def on_pcs(ev: str, obj: Obj, _old) -> None:
             md = obj["metadata"]
-            self.c_pcs.enqueue(md.get("namespace", "default"), md["name"])
+            ns = md.get("namespace", "default")
+            self.c_pcs.enqueue(ns, md["name"])
+            # spec changes flow to child PCSGs (template propagation — e.g. an
+            # OnDelete template change must reach member PCLQs without any PCSG
+            # spec change; reference podcliquescalinggroup watch predicates)
+            if _old is not None and md.get("generation") !=                     _old["metadata"].get("generation"):
+                for g in self.store.list(c.KIND_PCSG, ns,
+                                         {c.LABEL_PART_OF: md["name"]},
+                                         copy_objects=False):
+                    self.c_pcsg.enqueue(ns, g["metadata"]["name"])
 
         def on_pclq(ev: str, obj: Obj, _old) -> None:
             md = obj["metadata"]

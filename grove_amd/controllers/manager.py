@@ -209,20 +209,31 @@ class Manager:
 
     def watch(self, kind: str, handler: Callable[[str, Obj, Optional[Obj]], None],
               seed: bool = True) -> None:
-        """handler(event_type, obj, old_obj) — old_obj currently unused (None)."""
+        """handler(event_type, obj, old_obj). old_obj is the previous version seen
+        by THIS watch (informer-cache semantics: predicates compare new vs old —
+        e.g. generation changes, scheduledReplicas transitions). Objects are
+        immutable store references, so the cache holds refs, not copies."""
         w = self.store.watch(kind, seed=seed)
         self._watches.append(w)
 
         def run() -> None:
+            last: Dict[tuple, Obj] = {}
             while not self._stop.is_set():
                 try:
                     ev, obj = w.queue.get(timeout=0.2)
                 except Exception:
                     continue
+                md = obj.get("metadata", {})
+                key = (md.get("namespace", ""), md.get("name", ""))
+                old = last.get(key)
                 try:
-                    handler(ev, obj, None)
+                    handler(ev, obj, old)
                 except Exception:
                     log.debug("watch handler for %s failed:\n%s", kind, traceback.format_exc())
+                if ev == "DELETED":
+                    last.pop(key, None)
+                else:
+                    last[key] = obj
 
         t = threading.Thread(target=run, name=f"watch-{kind}-{len(self._watch_threads)}",
                              daemon=True)

@@ -155,6 +155,17 @@ class PCSGReconciler:
                     continue
                 if cur["metadata"].get("deletionTimestamp"):
                     continue
+                want_strategy = (pcs["spec"].get("updateStrategy") or {}).get(
+                    "type", c.UPDATE_ROLLING_RECREATE)
+                if cur["spec"].get("updateStrategy") != want_strategy:
+                    with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                                  benign=groveerr.BENIGN_UPDATE,
+                                  detail=f"propagate updateStrategy to {fqn}"):
+                        self.store.patch(
+                            c.KIND_PCLQ, ns, fqn,
+                            lambda o: o["spec"].update(
+                                updateStrategy=want_strategy),
+                            return_copy=False)
                 new_hash = pod_template_hash(
                     mn, cl["spec"].get("podSpec", {}),
                     pcs["spec"]["template"].get("priorityClassName", ""),

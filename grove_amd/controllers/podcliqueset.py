@@ -273,6 +273,16 @@ class PodCliqueSetReconciler:
                                      pcs["spec"]["template"].get("priorityClassName", ""),
                                      clique_tmpl.get("labels"),
                                      clique_tmpl.get("annotations"))
+        want_strategy = (pcs["spec"].get("updateStrategy") or {}).get(
+            "type", c.UPDATE_ROLLING_RECREATE)
+        if cur["spec"].get("updateStrategy") != want_strategy:
+            # strategy transitions (OD9: OnDelete -> RollingRecreate) propagate
+            # immediately and independently of template changes
+            with rec.step(groveerr.ERR_SYNC_PODCLIQUE,
+                          benign=groveerr.BENIGN_UPDATE,
+                          detail=f"propagate updateStrategy to {fqn}"):
+                self.store.patch(c.KIND_PCLQ, ns, fqn, lambda o: o["spec"].update(
+                    updateStrategy=want_strategy), return_copy=False)
         if cur["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) != new_hash \
                 and self._replica_selected_for_update(pcs, r):
             def upd(o: Obj) -> None:

@@ -169,6 +169,10 @@ class Store:
         self._tables: Dict[str, _KindTable] = {}
         self._rv = itertools.count(1)
         self._uid_index: Dict[str, Tuple[str, str, str]] = {}  # uid -> (kind, ns, name)
+        # owner uid -> set of (kind, ns, name) owned — makes cascade GC O(children)
+        # instead of a full-store scan per delete (the 10k-pod delete path was
+        # quadratic without it: 181 s for a 10k-pod tree)
+        self._owned_by: Dict[str, set] = {}
         # admission chains keyed by kind
         self._mutators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
         self._validators: Dict[str, List[Callable[[Obj, Optional[Obj]], None]]] = {}
@@ -247,6 +251,10 @@ class Store:
             tbl.objects[key] = obj
             tbl.index_add(key, obj)
             self._uid_index[m["uid"]] = (kind, ns, m["name"])
+            for ref in m.get("ownerReferences") or []:
+                if ref.get("uid"):
+                    self._owned_by.setdefault(ref["uid"], set()).add(
+                        (kind, ns, m["name"]))
             self._notify(tbl, ADDED, obj)
         return json_copy(obj)  # caller gets a private copy; stored one is immutable
 
@@ -373,6 +381,15 @@ class Store:
                 if not status_only and new["metadata"].get("labels") != cur_m.get("labels"):
                     tbl.index_remove(key, cur)
                     tbl.index_add(key, new)
+                if not status_only and new["metadata"].get("ownerReferences") \
+                        != cur_m.get("ownerReferences"):
+                    loc = (kind, ns, m["name"])
+                    for ref in cur_m.get("ownerReferences") or []:
+                        if ref.get("uid"):
+                            self._owned_by.get(ref["uid"], set()).discard(loc)
+                    for ref in new["metadata"].get("ownerReferences") or []:
+                        if ref.get("uid"):
+                            self._owned_by.setdefault(ref["uid"], set()).add(loc)
                 tbl.objects[key] = new
                 self._notify(tbl, MODIFIED, new)
                 # finalizer removal on a deleting object may allow actual deletion
@@ -457,20 +474,18 @@ class Store:
             uid = obj["metadata"].get("uid")
             if uid:
                 self._uid_index.pop(uid, None)
+            loc = (kind, ns, name)
+            for ref in obj["metadata"].get("ownerReferences") or []:
+                if ref.get("uid"):
+                    self._owned_by.get(ref["uid"], set()).discard(loc)
             self._notify(tbl, DELETED, obj)
             if cascade and uid:
                 self._cascade(uid)
 
     def _cascade(self, owner_uid: str) -> None:
-        """Delete all objects whose ownerReferences include owner_uid (kube GC stand-in)."""
+        """Delete all objects owned by owner_uid (kube GC stand-in) — indexed."""
         with self._lock:
-            victims: List[Tuple[str, str, str]] = []
-            for kind, tbl in self._tables.items():
-                for (ns, name), obj in tbl.objects.items():
-                    for ref in obj["metadata"].get("ownerReferences", []) or []:
-                        if ref.get("uid") == owner_uid:
-                            victims.append((kind, ns, name))
-                            break
+            victims = list(self._owned_by.pop(owner_uid, ()) or ())
             for kind, ns, name in victims:
                 try:
                     self.delete(kind, ns or None, name)

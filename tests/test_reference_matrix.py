@@ -1,0 +1,460 @@
+"""Reference e2e matrix coverage: scenario tests mapped 1:1 to
+the reference's e2e case IDs (GS*/SO*/GT*/OD*/RU*/TAS*; see docs/E2E_MATRIX.md for
+the full test-ID → test-function mapping). Each docstring names the reference test
+function it mirrors (behavior spec, fresh implementation on the in-process cluster
+with virtual nodes standing in for KWOK)."""
+import time
+
+import pytest
+
+from grove_amd.api import constants as c
+from grove_amd.utils import conditions as cond
+
+
+def _pcs(name, replicas=1, cliques=(("a", 1, 1),), sg=None, startup=None,
+         termination_delay="1h", strategy=None, gpus=0):
+    cl = []
+    for (n, r, m) in cliques:
+        spec = {"roleName": n, "replicas": r, "minAvailable": m,
+                "podSpec": {"containers": [{
+                    "name": "m", "image": "img:v1",
+                    "resources": {"requests": ({"cpu": "1", c.AMD_GPU_RESOURCE:
+                                                str(gpus)} if gpus else
+                                               {"cpu": "1"})}}]}}
+        cl.append({"name": n, "spec": spec})
+    tmpl = {"cliques": cl, "terminationDelay": termination_delay}
+    if sg:
+        tmpl["podCliqueScalingGroups"] = sg
+    if startup:
+        tmpl["cliqueStartupType"] = startup
+    spec = {"replicas": replicas, "template": tmpl}
+    if strategy:
+        spec["updateStrategy"] = {"type": strategy}
+    return {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+            "metadata": {"name": name}, "spec": spec}
+
+
+def _pods(cluster, selector, ns="default"):
+    return cluster.store.list("Pod", ns, selector)
+
+
+def _cordon_all(cluster, flag=True):
+    for n in cluster.store.list("Node"):
+        cluster.store.patch("Node", None, n["metadata"]["name"],
+                            lambda o: o["spec"].update(unschedulable=flag))
+
+
+# --------------------------------------------------------------------------- GS
+def test_gs4_pcs_and_pcsg_scaling_full_replicas(cluster):
+    """GS4 (gang_scheduling_test.go:217): PCS replicas scaled up AND the PCSG
+    scaled up — every new replica tree gets its own gangs and reaches ready."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 1, "minAvailable": 1}]
+    cluster.add_virtual_nodes(4, cpu="64", pods=64)
+    cluster.apply(_pcs("gs4", cliques=(("a", 1, 1), ("b", 1, 1)), sg=sg))
+    cluster.wait_pcs_available("gs4", timeout=20)
+    # scale PCS 1 -> 2 and the replica-0 PCSG 1 -> 2
+    cluster.store.patch(c.KIND_PCS, "default", "gs4",
+                        lambda o: o["spec"].update(replicas=2))
+    cluster.store.patch(c.KIND_PCSG, "default", "gs4-0-sg",
+                        lambda o: o["spec"].update(replicas=2))
+    cluster.wait_pcs_available("gs4", timeout=30)
+    cluster.wait_for(
+        lambda: len(_pods(cluster, {c.LABEL_PART_OF: "gs4"})) == 5,
+        timeout=20, desc="2 replicas x (a + sg members) + scaled member")
+    # scaled PCSG replica got its own scaled PodGang
+    assert cluster.store.try_get(c.KIND_PODGANG, "default", "gs4-0-sg-0") \
+        is not None
+    assert cluster.store.try_get(c.KIND_PODGANG, "default", "gs4-1") is not None
+
+
+def test_gs5_gang_schedules_at_min_replicas(cluster):
+    """GS5 (gang_scheduling_test.go:283): a clique with minAvailable < replicas is
+    gang-admitted once capacity fits the MINIMUM; the rest schedule as capacity
+    arrives."""
+    cluster.add_virtual_nodes(1, cpu="3", pods=64)  # fits exactly 3 x cpu:1 pods
+    cluster.apply(_pcs("gs5", cliques=(("w", 10, 3),)))
+    # min gang (3 pods) must schedule + become ready; 7 stay pending
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "gs5-0-w"}, 3, timeout=20)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "gs5-0-w"})
+    assert sum(1 for p in pods if p["spec"].get("nodeName")) == 3
+    assert len(pods) == 10
+    # capacity arrives -> everyone lands
+    cluster.add_virtual_nodes(7, cpu="1", pods=8, prefix="late")
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "gs5-0-w"}, 10, timeout=20)
+
+
+def test_gs6_pcsg_min_available_scaled_gangs(cluster):
+    """GS6/GS7 (gang_scheduling_test.go:351,463): a PCSG with minAvailable=1,
+    replicas=3 forms ONE base gang (replica 0) and TWO scaled gangs, each
+    independently schedulable; scaled gangs wait for the base gang."""
+    sg = [{"name": "sx", "cliqueNames": ["b", "c"], "replicas": 3,
+           "minAvailable": 1}]
+    cluster.add_virtual_nodes(8, cpu="8", pods=64)
+    cluster.apply(_pcs("gs6", cliques=(("a", 1, 1), ("b", 1, 1), ("c", 1, 1)),
+                       sg=sg))
+    cluster.wait_pcs_available("gs6", timeout=30)
+    gangs = {g["metadata"]["name"]
+             for g in cluster.store.list(c.KIND_PODGANG, "default",
+                                         {c.LABEL_PART_OF: "gs6"})}
+    assert gangs == {"gs6-0", "gs6-0-sx-0", "gs6-0-sx-1"}
+    for sgname in ("gs6-0-sx-0", "gs6-0-sx-1"):
+        pg = cluster.store.get(c.KIND_PODGANG, "default", sgname)
+        assert pg["metadata"]["labels"][c.LABEL_BASE_PODGANG] == "gs6-0"
+        assert cond.condition_true(pg, c.PODGANG_COND_SCHEDULED)
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "gs6"}, 7, timeout=20)
+
+
+def test_gs9_pcs_scaling_with_min_replicas(cluster):
+    """GS9 (gang_scheduling_test.go:679): scaling the PCS adds whole replica trees
+    whose gangs admit at clique minAvailable while capacity is scarce."""
+    cluster.add_virtual_nodes(1, cpu="2", pods=64)
+    cluster.apply(_pcs("gs9", replicas=1, cliques=(("w", 2, 2),)))
+    cluster.wait_pcs_available("gs9", timeout=20)
+    cluster.store.patch(c.KIND_PCS, "default", "gs9",
+                        lambda o: o["spec"].update(replicas=2))
+    # replica 1's gang cannot fit -> its pods stay unscheduled, replica 0 untouched
+    cluster.wait_for(
+        lambda: len(_pods(cluster, {c.LABEL_PART_OF: "gs9"})) == 4,
+        timeout=20, desc="4 pods exist")
+    time.sleep(0.3)
+    r1 = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})
+    assert all(not p["spec"].get("nodeName") for p in r1), \
+        "partial gang must not schedule"
+    cluster.add_virtual_nodes(1, cpu="2", pods=64, prefix="extra")
+    cluster.wait_pcs_available("gs9", timeout=20, min_available=2)
+
+
+# --------------------------------------------------------------------------- SO
+def test_so2_inorder_startup_with_min_replicas(cluster):
+    """SO2/SO4 (startup_ordering_test.go:108,218): with InOrder startup and
+    minAvailable < replicas, the dependent clique starts once the parent reaches
+    minAvailable ready pods (not full replicas)."""
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)
+    pcs = _pcs("so2", cliques=(("a", 3, 2), ("b", 1, 1)),
+               startup=c.STARTUP_IN_ORDER)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("so2", timeout=20)
+    b = cluster.store.get(c.KIND_PCLQ, "default", "so2-0-b")
+    assert b["spec"]["startsAfter"] == ["so2-0-a"]
+    b_pods = _pods(cluster, {c.LABEL_PODCLIQUE: "so2-0-b"})
+    ic = b_pods[0]["spec"]["initContainers"][0]
+    # waits for the parent's minAvailable (2), not its full replicas (3)
+    assert "--podcliques=so2-0-a:2" in ic["args"]
+
+
+# --------------------------------------------------------------------------- GT
+def test_gt2_pcsg_owned_breach_terminates_pcs_replica(cluster):
+    """GT2 (gang_termination_test.go:57): a breach in a PCSG member clique (the
+    PCSG's MinAvailableBreached, not a standalone clique's) gang-terminates the
+    whole PCS replica after terminationDelay — standalone cliques of the same
+    replica are recreated too."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 2}]
+    cluster.add_virtual_nodes(3, cpu="8", pods=64)
+    cluster.apply(_pcs("gt2", cliques=(("a", 1, 1), ("b", 1, 1)), sg=sg,
+                       termination_delay="300ms"))
+    cluster.wait_pcs_available("gt2", timeout=20)
+    a_uid = cluster.store.get(c.KIND_PCLQ, "default", "gt2-0-a")["metadata"]["uid"]
+    _cordon_all(cluster)
+    # kill a PCSG member pod -> PCSG availableReplicas drops below minAvailable
+    victim = _pods(cluster, {c.LABEL_PODCLIQUE: "gt2-0-sg-0-b"})[0]
+    cluster.store.delete("Pod", "default", victim["metadata"]["name"])
+
+    def pcsg_breached():
+        g = cluster.store.get(c.KIND_PCSG, "default", "gt2-0-sg")
+        return cond.condition_true(g, c.COND_MIN_AVAILABLE_BREACHED)
+    cluster.wait_for(pcsg_breached, timeout=20, desc="PCSG MinAvailableBreached")
+
+    # nodes stay cordoned so the breach cannot heal before terminationDelay —
+    # the whole replica (including standalone clique a) must be recreated
+    def replica_recreated():
+        q = cluster.store.try_get(c.KIND_PCLQ, "default", "gt2-0-a")
+        return q is not None and q["metadata"]["uid"] != a_uid
+    cluster.wait_for(replica_recreated, timeout=30,
+                     desc="standalone clique recreated by gang termination")
+    _cordon_all(cluster, False)
+    cluster.wait_pcs_available("gt2", timeout=30)
+
+
+def test_gt6_scaled_pod_deletion_does_not_terminate_pcs_replica(cluster):
+    """GT6 (gang_termination_test.go:307): deleting a pod of a SCALED PCSG replica
+    (base gang minimum still met) must not fire PCS-scope gang termination — the
+    standalone cliques keep their identity and only the scaled pod is replaced."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 1}]
+    cluster.add_virtual_nodes(3, cpu="8", pods=64)
+    cluster.apply(_pcs("gt6", cliques=(("a", 1, 1), ("b", 1, 1)), sg=sg,
+                       termination_delay="300ms"))
+    cluster.wait_pcs_available("gt6", timeout=20)
+    a_uid = cluster.store.get(c.KIND_PCLQ, "default", "gt6-0-a")["metadata"]["uid"]
+    # scaled replica j=1 (>= minAvailable) pod
+    victim = _pods(cluster, {c.LABEL_PODCLIQUE: "gt6-0-sg-1-b"})[0]
+    cluster.store.delete("Pod", "default", victim["metadata"]["name"])
+    time.sleep(0.9)  # > terminationDelay
+    # standalone clique survived (no PCS-scope termination)
+    a = cluster.store.get(c.KIND_PCLQ, "default", "gt6-0-a")
+    assert a["metadata"]["uid"] == a_uid
+    # the scaled pod was replaced and the set is whole again
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "gt6-0-sg-1-b"}, 1, timeout=20)
+
+
+# --------------------------------------------------------------------------- OD
+def test_od2_manual_deletion_creates_updated_pod(cluster):
+    """OD2 (update/ondelete_test.go:72): with OnDelete, a template change does not
+    touch running pods; deleting one manually yields a replacement built from the
+    NEW template (new pod-template-hash)."""
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)
+    cluster.apply(_pcs("od2", cliques=(("w", 2, 1),), strategy=c.UPDATE_ON_DELETE))
+    cluster.wait_pcs_available("od2", timeout=20)
+    old_pods = _pods(cluster, {c.LABEL_PODCLIQUE: "od2-0-w"})
+    old_hash = old_pods[0]["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH]
+    cur = cluster.store.get(c.KIND_PCS, "default", "od2")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+
+    def pclq_updated():
+        q = cluster.store.get(c.KIND_PCLQ, "default", "od2-0-w")
+        return q["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] != old_hash
+    cluster.wait_for(pclq_updated, timeout=20, desc="PCLQ spec propagated")
+    time.sleep(0.3)
+    # no automatic pod replacement (OD1 semantics)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "od2-0-w"})
+    assert {p["metadata"]["uid"] for p in pods} == \
+        {p["metadata"]["uid"] for p in old_pods}
+    # manual delete -> replacement carries the new hash + image
+    cluster.store.delete("Pod", "default", old_pods[0]["metadata"]["name"])
+
+    def replaced():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "od2-0-w"})
+        return len(ps) == 2 and any(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] != old_hash
+            for p in ps)
+    cluster.wait_for(replaced, timeout=20, desc="updated replacement pod")
+    new_pod = [p for p in _pods(cluster, {c.LABEL_PODCLIQUE: "od2-0-w"})
+               if p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] != old_hash][0]
+    assert new_pod["spec"]["containers"][0]["image"] == "img:v2"
+
+
+def test_od5_pcsg_manual_deletion_creates_updated_replica(cluster):
+    """OD5 (update/ondelete_test.go:293): OnDelete at PCSG scope — member pods
+    survive the template change until manually deleted, then rebuild updated."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 1}]
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)
+    cluster.apply(_pcs("od5", cliques=(("b", 1, 1),), sg=sg,
+                       strategy=c.UPDATE_ON_DELETE))
+    cluster.wait_pcs_available("od5", timeout=20)
+    old = _pods(cluster, {c.LABEL_PODCLIQUE: "od5-0-sg-1-b"})[0]
+    old_hash = old["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH]
+    cur = cluster.store.get(c.KIND_PCS, "default", "od5")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+
+    def member_pclq_updated():
+        q = cluster.store.get(c.KIND_PCLQ, "default", "od5-0-sg-1-b")
+        return q["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] != old_hash
+    cluster.wait_for(member_pclq_updated, timeout=20, desc="member PCLQ updated")
+    time.sleep(0.2)
+    assert _pods(cluster, {c.LABEL_PODCLIQUE: "od5-0-sg-1-b"})[0][
+        "metadata"]["uid"] == old["metadata"]["uid"]
+    cluster.store.delete("Pod", "default", old["metadata"]["name"])
+
+    def replaced():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "od5-0-sg-1-b"})
+        return len(ps) == 1 and ps[0]["metadata"]["labels"][
+            c.LABEL_POD_TEMPLATE_HASH] != old_hash
+    cluster.wait_for(replaced, timeout=20, desc="updated member replacement")
+
+
+def test_od9_strategy_transition_resumes_update(cluster):
+    """OD9 (update/ondelete_test.go:567): switching OnDelete -> RollingRecreate
+    with a pending template change starts replacing pods automatically."""
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)
+    cluster.apply(_pcs("od9", cliques=(("w", 2, 1),), strategy=c.UPDATE_ON_DELETE))
+    cluster.wait_pcs_available("od9", timeout=20)
+    old_hash = _pods(cluster, {c.LABEL_PODCLIQUE: "od9-0-w"})[0][
+        "metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH]
+    cur = cluster.store.get(c.KIND_PCS, "default", "od9")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    time.sleep(0.3)
+    assert all(p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == old_hash
+               for p in _pods(cluster, {c.LABEL_PODCLIQUE: "od9-0-w"}))
+    cur = cluster.store.get(c.KIND_PCS, "default", "od9")
+    cur["spec"]["updateStrategy"] = {"type": c.UPDATE_ROLLING_RECREATE}
+    cluster.apply(cur)
+
+    def all_updated():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "od9-0-w"})
+        return len(ps) == 2 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] != old_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(all_updated, timeout=30,
+                     desc="rolling update resumed after strategy switch")
+
+
+# --------------------------------------------------------------------------- RU
+def test_ru11_scale_out_during_update_gets_new_template(cluster):
+    """RU11 (update/rolling_recreate_test.go:293): a PCS replica added while a
+    rolling update is in flight is created directly from the NEW template."""
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru11", replicas=2, cliques=(("w", 2, 1),)))
+    cluster.wait_pcs_available("ru11", timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru11")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cur["spec"]["replicas"] = 3
+    cluster.apply(cur)
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "w", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def replica2_new():
+        ps = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "2"})
+        return len(ps) == 2 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            for p in ps)
+    cluster.wait_for(replica2_new, timeout=25,
+                     desc="scale-out replica born on the new template")
+    cluster.wait_pcs_available("ru11", timeout=40)
+    # eventually every pod carries the new template
+    def all_new():
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru11"})
+        return len(ps) == 6 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            for p in ps)
+    cluster.wait_for(all_new, timeout=40, desc="rolling update completed")
+
+
+def test_ru12_scale_in_during_update_completes(cluster):
+    """RU12/RU13 (rolling_recreate_test.go:345,412): scaling the PCS in while an
+    update is running removes the excess replica trees and the update still ends
+    (updateProgress.updateEndedAt set)."""
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru12", replicas=3, cliques=(("w", 2, 1),)))
+    cluster.wait_pcs_available("ru12", timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru12")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cur["spec"]["replicas"] = 1
+    cluster.apply(cur)
+
+    def done():
+        pcs = cluster.store.get(c.KIND_PCS, "default", "ru12")
+        st = pcs.get("status") or {}
+        prog = st.get("updateProgress") or {}
+        return bool(prog.get("updateEndedAt")) \
+            and int(st.get("availableReplicas", 0)) >= 1
+    cluster.wait_for(done, timeout=40, desc="update finished after scale-in")
+    assert len(_pods(cluster, {c.LABEL_PART_OF: "ru12"})) == 2
+    assert cluster.store.try_get(c.KIND_PCLQ, "default", "ru12-2-w") is None
+
+
+def test_ru14_pcsg_scale_out_during_update(cluster):
+    """RU14/RU15 (rolling_recreate_test.go:465,522): PCSG replicas added around an
+    update come up on the new template; existing members roll."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 1, "minAvailable": 1}]
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru14", cliques=(("b", 1, 1),), sg=sg))
+    cluster.wait_pcs_available("ru14", timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru14")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    cluster.store.patch(c.KIND_PCSG, "default", "ru14-0-sg",
+                        lambda o: o["spec"].update(replicas=2))
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "b", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def all_new():
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru14"})
+        return len(ps) == 2 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(all_new, timeout=40,
+                     desc="scaled member + rolled member both on new template")
+
+
+# --------------------------------------------------------------------------- TAS
+def _rack_nodes(cluster, racks, per_rack, gpus=8):
+    from grove_amd.kubelet.virtual import make_virtual_node
+    for r in range(racks):
+        for i in range(per_rack):
+            n = make_virtual_node(f"r{r}n{i}", gpus=gpus, cpu="64", pods=64)
+            n["metadata"]["labels"]["topology.kubernetes.io/rack"] = f"rack{r}"
+            cluster.store.create(n)
+
+
+_CTB = {"apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+        "metadata": {"name": "default-topology"},
+        "spec": {"levels": [
+            {"domain": "rack", "key": "topology.kubernetes.io/rack"},
+            {"domain": "host", "key": "kubernetes.io/hostname"}]}}
+
+
+def test_tas9_pcs_plus_clique_constraint(cluster):
+    """TAS9 (topology_test.go:585): gang-level pack (rack) + clique-level pack
+    (host) — the clique's pods pack onto one host inside the gang's rack."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=2, per_rack=2, gpus=4)
+    pcs = _pcs("tas9", cliques=(("w", 2, 2),), gpus=1)
+    pcs["spec"]["template"]["topologyConstraint"] = {"pack": {"required": "rack"}}
+    pcs["spec"]["template"]["cliques"][0]["topologyConstraint"] = {
+        "pack": {"required": "host"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas9", timeout=20)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "tas9-0-w"})
+    hosts = {p["spec"]["nodeName"] for p in pods}
+    assert len(hosts) == 1  # clique packed to one host (inside one rack a fortiori)
+
+
+def test_tas10_pcsg_scaling_with_constraint(cluster):
+    """TAS10 (topology_test.go:635): each PCSG replica packs within its own
+    domain of the PCSG's pack key; scaled replicas get the same constraint."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=3, per_rack=1, gpus=4)
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 1,
+           "topologyConstraint": {"pack": {"required": "host"}}}]
+    pcs = _pcs("tas10", cliques=(("b", 2, 2),), sg=sg, gpus=1)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas10", timeout=20)
+    for j in (0, 1):
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: f"tas10-0-sg-{j}-b"})
+        assert len({p["spec"]["nodeName"] for p in ps}) == 1, \
+            f"PCSG replica {j} spans hosts"
+    # scaled gang carries the translated constraint
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "tas10-0-sg-0")
+    assert pg["spec"]["topologyConstraint"]["packConstraint"]["required"] == \
+        "kubernetes.io/hostname"
+
+
+def test_tas14_multi_replica_rack_constraint(cluster):
+    """TAS14/TAS16 (topology_test.go:927,1099): every PCS replica's gang packs
+    into ONE rack; different replicas may use different racks."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=2, per_rack=2, gpus=2)
+    pcs = _pcs("tas14", replicas=2, cliques=(("w", 3, 3),), gpus=1)
+    pcs["spec"]["template"]["topologyConstraint"] = {"pack": {"required": "rack"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas14", timeout=20)
+    node_rack = {n["metadata"]["name"]:
+                 n["metadata"]["labels"]["topology.kubernetes.io/rack"]
+                 for n in cluster.store.list("Node")}
+    for r in (0, 1):
+        ps = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: str(r)})
+        racks = {node_rack[p["spec"]["nodeName"]] for p in ps}
+        assert len(racks) == 1, f"replica {r} spans racks {racks}"
+
+
+def test_tas17_heterogeneous_gpu_cluster(cluster):
+    """TAS17 (topology_test.go:1200): nodes with different GPU counts — an 8-GPU
+    gang must land on the node that actually has 8 free GPUs."""
+    from grove_amd.kubelet.virtual import make_virtual_node
+    cluster.store.create(make_virtual_node("small", gpus=2, cpu="64", pods=64))
+    cluster.store.create(make_virtual_node("big", gpus=8, cpu="64", pods=64))
+    pcs = _pcs("tas17", cliques=(("w", 8, 8),), gpus=1)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas17", timeout=20)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "tas17-0-w"})
+    assert all(p["spec"]["nodeName"] == "big" for p in pods)

@@ -458,3 +458,64 @@ def test_tas17_heterogeneous_gpu_cluster(cluster):
     cluster.wait_pcs_available("tas17", timeout=20)
     pods = _pods(cluster, {c.LABEL_PODCLIQUE: "tas17-0-w"})
     assert all(p["spec"]["nodeName"] == "big" for p in pods)
+
+
+def test_ru16_pcsg_scale_in_during_update(cluster):
+    """RU16/RU17 (rolling_recreate_test.go:580,640): scaling the PCSG in while a
+    rolling update runs removes the excess member trees and the update still
+    completes on the survivors."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 3, "minAvailable": 1}]
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru16", cliques=(("b", 1, 1),), sg=sg))
+    cluster.wait_pcs_available("ru16", timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru16")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    cluster.store.patch(c.KIND_PCSG, "default", "ru16-0-sg",
+                        lambda o: o["spec"].update(replicas=1))
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "b", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def settled():
+        if cluster.store.try_get(c.KIND_PCLQ, "default", "ru16-0-sg-2-b"):
+            return False
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru16"})
+        return len(ps) == 1 and ps[0]["metadata"]["labels"][
+            c.LABEL_POD_TEMPLATE_HASH] == new_hash and cond.pod_is_ready(ps[0])
+    cluster.wait_for(settled, timeout=40,
+                     desc="scale-in applied and survivor updated")
+
+
+def test_ru18_clique_scale_out_during_update(cluster):
+    """RU18/RU20 (rolling_recreate_test.go:704,832): scaling a PodClique's
+    replicas (the HPA path — PCLQ spec patched directly) while the template
+    update is in flight: new pods are born on the NEW template and the update
+    finishes across the larger set."""
+    cluster.add_virtual_nodes(2, cpu="16", pods=64)
+    pcs = _pcs("ru18", cliques=(("w", 2, 1),))
+    # autoscaled clique: replica preservation keeps the HPA's scale decisions
+    # (podclique.go:284 parity) while the template rolls
+    pcs["spec"]["template"]["cliques"][0]["spec"]["autoScalingConfig"] = {
+        "minReplicas": 1, "maxReplicas": 5}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("ru18", timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru18")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    # HPA-style direct PCLQ scale mid-update
+    cluster.store.patch(c.KIND_PCLQ, "default", "ru18-0-w",
+                        lambda o: o["spec"].update(replicas=3))
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "w", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def all_updated():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "ru18-0-w"})
+        return len(ps) == 3 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(all_updated, timeout=40,
+                     desc="3 pods, all on the new template")

@@ -7,6 +7,7 @@ constraints :573-589, immutable-field update rules :643-702). Fresh implementati
 """
 from __future__ import annotations
 
+import re
 from typing import Any, Dict, List, Optional
 
 from . import constants as c
@@ -80,6 +81,84 @@ def _check_startup_dag(cliques: List[Obj], startup_type: str) -> None:
             visit(n)
 
 
+
+
+_DNS1123_LABEL = re.compile(r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?$")
+_DNS_SUBDOMAIN = re.compile(
+    r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?(\.[a-z0-9]([-a-z0-9]*[a-z0-9])?)*$")
+_ENV_VAR_NAME = re.compile(r"^[-._a-zA-Z][-._a-zA-Z0-9]*$")
+
+
+def _validate_pod_spec(podspec: Obj, path: str, is_create: bool) -> None:
+    """PodSpec sanity (validation/podcliqueset.go:591-642): operator-managed
+    fields must not be user-set; env var names must be valid and unique."""
+    if is_create:
+        if podspec.get("topologySpreadConstraints"):
+            raise _err(f"{path}.topologySpreadConstraints",
+                       "must not be set (the gang scheduler owns placement)")
+        if podspec.get("nodeName"):
+            raise _err(f"{path}.nodeName",
+                       "must not be set (the gang scheduler owns placement)")
+    for kind_key in ("containers", "initContainers"):
+        for ctr in podspec.get(kind_key) or []:
+            seen_env = set()
+            for ev in ctr.get("env") or []:
+                nm = str(ev.get("name", ""))
+                if not _ENV_VAR_NAME.match(nm):
+                    raise _err(f"{path}.{kind_key}[{ctr.get('name')}].env",
+                               f"invalid environment variable name {nm!r}")
+                if nm in seen_env:
+                    raise _err(f"{path}.{kind_key}[{ctr.get('name')}].env",
+                               f"duplicate environment variable {nm!r}")
+                seen_env.add(nm)
+
+
+def _validate_resource_sharing(refs: List[Obj], template_names: set, path: str,
+                               clique_names: Optional[set] = None,
+                               group_names: Optional[set] = None,
+                               allow_group_filter: bool = False) -> None:
+    """ResourceSharing entries (validation/podcliqueset.go:139-233): name
+    required+unique, scope enum, namespace only for external templates, filter
+    children must exist within the declaring scope."""
+    seen = set()
+    for ref in refs:
+        nm = ref.get("name", "")
+        if not nm:
+            raise _err(f"{path}.name", "reference name is required")
+        if nm in seen:
+            raise _err(f"{path}.name", f"duplicate reference {nm!r}")
+        seen.add(nm)
+        if ref.get("namespace") and nm in template_names:
+            raise _err(f"{path}[{nm}].namespace",
+                       "namespace must be empty when name matches an internal "
+                       "resourceClaimTemplate")
+        if ref.get("scope") not in ("AllReplicas", "PerReplica"):
+            raise _err(f"{path}[{nm}].scope",
+                       f"unsupported scope {ref.get('scope')!r}; "
+                       "supported: AllReplicas, PerReplica")
+        filt = ref.get("filter")
+        if filt is None:
+            continue
+        child_cl = filt.get("childCliqueNames") or []
+        child_sg = filt.get("childScalingGroupNames") or []
+        if not child_cl and not (child_sg if allow_group_filter else ()):
+            raise _err(f"{path}[{nm}].filter",
+                       "filter must specify at least one child entry")
+        if clique_names is not None:
+            for cn in child_cl:
+                if cn not in clique_names:
+                    raise _err(f"{path}[{nm}].filter.childCliqueNames",
+                               f"unknown clique {cn!r}")
+        if not allow_group_filter and child_sg:
+            raise _err(f"{path}[{nm}].filter.childScalingGroupNames",
+                       "not allowed at this scope")
+        if allow_group_filter and group_names is not None:
+            for gn in child_sg:
+                if gn not in group_names:
+                    raise _err(f"{path}[{nm}].filter.childScalingGroupNames",
+                               f"unknown scaling group {gn!r}")
+
+
 def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
     name = pcs.get("metadata", {}).get("name", "")
     spec = pcs.get("spec") or {}
@@ -118,6 +197,9 @@ def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
             if ma > reps:
                 raise _err(f"spec.template.cliques[{cn}].spec.minAvailable",
                            "must not be greater than replicas")
+        if not _DNS1123_LABEL.match(str(cn)):
+            raise _err(f"spec.template.cliques[{cn}].name",
+                       "must be a valid DNS-1123 label")
         podspec = cs.get("podSpec") or {}
         if not podspec.get("containers"):
             raise _err(f"spec.template.cliques[{cn}].spec.podSpec.containers",
@@ -127,6 +209,8 @@ def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
                 if str(ev.get("name", "")).startswith("GROVE_"):
                     raise _err(f"spec.template.cliques[{cn}].spec.podSpec",
                                f"env var {ev.get('name')!r} uses the reserved GROVE_ prefix")
+        _validate_pod_spec(podspec, f"spec.template.cliques[{cn}].spec.podSpec",
+                           is_create=(old is None))
         asc = cs.get("autoScalingConfig")
         if asc is not None:
             _validate_scale_config(asc, cs.get("minAvailable", reps),
@@ -140,6 +224,9 @@ def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
             raise _err("spec.template.podCliqueScalingGroups", "scaling group name is required")
         if sgn in sg_names:
             raise _err("spec.template.podCliqueScalingGroups", f"duplicate scaling group {sgn!r}")
+        if not _DNS_SUBDOMAIN.match(str(sgn)):
+            raise _err(f"spec.template.podCliqueScalingGroups[{sgn}].name",
+                       "must be a valid DNS subdomain")
         sg_names.add(sgn)
         members = sg.get("cliqueNames") or []
         if not members:
@@ -173,6 +260,44 @@ def validate_podcliqueset(pcs: Obj, old: Optional[Obj] = None) -> None:
     for cl in cliques:
         if cl.get("name") not in sg_member_cliques:
             _check_name_budget(name, "", cl.get("name", ""))
+
+    # resource claim templates: required name (DNS subdomain), unique, device
+    # requests present (validation/podcliqueset.go:121-137)
+    template_names: set = set()
+    for rct in tmpl.get("resourceClaimTemplates") or []:
+        rn = rct.get("name", "")
+        if not rn:
+            raise _err("spec.template.resourceClaimTemplates.name",
+                       "template name is required")
+        if not _DNS_SUBDOMAIN.match(rn):
+            raise _err(f"spec.template.resourceClaimTemplates[{rn}].name",
+                       "must be a valid DNS subdomain")
+        if rn in template_names:
+            raise _err("spec.template.resourceClaimTemplates.name",
+                       f"duplicate template {rn!r}")
+        template_names.add(rn)
+        reqs = (((rct.get("templateSpec") or {}).get("spec") or {})
+                .get("devices") or {}).get("requests") or []
+        if not reqs:
+            raise _err(f"spec.template.resourceClaimTemplates[{rn}]"
+                       ".templateSpec.spec.devices.requests",
+                       "at least one device request is required")
+
+    clique_names = set(seen)
+    _validate_resource_sharing(tmpl.get("resourceSharing") or [], template_names,
+                               "spec.template.resourceSharing",
+                               clique_names=clique_names, group_names=sg_names,
+                               allow_group_filter=True)
+    for cl in cliques:
+        _validate_resource_sharing(
+            cl.get("resourceSharing") or [], template_names,
+            f"spec.template.cliques[{cl.get('name')}].resourceSharing")
+    for sg in tmpl.get("podCliqueScalingGroups") or []:
+        _validate_resource_sharing(
+            sg.get("resourceSharing") or [], template_names,
+            f"spec.template.podCliqueScalingGroups[{sg.get('name')}]"
+            ".resourceSharing",
+            clique_names=set(sg.get("cliqueNames") or []))
 
     _check_startup_dag(cliques, startup)
 

@@ -194,3 +194,97 @@ def test_validation_immutable_updates():
     default_podcliqueset(new)
     with pytest.raises(ApiError):
         validate_podcliqueset(new, old)
+
+
+def test_validation_podspec_sanity():
+    """validatePodSpec parity (validation/podcliqueset.go:591-642): operator-owned
+    scheduling fields rejected on create; env names validated and deduped."""
+    from grove_amd import Cluster
+    from grove_amd.kubecore.store import ApiError
+
+    def pcs(podspec_extra=None, env=None):
+        ps = {"containers": [{"name": "m", "image": "i",
+                              "env": env or []}]}
+        ps.update(podspec_extra or {})
+        return {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+                "metadata": {"name": "vps"},
+                "spec": {"replicas": 1, "template": {"cliques": [{
+                    "name": "w", "spec": {"roleName": "w", "replicas": 1,
+                                          "podSpec": ps}}]}}}
+    cl = Cluster()
+    try:
+        with pytest.raises(ApiError, match="nodeName"):
+            cl.store.create(pcs({"nodeName": "n1"}))
+        with pytest.raises(ApiError, match="topologySpreadConstraints"):
+            cl.store.create(pcs({"topologySpreadConstraints": [
+                {"maxSkew": 1, "topologyKey": "zone",
+                 "whenUnsatisfiable": "DoNotSchedule"}]}))
+        with pytest.raises(ApiError, match="invalid environment variable"):
+            cl.store.create(pcs(env=[{"name": "1BAD", "value": "x"}]))
+        with pytest.raises(ApiError, match="duplicate environment variable"):
+            cl.store.create(pcs(env=[{"name": "A", "value": "1"},
+                                     {"name": "A", "value": "2"}]))
+        cl.store.create(pcs(env=[{"name": "GOOD_ONE", "value": "1"}]))
+    finally:
+        cl.stop()
+
+
+def test_validation_resource_sharing_rules():
+    """validateResourceSharingSpecs/filters parity (podcliqueset.go:139-233)."""
+    from grove_amd import Cluster
+    from grove_amd.kubecore.store import ApiError
+
+    def pcs(sharing=None, sg_sharing=None, templates=None):
+        tmpl = {"cliques": [
+            {"name": "a", "spec": {"roleName": "a", "replicas": 1,
+                                   "podSpec": {"containers": [
+                                       {"name": "m", "image": "i"}]}}},
+            {"name": "b", "spec": {"roleName": "b", "replicas": 1,
+                                   "podSpec": {"containers": [
+                                       {"name": "m", "image": "i"}]}}}],
+            "podCliqueScalingGroups": [{
+                "name": "sg", "cliqueNames": ["b"],
+                **({"resourceSharing": sg_sharing} if sg_sharing else {})}]}
+        if templates is not None:
+            tmpl["resourceClaimTemplates"] = templates
+        if sharing is not None:
+            tmpl["resourceSharing"] = sharing
+        return {"apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+                "metadata": {"name": "vrs"}, "spec": {"replicas": 1,
+                                                      "template": tmpl}}
+    tpl = [{"name": "t1", "templateSpec": {"spec": {"devices": {"requests": [
+        {"name": "d", "deviceClassName": "x"}]}}}}]
+    cl = Cluster()
+    try:
+        with pytest.raises(ApiError, match="scope"):
+            cl.store.create(pcs(sharing=[{"name": "t1", "scope": "Bogus"}],
+                                templates=tpl))
+        with pytest.raises(ApiError, match="duplicate reference"):
+            cl.store.create(pcs(sharing=[
+                {"name": "t1", "scope": "AllReplicas"},
+                {"name": "t1", "scope": "PerReplica"}], templates=tpl))
+        with pytest.raises(ApiError, match="namespace must be empty"):
+            cl.store.create(pcs(sharing=[{"name": "t1", "namespace": "other",
+                                          "scope": "AllReplicas"}],
+                                templates=tpl))
+        with pytest.raises(ApiError, match="unknown clique"):
+            cl.store.create(pcs(sharing=[{
+                "name": "t1", "scope": "AllReplicas",
+                "filter": {"childCliqueNames": ["nope"]}}], templates=tpl))
+        # PCSG-scope filter may only reference member cliques
+        with pytest.raises(ApiError, match="unknown clique"):
+            cl.store.create(pcs(sg_sharing=[{
+                "name": "t1", "scope": "PerReplica",
+                "filter": {"childCliqueNames": ["a"]}}], templates=tpl))
+        # claim template device requests required
+        with pytest.raises(ApiError, match="device request"):
+            cl.store.create(pcs(templates=[
+                {"name": "t2", "templateSpec": {"spec": {}}}]))
+        with pytest.raises(ApiError, match="duplicate template"):
+            cl.store.create(pcs(templates=tpl + tpl))
+        cl.store.create(pcs(sharing=[{
+            "name": "t1", "scope": "AllReplicas",
+            "filter": {"childCliqueNames": ["a"],
+                       "childScalingGroupNames": ["sg"]}}], templates=tpl))
+    finally:
+        cl.stop()

@@ -117,6 +117,12 @@ class PCSGReconciler:
         xgmi_groups = resourceclaims.effective_xgmi_groups(
             pcs, self.auto_xgmi_domain)
         existing = {q["metadata"]["name"]: q for q in self._member_pclqs(pcsg)}
+        member_hash = {mn: pod_template_hash(
+            mn, (cliques.get(mn) or {}).get("spec", {}).get("podSpec", {}),
+            pcs["spec"]["template"].get("priorityClassName", ""),
+            (cliques.get(mn) or {}).get("labels"),
+            (cliques.get(mn) or {}).get("annotations"))
+            for mn in member_names if mn in cliques}
         expected: set = set()
         for j in range(replicas):
             pg_name = namegen.podgang_name_for_pclq_in_pcsg(
@@ -166,10 +172,7 @@ class PCSGReconciler:
                             lambda o: o["spec"].update(
                                 updateStrategy=want_strategy),
                             return_copy=False)
-                new_hash = pod_template_hash(
-                    mn, cl["spec"].get("podSpec", {}),
-                    pcs["spec"]["template"].get("priorityClassName", ""),
-                    cl.get("labels"), cl.get("annotations"))
+                new_hash = member_hash[mn]
                 if cur["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) != new_hash \
                         and self._replica_selected_for_update(pcs, pcs_replica):
                     def upd(o: Obj) -> None:

@@ -189,6 +189,13 @@ class PodCliqueSetReconciler:
         self._sync_hpas(pcs, rec)
 
         # ---- G2: standalone PodCliques per replica (+ shared ResourceClaims)
+        # template hashes are per-CLIQUE, identical across replicas — computing
+        # them once per pass instead of per replica removes 5000 sha256 calls per
+        # reconcile of a 5000-replica PCS
+        clique_hash = {cl["name"]: pod_template_hash(
+            cl["name"], cl["spec"].get("podSpec", {}),
+            tmpl.get("priorityClassName", ""), cl.get("labels"),
+            cl.get("annotations")) for cl in tmpl.get("cliques") or []}
         pclq_by_name = {q["metadata"]["name"]: q for q in self.store.list(
             c.KIND_PCLQ, ns, {c.LABEL_PART_OF: name,
                               c.LABEL_COMPONENT: c.COMPONENT_PCS_PODCLIQUE},
@@ -212,7 +219,8 @@ class PodCliqueSetReconciler:
                 refs = refs + resourceclaims.claim_refs_for_clique(
                     cl_claims, cl["name"])
                 self._sync_pclq(pcs, r, cl, fqn, owner=pcs, claim_refs=refs,
-                                cur=pclq_by_name.get(fqn), rec=rec)
+                                cur=pclq_by_name.get(fqn), rec=rec,
+                                new_hash=clique_hash[cl["name"]])
         # GC excess standalone PCLQs (scale-in / replica removal)
         for pclq in pclq_by_name.values():
             if pclq["metadata"]["name"] not in expected_pclqs:
@@ -248,7 +256,8 @@ class PodCliqueSetReconciler:
 
     def _sync_pclq(self, pcs: Obj, r: int, clique_tmpl: Obj, fqn: str, owner: Obj,
                    claim_refs=None, cur="__lookup__",
-                   rec: Optional[groveerr.StepRecorder] = None) -> None:
+                   rec: Optional[groveerr.StepRecorder] = None,
+                   new_hash: Optional[str] = None) -> None:
         ns = pcs["metadata"].get("namespace", "default")
         rec = rec or groveerr.StepRecorder(self.store, c.KIND_PCS, ns,
                                            pcs["metadata"]["name"])
@@ -268,11 +277,11 @@ class PodCliqueSetReconciler:
         if cur["metadata"].get("deletionTimestamp"):
             return
         # propagate template changes only to replicas selected for update (GREP-393)
-        new_hash = pod_template_hash(clique_tmpl["name"],
-                                     clique_tmpl["spec"].get("podSpec", {}),
-                                     pcs["spec"]["template"].get("priorityClassName", ""),
-                                     clique_tmpl.get("labels"),
-                                     clique_tmpl.get("annotations"))
+        if new_hash is None:
+            new_hash = pod_template_hash(
+                clique_tmpl["name"], clique_tmpl["spec"].get("podSpec", {}),
+                pcs["spec"]["template"].get("priorityClassName", ""),
+                clique_tmpl.get("labels"), clique_tmpl.get("annotations"))
         want_strategy = (pcs["spec"].get("updateStrategy") or {}).get(
             "type", c.UPDATE_ROLLING_RECREATE)
         if cur["spec"].get("updateStrategy") != want_strategy:
@@ -616,14 +625,20 @@ class PodCliqueSetReconciler:
         st = pcs.get("status") or {}
         prog = st.get("updateProgress")
 
+        _hash_by_clique = {cl["name"]: pod_template_hash(
+            cl["name"], cl["spec"].get("podSpec", {}),
+            tmpl.get("priorityClassName", ""), cl.get("labels"),
+            cl.get("annotations")) for cl in tmpl.get("cliques") or []}
+        _hash_by_fqn: Dict[str, str] = {}
+
         def expected_hash_of(q: Obj) -> str:
-            cl = builders.match_by_fqn_suffix(q["metadata"]["name"],
-                                              tmpl.get("cliques") or [])
-            if cl is None:
-                return ""
-            return pod_template_hash(cl["name"], cl["spec"].get("podSpec", {}),
-                                     tmpl.get("priorityClassName", ""),
-                                     cl.get("labels"), cl.get("annotations"))
+            fqn = q["metadata"]["name"]
+            h = _hash_by_fqn.get(fqn)
+            if h is None:
+                cl = builders.match_by_fqn_suffix(fqn, tmpl.get("cliques") or [])
+                h = _hash_by_clique.get(cl["name"], "") if cl is not None else ""
+                _hash_by_fqn[fqn] = h
+            return h
 
         def pclq_hash_current(q: Obj) -> bool:
             return q["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH) \

@@ -57,14 +57,15 @@ class GangScheduler:
         self.use_native = use_native
         self.passes = 0
         self.gangs_scheduled = 0
+        # pod resource-request parse cache (requests are immutable for a pod's
+        # lifetime; parsing quantities for every bound pod on every pass was the
+        # dominant scheduler cost at 10k-pod scale)
+        self._req_cache: Dict[str, PodRequest] = {}
 
     # ------------------------------------------------------------------ pass
     def reconcile(self, _ns: str = "", _name: str = "") -> None:
         """One scheduling pass over the cluster."""
         self.passes += 1
-        nodes = self._build_node_views()
-        if not nodes:
-            return
         pods = self.store.list("Pod", copy_objects=False)
         bound: List[Obj] = []
         pending_by_gang: Dict[Tuple[str, str], List[Obj]] = {}
@@ -85,11 +86,20 @@ class GangScheduler:
                 pending_by_gang.setdefault(key, []).append(p)
             else:
                 pending_single.append(p)
+        pods_by_name = {(p["metadata"].get("namespace", "default"),
+                         p["metadata"]["name"]): p for p in pods}
+        if not pending_by_gang and not pending_single:
+            # nothing to place: skip the expensive node-view build/subtract
+            # entirely (the common pass during ready-wait churn at scale) —
+            # only the gang Ready rollup needs to run
+            self._rollup_ready(pods_by_name)
+            return
+        nodes = self._build_node_views()
+        if not nodes:
+            return
 
         self._subtract_bound(nodes, bound)
         node_list = list(nodes.values())
-        pods_by_name = {(p["metadata"].get("namespace", "default"),
-                         p["metadata"]["name"]): p for p in pods}
 
         # ---- gang scheduling, FIFO by PodGang creation
         gangs: List[Obj] = []
@@ -199,8 +209,11 @@ class GangScheduler:
             elif req.gpus:
                 del node.gpu_ids[: req.gpus]
 
-    @staticmethod
-    def _pod_request(pod: Obj) -> PodRequest:
+    def _pod_request(self, pod: Obj) -> PodRequest:
+        uid = pod["metadata"].get("uid", "")
+        cached = self._req_cache.get(uid)
+        if cached is not None:
+            return cached
         cpu = 0
         mem = 0.0
         gpus = 0
@@ -210,7 +223,12 @@ class GangScheduler:
             cpu += cpu_millis(r.get("cpu", 0))
             mem += parse_quantity(r.get("memory", 0))
             gpus += int(parse_quantity(r.get(c.AMD_GPU_RESOURCE, 0)))
-        return PodRequest(pod["metadata"]["name"], cpu, mem, gpus)
+        req = PodRequest(pod["metadata"]["name"], cpu, mem, gpus)
+        if uid:
+            if len(self._req_cache) > 50000:
+                self._req_cache.clear()  # bound memory across churny lifetimes
+            self._req_cache[uid] = req
+        return req
 
     def _gang_nodes(self, pg: Obj, pods_by_name: Dict) -> Set[str]:
         ns = pg["metadata"].get("namespace", "default")

@@ -66,10 +66,18 @@ def run(pods: int, workers: int = 8, profile: bool = False):
         if time.monotonic() - t0 > 900:
             break
     stop[0] = True
+    import os
+    t_spin = time.perf_counter()
+    x = 0
+    for i in range(2_000_000):
+        x += i * 3 // 7
+    spin_ms = (time.perf_counter() - t_spin) * 1000
     rec = {"pods": pods, "gangs": gangs, "nodes": nodes,
            "created_s": round(created, 1) if created else None,
            "all_ready_s": round(ready, 1) if ready else None,
-           "workers": workers}
+           "workers": workers,
+           "calibration": {"spin_ms": round(spin_ms, 1),
+                           "loadavg_1m": round(os.getloadavg()[0], 1)}}
     print(json.dumps(rec), flush=True)
     if profile:
         for k, v in counts.most_common(15):

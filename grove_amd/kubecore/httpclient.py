@@ -43,6 +43,38 @@ class HttpStoreClient:
         self.timeout = timeout
         self.token = token
         self._ctx = ssl.create_default_context(cafile=cafile) if cafile else None
+        # keep-alive connection per thread (urllib opens a fresh TCP connection
+        # per request — measured as the dominant wire-path overhead); watch
+        # streams still use urllib (dedicated long-lived connection)
+        import threading as _threading
+        import urllib.parse as _parse
+        u = _parse.urlsplit(self.base_url)
+        self._scheme = u.scheme
+        self._netloc = u.netloc
+        self._prefix = u.path.rstrip("/")
+        self._local = _threading.local()
+
+    def _conn(self):
+        import http.client
+        conn = getattr(self._local, "conn", None)
+        if conn is None:
+            if self._scheme == "https":
+                conn = http.client.HTTPSConnection(
+                    self._netloc, timeout=self.timeout, context=self._ctx)
+            else:
+                conn = http.client.HTTPConnection(self._netloc,
+                                                  timeout=self.timeout)
+            self._local.conn = conn
+        return conn
+
+    def _drop_conn(self):
+        conn = getattr(self._local, "conn", None)
+        if conn is not None:
+            try:
+                conn.close()
+            except Exception:
+                pass
+            self._local.conn = None
 
     def _headers(self, content: bool = False) -> Dict[str, str]:
         h: Dict[str, str] = {}
@@ -69,21 +101,34 @@ class HttpStoreClient:
             url += f"?{query}"
         return url
 
-    def _request(self, method: str, url: str, body: Optional[Obj] = None) -> Obj:
+    def _request(self, method: str, url: str, body: Optional[Obj] = None,
+                 content_type: str = "application/json") -> Obj:
         data = json.dumps(body).encode() if body is not None else None
-        req = urllib.request.Request(url, data=data, method=method,
-                                     headers=self._headers(content=True))
-        try:
-            with urllib.request.urlopen(req, timeout=self.timeout,
-                                        context=self._ctx) as r:
-                return json.loads(r.read())
-        except urllib.error.HTTPError as e:
+        path = url[len(f"{self._scheme}://{self._netloc}"):] \
+            if url.startswith(f"{self._scheme}://{self._netloc}") else url
+        headers = self._headers()
+        headers["Content-Type"] = content_type
+        for attempt in (0, 1):  # one retry on a stale keep-alive connection
+            conn = self._conn()
             try:
-                payload = json.loads(e.read())
+                conn.request(method, path, body=data, headers=headers)
+                resp = conn.getresponse()
+                raw = resp.read()
+            except ApiError:
+                raise
             except Exception:
-                payload = {}
-            raise ApiError(e.code, payload.get("reason", "HTTPError"),
-                           payload.get("message", str(e)))
+                self._drop_conn()
+                if attempt == 1:
+                    raise
+                continue
+            if resp.status >= 400:
+                try:
+                    payload = json.loads(raw)
+                except Exception:
+                    payload = {}
+                raise ApiError(resp.status, payload.get("reason", "HTTPError"),
+                               payload.get("message", f"HTTP {resp.status}"))
+            return json.loads(raw)
 
     # ------------------------------------------------------------------ store API
     def get(self, kind: str, namespace: Optional[str], name: str,
@@ -156,24 +201,9 @@ class HttpStoreClient:
                               patch: Obj) -> Obj:
         """client-go Patch(types.StrategicMergePatchType): merge-by-key list
         semantics for pod specs (containers by name etc.)."""
-        import json as _json
-        import urllib.request as _rq
-        url = self._url(kind, namespace, name)
-        req = _rq.Request(url, data=_json.dumps(patch).encode(), method="PATCH",
-                          headers={**self._headers(),
-                                   "Content-Type":
-                                       "application/strategic-merge-patch+json"})
-        import urllib.error as _er
-        try:
-            with _rq.urlopen(req, timeout=self.timeout, context=self._ctx) as r:
-                return _json.loads(r.read())
-        except _er.HTTPError as e:
-            try:
-                payload = _json.loads(e.read())
-            except Exception:
-                payload = {}
-            raise ApiError(e.code, payload.get("reason", "HTTPError"),
-                           payload.get("message", str(e)))
+        return self._request(
+            "PATCH", self._url(kind, namespace, name), patch,
+            content_type="application/strategic-merge-patch+json")
 
     def list_page(self, kind: str, namespace: Optional[str] = None,
                   label_selector: Optional[Dict[str, str]] = None,

@@ -61,6 +61,19 @@ class GangScheduler:
         # lifetime; parsing quantities for every bound pod on every pass was the
         # dominant scheduler cost at 10k-pod scale)
         self._req_cache: Dict[str, PodRequest] = {}
+        # scheduler cache (kube scheduler-cache analog): the live free-capacity
+        # view survives across passes; bound pods are subtracted INCREMENTALLY by
+        # uid instead of rebuilding + re-subtracting O(pods) node views every pass
+        # (that rebuild made total scheduling work O(pods^2) at 10k-pod scale).
+        # Invalidation: any Node add/remove/update (rv signature) or every
+        # _REBUILD_EVERY passes forces a full rebuild.
+        self._view: Optional[Dict[str, NodeFree]] = None
+        self._node_sig: Optional[tuple] = None
+        # pod uid -> (pool_key, PodRequest, gpu_ids) for release on termination
+        self._consumed: Dict[str, tuple] = {}
+        self._passes_since_rebuild = 0
+
+    _REBUILD_EVERY = 100
 
     # ------------------------------------------------------------------ pass
     def reconcile(self, _ns: str = "", _name: str = "") -> None:
@@ -94,11 +107,9 @@ class GangScheduler:
             # only the gang Ready rollup needs to run
             self._rollup_ready(pods_by_name)
             return
-        nodes = self._build_node_views()
+        nodes = self._current_view(bound)
         if not nodes:
             return
-
-        self._subtract_bound(nodes, bound)
         node_list = list(nodes.values())
 
         # ---- gang scheduling, FIFO by PodGang creation
@@ -135,6 +146,90 @@ class GangScheduler:
         self._rollup_ready(pods_by_name)
 
     # ------------------------------------------------------------------ node views
+    def _current_view(self, bound: List[Obj]) -> Dict[str, NodeFree]:
+        """Return the live free-capacity view, incrementally maintained."""
+        node_objs = self.store.list("Node", copy_objects=False)
+        sig = tuple(sorted((n["metadata"]["name"],
+                            n["metadata"].get("resourceVersion", ""))
+                           for n in node_objs))
+        self._passes_since_rebuild += 1
+        if (self._view is None or sig != self._node_sig
+                or self._passes_since_rebuild >= self._REBUILD_EVERY):
+            self._view = self._build_node_views()
+            self._node_sig = sig
+            self._consumed = {}
+            self._passes_since_rebuild = 0
+            self._subtract_bound(self._view, bound)
+            for p in bound:
+                self._note_consumed(p, None)
+            return self._view
+        pools_by_node: Optional[Dict[str, List[NodeFree]]] = None
+        active_uids = set()
+        for p in bound:
+            uid = p["metadata"].get("uid", "")
+            active_uids.add(uid)
+            if uid in self._consumed:
+                continue
+            if pools_by_node is None:
+                pools_by_node = {}
+                for nf in self._view.values():
+                    pools_by_node.setdefault(nf.node_name, []).append(nf)
+            self._subtract_one(pools_by_node, p)
+            self._note_consumed(p, None)
+        # release terminal/deleted pods' resources back to their pool
+        for uid in [u for u in self._consumed if u not in active_uids]:
+            pool_key, req, gpu_ids = self._consumed.pop(uid)
+            nf = self._view.get(pool_key)
+            if nf is None:
+                continue
+            nf.cpu_milli += req.cpu_milli
+            nf.mem_bytes += req.mem_bytes
+            nf.pods += 1
+            nf.gpu_ids = sorted(set(nf.gpu_ids) | set(gpu_ids))
+        return self._view
+
+    def _note_consumed(self, pod: Obj, pool_key: Optional[str]) -> None:
+        uid = pod["metadata"].get("uid", "")
+        if not uid:
+            return
+        req = self._pod_request(pod)
+        ids_str = (pod["metadata"].get("annotations") or {}).get(
+            GPU_IDS_ANNOTATION, "")
+        gpu_ids = [int(x) for x in ids_str.split(",") if x != ""]
+        if pool_key is None:
+            node = pod.get("spec", {}).get("nodeName", "")
+            pool_key = node
+            if self._view is not None and gpu_ids:
+                for nf in self._view.values():
+                    if nf.node_name == node and (set(gpu_ids) & set(nf.gpu_ids)
+                                                 or not nf.gpu_ids):
+                        pool_key = nf.name
+                        break
+        self._consumed[uid] = (pool_key, req, gpu_ids)
+
+    def _subtract_one(self, pools_by_node: Dict[str, List[NodeFree]],
+                      p: Obj) -> None:
+        pools = pools_by_node.get(p["spec"].get("nodeName", ""))
+        if not pools:
+            return
+        req = self._pod_request(p)
+        ids = (p["metadata"].get("annotations") or {}).get(GPU_IDS_ANNOTATION, "")
+        taken = {int(x) for x in ids.split(",") if x != ""} if ids else set()
+        node = pools[0]
+        if taken:
+            for nf in pools:
+                if taken & set(nf.gpu_ids):
+                    node = nf
+                    break
+        node.cpu_milli -= req.cpu_milli
+        node.mem_bytes -= req.mem_bytes
+        node.pods -= 1
+        if taken:
+            for nf in pools:
+                nf.gpu_ids = [g for g in nf.gpu_ids if g not in taken]
+        elif req.gpus:
+            del node.gpu_ids[: req.gpus]
+
     def _build_node_views(self) -> Dict[str, NodeFree]:
         """One NodeFree POOL per xGMI hive (schedule from the
         DISCOVERED fabric). The topology agent publishes the hive partition in the
@@ -424,11 +519,24 @@ class GangScheduler:
                     del n.gpu_ids[: req.gpus]
                     self._bind(p, Assignment(p["metadata"]["name"], n.node_name,
                                              gpu_ids))
+                    uid = p["metadata"].get("uid", "")
+                    if uid:
+                        # record against the POOL (n.name), not the node — a
+                        # multi-hive node's release must return GPUs to the
+                        # right hive pool
+                        self._consumed[uid] = (n.name, req, list(gpu_ids))
                     break
 
     # ------------------------------------------------------------------ bind
     def _bind(self, pod: Obj, a: Assignment) -> None:
         ns = pod["metadata"].get("namespace", "default")
+        # consumption was already applied to the live view by the placement —
+        # register it under the pod uid so the incremental pass never
+        # double-subtracts this pod when it shows up bound
+        uid = pod["metadata"].get("uid", "")
+        if uid:
+            self._consumed[uid] = (a.node, self._pod_request(pod),
+                                   list(a.gpu_ids))
 
         def apply(o: Obj) -> None:
             if o["metadata"].get("deletionTimestamp"):

@@ -452,3 +452,49 @@ def test_agent_multi_hive_labeling():
     with mock.patch.object(ag, "probe", return_value=info2):
         node2 = ag.discover_node("tn2")
     assert node2["metadata"]["labels"][c.NODE_LABEL_XGMI_HIVE] == "tn2-hive0"
+
+
+def test_scheduler_cache_matches_fresh_rebuild(cluster):
+    """The incremental scheduler cache (bound-pod accounting by uid) must agree
+    with a from-scratch node-view rebuild after churn: schedule gangs, complete
+    some pods, kill others, then compare free capacity pool by pool."""
+    cluster.add_virtual_nodes(3, gpus=4, cpu="64", pods=64)
+    for i in range(3):
+        cluster.apply({
+            "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
+            "metadata": {"name": f"cache{i}"},
+            "spec": {"replicas": 1, "template": {"cliques": [{
+                "name": "w", "spec": {
+                    "roleName": "w", "replicas": 2, "minAvailable": 2,
+                    "podSpec": {"containers": [{
+                        "name": "m", "image": "i",
+                        "resources": {"requests": {
+                            "cpu": "1", c.AMD_GPU_RESOURCE: "1"}}}]}}}]}}})
+        cluster.wait_pcs_available(f"cache{i}", timeout=20)
+    # churn: delete one whole PCS (releases), kill one pod (replaced)
+    cluster.delete_pcs("cache1")
+    cluster.wait_deleted(c.KIND_PCS, "cache1", timeout=20)
+    victim = cluster.store.list("Pod", "default",
+                                {c.LABEL_PODCLIQUE: "cache2-0-w"})[0]
+    cluster.store.delete("Pod", "default", victim["metadata"]["name"])
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "cache2-0-w"}, 2, timeout=20)
+    import time as _t
+    _t.sleep(0.3)  # let the scheduler run its incremental passes
+
+    sched = cluster.scheduler
+    pods = cluster.store.list("Pod", copy_objects=False)
+    bound = [p for p in pods
+             if p.get("spec", {}).get("nodeName")
+             and not p["metadata"].get("deletionTimestamp")
+             and (p.get("status") or {}).get("phase") not in ("Succeeded",
+                                                              "Failed")]
+    live = sched._current_view(bound)
+    fresh = sched._build_node_views()
+    sched._subtract_bound(fresh, bound)
+    assert set(live) == set(fresh)
+    for key in fresh:
+        a, b = live[key], fresh[key]
+        assert sorted(a.gpu_ids) == sorted(b.gpu_ids), \
+            f"{key}: gpu {sorted(a.gpu_ids)} != {sorted(b.gpu_ids)}"
+        assert a.cpu_milli == b.cpu_milli, f"{key}: cpu {a.cpu_milli} != {b.cpu_milli}"
+        assert a.pods == b.pods, f"{key}: pods {a.pods} != {b.pods}"

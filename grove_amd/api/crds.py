@@ -171,19 +171,62 @@ def write_crds(directory: str) -> List[str]:
     return paths
 
 
+_SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
 def install_crds(server_url: str) -> int:
-    """Apply rendered CRDs to an apiextensions-speaking apiserver over HTTP."""
+    """Apply rendered CRDs to an apiextensions-speaking apiserver over HTTP.
+
+    Running in-cluster (the install-crds init container, crdinstaller/installer.go
+    parity) it authenticates with the pod's projected ServiceAccount token and
+    trusts the cluster CA; AlreadyExists falls back to server-side-apply-style
+    PUT so upgrades replace the schema (GREP-436)."""
+    import os
+    import ssl
+    import urllib.error
     import urllib.request
+    headers = {"Content-Type": "application/json"}
+    ctx = None
+    token_path = os.path.join(_SA_DIR, "token")
+    if os.path.exists(token_path):
+        with open(token_path) as f:
+            headers["Authorization"] = f"Bearer {f.read().strip()}"
+    ca_path = os.path.join(_SA_DIR, "ca.crt")
+    if os.path.exists(ca_path):
+        ctx = ssl.create_default_context(cafile=ca_path)
+
+    def call(method: str, url: str, body) -> bool:
+        req = urllib.request.Request(url, data=json.dumps(body).encode(),
+                                     method=method, headers=headers)
+        try:
+            urllib.request.urlopen(req, timeout=15, context=ctx)
+            return True
+        except urllib.error.HTTPError as e:
+            if method == "POST" and e.code == 409:
+                return False  # exists — caller retries as update
+            return False
+        except Exception:
+            return False
+
+    base = f"{server_url}/apis/apiextensions.k8s.io/v1/customresourcedefinitions"
     n = 0
     for crd in render_all():
-        req = urllib.request.Request(
-            f"{server_url}/apis/apiextensions.k8s.io/v1/customresourcedefinitions"
-            f"?fieldManager={FIELD_MANAGER}",
-            data=json.dumps(crd).encode(), method="POST",
-            headers={"Content-Type": "application/json"})
-        try:
-            urllib.request.urlopen(req, timeout=10)
+        name = crd["metadata"]["name"]
+        if call("POST", f"{base}?fieldManager={FIELD_MANAGER}", crd):
             n += 1
+            continue
+        # update path: fetch current resourceVersion, then PUT the new schema
+        try:
+            with urllib.request.urlopen(
+                    urllib.request.Request(f"{base}/{name}", headers=headers),
+                    timeout=15, context=ctx) as r:
+                cur = json.loads(r.read())
+            crd = dict(crd)
+            crd["metadata"] = dict(crd["metadata"],
+                                   resourceVersion=cur["metadata"]
+                                   ["resourceVersion"])
+            if call("PUT", f"{base}/{name}?fieldManager={FIELD_MANAGER}", crd):
+                n += 1
         except Exception:
-            pass  # already exists / server applies its own merge
+            pass
     return n

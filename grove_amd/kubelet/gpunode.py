@@ -94,7 +94,8 @@ def run_payload_descriptor(kind: str, dims: Tuple[int, ...], device: int) -> Dic
             tflops = ext.burn_gemm(m, n, k, iters)
             return {"device": device, "kind": kind, "tflops": tflops}
         if kind == "stream":
-            n_floats, iters = dims
+            n_floats = dims[0]
+            iters = dims[1] if len(dims) > 1 else 2
             gbps = ext.stream_triad(n_floats, iters)
             return {"device": device, "kind": kind, "gbps": gbps}
         if kind == "decode":

@@ -173,6 +173,12 @@ class Cluster:
             pclq = md.get("labels", {}).get(c.LABEL_PODCLIQUE)
             if pclq:
                 self.c_pclq.enqueue(ns, pclq)
+            if ev in ("ADDED", "DELETED"):
+                pcs_name = md.get("labels", {}).get(c.LABEL_PART_OF)
+                if pcs_name:
+                    # pod-set membership changed -> the PCS structural sync
+                    # (PodGang podReferences etc.) must run again
+                    self.pcs_rec.bump_pod_epoch(ns, pcs_name)
             self.c_sched.enqueue("", "pass")
             if obj.get("spec", {}).get("nodeName"):
                 self.c_kubelet.enqueue(ns, md["name"])
@@ -199,6 +205,7 @@ class Cluster:
         def on_ctb(ev: str, obj: Obj, _old) -> None:
             self.c_ctb.enqueue("", obj["metadata"]["name"])
             # topology translation feeds PodGang specs: re-sync every PCS
+            self.pcs_rec.invalidate_sync_fingerprints()
             for p in self.store.list(c.KIND_PCS):
                 self.c_pcs.enqueue(p["metadata"].get("namespace", "default"),
                                    p["metadata"]["name"])

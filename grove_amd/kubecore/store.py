@@ -179,6 +179,8 @@ class Store:
         self._delete_validators: Dict[str, List[Callable[[Obj], None]]] = {}
         self.events: List[Obj] = []
         self._events_lock = threading.Lock()
+        import collections as _collections
+        self._pending_notify: "_collections.deque" = _collections.deque()
 
     # ------------------------------------------------------------------ admission
     def register_mutator(self, kind: str, fn: Callable[[Obj, Optional[Obj]], None]) -> None:
@@ -199,7 +201,11 @@ class Store:
 
     def _notify(self, tbl: _KindTable, ev: str, obj: Obj) -> None:
         # stored objects are immutable after insert; watchers share the reference
-        # (read-only contract, same as list(copy_objects=False))
+        # (read-only contract, same as list(copy_objects=False)).
+        # Called UNDER the store lock: only the resume history (rv-ordered) is
+        # appended here; the per-watcher queue fanout is deferred to
+        # _drain_notifications OUTSIDE the lock — queue.put wakes waiter threads
+        # and was a measured store-lock hold at 10k-pod scale.
         try:
             rv = int(obj["metadata"].get("resourceVersion", "0"))
         except (TypeError, ValueError):
@@ -207,8 +213,19 @@ class Store:
         tbl.history.append((rv, ev, obj))
         if len(tbl.history) > tbl.HISTORY_DEPTH:
             del tbl.history[: tbl.HISTORY_DEPTH // 2]
-        for w in list(tbl.watchers):
-            w.queue.put((ev, obj))
+        self._pending_notify.append((tbl, ev, obj))
+
+    def _drain_notifications(self) -> None:
+        """Deliver queued watch events outside the store lock. Per-watcher FIFO
+        order is preserved for events produced under the same lock hold; events
+        from different writers may interleave (level-triggered consumers)."""
+        while True:
+            try:
+                tbl, ev, obj = self._pending_notify.popleft()
+            except IndexError:
+                return
+            for w in list(tbl.watchers):
+                w.queue.put((ev, obj))
 
     def _next_rv(self) -> str:
         return str(next(self._rv))
@@ -256,6 +273,7 @@ class Store:
                     self._owned_by.setdefault(ref["uid"], set()).add(
                         (kind, ns, m["name"]))
             self._notify(tbl, ADDED, obj)
+        self._drain_notifications()
         return json_copy(obj)  # caller gets a private copy; stored one is immutable
 
     def get(self, kind: str, namespace: Optional[str], name: str,
@@ -310,14 +328,17 @@ class Store:
                                       if k in tbl.objects]
                         break
             if candidates is None:
+                # snapshot only (C-level dict copy); the python-level namespace/
+                # label filter loop runs OUTSIDE the lock — full-table scans at
+                # 10k objects were the dominant store-lock hold
                 candidates = list(tbl.objects.items())
-            refs = []
-            for (ns, _name), obj in candidates:
-                if namespace is not None and ns != namespace:
-                    continue
-                if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
-                    continue
-                refs.append(obj)
+        refs = []
+        for (ns, _name), obj in candidates:
+            if namespace is not None and ns != namespace:
+                continue
+            if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
+                continue
+            refs.append(obj)
         # filtering callbacks + copies run OUTSIDE the lock (objects immutable)
         if filter_fn is not None:
             refs = [o for o in refs if filter_fn(o)]
@@ -396,6 +417,7 @@ class Store:
                 if new["metadata"].get("deletionTimestamp") \
                         and not new["metadata"].get("finalizers"):
                     self._finalize_delete(kind, ns, m["name"])
+            self._drain_notifications()
             return json_copy(new) if return_copy else None
         raise conflict(kind, m["name"], "persistent write interleaving")
 
@@ -447,8 +469,9 @@ class Store:
                     marked["metadata"]["resourceVersion"] = self._next_rv()
                     tbl.objects[(ns, name)] = marked
                     self._notify(tbl, MODIFIED, marked)
-                return
-            self._finalize_delete(kind, ns, name, cascade=cascade)
+            else:
+                self._finalize_delete(kind, ns, name, cascade=cascade)
+        self._drain_notifications()
 
     def delete_collection(self, kind: str, namespace: Optional[str],
                           label_selector: Optional[Dict[str, str]] = None) -> int:

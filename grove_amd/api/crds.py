@@ -164,7 +164,14 @@ def write_crds(directory: str) -> List[str]:
     for crd in render_all():
         path = f"{directory}/{crd['metadata']['name']}.yaml"
         with open(path, "w") as f:
-            f.write("---\n")
+            f.write("# GENERATED FILE — do not edit. Rendered by grove_amd.api.crds\n"
+                    "# (python -m grove_amd install-crds --output-dir crds) from the\n"
+                    "# structural schemas in grove_amd/api/openapi.py; the embedded\n"
+                    "# standard-Kubernetes subtrees are vendored controller-gen\n"
+                    "# output (scripts/vendor_k8s_schemas.py). Byte-compatibility\n"
+                    "# with the reference CRD schema is the contract\n"
+                    "# (tests/test_crd_parity.py holds the semantic diff empty).\n"
+                    "---\n")
             yaml.dump(crd, f, Dumper=_NoAliasDumper, sort_keys=True,
                       default_flow_style=False)
         paths.append(path)

@@ -179,6 +179,7 @@ class Cluster:
                     # pod-set membership changed -> the PCS structural sync
                     # (PodGang podReferences etc.) must run again
                     self.pcs_rec.bump_pod_epoch(ns, pcs_name)
+            self.scheduler.note_pod_event(ev, obj, _old)
             self.c_sched.enqueue("", "pass")
             if obj.get("spec", {}).get("nodeName"):
                 self.c_kubelet.enqueue(ns, md["name"])
@@ -186,6 +187,7 @@ class Cluster:
         def on_podgang(ev: str, obj: Obj, _old) -> None:
             md = obj["metadata"]
             ns = md.get("namespace", "default")
+            self.scheduler.note_placement_event()
             self.c_sched.enqueue("", "pass")
             if ev != "DELETED":
                 self.c_podgang.enqueue(ns, md["name"])
@@ -198,6 +200,7 @@ class Cluster:
                 pass
 
         def on_node(ev: str, obj: Obj, _old) -> None:
+            self.scheduler.note_placement_event()
             self.c_sched.enqueue("", "pass")
             if ev in ("DELETED", "MODIFIED"):
                 self.c_node.enqueue("", obj["metadata"]["name"])

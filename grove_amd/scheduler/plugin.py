@@ -72,13 +72,54 @@ class GangScheduler:
         # pod uid -> (pool_key, PodRequest, gpu_ids) for release on termination
         self._consumed: Dict[str, tuple] = {}
         self._passes_since_rebuild = 0
+        # event classification (fed by the cluster watch): a pass that was
+        # triggered ONLY by pods becoming Ready does a per-gang rollup instead of
+        # the O(all pods) classification scan — ready-churn dominates pass volume
+        # at 10k-pod scale
+        self._dirty_placement = True
+        self._dirty_ready_gangs: set = set()
+        self._rollup_forced = 0
 
     _REBUILD_EVERY = 100
+
+    # ------------------------------------------------------------------ events
+    def note_pod_event(self, ev: str, obj, old) -> None:
+        spec = obj.get("spec") or {}
+        st = obj.get("status") or {}
+        labels = obj["metadata"].get("labels") or {}
+        if ev == "DELETED" or st.get("phase") in ("Succeeded", "Failed"):
+            if spec.get("nodeName"):
+                self._dirty_placement = True  # capacity released
+            return
+        if not spec.get("nodeName"):
+            self._dirty_placement = True  # new/ungated pod may be placeable
+            return
+        was_ready = old is not None and cond.pod_is_ready(old)
+        if cond.pod_is_ready(obj) and not was_ready:
+            gang = labels.get(c.LABEL_PODGANG)
+            if gang:
+                self._dirty_ready_gangs.add(
+                    (obj["metadata"].get("namespace", "default"), gang))
+
+    def note_placement_event(self) -> None:
+        """PodGang/Node change: placement inputs moved."""
+        self._dirty_placement = True
 
     # ------------------------------------------------------------------ pass
     def reconcile(self, _ns: str = "", _name: str = "") -> None:
         """One scheduling pass over the cluster."""
         self.passes += 1
+        self._rollup_forced += 1
+        if not self._dirty_placement and self._rollup_forced < 50:
+            # ready-churn only: per-gang rollup, no cluster-wide scan
+            gangs = self._dirty_ready_gangs
+            self._dirty_ready_gangs = set()
+            if gangs:
+                self._rollup_gangs(gangs)
+            return
+        self._dirty_placement = False
+        self._dirty_ready_gangs = set()
+        self._rollup_forced = 0
         pods = self.store.list("Pod", copy_objects=False)
         bound: List[Obj] = []
         pending_by_gang: Dict[Tuple[str, str], List[Obj]] = {}
@@ -564,6 +605,35 @@ class GangScheduler:
                              "bind pod", e)
 
     # ------------------------------------------------------------------ ready rollup
+    def _rollup_gangs(self, gangs: set) -> None:
+        """Targeted Ready rollup for gangs whose pods just became Ready."""
+        for (ns, gname) in gangs:
+            pg = self.store.try_get(c.KIND_PODGANG, ns, gname, copy=False)
+            if pg is None or not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED) \
+                    or cond.condition_true(pg, c.PODGANG_COND_READY):
+                continue
+            pods = {p["metadata"]["name"]: p for p in self.store.list(
+                "Pod", ns, {c.LABEL_PODGANG: gname}, copy_objects=False)}
+            ready = True
+            for group in (pg.get("spec") or {}).get("podgroups") or []:
+                n = sum(1 for ref in group.get("podReferences") or []
+                        if (p := pods.get(ref.get("name", ""))) is not None
+                        and cond.pod_is_ready(p))
+                if n < int(group.get("minReplicas", 0)):
+                    ready = False
+                    break
+            if ready:
+                def mark(o: Obj) -> None:
+                    cond.set_condition(o, c.PODGANG_COND_READY, True,
+                                       "AllPodGroupsReady")
+                    o["status"]["phase"] = "Running"
+                try:
+                    self.store.patch(c.KIND_PODGANG, ns, gname, mark,
+                                     status=True, return_copy=False)
+                except ApiError as e:
+                    report_api_error(self.store, c.KIND_PODGANG, ns, gname,
+                                     "mark gang ready", e)
+
     def _rollup_ready(self, pods_by_name: Dict) -> None:
         for pg in self.store.list(c.KIND_PODGANG, copy_objects=False):
             if not cond.condition_true(pg, c.PODGANG_COND_SCHEDULED):

@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import logging
 import time
-from typing import List, Optional
+from typing import List, Optional, Tuple
 
 from ..api import constants as c
 from ..kubecore.store import Store, Obj, ApiError
@@ -38,8 +38,8 @@ class PodCliqueReconciler:
         if pclq["metadata"].get("deletionTimestamp"):
             return self._reconcile_delete(pclq)
         rec = groveerr.StepRecorder(self.store, c.KIND_PCLQ, namespace, name)
-        res = self._reconcile_spec(pclq, rec)
-        self._reconcile_status(namespace, name)
+        res, pods = self._reconcile_spec(pclq, rec)
+        self._reconcile_status(namespace, name, pods=pods)
         rec.flush()
         return res
 
@@ -66,8 +66,8 @@ class PodCliqueReconciler:
             "Pod", pclq["metadata"].get("namespace"),
             {c.LABEL_PODCLIQUE: pclq["metadata"]["name"]}, copy_objects=False)
 
-    def _reconcile_spec(self, pclq: Obj,
-                        rec: groveerr.StepRecorder) -> Result:
+    def _reconcile_spec(self, pclq: Obj, rec: groveerr.StepRecorder
+                        ) -> Tuple[Result, List[Obj]]:
         ns = pclq["metadata"].get("namespace")
         desired = int(pclq["spec"].get("replicas", 1))
         pods = self._owned_pods(pclq)
@@ -101,7 +101,7 @@ class PodCliqueReconciler:
             in_use = [int(p["metadata"]["labels"].get(c.LABEL_POD_INDEX, -1)) for p in pods]
             pcs = self._find_pcs(pclq)
             if pcs is None:
-                return Result(requeue_after=0.1)
+                return Result(requeue_after=0.1), pods
             num_pods = self._pcsg_template_num_pods(pcs, pclq)
             indices = available_indices([i for i in in_use if i >= 0], desired - n)
             if len(indices) <= 2:
@@ -137,9 +137,10 @@ class PodCliqueReconciler:
             from .podgang_component import try_complete_podgang
             try_complete_podgang(self.store, ns, gang_name)
         # the maintained pod list (creates appended, deletes removed) saves one
-        # store list per reconcile — a measured hotspot at 10k-pod scale
+        # store list per reconcile — a measured hotspot at 10k-pod scale; the
+        # status pass reuses the same list
         self._remove_scheduling_gates(pclq, rec, pods=pods)
-        return Result.DONE
+        return Result.DONE, pods
 
     @staticmethod
     def _deletion_order(pods: List[Obj]) -> List[Obj]:
@@ -246,11 +247,13 @@ class PodCliqueReconciler:
         return True
 
     # ------------------------------------------------------------------ status
-    def _reconcile_status(self, namespace: str, name: str) -> None:
+    def _reconcile_status(self, namespace: str, name: str,
+                          pods: Optional[List[Obj]] = None) -> None:
         pclq = self.store.try_get(c.KIND_PCLQ, namespace, name, copy=False)
         if pclq is None or pclq["metadata"].get("deletionTimestamp"):
             return
-        pods = self._owned_pods(pclq)
+        if pods is None:
+            pods = self._owned_pods(pclq)
         tmpl_hash = pclq["metadata"]["labels"].get(c.LABEL_POD_TEMPLATE_HASH, "")
         n_total = len(pods)
         n_ready = sum(1 for p in pods if cond.pod_is_ready(p))

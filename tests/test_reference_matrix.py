@@ -519,3 +519,64 @@ def test_ru18_clique_scale_out_during_update(cluster):
             and cond.pod_is_ready(p) for p in ps)
     cluster.wait_for(all_updated, timeout=40,
                      desc="3 pods, all on the new template")
+
+
+def test_gs11_pcs_and_pcsg_scaling_with_min_replicas(cluster):
+    """GS11 (gang_scheduling_test.go:886): PCS replicas AND PCSG replicas scaled
+    while capacity only fits the gang minima — base gangs admit at minAvailable,
+    scaled gangs wait, everything lands once capacity arrives."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 1}]
+    cluster.add_virtual_nodes(1, cpu="2", pods=64)
+    cluster.apply(_pcs("gs11", replicas=1, cliques=(("a", 1, 1), ("b", 1, 1)),
+                       sg=sg))
+    # base gang (a + sg replica 0) = 2 pods fits exactly; scaled replica 1 pends
+    cluster.wait_pods_ready({c.LABEL_PCS_REPLICA_INDEX: "0"}, 2, timeout=20)
+    sc = _pods(cluster, {c.LABEL_PODCLIQUE: "gs11-0-sg-1-b"})
+    assert sc and all(not p["spec"].get("nodeName") for p in sc)
+    # scale the PCS out too: replica 1's tree pends entirely
+    cluster.store.patch(c.KIND_PCS, "default", "gs11",
+                        lambda o: o["spec"].update(replicas=2))
+    cluster.wait_for(
+        lambda: len(_pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})) == 3,
+        timeout=20, desc="replica 1 pods created")
+    time.sleep(0.3)
+    assert all(not p["spec"].get("nodeName")
+               for p in _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"}))
+    cluster.add_virtual_nodes(3, cpu="4", pods=64, prefix="cap")
+    cluster.wait_pcs_available("gs11", timeout=30, min_available=2)
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "gs11"}, 6, timeout=30)
+
+
+def test_od7_on_delete_multiple_replicas(cluster):
+    """OD7 (update/ondelete_test.go:447): OnDelete across several PCS replicas —
+    each replica's pods wait for their own manual deletion; deleting one
+    replica's pod must not touch the others."""
+    cluster.add_virtual_nodes(2, cpu="16", pods=64)
+    cluster.apply(_pcs("od7", replicas=3, cliques=(("w", 1, 1),),
+                       strategy=c.UPDATE_ON_DELETE))
+    cluster.wait_pcs_available("od7", timeout=20)
+    pods0 = {p["metadata"]["uid"]: p
+             for p in _pods(cluster, {c.LABEL_PART_OF: "od7"})}
+    old_hash = next(iter(pods0.values()))["metadata"]["labels"][
+        c.LABEL_POD_TEMPLATE_HASH]
+    cur = cluster.store.get(c.KIND_PCS, "default", "od7")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    time.sleep(0.4)
+    assert {p["metadata"]["uid"]
+            for p in _pods(cluster, {c.LABEL_PART_OF: "od7"})} == set(pods0)
+    # delete replica 1's pod only
+    victim = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})[0]
+    cluster.store.delete("Pod", "default", victim["metadata"]["name"])
+
+    def replica1_updated():
+        ps = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})
+        return len(ps) == 1 and ps[0]["metadata"]["labels"][
+            c.LABEL_POD_TEMPLATE_HASH] != old_hash and cond.pod_is_ready(ps[0])
+    cluster.wait_for(replica1_updated, timeout=20, desc="replica 1 updated")
+    # replicas 0 and 2 untouched (still old template, same uids)
+    for r in ("0", "2"):
+        ps = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: r})
+        assert ps[0]["metadata"]["uid"] in pods0
+        assert ps[0]["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == old_hash

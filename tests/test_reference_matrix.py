@@ -580,3 +580,33 @@ def test_od7_on_delete_multiple_replicas(cluster):
         ps = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: r})
         assert ps[0]["metadata"]["uid"] in pods0
         assert ps[0]["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == old_hash
+
+
+def test_gt3_min_replicas_pcs_owned_termination(cluster):
+    """GT3 (gang_termination_test.go:119): a clique running at replicas=3,
+    minAvailable=2 tolerates ONE pod loss (no termination), but a second loss
+    breaches minAvailable and gang-terminates the replica after the delay."""
+    cluster.add_virtual_nodes(3, cpu="8", pods=64)
+    cluster.apply(_pcs("gt3", cliques=(("w", 3, 2),),
+                       termination_delay="400ms"))
+    cluster.wait_pcs_available("gt3", timeout=20)
+    uid0 = cluster.store.get(c.KIND_PCLQ, "default", "gt3-0-w")["metadata"]["uid"]
+    _cordon_all(cluster)
+    # one loss: 2 >= minAvailable -> tolerated
+    victims = _pods(cluster, {c.LABEL_PODCLIQUE: "gt3-0-w"})
+    cluster.store.delete("Pod", "default", victims[0]["metadata"]["name"])
+    time.sleep(0.9)  # > terminationDelay
+    assert cluster.store.get(c.KIND_PCLQ, "default",
+                             "gt3-0-w")["metadata"]["uid"] == uid0, \
+        "one pod loss within minAvailable must not gang-terminate"
+    # second loss: 1 < minAvailable -> breach -> termination fires
+    left = [p for p in _pods(cluster, {c.LABEL_PODCLIQUE: "gt3-0-w"})
+            if p["spec"].get("nodeName")]
+    cluster.store.delete("Pod", "default", left[0]["metadata"]["name"])
+
+    def recreated():
+        q = cluster.store.try_get(c.KIND_PCLQ, "default", "gt3-0-w")
+        return q is not None and q["metadata"]["uid"] != uid0
+    cluster.wait_for(recreated, timeout=25, desc="gang termination after breach")
+    _cordon_all(cluster, False)
+    cluster.wait_pcs_available("gt3", timeout=30)

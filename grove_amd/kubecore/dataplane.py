@@ -110,6 +110,12 @@ class DataPlane:
         headers = dict(scope.get("headers") or [])
         try:
             kind = self._kind(plural)
+            if status_sub and method in ("POST", "DELETE"):
+                # the status subresource only supports read/update/patch —
+                # a DELETE here must never delete the parent object
+                raise ApiError(405, "MethodNotAllowed",
+                               f"{method} is not supported on the status "
+                               "subresource")
             if method == "GET" and name is None:
                 if q.get("watch") in ("true", "1"):
                     await self._watch(send, kind, q)

@@ -162,3 +162,18 @@ def test_strategic_merge_env_and_volume_mounts(served):
     env = {e["name"]: e["value"] for e in main["env"]}
     assert env == {"A": "1", "B": "override", "C": "3"}
     assert sorted(m["mountPath"] for m in main["volumeMounts"]) == ["/a", "/b"]
+
+
+def test_status_subresource_method_guard(served):
+    """DELETE/POST on .../status must be rejected (405), never touch the parent."""
+    cluster, api, client = served
+    cluster.store.create(_mkpod(0))
+    req = urllib.request.Request(
+        f"{api.url}/api/v1/namespaces/default/pods/wp000/status",
+        method="DELETE")
+    try:
+        urllib.request.urlopen(req, timeout=5)
+        assert False, "expected 405"
+    except urllib.error.HTTPError as e:
+        assert e.code == 405
+    assert cluster.store.try_get("Pod", "default", "wp000") is not None

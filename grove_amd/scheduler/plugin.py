@@ -79,6 +79,7 @@ class GangScheduler:
         self._dirty_placement = True
         self._dirty_ready_gangs: set = set()
         self._rollup_forced = 0
+        self._pool_of_gpu: Dict[tuple, str] = {}
 
     _REBUILD_EVERY = 100
 
@@ -200,6 +201,12 @@ class GangScheduler:
             self._node_sig = sig
             self._consumed = {}
             self._passes_since_rebuild = 0
+            # authoritative (node, gpu) -> pool map BEFORE subtraction: release
+            # must return a pod's GPUs to the hive pool they came from, which the
+            # post-subtraction free sets can no longer tell us
+            self._pool_of_gpu = {
+                (nf.node_name, g): key
+                for key, nf in self._view.items() for g in nf.gpu_ids}
             self._subtract_bound(self._view, bound)
             for p in bound:
                 self._note_consumed(p, None)
@@ -240,12 +247,9 @@ class GangScheduler:
         if pool_key is None:
             node = pod.get("spec", {}).get("nodeName", "")
             pool_key = node
-            if self._view is not None and gpu_ids:
-                for nf in self._view.values():
-                    if nf.node_name == node and (set(gpu_ids) & set(nf.gpu_ids)
-                                                 or not nf.gpu_ids):
-                        pool_key = nf.name
-                        break
+            if gpu_ids:
+                pool_key = getattr(self, "_pool_of_gpu", {}).get(
+                    (node, gpu_ids[0]), node)
         self._consumed[uid] = (pool_key, req, gpu_ids)
 
     def _subtract_one(self, pools_by_node: Dict[str, List[NodeFree]],

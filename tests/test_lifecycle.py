@@ -188,7 +188,7 @@ class TestRollingUpdate:
         cluster.c_pcs.enqueue("default", "ru3")
         # observe: never more than one replica concurrently updating
         max_concurrent = 0
-        deadline = time.monotonic() + 30
+        deadline = time.monotonic() + 60  # generous: suite may share a loaded box
         while time.monotonic() < deadline:
             p = cluster.store.get(c.KIND_PCS, "default", "ru3")
             prog = (p.get("status") or {}).get("updateProgress") or {}

@@ -610,3 +610,37 @@ def test_gt3_min_replicas_pcs_owned_termination(cluster):
     cluster.wait_for(recreated, timeout=25, desc="gang termination after breach")
     _cordon_all(cluster, False)
     cluster.wait_pcs_available("gt3", timeout=30)
+
+
+def test_gt4_min_replicas_pcsg_owned_termination(cluster):
+    """GT4 (gang_termination_test.go:170): a PCSG at replicas=3, minAvailable=2
+    tolerates losing ONE member replica; losing a second drops
+    availableReplicas below minAvailable and gang-terminates the PCS replica."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 3, "minAvailable": 2}]
+    cluster.add_virtual_nodes(4, cpu="8", pods=64)
+    cluster.apply(_pcs("gt4", cliques=(("b", 1, 1),), sg=sg,
+                       termination_delay="400ms"))
+    cluster.wait_pcs_available("gt4", timeout=20)
+    sg_uid = cluster.store.get(c.KIND_PCSG, "default", "gt4-0-sg")["metadata"]["uid"]
+    _cordon_all(cluster)
+    # lose replica 2 (scaled): 2 available >= minAvailable -> recycle only, no
+    # PCS-scope termination
+    v2 = _pods(cluster, {c.LABEL_PODCLIQUE: "gt4-0-sg-2-b"})[0]
+    cluster.store.delete("Pod", "default", v2["metadata"]["name"])
+    time.sleep(0.9)
+    assert cluster.store.get(c.KIND_PCSG, "default",
+                             "gt4-0-sg")["metadata"]["uid"] == sg_uid
+    # lose a base replica too: 1 available < minAvailable -> breach -> the PCS
+    # replica is gang-terminated (member PCLQs recreated with fresh uids)
+    uid_b0 = cluster.store.get(c.KIND_PCLQ, "default",
+                               "gt4-0-sg-0-b")["metadata"]["uid"]
+    v0 = _pods(cluster, {c.LABEL_PODCLIQUE: "gt4-0-sg-0-b"})[0]
+    cluster.store.delete("Pod", "default", v0["metadata"]["name"])
+
+    def terminated():
+        q = cluster.store.try_get(c.KIND_PCLQ, "default", "gt4-0-sg-0-b")
+        return q is not None and q["metadata"]["uid"] != uid_b0
+    cluster.wait_for(terminated, timeout=25,
+                     desc="PCS-scope termination after PCSG minAvailable breach")
+    _cordon_all(cluster, False)
+    cluster.wait_pcs_available("gt4", timeout=30)

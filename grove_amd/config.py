@@ -15,7 +15,7 @@ import yaml
 from .api import constants as c
 from .kubecore.store import invalid
 
-ALLOWED_SCHEDULERS = (c.SCHEDULER_AMD_GANG, c.SCHEDULER_DEFAULT)
+ALLOWED_SCHEDULERS = (c.SCHEDULER_AMD_GANG, c.SCHEDULER_DEFAULT, "lpx-scheduler")
 
 
 @dataclasses.dataclass

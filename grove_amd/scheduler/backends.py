@@ -130,13 +130,41 @@ class AmdGangBackend(Backend):
         return None
 
 
+class LpxBackend(Backend):
+    """Name-only backend (scheduler/lpx/backend.go:86): sets schedulerName and
+    REJECTS grove topology constraints at admission (backend.go:68-86) — lpx has
+    no topology support, so constrained workloads must not silently lose their
+    packing guarantee."""
+
+    name = "lpx-scheduler"
+
+    def sync_podgang(self, podgang: Obj) -> None:
+        return  # no gang resource; lpx consumes pods directly
+
+    def validate_podcliqueset(self, pcs: Obj) -> None:
+        tmpl = (pcs.get("spec") or {}).get("template") or {}
+        sites = [("spec.template.topologyConstraint",
+                  tmpl.get("topologyConstraint"))]
+        sites += [(f"spec.template.cliques[{cl.get('name')}].topologyConstraint",
+                   cl.get("topologyConstraint"))
+                  for cl in tmpl.get("cliques") or []]
+        sites += [(f"spec.template.podCliqueScalingGroups[{sg.get('name')}]"
+                   ".topologyConstraint", sg.get("topologyConstraint"))
+                  for sg in tmpl.get("podCliqueScalingGroups") or []]
+        for path, tc in sites:
+            if tc:
+                raise invalid(
+                    f"{path}: topology constraints are not supported by the "
+                    f"{self.name} backend")
+
+
 class Registry:
     """scheduler/registry/registry.go:45-115 equivalent."""
 
     def __init__(self, store: Store, default: str = c.SCHEDULER_AMD_GANG,
                  backends: Optional[List[Backend]] = None):
         self._by_name: Dict[str, Backend] = {}
-        for b in backends or [AmdGangBackend(), KubeBackend()]:
+        for b in backends or [AmdGangBackend(), KubeBackend(), LpxBackend()]:
             b.init(store)
             self._by_name[b.name] = b
         if default not in self._by_name:

@@ -464,3 +464,32 @@ def test_http_auth_identity(served_cluster, simple1_yaml):
             assert r.status == 200
     finally:
         api2.stop()
+
+
+def test_lpx_backend_rejects_topology_constraints():
+    """lpx backend parity (scheduler/lpx/backend.go:68-86): schedulerName-only
+    dispatch; workloads with topology constraints are rejected at validation."""
+    from grove_amd.kubecore.store import Store, ApiError
+    from grove_amd.scheduler.backends import Registry
+    st = Store()
+    reg = Registry(st, default="lpx-scheduler")
+    lpx = reg.get("lpx-scheduler")
+    pod = {"spec": {}}
+    lpx.prepare_pod(pod)
+    assert pod["spec"]["schedulerName"] == "lpx-scheduler"
+    ok_pcs = {"spec": {"template": {"cliques": [{"name": "a", "spec": {}}]}}}
+    lpx.validate_podcliqueset(ok_pcs)  # no constraints -> fine
+    bad = {"spec": {"template": {
+        "cliques": [{"name": "a", "spec": {},
+                     "topologyConstraint": {"pack": {"required": "rack"}}}]}}}
+    with pytest.raises(ApiError, match="not supported by the lpx-scheduler"):
+        lpx.validate_podcliqueset(bad)
+    # config accepts the lpx profile
+    from grove_amd.config import load_configuration
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".yaml", delete=False) as f:
+        f.write("scheduler: {default: lpx-scheduler, "
+                "profiles: [lpx-scheduler, amd-gang-scheduler]}\n")
+        path = f.name
+    cfg = load_configuration(path)
+    assert cfg.default_scheduler == "lpx-scheduler"

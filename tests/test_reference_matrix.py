@@ -644,3 +644,54 @@ def test_gt4_min_replicas_pcsg_owned_termination(cluster):
                      desc="PCS-scope termination after PCSG minAvailable breach")
     _cordon_all(cluster, False)
     cluster.wait_pcs_available("gt4", timeout=30)
+
+
+def test_tas12_large_scaling_ratio(cluster):
+    """TAS12 (topology_test.go:774): a PCSG scaled to MANY replicas under a
+    host-pack constraint — every one of the 6 replicas (each 2 pods) must pack
+    onto one host; the scheduler fans them across hosts without violating any
+    per-replica constraint."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=3, per_rack=2, gpus=4)
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 6, "minAvailable": 2,
+           "topologyConstraint": {"pack": {"required": "host"}}}]
+    pcs = _pcs("tas12", cliques=(("b", 2, 2),), sg=sg, gpus=1)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas12", timeout=30)
+    # PCS availability only needs minAvailable PCSG replicas — wait for ALL 12
+    # pods (scaled replicas included) before asserting per-replica placement
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "tas12"}, 12, timeout=30)
+    for j in range(6):
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: f"tas12-0-sg-{j}-b"})
+        assert len(ps) == 2
+        assert len({p["spec"]["nodeName"] for p in ps}) == 1, \
+            f"PCSG replica {j} spans hosts"
+
+
+def test_ru21_clique_scale_in_before_update(cluster):
+    """RU20/RU21 (rolling_recreate_test.go:832,897): scaling a PodClique IN just
+    before/while the template rolls — the update completes across the reduced
+    set and no orphan pods survive."""
+    cluster.add_virtual_nodes(2, cpu="16", pods=64)
+    pcs = _pcs("ru21", cliques=(("w", 4, 1),))
+    pcs["spec"]["template"]["cliques"][0]["spec"]["autoScalingConfig"] = {
+        "minReplicas": 1, "maxReplicas": 6}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("ru21", timeout=20)
+    # HPA-style scale-in first, then the template change
+    cluster.store.patch(c.KIND_PCLQ, "default", "ru21-0-w",
+                        lambda o: o["spec"].update(replicas=2))
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru21")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "w", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def settled():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "ru21-0-w"})
+        return len(ps) == 2 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(settled, timeout=40, desc="2 pods, all on new template")

@@ -593,8 +593,12 @@ def _drive(plane, comm: Comm, args, gang_size: int,
     # priming (setup, before the W official warmup steps): a cold box pays
     # page-cache/HIP-module/thread-pool costs on the first gangs — measured 222 vs
     # 330 gangs/s first-run-vs-steady on one box. A fixed handful of priming gangs
-    # reaches steady state regardless of the driver's chosen -W.
-    run_steps(4, "prime")
+    # reaches steady state regardless of the driver's chosen -W. The wire plane
+    # needs a longer ramp (TCP/uvicorn/operator-process warmup: a 2000-step soak
+    # sustains 194 gangs/s where 4-prime short runs report ~120).
+    prime = int(os.environ.get("GROVE_BENCH_PRIME",
+                               "16" if plane.transport == "http" else "4"))
+    run_steps(prime, "prime")
     run_steps(args.warmup, "warm")
 
     # RCCL-over-xGMI evidence: bus bandwidth of a 256 MB all-reduce across all ranks

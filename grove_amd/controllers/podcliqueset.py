@@ -140,6 +140,8 @@ class PodCliqueSetReconciler:
     # ------------------------------------------------------------------ delete
     def _reconcile_delete(self, pcs: Obj) -> Result:
         ns, name = pcs["metadata"].get("namespace"), pcs["metadata"]["name"]
+        self._sync_fp.pop(f"{ns}/{name}", None)
+        self.pod_epoch.pop(f"{ns}/{name}", None)
         rec = groveerr.StepRecorder(self.store, c.KIND_PCS, ns, name)
         sel = {c.LABEL_PART_OF: name}
         remaining = 0

@@ -368,3 +368,23 @@ def test_pcsg_startup_dependency_scoping(cluster):
     for j in range(3):
         ld = cluster.store.get(c.KIND_PCLQ, "default", f"dep-0-sg-{j}-ld")
         assert ld["spec"]["startsAfter"] == []
+
+
+def test_child_drift_repaired_despite_sync_fingerprint(cluster):
+    """The structural-sync fingerprint must never mask child drift: deleting a
+    PodClique out from under the PCS (a structural change with no PCS spec
+    change) gets repaired — the fingerprint includes child generations, so the
+    vanished child forces a full resync."""
+    cluster.add_virtual_nodes(1)
+    pcs = _gpu_pcs("drift", cliques=(("a", 1, 1), ("b", 1, 1)), gpus_per_pod=0)
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("drift", timeout=20)
+    uid = cluster.store.get(c.KIND_PCLQ, "default", "drift-0-b")["metadata"]["uid"]
+    cluster.store.delete(c.KIND_PCLQ, "default", "drift-0-b")
+
+    def repaired():
+        q = cluster.store.try_get(c.KIND_PCLQ, "default", "drift-0-b")
+        return q is not None and q["metadata"]["uid"] != uid \
+            and not q["metadata"].get("deletionTimestamp")
+    cluster.wait_for(repaired, timeout=20, desc="deleted PodClique recreated")
+    cluster.wait_pcs_available("drift", timeout=20)

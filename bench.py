@@ -315,22 +315,9 @@ class HttpPlane:
         self._deferred: List[Obj] = []  # startup-dep-blocked pods, re-checked in drain
         self._seen: set = set()
         self._stop = threading.Event()
-        self.tracker = RemoteTracker(self.client).start()
-        self._watch_thread = threading.Thread(target=self._pump_pods, daemon=True)
-        self._watch_thread.start()
-
-    # -- node-agent side: collect dispatchable pods from the wire watch
-    def _pump_pods(self) -> None:
-        try:
-            for ev, pod in self.client.watch_events("Pod", None, seed=True):
-                if self._stop.is_set():
-                    return
-                try:
-                    self._on_pod(ev, pod)
-                except Exception:
-                    pass
-        except Exception:
-            pass
+        # one wire stream serves both the dispatch collector (tap) and the
+        # tracker's milestone handler
+        self.tracker = RemoteTracker(self.client, pod_tap=self._on_pod).start()
 
     def _on_pod(self, ev: str, pod: Obj) -> None:
         if ev == "DELETED":

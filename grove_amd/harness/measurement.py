@@ -117,17 +117,31 @@ class RemoteTracker(Tracker):
     of the deployable (multi-process) shape, like the reference's client-go-based
     e2e/measurement/measurement.go:29-104 harness."""
 
-    def __init__(self, client):
+    def __init__(self, client, pod_tap=None):
         super().__init__(store=None)  # type: ignore[arg-type]
         self.client = client
+        # optional tap: every Pod event is ALSO forwarded here (lets a co-located
+        # consumer — e.g. the bench node agent's dispatch collector — share one
+        # wire stream instead of opening its own)
+        self._pod_tap = pod_tap
 
     def start(self) -> "RemoteTracker":
-        for kind, handler in (("Pod", self._on_pod), (c.KIND_PODGANG, self._on_gang)):
+        handlers = [("Pod", self._pod_with_tap),
+                    (c.KIND_PODGANG, self._on_gang)]
+        for kind, handler in handlers:
             t = threading.Thread(target=self._pump_remote, args=(kind, handler),
                                  daemon=True)
             t.start()
             self._threads.append(t)
         return self
+
+    def _pod_with_tap(self, ev, obj):
+        if self._pod_tap is not None:
+            try:
+                self._pod_tap(ev, obj)
+            except Exception:
+                pass
+        self._on_pod(ev, obj)
 
     def _pump_remote(self, kind, handler):
         try:

@@ -111,6 +111,34 @@ class Tracker:
             w.stop()
 
 
+    # ---- reporting ----
+    def wait_all_running(self, timeout: float = 60.0) -> bool:
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            with self._lock:
+                if all(t.running is not None for t in self.gangs.values()):
+                    return True
+            time.sleep(0.002)
+        return False
+
+    def summary(self) -> Dict[str, Any]:
+        with self._lock:
+            ttr = [(t.running - t.submitted) * 1000 for t in self.gangs.values()
+                   if t.running is not None]
+            tts = [(t.scheduled - t.submitted) * 1000 for t in self.gangs.values()
+                   if t.scheduled is not None]
+            n = len(self.gangs)
+            done = sum(1 for t in self.gangs.values() if t.running is not None)
+        return {
+            "gangs_total": n,
+            "gangs_running": done,
+            "p50_time_to_running_ms": percentile(ttr, 50),
+            "p95_time_to_running_ms": percentile(ttr, 95),
+            "max_time_to_running_ms": max(ttr) if ttr else None,
+            "p50_time_to_scheduled_ms": percentile(tts, 50),
+        }
+
+
 class RemoteTracker(Tracker):
     """Tracker over the wire: pumps the apiserver's ndjson watch streams through an
     HttpStoreClient instead of in-process store watches — the measurement instrument
@@ -155,29 +183,3 @@ class RemoteTracker(Tracker):
         except Exception:
             pass  # stream torn down (server stop / tracker stop)
 
-    # ---- reporting ----
-    def wait_all_running(self, timeout: float = 60.0) -> bool:
-        deadline = time.monotonic() + timeout
-        while time.monotonic() < deadline:
-            with self._lock:
-                if all(t.running is not None for t in self.gangs.values()):
-                    return True
-            time.sleep(0.002)
-        return False
-
-    def summary(self) -> Dict[str, Any]:
-        with self._lock:
-            ttr = [(t.running - t.submitted) * 1000 for t in self.gangs.values()
-                   if t.running is not None]
-            tts = [(t.scheduled - t.submitted) * 1000 for t in self.gangs.values()
-                   if t.scheduled is not None]
-            n = len(self.gangs)
-            done = sum(1 for t in self.gangs.values() if t.running is not None)
-        return {
-            "gangs_total": n,
-            "gangs_running": done,
-            "p50_time_to_running_ms": percentile(ttr, 50),
-            "p95_time_to_running_ms": percentile(ttr, 95),
-            "max_time_to_running_ms": max(ttr) if ttr else None,
-            "p50_time_to_scheduled_ms": percentile(tts, 50),
-        }

@@ -560,6 +560,11 @@ class Store:
         (namespace, name) order; returns (items, next_continue, resourceVersion)."""
         import base64
         import json as _json
+        if limit is None and continue_token is None:
+            # unchunked: use the indexed list path (no full-table sort; the
+            # inverted label indexes serve selector queries)
+            return (self.list(kind, namespace, label_selector), None,
+                    self.current_rv())
         start_after: Optional[Tuple[str, str]] = None
         if continue_token:
             try:

@@ -369,7 +369,11 @@ class Store:
             if status_only:
                 if cur.get("status", {}) == obj.get("status", {}):
                     return json_copy(cur) if return_copy else None  # no-op
-                new = json_copy(cur)
+                # structural sharing: spec (the object's bulk — podSpecs) is
+                # immutable across a status write, so the new stored object
+                # shares it; only metadata (rv bump) + status are fresh
+                new = dict(cur)
+                new["metadata"] = json_copy(cur["metadata"])
                 new["status"] = json_copy(obj.get("status", {}))
             else:
                 # admission on spec/metadata updates
@@ -439,7 +443,16 @@ class Store:
         last: Optional[ApiError] = None
         for _ in range(retries):
             cur = self.get(kind, namespace, name, copy=False)
-            obj = json_copy(cur)
+            if status:
+                # status patches may only mutate obj["status"] (and read the
+                # rest) — the working copy shares the immutable spec subtree,
+                # skipping the podSpec deep copy that dominated status-write
+                # cost at 10k-pod scale
+                obj = dict(cur)
+                obj["metadata"] = json_copy(cur["metadata"])
+                obj["status"] = json_copy(cur.get("status") or {})
+            else:
+                obj = json_copy(cur)
             fn(obj)
             try:
                 return self._apply_update(obj, status_only=status, owned=True,

@@ -172,6 +172,14 @@ class DataPlane:
             await _json(send, e.code, {
                 "kind": "Status", "status": "Failure", "reason": e.reason,
                 "message": e.message, "code": e.code})
+        except Exception as e:  # internal fault -> kube-style 500, not a dead socket
+            try:
+                await _json(send, 500, {
+                    "kind": "Status", "status": "Failure",
+                    "reason": "InternalError",
+                    "message": f"{type(e).__name__}: {e}", "code": 500})
+            except Exception:
+                pass  # response already started (watch stream teardown)
 
     async def _read_body(self, receive) -> Obj:
         chunks: List[bytes] = []

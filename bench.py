@@ -632,6 +632,15 @@ def _run_rank0(comm: Comm, args, n_gpus: int, gang_size: int) -> None:
     # default 5 ms GIL switch interval throttles those handoffs (same-box A/B:
     # +6-11% gangs/s at 0.2 ms). The http-mode operator process sets its own.
     sys.setswitchinterval(0.0002)
+    # GC tuning for the latency tail: the default gen0 threshold (700) fires
+    # collections mid-gang at high object churn; raising thresholds + freezing
+    # startup objects moves collections off the critical path (A/B on tail
+    # percentiles; the store's object graph is acyclic dicts, so cycles are rare)
+    if os.environ.get("GROVE_GC_TUNE", "1") != "0":
+        import gc
+        gc.collect()
+        gc.freeze()
+        gc.set_threshold(100000, 50, 50)
 
     results: Dict[str, Dict[str, Any]] = {}
     order = {"both": ["inproc", "http"], "http": ["http"],

@@ -19,6 +19,12 @@ from . import __version__
 def cmd_operator(args) -> int:
     import sys as _sys
     _sys.setswitchinterval(0.0002)  # see bench.py: watch-chain handoff latency
+    import os as _os0
+    if _os0.environ.get("GROVE_GC_TUNE", "1") != "0":
+        import gc as _gc
+        _gc.collect()
+        _gc.freeze()
+        _gc.set_threshold(100000, 50, 50)  # keep gen0 GC off the reconcile path
     from .cluster import Cluster
     from .config import load_configuration
     from .kubecore.apiserver import ApiServer

@@ -695,3 +695,43 @@ def test_ru21_clique_scale_in_before_update(cluster):
             p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
             and cond.pod_is_ready(p) for p in ps)
     cluster.wait_for(settled, timeout=40, desc="2 pods, all on new template")
+
+
+def test_tas16_three_level_hierarchy(cluster):
+    """TAS16 (topology_test.go:1099): zone > rack > host CTB; the gang requires
+    zone packing and prefers host packing — with a host that fits, everything
+    lands on one host; when no single host fits, it falls back within one zone
+    but never crosses zones."""
+    ctb = {"apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+           "metadata": {"name": "default-topology"},
+           "spec": {"levels": [
+               {"domain": "zone", "key": "topology.kubernetes.io/zone"},
+               {"domain": "rack", "key": "topology.kubernetes.io/rack"},
+               {"domain": "host", "key": "kubernetes.io/hostname"}]}}
+    cluster.store.create(ctb)
+    from grove_amd.kubelet.virtual import make_virtual_node
+    for z in range(2):
+        for r in range(2):
+            for h in range(2):
+                n = make_virtual_node(f"z{z}r{r}h{h}", gpus=2, cpu="64", pods=64)
+                n["metadata"]["labels"]["topology.kubernetes.io/zone"] = f"z{z}"
+                n["metadata"]["labels"]["topology.kubernetes.io/rack"] = f"z{z}r{r}"
+                cluster.store.create(n)
+    # 2-GPU gang: fits one host -> host-preferred packing
+    pcs = _pcs("t16a", cliques=(("w", 2, 2),), gpus=1)
+    pcs["spec"]["template"]["topologyConstraint"] = {
+        "pack": {"required": "zone", "preferred": "host"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("t16a", timeout=20)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "t16a-0-w"})
+    assert len({p["spec"]["nodeName"] for p in pods}) == 1
+    # 6-GPU gang: no host fits (2 GPUs each) -> falls back inside ONE zone
+    pcs = _pcs("t16b", cliques=(("w", 6, 6),), gpus=1)
+    pcs["spec"]["template"]["topologyConstraint"] = {
+        "pack": {"required": "zone", "preferred": "host"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("t16b", timeout=20)
+    pods = _pods(cluster, {c.LABEL_PODCLIQUE: "t16b-0-w"})
+    zones = {p["spec"]["nodeName"][:2] for p in pods}
+    assert len(zones) == 1, f"gang crossed zones: {zones}"
+    assert len({p["spec"]["nodeName"] for p in pods}) == 3  # 3 hosts x 2 GPUs

@@ -735,3 +735,23 @@ def test_tas16_three_level_hierarchy(cluster):
     zones = {p["spec"]["nodeName"][:2] for p in pods}
     assert len(zones) == 1, f"gang crossed zones: {zones}"
     assert len({p["spec"]["nodeName"] for p in pods}) == 3  # 3 hosts x 2 GPUs
+
+
+def test_so4_explicit_startup_with_min_replicas(cluster):
+    """SO4 (startup_ordering_test.go:218): Explicit startsAfter DAG with
+    minAvailable < replicas — the dependent waits on the parent's minAvailable,
+    and the grove-initc flag carries exactly that threshold."""
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)
+    pcs = _pcs("so4", cliques=(("a", 4, 2), ("b", 2, 1), ("z", 1, 1)),
+               startup=c.STARTUP_EXPLICIT)
+    pcs["spec"]["template"]["cliques"][1]["spec"]["startsAfter"] = ["a"]
+    pcs["spec"]["template"]["cliques"][2]["spec"]["startsAfter"] = ["a", "b"]
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("so4", timeout=20)
+    z_pods = _pods(cluster, {c.LABEL_PODCLIQUE: "so4-0-z"})
+    ic = z_pods[0]["spec"]["initContainers"][0]
+    assert "--podcliques=so4-0-a:2" in ic["args"]  # parent a's minAvailable
+    assert "--podcliques=so4-0-b:1" in ic["args"]  # parent b's minAvailable
+    assert cluster.store.get(c.KIND_PCLQ, "default",
+                             "so4-0-z")["spec"]["startsAfter"] == \
+        ["so4-0-a", "so4-0-b"]

@@ -33,6 +33,10 @@ class PCSGReconciler:
         self.store = store
         self.scheduler_name = scheduler_name
         self.auto_xgmi_domain = auto_xgmi_domain
+        # structural-sync fingerprint (PCS reconciler pattern): member-PCLQ sync
+        # is skipped while only statuses churn — PCSG/PCS spec generations, the
+        # rolling-update state and member generations are the structural inputs
+        self._sync_fp: Dict[str, tuple] = {}
 
     def reconcile(self, namespace: str, name: str) -> Result:
         pcsg = self.store.try_get(c.KIND_PCSG, namespace, name)
@@ -44,7 +48,23 @@ class PCSGReconciler:
         if pcs is None:
             return Result(requeue_after=0.1)
         rec = groveerr.StepRecorder(self.store, c.KIND_PCSG, namespace, name)
-        res = self._sync_member_pclqs(pcs, pcsg, rec)
+        key = f"{namespace}/{name}"
+        members_sig = tuple(sorted(
+            (q["metadata"]["name"], q["metadata"].get("generation", 0))
+            for q in self._member_pclqs(pcsg)))
+        st = pcs.get("status") or {}
+        prog = st.get("updateProgress") or {}
+        fp = (pcsg["metadata"].get("generation"),
+              pcs["metadata"].get("generation"),
+              st.get("currentGenerationHash"),
+              tuple(sorted(_currently_updating_indices(prog))),
+              hash(members_sig))
+        if self._sync_fp.get(key) != fp:
+            res = self._sync_member_pclqs(pcs, pcsg, rec)
+            if not rec.errors:
+                self._sync_fp[key] = fp
+        else:
+            res = Result.DONE
         recycle_wait = self._replica_recycle(pcs, pcsg, rec)
         self._reconcile_status(namespace, name)
         rec.flush()
@@ -77,6 +97,7 @@ class PCSGReconciler:
     # ------------------------------------------------------------------ delete
     def _reconcile_delete(self, pcsg: Obj) -> Result:
         ns, name = pcsg["metadata"].get("namespace"), pcsg["metadata"]["name"]
+        self._sync_fp.pop(f"{ns}/{name}", None)
         remaining = 0
         for q in self._member_pclqs(pcsg):
             remaining += 1

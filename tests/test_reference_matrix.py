@@ -192,8 +192,10 @@ def test_gt6_scaled_pod_deletion_does_not_terminate_pcs_replica(cluster):
     # standalone clique survived (no PCS-scope termination)
     a = cluster.store.get(c.KIND_PCLQ, "default", "gt6-0-a")
     assert a["metadata"]["uid"] == a_uid
-    # the scaled pod was replaced and the set is whole again
-    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "gt6-0-sg-1-b"}, 1, timeout=20)
+    # the scaled pod was replaced and the set is whole again (generous timeout:
+    # the replacement traverses podReferences refresh + ungate + schedule, and
+    # the suite may share a heavily loaded box)
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "gt6-0-sg-1-b"}, 1, timeout=45)
 
 
 # --------------------------------------------------------------------------- OD

@@ -142,17 +142,48 @@ def cmd_agent(args) -> int:
     signal.signal(signal.SIGINT, on_sig)
     signal.signal(signal.SIGTERM, on_sig)
     log.info("agent serving node %s against %s", name, args.server)
+
+    # watch-driven (kubelet informer analog): pod events for this node reconcile
+    # immediately; a periodic full relist catches anything a dropped stream missed
+    import queue as _queue
+    import threading as _threading
+    events: "_queue.Queue" = _queue.Queue()
+
+    def pump() -> None:
+        while not stop["flag"]:
+            try:
+                for ev, pod in client.watch_events("Pod", args.namespace,
+                                                   seed=True):
+                    if stop["flag"]:
+                        return
+                    if pod.get("spec", {}).get("nodeName") == name:
+                        events.put(pod)
+            except Exception as e:
+                log.warning("pod watch stream ended (%s); reconnecting", e)
+                _time.sleep(min(2.0, args.poll_interval * 5))
+    _threading.Thread(target=pump, daemon=True).start()
+
+    last_resync = 0.0
     while not stop["flag"]:
         try:
-            pods = client.list("Pod", args.namespace,
-                               filter_fn=lambda p: p.get("spec", {})
-                               .get("nodeName") == name)
-            for p in pods:
-                kubelet.reconcile(p["metadata"].get("namespace", "default"),
-                                  p["metadata"]["name"])
+            pod = events.get(timeout=args.poll_interval)
+            kubelet.reconcile(pod["metadata"].get("namespace", "default"),
+                              pod["metadata"]["name"])
+        except _queue.Empty:
+            pass
         except Exception as e:
-            log.warning("agent pass failed: %s", e)
-        _time.sleep(args.poll_interval)
+            log.warning("agent reconcile failed: %s", e)
+        now = _time.monotonic()
+        if now - last_resync > max(10.0, args.poll_interval * 50):
+            last_resync = now
+            try:
+                for p in client.list("Pod", args.namespace,
+                                     filter_fn=lambda p: p.get("spec", {})
+                                     .get("nodeName") == name):
+                    kubelet.reconcile(p["metadata"].get("namespace", "default"),
+                                      p["metadata"]["name"])
+            except Exception as e:
+                log.warning("agent resync failed: %s", e)
     kubelet.shutdown()
     return 0
 

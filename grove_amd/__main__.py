@@ -152,12 +152,12 @@ def cmd_agent(args) -> int:
     def pump() -> None:
         while not stop["flag"]:
             try:
-                for ev, pod in client.watch_events("Pod", args.namespace,
-                                                   seed=True):
+                for ev, pod in client.watch_events(
+                        "Pod", args.namespace, seed=True,
+                        field_selector={"spec.nodeName": name}):
                     if stop["flag"]:
                         return
-                    if pod.get("spec", {}).get("nodeName") == name:
-                        events.put(pod)
+                    events.put(pod)
             except Exception as e:
                 log.warning("pod watch stream ended (%s); reconnecting", e)
                 _time.sleep(min(2.0, args.poll_interval * 5))
@@ -178,8 +178,7 @@ def cmd_agent(args) -> int:
             last_resync = now
             try:
                 for p in client.list("Pod", args.namespace,
-                                     filter_fn=lambda p: p.get("spec", {})
-                                     .get("nodeName") == name):
+                                     field_selector={"spec.nodeName": name}):
                     kubelet.reconcile(p["metadata"].get("namespace", "default"),
                                       p["metadata"]["name"])
             except Exception as e:

@@ -29,6 +29,17 @@ from .store import Store, ApiError
 Obj = Dict[str, Any]
 
 
+def parse_field_selector(sel: Optional[str]) -> Optional[Dict[str, str]]:
+    if not sel:
+        return None
+    out = {}
+    for part in sel.split(","):
+        if "=" in part:
+            k, v = part.split("=", 1)
+            out[k.strip().lstrip("=")] = v.strip()
+    return out
+
+
 def parse_selector(sel: Optional[str]) -> Optional[Dict[str, str]]:
     if not sel:
         return None
@@ -198,7 +209,15 @@ class DataPlane:
     async def _list(self, send, kind: str, ns: Optional[str],
                     q: Dict[str, str]) -> None:
         selector = parse_selector(q.get("labelSelector"))
+        fields = parse_field_selector(q.get("fieldSelector"))
         limit = int(q["limit"]) if q.get("limit") else None
+        if fields and not limit and not q.get("continue"):
+            items = self.store.list(kind, ns, selector, field_selector=fields)
+            await _json(send, 200, {
+                "kind": f"{kind}List", "apiVersion": "v1",
+                "metadata": {"resourceVersion": self.store.current_rv()},
+                "items": items})
+            return
         items, next_cont, rv = self.store.list_page(
             kind, ns, selector, limit, q.get("continue") or None)
         meta: Dict[str, Any] = {"resourceVersion": rv}
@@ -213,7 +232,8 @@ class DataPlane:
         w = self.store.watch(
             kind,
             seed=(since_rv is None and q.get("seed", "true") in ("true", "1")),
-            since_rv=since_rv)
+            since_rv=since_rv,
+            field_selector=parse_field_selector(q.get("fieldSelector")))
         await send({"type": "http.response.start", "status": 200,
                     "headers": [(b"content-type",
                                  b"application/x-ndjson")]})

@@ -147,11 +147,16 @@ class HttpStoreClient:
     def list(self, kind: str, namespace: Optional[str] = None,
              label_selector: Optional[Dict[str, str]] = None,
              filter_fn: Optional[Callable[[Obj], bool]] = None,
-             copy_objects: bool = True) -> List[Obj]:
-        query = ""
+             copy_objects: bool = True,
+             field_selector: Optional[Dict[str, str]] = None) -> List[Obj]:
+        parts = []
         if label_selector:
             sel = ",".join(f"{k}={v}" for k, v in label_selector.items())
-            query = f"labelSelector={urllib.parse.quote(sel)}"
+            parts.append(f"labelSelector={urllib.parse.quote(sel)}")
+        if field_selector:
+            fs = ",".join(f"{k}={v}" for k, v in field_selector.items())
+            parts.append(f"fieldSelector={urllib.parse.quote(fs)}")
+        query = "&".join(parts)
         items = self._request("GET", self._url(kind, namespace, query=query))["items"]
         if filter_fn is not None:
             items = [o for o in items if filter_fn(o)]
@@ -230,7 +235,9 @@ class HttpStoreClient:
 
     def watch_events(self, kind: str, namespace: Optional[str] = None,
                      seed: bool = True, resource_version: Optional[str] = None,
-                     bookmarks: bool = False) -> Iterator[Tuple[str, Obj]]:
+                     bookmarks: bool = False,
+                     field_selector: Optional[Dict[str, str]] = None
+                     ) -> Iterator[Tuple[str, Obj]]:
         """ndjson watch stream; yields (event_type, object). resource_version
         resumes from that RV (replay + live); bookmarks=True interleaves BOOKMARK
         progress events carrying the current resourceVersion."""
@@ -239,6 +246,10 @@ class HttpStoreClient:
             q += f"&resourceVersion={resource_version}"
         if bookmarks:
             q += "&allowWatchBookmarks=true"
+        if field_selector:
+            import urllib.parse as _p
+            fs = ",".join(f"{k}={v}" for k, v in field_selector.items())
+            q += f"&fieldSelector={_p.quote(fs)}"
         url = self._url(kind, namespace, query=q)
         req = urllib.request.Request(url, headers=self._headers())
         with urllib.request.urlopen(req, timeout=3600, context=self._ctx) as r:

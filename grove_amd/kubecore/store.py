@@ -136,12 +136,31 @@ class _KindTable:
                     s.discard(key)
 
 
+def match_fields(obj: Obj, field_selector: Optional[Dict[str, str]]) -> bool:
+    """kube fieldSelector subset: the fields real components select on —
+    spec.nodeName (kubelet), metadata.name, metadata.namespace, status.phase."""
+    if not field_selector:
+        return True
+    for path, want in field_selector.items():
+        cur: Any = obj
+        for part in path.split("."):
+            if not isinstance(cur, dict):
+                cur = None
+                break
+            cur = cur.get(part)
+        if cur != want:
+            return False
+    return True
+
+
 class Watch:
     """A subscription to one kind's events. Iterate or poll `.queue`."""
 
-    def __init__(self, store: "Store", kind: str):
+    def __init__(self, store: "Store", kind: str,
+                 field_selector: Optional[Dict[str, str]] = None):
         self.store = store
         self.kind = kind
+        self.field_selector = field_selector
         self.queue: "queue.Queue[Tuple[str, Obj]]" = queue.Queue()
         self._stopped = False
 
@@ -225,7 +244,8 @@ class Store:
             except IndexError:
                 return
             for w in list(tbl.watchers):
-                w.queue.put((ev, obj))
+                if w.field_selector is None or match_fields(obj, w.field_selector):
+                    w.queue.put((ev, obj))
 
     def _next_rv(self) -> str:
         return str(next(self._rv))
@@ -306,7 +326,8 @@ class Store:
     def list(self, kind: str, namespace: Optional[str] = None,
              label_selector: Optional[Dict[str, str]] = None,
              filter_fn: Optional[Callable[[Obj], bool]] = None,
-             copy_objects: bool = True) -> List[Obj]:
+             copy_objects: bool = True,
+             field_selector: Optional[Dict[str, str]] = None) -> List[Obj]:
         """List objects. copy_objects=False returns direct references for read-only
         consumers (status rollups, the scheduler pass) — callers MUST NOT mutate."""
         with self._lock:
@@ -337,6 +358,8 @@ class Store:
             if namespace is not None and ns != namespace:
                 continue
             if not match_labels(obj.get("metadata", {}).get("labels"), label_selector):
+                continue
+            if field_selector and not match_fields(obj, field_selector):
                 continue
             refs.append(obj)
         # filtering callbacks + copies run OUTSIDE the lock (objects immutable)
@@ -530,12 +553,13 @@ class Store:
 
     # ------------------------------------------------------------------ watches & events
     def watch(self, kind: str, seed: bool = False,
-              since_rv: Optional[str] = None) -> Watch:
+              since_rv: Optional[str] = None,
+              field_selector: Optional[Dict[str, str]] = None) -> Watch:
         """since_rv: resume semantics — replay retained events with
         resourceVersion > since_rv, then stream live (client-go ListAndWatch /
         watch cache parity). Raises 410 Expired if since_rv predates the
         retained window."""
-        w = Watch(self, kind)
+        w = Watch(self, kind, field_selector)
         with self._lock:
             tbl = self._table(kind)
             if since_rv is not None:
@@ -549,13 +573,14 @@ class Store:
                                    f"resourceVersion {since_rv} is too old")
                 tbl.watchers.append(w)
                 for (erv, ev, obj) in tbl.history:
-                    if erv > rv:
+                    if erv > rv and match_fields(obj, field_selector):
                         w.queue.put((ev, obj))
                 return w
             tbl.watchers.append(w)
             if seed:
                 for obj in tbl.objects.values():
-                    w.queue.put((ADDED, json_copy(obj)))
+                    if match_fields(obj, field_selector):
+                        w.queue.put((ADDED, json_copy(obj)))
         return w
 
     def current_rv(self) -> str:

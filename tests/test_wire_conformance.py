@@ -177,3 +177,38 @@ def test_status_subresource_method_guard(served):
     except urllib.error.HTTPError as e:
         assert e.code == 405
     assert cluster.store.try_get("Pod", "default", "wp000") is not None
+
+
+def test_field_selector_list_and_watch(served):
+    """fieldSelector on list + watch (the kubelet's spec.nodeName pattern): the
+    server filters, so a node agent's watch only carries its own pods."""
+    cluster, api, client = served
+    for i, node in enumerate(["n0", "n1", "n0"]):
+        p = _mkpod(i)
+        p["spec"]["nodeName"] = node
+        cluster.store.create(p)
+    items = client.list("Pod", "default",
+                        field_selector={"spec.nodeName": "n0"})
+    assert sorted(p["metadata"]["name"] for p in items) == ["wp000", "wp002"]
+    got = []
+    done = threading.Event()
+
+    def consume():
+        for ev, obj in client.watch_events(
+                "Pod", "default", seed=True,
+                field_selector={"spec.nodeName": "n1"}):
+            got.append(obj["metadata"]["name"])
+            if len(got) >= 2:
+                done.set()
+                return
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    p = _mkpod(9)
+    p["spec"]["nodeName"] = "n1"
+    cluster.store.create(p)
+    p = _mkpod(8)
+    p["spec"]["nodeName"] = "n0"  # must NOT appear on the n1 stream
+    cluster.store.create(p)
+    assert done.wait(10)
+    assert got == ["wp001", "wp009"]

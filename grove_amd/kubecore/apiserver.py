@@ -64,6 +64,7 @@ def build_app(store: Store, metrics_fn=None,
                 return
         await aux(scope, receive, send)
 
+    app.dataplane = plane  # ApiServer.stop() signals open watch streams
     return app
 
 
@@ -178,6 +179,7 @@ class ApiServer:
     def start(self) -> "ApiServer":
         import uvicorn
         app = build_app(self.store, self.metrics_fn, self.auth_tokens)
+        self._app = app
         config = uvicorn.Config(app, host=self.host, port=self.port,
                                 log_level="warning", lifespan="off",
                                 ssl_certfile=self.ssl_certfile,
@@ -206,6 +208,9 @@ class ApiServer:
         return f"{scheme}://{self.host}:{self.port}"
 
     def stop(self) -> None:
+        plane = getattr(getattr(self, "_app", None), "dataplane", None)
+        if plane is not None:
+            plane.shutdown.set()  # end open watch streams promptly
         if self._server is not None:
             self._server.should_exit = True
         if self._thread is not None:

@@ -58,6 +58,10 @@ class DataPlane:
         self.plurals = plurals
         self.cluster_scoped = cluster_scoped
         self.auth_tokens = auth_tokens or {}
+        # set by ApiServer.stop(): open watch streams end their responses so
+        # uvicorn's graceful shutdown does not wait out its timeout
+        import threading
+        self.shutdown = threading.Event()
 
     # ------------------------------------------------------------------ routing
     def parse(self, path: str) -> Optional[Tuple[str, Optional[str],
@@ -238,14 +242,19 @@ class DataPlane:
                     "headers": [(b"content-type",
                                  b"application/x-ndjson")]})
         loop = asyncio.get_event_loop()
+        idle = 0
         try:
-            while True:
+            while not self.shutdown.is_set():
                 try:
                     ev, obj = await loop.run_in_executor(
-                        None, w.queue.get, True, 1.0)
+                        None, w.queue.get, True, 0.2)
                 except RuntimeError:
                     return  # event loop / executor shutting down
                 except queue.Empty:
+                    idle += 1
+                    if idle < 5:
+                        continue  # keepalive cadence stays ~1 s
+                    idle = 0
                     if bookmarks:
                         payload = json.dumps({"type": "BOOKMARK", "object": {
                             "kind": kind, "metadata": {
@@ -256,10 +265,14 @@ class DataPlane:
                     await send({"type": "http.response.body",
                                 "body": payload.encode(), "more_body": True})
                     continue
+                idle = 0
                 await send({"type": "http.response.body",
                             "body": (json.dumps({"type": ev, "object": obj})
                                      + "\n").encode(),
                             "more_body": True})
+            # graceful end-of-stream on shutdown
+            await send({"type": "http.response.body", "body": b"",
+                        "more_body": False})
         finally:
             w.stop()
 

@@ -61,7 +61,7 @@ class PCSGReconciler:
               hash(members_sig))
         if self._sync_fp.get(key) != fp:
             res = self._sync_member_pclqs(pcs, pcsg, rec)
-            if not rec.errors:
+            if not rec.errors and not rec.retry_needed:
                 self._sync_fp[key] = fp
         else:
             res = Result.DONE

@@ -126,7 +126,7 @@ class PodCliqueSetReconciler:
               bool(prog) and not prog.get("updateEndedAt"))
         if self._sync_fp.get(key) != fp:
             res = self._sync_resources(pcs, rec)
-            if not rec.errors:
+            if not rec.errors and not rec.retry_needed:
                 self._sync_fp[key] = fp
         else:
             res = Result.DONE

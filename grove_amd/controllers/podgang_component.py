@@ -225,6 +225,7 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str,
                                f"{e.message}")
                 cur = store.try_get(c.KIND_PODGANG, ns, gang.name)
                 if cur is None:
+                    rec.retry_needed = True
                     continue
 
         def upd(o: Obj) -> None:
@@ -240,7 +241,9 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str,
             try:
                 cur = store.patch(c.KIND_PODGANG, ns, gang.name, upd)
             except ApiError as e:
-                if e.reason not in groveerr.BENIGN_UPDATE:
+                if e.reason == "Conflict":
+                    rec.retry_needed = True  # refs not written; do not latch
+                elif e.reason not in groveerr.BENIGN_UPDATE:
                     rec.record(groveerr.ERR_SYNC_PODGANG,
                                f"patch PodGang {gang.name}: {e.reason}: "
                                f"{e.message}")

@@ -53,6 +53,10 @@ class StepRecorder:
         self.namespace = namespace
         self.name = name
         self.errors: list = []
+        # a step swallowed a retryable race (exhausted Conflict): the pass did
+        # NOT fully converge — callers that cache "converged" state (the PCS/PCSG
+        # structural-sync fingerprints) must not latch it
+        self.retry_needed = False
 
     @_contextmanager
     def step(self, code: str, benign: tuple = BENIGN_CREATE + BENIGN_DELETE,
@@ -61,6 +65,8 @@ class StepRecorder:
             yield
         except ApiError as e:
             if e.reason in benign:
+                if e.reason == "Conflict":
+                    self.retry_needed = True  # work skipped, must re-run
                 return
             msg = f"{detail + ': ' if detail else ''}{e.reason}: {e.message}"
             self.errors.append((code, msg))

@@ -43,6 +43,12 @@ def cluster(request):
         import sys
         print("\n=== cluster diagnostics (on failure) ===", file=sys.stderr)
         print("store:", cl.store.stats(), file=sys.stderr)
+        for pg in cl.store.list("PodGang"):
+            refs = {g.get("name"): [r.get("name") for r in
+                                    g.get("podReferences") or []]
+                    for g in (pg.get("spec") or {}).get("podgroups") or []}
+            print(f"  PodGang {pg['metadata']['name']} podReferences: {refs}",
+                  file=sys.stderr)
         for kind in ("PodCliqueSet", "PodClique", "PodCliqueScalingGroup", "PodGang"):
             for o in cl.store.list(kind):
                 st = o.get("status") or {}

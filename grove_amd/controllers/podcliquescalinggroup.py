@@ -68,6 +68,8 @@ class PCSGReconciler:
         recycle_wait = self._replica_recycle(pcs, pcsg, rec)
         self._reconcile_status(namespace, name)
         rec.flush()
+        if rec.retry_needed:
+            return Result(requeue_after=0.05)
         if recycle_wait is not None:
             return Result(requeue_after=recycle_wait)
         return res

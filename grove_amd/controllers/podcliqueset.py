@@ -133,6 +133,10 @@ class PodCliqueSetReconciler:
         term = self._gang_termination(pcs, rec)
         self._orchestrate_rolling_update(pcs, rec)
         self._reconcile_status(namespace, name, rec)
+        if rec.retry_needed:
+            # a step lost an optimistic race and was skipped — nothing else may
+            # retrigger this PCS (PodGang events don't map back), so requeue
+            return Result(requeue_after=0.05)
         if term is not None:
             return Result(requeue_after=term)
         return res

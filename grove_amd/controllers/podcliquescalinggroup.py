@@ -257,12 +257,39 @@ class PCSGReconciler:
                 next_wait = remaining if next_wait is None else min(next_wait, remaining)
                 continue
             log.info("PCSG %s/%s recycling replica %d", ns, pcsg["metadata"]["name"], j)
+            self._reset_gang_for_recycle(pcs, pcsg, j, rec)
             for q in qs:
                 with rec.step(groveerr.ERR_GANG_TERMINATION,
                               benign=groveerr.BENIGN_DELETE,
                               detail=f"recycle PodClique {q['metadata']['name']}"):
                     self.store.delete(c.KIND_PCLQ, ns, q["metadata"]["name"])
         return next_wait
+
+    def _reset_gang_for_recycle(self, pcs: Obj, pcsg: Obj, j: int,
+                                rec: groveerr.StepRecorder) -> None:
+        """Reset the lifecycle conditions of the gang covering PCSG replica j before
+        its cliques are deleted: a recycled replica re-runs init → permit → ready, and
+        a stale Initialized=True would short-circuit the inline podReferences refill
+        and leave the recreated pods schedule-gated (same wedge as PCS-scope gang
+        termination)."""
+        ns = pcsg["metadata"].get("namespace", "default")
+        min_avail = int(pcsg["spec"].get("minAvailable", 1))
+        if j >= min_avail:
+            gang = namegen.scaled_podgang_name(pcsg["metadata"]["name"], j - min_avail)
+        else:
+            pcs_replica = int(
+                pcsg["metadata"]["labels"].get(c.LABEL_PCS_REPLICA_INDEX, 0))
+            gang = namegen.base_podgang_name(pcs["metadata"]["name"], pcs_replica)
+
+        def reset(o: Obj) -> None:
+            cond.set_condition(o, c.PODGANG_COND_INITIALIZED, False, "GangTerminated")
+            cond.set_condition(o, c.PODGANG_COND_SCHEDULED, False, "GangTerminated")
+            cond.set_condition(o, c.PODGANG_COND_READY, False, "GangTerminated")
+            o.setdefault("status", {})["phase"] = "Pending"
+        with rec.step(groveerr.ERR_GANG_TERMINATION,
+                      benign=groveerr.BENIGN_UPDATE,
+                      detail=f"reset gang {gang} for recycle"):
+            self.store.patch(c.KIND_PODGANG, ns, gang, reset, status=True)
 
     # ------------------------------------------------------------------ status
     def _reconcile_status(self, namespace: str, name: str) -> None:

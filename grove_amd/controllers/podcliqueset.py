@@ -488,6 +488,18 @@ class PodCliqueSetReconciler:
                 def mark_disrupted(o: Obj) -> None:
                     cond.set_condition(o, c.PODGANG_COND_DISRUPTION_TARGET, True,
                                        "GangTerminated")
+                    # reset the gang's lifecycle for the recycle: a recreated
+                    # replica must re-run init → permit → ready from scratch. A
+                    # stale Initialized=True would short-circuit the inline
+                    # podReferences refill (try_complete_podgang) and leave the
+                    # recreated pods schedule-gated forever (observed wedge).
+                    cond.set_condition(o, c.PODGANG_COND_INITIALIZED, False,
+                                       "GangTerminated")
+                    cond.set_condition(o, c.PODGANG_COND_SCHEDULED, False,
+                                       "GangTerminated")
+                    cond.set_condition(o, c.PODGANG_COND_READY, False,
+                                       "GangTerminated")
+                    o.setdefault("status", {})["phase"] = "Pending"
                 with rec.step(groveerr.ERR_GANG_TERMINATION,
                               benign=groveerr.BENIGN_UPDATE,
                               detail=f"mark DisruptionTarget on {pg_name}"):

@@ -100,7 +100,13 @@ def test_gs6_pcsg_min_available_scaled_gangs(cluster):
     for sgname in ("gs6-0-sx-0", "gs6-0-sx-1"):
         pg = cluster.store.get(c.KIND_PODGANG, "default", sgname)
         assert pg["metadata"]["labels"][c.LABEL_BASE_PODGANG] == "gs6-0"
-        assert cond.condition_true(pg, c.PODGANG_COND_SCHEDULED)
+        # PCS availability only needs PCSG minAvailable=1, so the scaled gangs
+        # may still be mid-bind here — wait for their Scheduled stamp
+        cluster.wait_for(
+            lambda n=sgname: cond.condition_true(
+                cluster.store.get(c.KIND_PODGANG, "default", n),
+                c.PODGANG_COND_SCHEDULED),
+            timeout=15, desc=f"{sgname} Scheduled")
     cluster.wait_pods_ready({c.LABEL_PART_OF: "gs6"}, 7, timeout=20)
 
 
@@ -422,6 +428,15 @@ def test_tas10_pcsg_scaling_with_constraint(cluster):
     cluster.apply(pcs)
     cluster.wait_pcs_available("tas10", timeout=20)
     for j in (0, 1):
+        # PCS availability only needs PCSG minAvailable=1 — the scaled replica
+        # (j=1) may still be binding; wait for both pods to land first
+        cluster.wait_for(
+            lambda j=j: all(
+                p["spec"].get("nodeName")
+                for p in _pods(cluster, {c.LABEL_PODCLIQUE: f"tas10-0-sg-{j}-b"})
+            ) and len(_pods(cluster,
+                            {c.LABEL_PODCLIQUE: f"tas10-0-sg-{j}-b"})) == 2,
+            timeout=15, desc=f"sg replica {j} bound")
         ps = _pods(cluster, {c.LABEL_PODCLIQUE: f"tas10-0-sg-{j}-b"})
         assert len({p["spec"]["nodeName"] for p in ps}) == 1, \
             f"PCSG replica {j} spans hosts"

@@ -41,6 +41,10 @@ class PodCliqueReconciler:
         res, pods = self._reconcile_spec(pclq, rec)
         self._reconcile_status(namespace, name, pods=pods)
         rec.flush()
+        if rec.retry_needed:
+            # a step (e.g. the ungate patch) lost an optimistic race and was
+            # swallowed as benign — nothing may retrigger this clique, so requeue
+            return Result(requeue_after=0.05)
         return res
 
     # ------------------------------------------------------------------ delete
@@ -206,11 +210,22 @@ class PodCliqueReconciler:
         base_name = pclq["metadata"]["labels"].get(c.LABEL_BASE_PODGANG)
         base_scheduled: Optional[bool] = None  # lazily computed
 
+        refs_repaired = False
         for p in (pods if pods is not None else self._owned_pods(pclq)):
             gates = p.get("spec", {}).get("schedulingGates") or []
             if not any(g.get("name") == c.POD_GANG_SCHEDULING_GATE for g in gates):
                 continue
             if p["metadata"]["name"] not in refs:
+                # gated pod not in an Initialized gang's refs = the gang still
+                # references a dead predecessor (pod replaced out-of-band) —
+                # repair the refs inline and retry this clique
+                if not refs_repaired and cond.condition_true(
+                        podgang, c.PODGANG_COND_INITIALIZED):
+                    from .podgang_component import try_complete_podgang
+                    try_complete_podgang(self.store, ns, podgang_name, rec=rec,
+                                         force=True)
+                    refs_repaired = True
+                    rec.retry_needed = True  # re-run with the repaired refs
                 continue
             if base_name:
                 if base_scheduled is None:

@@ -263,15 +263,21 @@ def sync_podgangs(store: Store, pcs: Obj, scheduler_name: str,
 
 
 def try_complete_podgang(store: Store, ns: str, gang_name: str,
-                         rec: "groveerr.StepRecorder" = None) -> None:
+                         rec: "groveerr.StepRecorder" = None,
+                         force: bool = False) -> None:
     """Latency fast-path: called inline from the PCLQ pass right
     after pod creation, so a gang whose pods all exist gets its podReferences filled
     and Initialized flipped in the SAME reconcile instead of waiting for the next
     PCS-scope sync_podgangs pass (saves two watch→queue→worker hops per gang on the
     serial time-to-running chain). The PCS pass remains the reconciling authority;
-    this only performs the monotonic completion step."""
+    this only performs the monotonic completion step.
+
+    force=True refreshes podReferences even on an already-Initialized gang — the
+    repair path for a replaced pod whose gang still references its dead predecessor
+    (the gate-removal wedge: gated pod not in refs + Initialized=True)."""
     pg = store.try_get(c.KIND_PODGANG, ns, gang_name, copy=False)
-    if pg is None or cond.condition_true(pg, c.PODGANG_COND_INITIALIZED):
+    if pg is None or (not force
+                      and cond.condition_true(pg, c.PODGANG_COND_INITIALIZED)):
         return
     groups = (pg.get("spec") or {}).get("podgroups") or []
     if not groups:
@@ -301,15 +307,22 @@ def try_complete_podgang(store: Store, ns: str, gang_name: str,
         for gs in groups_spec:
             if gs["name"] in cur_groups:
                 cur_groups[gs["name"]]["podReferences"] = gs["podReferences"]
-    try:
-        store.patch(c.KIND_PODGANG, ns, gang_name, fill)
-    except ApiError:
-        return
+    cur_refs = {g["name"]: [r.get("name") for r in g.get("podReferences") or []]
+                for g in groups}
+    if cur_refs != {gs["name"]: [r["name"] for r in gs["podReferences"]]
+                    for gs in groups_spec}:
+        try:
+            store.patch(c.KIND_PODGANG, ns, gang_name, fill)
+        except ApiError:
+            if rec is not None:
+                rec.retry_needed = True
+            return
 
-    def flip(o: Obj) -> None:
-        cond.set_condition(o, c.PODGANG_COND_INITIALIZED, True,
-                           "AllPodsAssociated")
-    try:
-        store.patch(c.KIND_PODGANG, ns, gang_name, flip, status=True)
-    except ApiError:
-        pass
+    if not cond.condition_true(pg, c.PODGANG_COND_INITIALIZED):
+        def flip(o: Obj) -> None:
+            cond.set_condition(o, c.PODGANG_COND_INITIALIZED, True,
+                               "AllPodsAssociated")
+        try:
+            store.patch(c.KIND_PODGANG, ns, gang_name, flip, status=True)
+        except ApiError:
+            pass

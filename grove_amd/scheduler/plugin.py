@@ -80,6 +80,8 @@ class GangScheduler:
         self._dirty_ready_gangs: set = set()
         self._rollup_forced = 0
         self._pool_of_gpu: Dict[tuple, str] = {}
+        # (ns, gang, key) already surfaced as UnsatisfiableTopologyConstraint
+        self._unsat_surfaced: set = set()
 
     _REBUILD_EVERY = 100
 
@@ -445,6 +447,7 @@ class GangScheduler:
             chosen = chosen2
             result = self._place_gang_pods(nodes, pg, chosen)
         if result is None:
+            self._surface_unsatisfiable(pg, ns)
             return  # Permit rollback: nothing bound
         assignments, score = result
         by_name = {p["metadata"]["name"]: p for p in chosen}
@@ -461,6 +464,26 @@ class GangScheduler:
         except ApiError as e:
             report_api_error(self.store, c.KIND_PODGANG, ns,
                              pg["metadata"]["name"], "mark gang scheduled", e)
+
+    def _surface_unsatisfiable(self, pg: Obj, ns: str) -> None:
+        """TAS20 parity (topology_test.go unavailable-level scenario): when a gang
+        cannot place because its REQUIRED pack key labels no node at all, say so —
+        a Warning Event on the PodGang (deduped per gang+key). Transient capacity
+        shortages are not surfaced here; only a structurally absent topology level."""
+        req = ((pg.get("spec") or {}).get("topologyConstraint") or {}) \
+            .get("packConstraint", {}).get("required")
+        if not req:
+            return
+        key = (ns, pg["metadata"]["name"], req)
+        if key in self._unsat_surfaced:
+            return
+        for n in self.store.list("Node", copy_objects=False):
+            if req in (n["metadata"].get("labels") or {}):
+                return  # level exists somewhere — normal capacity wait
+        self._unsat_surfaced.add(key)
+        self.store.record_event(
+            pg, "Warning", "UnsatisfiableTopologyConstraint",
+            f"required pack key {req!r} labels no node in the cluster")
 
     def _count_bound(self, ns: str, group: Obj, pods_by_name: Dict) -> int:
         n = 0

@@ -772,3 +772,79 @@ def test_so4_explicit_startup_with_min_replicas(cluster):
     assert cluster.store.get(c.KIND_PCLQ, "default",
                              "so4-0-z")["spec"]["startsAfter"] == \
         ["so4-0-a", "so4-0-b"]
+
+
+def test_gs7_scaled_gangs_schedule_independently(cluster):
+    """GS7 (gang_scheduling_test.go:463): with PCSG minAvailable=1, replicas=3,
+    each scaled gang admits independently — under scarcity one scaled gang binds
+    while the other stays pending without blocking it; relief schedules the rest."""
+    sg = [{"name": "sx", "cliqueNames": ["b"], "replicas": 3, "minAvailable": 1}]
+    cluster.add_virtual_nodes(1, cpu="3", pods=64)
+    cluster.apply(_pcs("gs7", cliques=(("a", 1, 1), ("b", 1, 1)), sg=sg))
+    cluster.wait_pcs_available("gs7", timeout=20)  # base gang (a + sx j0) fits
+    cluster.wait_for(lambda: len(_pods(cluster, {c.LABEL_PART_OF: "gs7"})) == 4,
+                     timeout=15, desc="4 pods exist")
+    # capacity 3 cpu, 4 one-cpu pods: exactly one scaled gang binds
+    cluster.wait_for(
+        lambda: sum(1 for p in _pods(cluster, {c.LABEL_PART_OF: "gs7"})
+                    if p["spec"].get("nodeName")) == 3,
+        timeout=15, desc="3 of 4 pods bound")
+    time.sleep(0.3)  # settle: the fourth must stay pending, not flap
+    unbound = [p for p in _pods(cluster, {c.LABEL_PART_OF: "gs7"})
+               if not p["spec"].get("nodeName")]
+    assert len(unbound) == 1, "exactly one scaled gang must remain pending"
+    cluster.add_virtual_nodes(1, cpu="2", pods=64, prefix="extra")
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "gs7"}, 4, timeout=20)
+
+
+def test_tas11_clique_constraint_without_parent(cluster):
+    """TAS11 (topology_test.go:697): a member clique carries a host pack with NO
+    PCSG- or PCS-level constraint — each PCSG replica's clique packs one host on
+    its own, and the scaled gang carries the clique constraint as a group config
+    rather than gang-level."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=2, per_rack=2, gpus=4)
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 1}]
+    pcs = _pcs("tas11", cliques=(("b", 2, 2),), sg=sg, gpus=1)
+    pcs["spec"]["template"]["cliques"][0]["topologyConstraint"] = {
+        "pack": {"required": "host"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas11", timeout=20)
+    for j in (0, 1):
+        sel = {c.LABEL_PODCLIQUE: f"tas11-0-sg-{j}-b"}
+        cluster.wait_for(
+            lambda s=sel: len(_pods(cluster, s)) == 2
+            and all(p["spec"].get("nodeName") for p in _pods(cluster, s)),
+            timeout=15, desc=f"sg replica {j} bound")
+        ps = _pods(cluster, sel)
+        assert len({p["spec"]["nodeName"] for p in ps}) == 1, \
+            f"sg replica {j} clique spans hosts"
+    pg = cluster.store.get(c.KIND_PODGANG, "default", "tas11-0-sg-0")
+    cfgs = pg["spec"].get("topologyConstraintGroupConfigs") or []
+    assert any(g["topologyConstraint"]["packConstraint"]["required"]
+               == "kubernetes.io/hostname" for g in cfgs), cfgs
+    assert not (pg["spec"].get("topologyConstraint") or {}), \
+        "no gang-level constraint expected"
+
+
+def test_tas20_unavailable_topology_level_surfaces(cluster):
+    """TAS20 (topology_test.go unavailable-level scenario): a required pack level
+    that labels NO node is surfaced as a Warning Event on the PodGang instead of a
+    silently pending gang; once nodes carry the level, the gang schedules."""
+    cluster.store.create(_CTB)
+    cluster.add_virtual_nodes(2, cpu="8", pods=64)  # nodes lack the rack label
+    pcs = _pcs("tas20", cliques=(("w", 2, 2),))
+    pcs["spec"]["template"]["topologyConstraint"] = {"pack": {"required": "rack"}}
+    cluster.apply(pcs)
+
+    def surfaced():
+        return any(e["reason"] == "UnsatisfiableTopologyConstraint"
+                   and e["involvedObject"]["name"] == "tas20-0"
+                   for e in cluster.store.events)
+    cluster.wait_for(surfaced, timeout=15, desc="unsatisfiable-level event")
+    for n in cluster.store.list("Node"):
+        cluster.store.patch(
+            "Node", None, n["metadata"]["name"],
+            lambda o: o["metadata"].setdefault("labels", {}).update(
+                {"topology.kubernetes.io/rack": "rack0"}))
+    cluster.wait_pcs_available("tas20", timeout=20)

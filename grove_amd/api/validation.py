@@ -442,6 +442,10 @@ class TopologyConstraintValidator:
                                  for cl in tmpl.get("cliques") or []] + \
                   [sg.get("topologyConstraint")
                    for sg in tmpl.get("podCliqueScalingGroups") or []]:
+            if tc is not None and not tc.get("pack") and not tc.get("packDomain"):
+                # CEL parity: has(self.pack) || has(self.packDomain)
+                raise _err("spec.template.topologyConstraint",
+                           "topologyConstraint must set pack or packDomain")
             if tc and tc.get("topologyName"):
                 names.add(tc["topologyName"])
         if len(names) > 1:
@@ -463,6 +467,20 @@ class TopologyConstraintValidator:
         known = self._known_domains()
 
         def check(tc, path):
+            # CRD CEL-rule equivalents (podcliqueset.go TopologyConstraint
+            # x-kubernetes-validations): a constraint must express SOMETHING,
+            # pack{} may not be empty, and the deprecated packDomain is mutually
+            # exclusive with pack.
+            if "pack" in tc and not (tc.get("pack") or {}).keys() & \
+                    {"required", "preferred"}:
+                raise _err(f"{path}.pack",
+                           "pack must set required and/or preferred")
+            if not tc.get("pack") and not tc.get("packDomain"):
+                raise _err(path, "topologyConstraint must set pack or packDomain")
+            if tc.get("packDomain") and (tc.get("pack") or {}).keys() & \
+                    {"required", "preferred"}:
+                raise _err(f"{path}.packDomain",
+                           "packDomain (deprecated) and pack are mutually exclusive")
             req, pref = self._domains_of(tc)
             for d, f in ((req, "required"), (pref, "preferred")):
                 if d is not None and d not in known:

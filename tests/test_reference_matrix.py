@@ -848,3 +848,27 @@ def test_tas20_unavailable_topology_level_surfaces(cluster):
             lambda o: o["metadata"].setdefault("labels", {}).update(
                 {"topology.kubernetes.io/rack": "rack0"}))
     cluster.wait_pcs_available("tas20", timeout=20)
+
+
+def test_tas22_topology_cel_validation(cluster):
+    """TAS22 (CRD CEL rules, podcliqueset.go x-kubernetes-validations): the
+    apiserver-side equivalents of the TopologyConstraint CEL rules — empty
+    constraint, empty pack, pack+packDomain mutual exclusion — reject on create;
+    constraints are immutable on update."""
+    cluster.store.create(_CTB)
+
+    def pcs(name, tc):
+        p = _pcs(name, cliques=(("w", 1, 1),))
+        p["spec"]["template"]["topologyConstraint"] = tc
+        return p
+
+    for tc in ({}, {"pack": {}},
+               {"pack": {"required": "rack"}, "packDomain": "rack"}):
+        with pytest.raises(Exception):
+            cluster.store.create(pcs("tas22-bad", tc))
+    cluster.store.create(pcs("tas22", {"pack": {"required": "rack"}}))
+    with pytest.raises(Exception):
+        cluster.store.patch(
+            c.KIND_PCS, "default", "tas22",
+            lambda o: o["spec"]["template"].update(
+                topologyConstraint={"pack": {"preferred": "rack"}}))

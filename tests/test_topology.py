@@ -290,11 +290,13 @@ def test_ctb_key_change_propagates_to_group_configs(cluster):
         "apiVersion": c.API_VERSION, "kind": c.KIND_PCS,
         "metadata": {"name": "gcfg"},
         "spec": {"replicas": 1, "template": {
-            # topologyName only — no gang-level pack, so the ONLY drift is in the
-            # clique-level group config (isolates the group-config update path)
-            "topologyConstraint": {"topologyName": "topo-gc"},
+            # no gang-level constraint (a bare topologyName violates the CRD CEL
+            # rule has(pack)||has(packDomain)) — the clique carries both the
+            # topologyName and the pack, so the ONLY drift is in the clique-level
+            # group config (isolates the group-config update path)
             "cliques": [{"name": "w",
-                         "topologyConstraint": {"pack": {"required": "rack"}},
+                         "topologyConstraint": {"topologyName": "topo-gc",
+                                                "pack": {"required": "rack"}},
                          "spec": {"roleName": "w", "replicas": 1,
                                   "podSpec": {"containers": [
                                       {"name": "m", "image": "x"}]}}}]}}}

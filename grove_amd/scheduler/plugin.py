@@ -508,11 +508,39 @@ class GangScheduler:
         if not cfgs:
             return place_gang_constrained(nodes, [self._pod_request(p) for p in chosen],
                                           req_key, pref_key, place_fn=self._place)
-        snap = snapshot(nodes)
         by_clique: Dict[str, List[Obj]] = {}
         for p in chosen:
             by_clique.setdefault(
                 p["metadata"]["labels"].get(c.LABEL_PODCLIQUE, ""), []).append(p)
+
+        if not req_key:
+            return self._place_cfg_groups(nodes, cfgs, by_clique, chosen,
+                                          req_key, pref_key)
+        # gang-level REQUIRED key + subgroup configs: the parent constraint binds
+        # ALL subgroups into ONE domain of req_key (KAI hierarchical-subgroup
+        # semantics) — try each domain pool, fullest-fitting first
+        domains: Dict[str, List[NodeFree]] = {}
+        for n in nodes:
+            v = n.labels.get(req_key)
+            if v is not None:
+                domains.setdefault(v, []).append(n)
+        order = sorted(domains, key=lambda v: (
+            -sum(len(n.gpu_ids) for n in domains[v]),
+            -sum(n.cpu_milli for n in domains[v]), v))
+        for v in order:
+            res = self._place_cfg_groups(domains[v], cfgs, by_clique, chosen,
+                                         req_key, pref_key)
+            if res is not None:
+                return res
+        return None
+
+    def _place_cfg_groups(self, nodes: List[NodeFree], cfgs: List[Obj],
+                          by_clique: Dict[str, List[Obj]], chosen: List[Obj],
+                          req_key, pref_key):
+        """Place each topologyConstraintGroupConfig subgroup (then the remainder)
+        within the given node pool, all-or-nothing with rollback."""
+        from .placement import place_gang_constrained, snapshot, restore
+        snap = snapshot(nodes)
         claimed: set = set()
         all_assignments: List[Assignment] = []
         worst_score = float("inf")

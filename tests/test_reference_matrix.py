@@ -872,3 +872,112 @@ def test_tas22_topology_cel_validation(cluster):
             c.KIND_PCS, "default", "tas22",
             lambda o: o["spec"]["template"].update(
                 topologyConstraint={"pack": {"preferred": "rack"}}))
+
+
+def test_gs10_pcs_scaling_min_replicas_advanced(cluster):
+    """GS10 (gang_scheduling_test.go:754): scaling the PCS under scarcity — the
+    NEW replica's gang admits at the clique minAvailable (2 of 3), the remainder
+    lands when capacity arrives; the original replica is never disturbed."""
+    cluster.add_virtual_nodes(1, cpu="5", pods=64)
+    cluster.apply(_pcs("gs10", replicas=1, cliques=(("w", 3, 2),)))
+    cluster.wait_pcs_available("gs10", timeout=20)
+    r0_uids = {p["metadata"]["uid"]
+               for p in _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "0"})}
+    cluster.store.patch(c.KIND_PCS, "default", "gs10",
+                        lambda o: o["spec"].update(replicas=2))
+    # 2 cpu left: replica 1 admits at minAvailable=2, third pod stays pending
+    cluster.wait_for(
+        lambda: sum(1 for p in _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})
+                    if p["spec"].get("nodeName")) == 2,
+        timeout=20, desc="replica 1 admitted at min")
+    time.sleep(0.3)
+    r1 = _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "1"})
+    assert len(r1) == 3
+    assert sum(1 for p in r1 if not p["spec"].get("nodeName")) == 1
+    assert {p["metadata"]["uid"]
+            for p in _pods(cluster, {c.LABEL_PCS_REPLICA_INDEX: "0"})} == r0_uids
+    cluster.add_virtual_nodes(1, cpu="1", pods=8, prefix="late")
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "gs10"}, 6, timeout=20)
+
+
+def test_tas6_standalone_clique_with_pcs_zone_constraint(cluster):
+    """TAS6 (topology_test.go:417): a PCS-level zone pack applies to a standalone
+    clique — all its pods land in ONE zone even though several zones have room."""
+    cluster.store.create({
+        "apiVersion": c.API_VERSION, "kind": c.KIND_CTB,
+        "metadata": {"name": "zoned"},
+        "spec": {"levels": [
+            {"domain": "zone", "key": "topology.kubernetes.io/zone"},
+            {"domain": "host", "key": "kubernetes.io/hostname"}]}})
+    from grove_amd.kubelet.virtual import make_virtual_node
+    for z in range(2):
+        for i in range(2):
+            n = make_virtual_node(f"z{z}n{i}", gpus=0, cpu="2", pods=16)
+            n["metadata"]["labels"]["topology.kubernetes.io/zone"] = f"zone{z}"
+            cluster.store.create(n)
+    pcs = _pcs("tas6", cliques=(("w", 4, 4),))
+    pcs["spec"]["template"]["topologyConstraint"] = {"pack": {"required": "zone"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas6", timeout=20)
+    zones = {cluster.store.get("Node", None, p["spec"]["nodeName"])
+             ["metadata"]["labels"]["topology.kubernetes.io/zone"]
+             for p in _pods(cluster, {c.LABEL_PODCLIQUE: "tas6-0-w"})}
+    assert len(zones) == 1, f"clique spans zones {zones}"
+
+
+def test_tas8_full_hierarchy_cascading_constraints(cluster):
+    """TAS8 (topology_test.go:529): three-level cascade — PCS packs the gang into
+    one rack, the PCSG packs each replica onto one host inside it, and every
+    level is satisfied simultaneously."""
+    cluster.store.create(_CTB)
+    _rack_nodes(cluster, racks=2, per_rack=2, gpus=4)
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 2, "minAvailable": 2,
+           "topologyConstraint": {"pack": {"required": "host"}}}]
+    pcs = _pcs("tas8", cliques=(("a", 1, 1), ("b", 2, 2)), sg=sg, gpus=1)
+    pcs["spec"]["template"]["topologyConstraint"] = {"pack": {"required": "rack"}}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("tas8", timeout=20)
+    node_rack = {n["metadata"]["name"]:
+                 n["metadata"]["labels"]["topology.kubernetes.io/rack"]
+                 for n in cluster.store.list("Node")}
+    pods = _pods(cluster, {c.LABEL_PART_OF: "tas8"})
+    assert len({node_rack[p["spec"]["nodeName"]] for p in pods}) == 1, \
+        "gang spans racks"
+    for j in (0, 1):
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: f"tas8-0-sg-{j}-b"})
+        assert len({p["spec"]["nodeName"] for p in ps}) == 1, \
+            f"sg replica {j} spans hosts"
+
+
+def test_od6_on_delete_mixed_cliques_and_pcsg(cluster):
+    """OD6 (update/ondelete_test.go:361): OnDelete with a standalone clique AND a
+    PCSG — a spec change replaces nothing; deleting a PCSG member pod recreates
+    it on the new template while the standalone clique keeps its old pod."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 1, "minAvailable": 1}]
+    cluster.add_virtual_nodes(2, cpu="16", pods=64)
+    cluster.apply(_pcs("od6", cliques=(("a", 1, 1), ("b", 1, 1)), sg=sg,
+                       strategy=c.UPDATE_ON_DELETE))
+    cluster.wait_pcs_available("od6", timeout=20)
+    pods0 = {p["metadata"]["name"]: p["metadata"]["labels"][
+        c.LABEL_POD_TEMPLATE_HASH]
+        for p in _pods(cluster, {c.LABEL_PART_OF: "od6"})}
+    cur = cluster.store.get(c.KIND_PCS, "default", "od6")
+    for cl in cur["spec"]["template"]["cliques"]:
+        cl["spec"]["podSpec"]["containers"][0]["image"] = "img:v2"
+    cluster.apply(cur)
+    time.sleep(0.4)
+    assert {p["metadata"]["name"]
+            for p in _pods(cluster, {c.LABEL_PART_OF: "od6"})} == set(pods0)
+    victim = _pods(cluster, {c.LABEL_PODCLIQUE: "od6-0-sg-0-b"})[0]
+    old_hash_b = victim["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH]
+    cluster.store.delete("Pod", "default", victim["metadata"]["name"])
+
+    def sg_pod_updated():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "od6-0-sg-0-b"})
+        return len(ps) == 1 and ps[0]["metadata"]["labels"][
+            c.LABEL_POD_TEMPLATE_HASH] != old_hash_b and cond.pod_is_ready(ps[0])
+    cluster.wait_for(sg_pod_updated, timeout=20, desc="sg pod on new template")
+    a_pods = _pods(cluster, {c.LABEL_PODCLIQUE: "od6-0-a"})
+    assert len(a_pods) == 1
+    assert a_pods[0]["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == \
+        pods0[a_pods[0]["metadata"]["name"]], "standalone clique must keep its pod"

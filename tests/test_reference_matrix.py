@@ -981,3 +981,125 @@ def test_od6_on_delete_mixed_cliques_and_pcsg(cluster):
     assert len(a_pods) == 1
     assert a_pods[0]["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == \
         pods0[a_pods[0]["metadata"]["name"]], "standalone clique must keep its pod"
+
+
+def test_ru15_pcsg_scale_out_before_update(cluster):
+    """RU15 (rolling_recreate_test.go:522): the PCSG is scaled OUT and settles
+    BEFORE the template update starts — the update then rolls every member,
+    including the previously scaled-out one."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 1, "minAvailable": 1}]
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru15", cliques=(("b", 1, 1),), sg=sg))
+    cluster.wait_pcs_available("ru15", timeout=20)
+    cluster.store.patch(c.KIND_PCSG, "default", "ru15-0-sg",
+                        lambda o: o["spec"].update(replicas=2))
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "ru15"}, 2, timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru15")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "b", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def all_new():
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru15"})
+        return len(ps) == 2 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(all_new, timeout=60, desc="both PCSG members rolled")
+
+
+def test_ru17_pcsg_scale_in_before_update(cluster):
+    """RU17 (rolling_recreate_test.go:640): the PCSG is scaled IN and settles
+    BEFORE the update — the update rolls only the survivors."""
+    sg = [{"name": "sg", "cliqueNames": ["b"], "replicas": 3, "minAvailable": 1}]
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru17", cliques=(("b", 1, 1),), sg=sg))
+    cluster.wait_pcs_available("ru17", timeout=20)
+    cluster.store.patch(c.KIND_PCSG, "default", "ru17-0-sg",
+                        lambda o: o["spec"].update(replicas=1))
+    cluster.wait_for(
+        lambda: cluster.store.try_get(c.KIND_PCLQ, "default", "ru17-0-sg-2-b")
+        is None and len(_pods(cluster, {c.LABEL_PART_OF: "ru17"})) == 1,
+        timeout=20, desc="scale-in settled")
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru17")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "b", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def survivor_new():
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru17"})
+        return len(ps) == 1 and ps[0]["metadata"]["labels"][
+            c.LABEL_POD_TEMPLATE_HASH] == new_hash and cond.pod_is_ready(ps[0])
+    cluster.wait_for(survivor_new, timeout=60, desc="survivor rolled")
+
+
+def test_ru19_clique_scale_out_before_update(cluster):
+    """RU19 (rolling_recreate_test.go:704): an autoscaled clique is scaled out
+    (HPA path, direct PCLQ patch) and settles BEFORE the update — the update then
+    rolls the full scaled set, preserving the HPA's replica decision."""
+    cluster.add_virtual_nodes(2, cpu="16", pods=64)
+    pcs = _pcs("ru19", cliques=(("w", 2, 1),))
+    pcs["spec"]["template"]["cliques"][0]["spec"]["autoScalingConfig"] = {
+        "minReplicas": 1, "maxReplicas": 5}
+    cluster.apply(pcs)
+    cluster.wait_pcs_available("ru19", timeout=20)
+    cluster.store.patch(c.KIND_PCLQ, "default", "ru19-0-w",
+                        lambda o: o["spec"].update(replicas=3))
+    cluster.wait_pods_ready({c.LABEL_PODCLIQUE: "ru19-0-w"}, 3, timeout=20)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru19")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "w", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def all_updated():
+        ps = _pods(cluster, {c.LABEL_PODCLIQUE: "ru19-0-w"})
+        return len(ps) == 3 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(all_updated, timeout=60,
+                     desc="3 pods rolled, HPA replicas preserved")
+
+
+def test_ru13_pcs_scale_in_of_final_updating_replica(cluster):
+    """RU13 (rolling_recreate_test.go:400): while the FINAL replica is the one
+    still updating, the PCS is scaled in past it — the update must end cleanly
+    with every surviving pod on the new template."""
+    cluster.add_virtual_nodes(3, cpu="16", pods=64)
+    cluster.apply(_pcs("ru13", replicas=2, cliques=(("w", 1, 1),)))
+    cluster.wait_pcs_available("ru13", timeout=20, min_available=2)
+    cur = cluster.store.get(c.KIND_PCS, "default", "ru13")
+    cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"]["containers"][0][
+        "image"] = "img:v2"
+    cluster.apply(cur)
+
+    # wait until exactly one replica has finished updating (the other is the
+    # final in-flight ordinal), then scale that final replica away
+    def one_done():
+        st = (cluster.store.get(c.KIND_PCS, "default", "ru13").get("status")
+              or {})
+        prog = st.get("updateProgress") or {}
+        return int(st.get("updatedReplicas") or 0) >= 1 and \
+            not prog.get("updateEndedAt")
+    cluster.wait_for(one_done, timeout=40, desc="first replica updated")
+    cluster.store.patch(c.KIND_PCS, "default", "ru13",
+                        lambda o: o["spec"].update(replicas=1))
+    from grove_amd.utils.hashing import pod_template_hash
+    new_hash = pod_template_hash(
+        "w", cur["spec"]["template"]["cliques"][0]["spec"]["podSpec"])
+
+    def ended_clean():
+        pcs_obj = cluster.store.get(c.KIND_PCS, "default", "ru13")
+        prog = (pcs_obj.get("status") or {}).get("updateProgress") or {}
+        ps = _pods(cluster, {c.LABEL_PART_OF: "ru13"})
+        return prog.get("updateEndedAt") and len(ps) == 1 and all(
+            p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
+            and cond.pod_is_ready(p) for p in ps)
+    cluster.wait_for(ended_clean, timeout=60, desc="update ended after scale-in")

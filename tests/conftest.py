@@ -62,6 +62,16 @@ def cluster(request):
         print(f"  pods={len(pods)} gated={gated} bound={bound}", file=sys.stderr)
         for ev in cl.store.events[-10:]:
             print("  event:", ev.get("reason"), ev.get("message"), file=sys.stderr)
+        # PCSG write history: rv-ordered spec.replicas transitions (who stomped?)
+        try:
+            tbl = cl.store._table("PodCliqueScalingGroup")
+            for (rv, ev2, o) in tbl.history[-40:]:
+                print(f"  pcsg-history rv={rv} {ev2} {o['metadata']['name']} "
+                      f"replicas={(o.get('spec') or {}).get('replicas')} "
+                      f"gen={o['metadata'].get('generation')}",
+                      file=sys.stderr)
+        except Exception as e:
+            print("  pcsg-history unavailable:", e, file=sys.stderr)
     cl.stop()
 
 

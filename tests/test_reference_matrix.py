@@ -1103,3 +1103,47 @@ def test_ru13_pcs_scale_in_of_final_updating_replica(cluster):
             p["metadata"]["labels"][c.LABEL_POD_TEMPLATE_HASH] == new_hash
             and cond.pod_is_ready(p) for p in ps)
     cluster.wait_for(ended_clean, timeout=60, desc="update ended after scale-in")
+
+
+def test_gs12_complex_multi_pcsg_scaling(cluster):
+    """GS12 (gang_scheduling_test.go:1009): two PCSGs with different minAvailable
+    scaled out and back in — base gang covers [0,minAvailable) of each group,
+    every scaled replica gets its own gang, scale-in GCs the scaled gangs and
+    their pods, and the set stays available throughout."""
+    sg = [{"name": "pf", "cliqueNames": ["p"], "replicas": 2, "minAvailable": 1},
+          {"name": "dc", "cliqueNames": ["d"], "replicas": 2, "minAvailable": 2}]
+    cluster.add_virtual_nodes(4, cpu="16", pods=64)
+    cluster.apply(_pcs("gs12", cliques=(("a", 1, 1), ("p", 1, 1), ("d", 1, 1)),
+                       sg=sg))
+    cluster.wait_pcs_available("gs12", timeout=20)
+    gangs = {g["metadata"]["name"]
+             for g in cluster.store.list(c.KIND_PODGANG, "default",
+                                         {c.LABEL_PART_OF: "gs12"})}
+    # base covers a + pf j0 + dc j0,j1; pf j1 is the one scaled gang
+    assert gangs == {"gs12-0", "gs12-0-pf-0"}, gangs
+    # scale pf 2->4 and dc 2->3: two more pf scaled gangs, one dc scaled gang
+    cluster.store.patch(c.KIND_PCSG, "default", "gs12-0-pf",
+                        lambda o: o["spec"].update(replicas=4))
+    cluster.store.patch(c.KIND_PCSG, "default", "gs12-0-dc",
+                        lambda o: o["spec"].update(replicas=3))
+    cluster.wait_for(
+        lambda: {g["metadata"]["name"]
+                 for g in cluster.store.list(c.KIND_PODGANG, "default",
+                                             {c.LABEL_PART_OF: "gs12"})}
+        == {"gs12-0", "gs12-0-pf-0", "gs12-0-pf-1", "gs12-0-pf-2",
+            "gs12-0-dc-0"},
+        timeout=20, desc="scaled gangs created")
+    cluster.wait_pods_ready({c.LABEL_PART_OF: "gs12"}, 8, timeout=30)
+    # scale pf back to 2: scaled gangs pf-1/pf-2 and their pods are GC'd
+    cluster.store.patch(c.KIND_PCSG, "default", "gs12-0-pf",
+                        lambda o: o["spec"].update(replicas=2))
+    cluster.wait_for(
+        lambda: {g["metadata"]["name"]
+                 for g in cluster.store.list(c.KIND_PODGANG, "default",
+                                             {c.LABEL_PART_OF: "gs12"})}
+        == {"gs12-0", "gs12-0-pf-0", "gs12-0-dc-0"},
+        timeout=20, desc="scaled gangs GCd")
+    cluster.wait_for(
+        lambda: len(_pods(cluster, {c.LABEL_PART_OF: "gs12"})) == 6,
+        timeout=20, desc="scaled pods removed")
+    cluster.wait_pcs_available("gs12", timeout=20)

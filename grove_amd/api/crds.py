@@ -228,6 +228,9 @@ def install_crds(server_url: str) -> int:
                     urllib.request.Request(f"{base}/{name}", headers=headers),
                     timeout=15, context=ctx) as r:
                 cur = json.loads(r.read())
+            if cur.get("spec") == crd["spec"]:
+                n += 1  # unchanged — idempotent re-run, no write
+                continue
             crd = dict(crd)
             crd["metadata"] = dict(crd["metadata"],
                                    resourceVersion=cur["metadata"]

@@ -46,8 +46,11 @@ PLURALS: Dict[str, str] = {
     "horizontalpodautoscalers": "HorizontalPodAutoscaler",
     "resourceclaims": "ResourceClaim",
     "nodes": "Node",
+    # apiextensions surface: lets the install-crds initc target THIS apiserver
+    "customresourcedefinitions": "CustomResourceDefinition",
 }
-CLUSTER_SCOPED_PLURALS = {"clustertopologybindings", "schedulertopologies", "nodes"}
+CLUSTER_SCOPED_PLURALS = {"clustertopologybindings", "schedulertopologies", "nodes",
+                          "customresourcedefinitions"}
 
 
 def build_app(store: Store, metrics_fn=None,

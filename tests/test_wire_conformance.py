@@ -212,3 +212,31 @@ def test_field_selector_list_and_watch(served):
     cluster.store.create(p)
     assert done.wait(10)
     assert got == ["wp001", "wp009"]
+
+
+def test_crd_installer_idempotent_over_wire():
+    """CRD3 (crd installer e2e): grove-install-crds applies every CRD to the
+    apiserver's apiextensions surface; a second run is a no-op (spec-equal CRDs
+    are skipped, resourceVersions untouched) — installer.go server-side-apply
+    idempotence parity."""
+    from grove_amd.kubecore.store import Store
+    from grove_amd.kubecore.apiserver import ApiServer
+    from grove_amd.api.crds import install_crds, render_all
+    import socket
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    store = Store()
+    srv = ApiServer(store, port=port).start()
+    try:
+        n1 = install_crds(srv.url)
+        assert n1 == len(render_all()) >= 5
+        rvs = {o["metadata"]["name"]: o["metadata"]["resourceVersion"]
+               for o in store.list("CustomResourceDefinition")}
+        assert len(rvs) == n1
+        n2 = install_crds(srv.url)
+        assert n2 == n1
+        rvs2 = {o["metadata"]["name"]: o["metadata"]["resourceVersion"]
+                for o in store.list("CustomResourceDefinition")}
+        assert rvs2 == rvs, "second install must not rewrite unchanged CRDs"
+    finally:
+        srv.stop()

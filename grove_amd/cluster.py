@@ -127,8 +127,24 @@ class Cluster:
             self.c_pcs.enqueue(ns, md["name"])
             # spec changes flow to child PCSGs (template propagation — e.g. an
             # OnDelete template change must reach member PCLQs without any PCSG
-            # spec change; reference podcliquescalinggroup watch predicates)
-            if _old is not None and md.get("generation") !=                     _old["metadata"].get("generation"):
+            # spec change; reference podcliquescalinggroup watch predicates).
+            # Rolling-update replica SELECTION is a status-only write (no
+            # generation bump) but gates the PCSG member-template propagation -
+            # without this edge a PCSG pass that ran just before selection
+            # latches its fingerprint and the selected replica's members never
+            # roll (observed stall).
+            def _upd_sig(o):
+                if o is None:
+                    return None
+                st = o.get("status") or {}
+                prog = st.get("updateProgress") or {}
+                return (st.get("currentGenerationHash"),
+                        tuple(sorted(int(e.get("replicaIndex", -1))
+                                     for e in prog.get("currentlyUpdating")
+                                     or [])))
+            if _old is not None and (
+                    md.get("generation") != _old["metadata"].get("generation")
+                    or _upd_sig(obj) != _upd_sig(_old)):
                 for g in self.store.list(c.KIND_PCSG, ns,
                                          {c.LABEL_PART_OF: md["name"]},
                                          copy_objects=False):

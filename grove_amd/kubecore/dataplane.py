@@ -141,7 +141,9 @@ class DataPlane:
             if method in ("POST", "PUT", "PATCH"):
                 body = await self._read_body(receive)
             if method == "GET":
-                await _json(send, 200, self.store.get(kind, ns, name))
+                # zero-copy read: the object is serialized immediately and store
+                # objects are immutable after insert, so no defensive copy needed
+                await _json(send, 200, self.store.get(kind, ns, name, copy=False))
                 return
             user = self._user(headers)
             if method == "POST":
@@ -216,14 +218,16 @@ class DataPlane:
         fields = parse_field_selector(q.get("fieldSelector"))
         limit = int(q["limit"]) if q.get("limit") else None
         if fields and not limit and not q.get("continue"):
-            items = self.store.list(kind, ns, selector, field_selector=fields)
+            items = self.store.list(kind, ns, selector, field_selector=fields,
+                                    copy_objects=False)  # serialized immediately
             await _json(send, 200, {
                 "kind": f"{kind}List", "apiVersion": "v1",
                 "metadata": {"resourceVersion": self.store.current_rv()},
                 "items": items})
             return
         items, next_cont, rv = self.store.list_page(
-            kind, ns, selector, limit, q.get("continue") or None)
+            kind, ns, selector, limit, q.get("continue") or None,
+            copy_objects=False)  # serialized immediately
         meta: Dict[str, Any] = {"resourceVersion": rv}
         if next_cont:
             meta["continue"] = next_cont

@@ -592,7 +592,8 @@ class Store:
     def list_page(self, kind: str, namespace: Optional[str] = None,
                   label_selector: Optional[Dict[str, str]] = None,
                   limit: Optional[int] = None,
-                  continue_token: Optional[str] = None
+                  continue_token: Optional[str] = None,
+                  copy_objects: bool = True
                   ) -> Tuple[List[Obj], Optional[str], str]:
         """Chunked list (apiserver limit/continue parity): deterministic
         (namespace, name) order; returns (items, next_continue, resourceVersion)."""
@@ -601,7 +602,8 @@ class Store:
         if limit is None and continue_token is None:
             # unchunked: use the indexed list path (no full-table sort; the
             # inverted label indexes serve selector queries)
-            return (self.list(kind, namespace, label_selector), None,
+            return (self.list(kind, namespace, label_selector,
+                              copy_objects=copy_objects), None,
                     self.current_rv())
         start_after: Optional[Tuple[str, str]] = None
         if continue_token:
@@ -633,7 +635,8 @@ class Store:
             refs = refs[:limit]
             nxt = base64.b64encode(
                 _json.dumps(list(refs[-1][0])).encode()).decode()
-        return [json_copy(o) for _k, o in refs], nxt, rv
+        return ([json_copy(o) for _k, o in refs] if copy_objects
+                else [o for _k, o in refs]), nxt, rv
 
     def record_event(self, involved: Obj, etype: str, reason: str, message: str) -> None:
         ev = {

@@ -61,6 +61,26 @@ __device__ __forceinline__ void tile_map(int bid, int num_pid_m, int tiles_n,
     tile_n = in_group / group_size;
 }
 
+// Rectangular per-XCD map for LARGE grids (mode 1): with the column map above, at
+// 8192³ each XCD's resident blocks share one M-row and walk ALL tiles_n, so every
+// XCD streams the whole B matrix from HBM each dispatch wave (~33 K-slabs/XCD).
+// Here a 256-block window gives the blocks resident on one XCD (bid % 8, the
+// hardware round-robin) an m_per_xcd x n_win RECTANGLE of tiles: the XCD's L2
+// re-serves m_per_xcd A-slabs + n_win B-slabs (e.g. 4+8=12 slabs instead of 33).
+// Host side guarantees divisibility before selecting mode 1.
+__device__ __forceinline__ void tile_map_rect(int bid, int tiles_m, int tiles_n,
+                                              int m_per_xcd, int n_win,
+                                              int& tile_m, int& tile_n) {
+    int mwin_tiles = 8 * m_per_xcd;          // M tiles covered per window
+    int win = bid >> 8;                      // 256 blocks per window
+    int idx = bid & 255;
+    int xcd = idx & 7;
+    int slot = idx >> 3;                     // 0 .. m_per_xcd*n_win - 1
+    int nwins = tiles_n / n_win;             // windows per M band
+    tile_m = (win / nwins) * mwin_tiles + xcd * m_per_xcd + slot % m_per_xcd;
+    tile_n = (win % nwins) * n_win + slot / m_per_xcd;
+}
+
 // Stage one (rows x BK) bf16 tile via 16-B LDS-DMA. The glds LDS destination is
 // wave-uniform-base + lane*16 and the LDS image is lane-linear in piece index, so the
 // XOR swizzle is applied on the SOURCE address (guide §5 rule 21: swizzled images via
@@ -93,14 +113,19 @@ __global__ __launch_bounds__(512, 1)
 void mfma_gemm_bf16_256_kernel(const bf16* __restrict__ A,   // [M][K] row-major
                                const bf16* __restrict__ Bt,  // [N][K] row-major
                                float* __restrict__ C,        // [M][N]
-                               int M, int N, int K) {
+                               int M, int N, int K,
+                               int m_per_xcd, int n_win) {   // 0,0 = column map
     constexpr int BM = 256, BN = 256;
     __shared__ __bf16 smem[2 * (BM * BK + BN * BK)];
     auto sAp = [&](int b) { return smem + b * (BM * BK + BN * BK); };
     auto sBp = [&](int b) { return smem + b * (BM * BK + BN * BK) + BM * BK; };
 
     int tile_m, tile_n;
-    tile_map(blockIdx.x, M / BM, N / BN, tile_m, tile_n);
+    if (m_per_xcd)
+        tile_map_rect(blockIdx.x, M / BM, N / BN, m_per_xcd, n_win,
+                      tile_m, tile_n);
+    else
+        tile_map(blockIdx.x, M / BM, N / BN, tile_m, tile_n);
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
     const int lane = tid % WAVE;
@@ -329,15 +354,29 @@ __global__ void stream_triad_kernel(const float4* __restrict__ a,
 // ---------------------------------------------------------------- host wrappers
 
 static void launch_gemm(const torch::Tensor& a, const torch::Tensor& bt,
-                        torch::Tensor& c, int64_t M, int64_t N, int64_t K) {
+                        torch::Tensor& c, int64_t M, int64_t N, int64_t K,
+                        int64_t map_mode = -1) {
     hipStream_t stream = at::hip::getCurrentHIPStream();
     // 256-tile needs >=256 workgroups to fill the 256-CU chip (one WG per CU at
     // 128 KB LDS); smaller grids run the 128-tile at 4x the block count.
     if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 256) {
+        int tiles_m = (int)(M / 256), tiles_n = (int)(N / 256);
+        // rectangular per-XCD map when the grid divides into 256-block windows
+        // (large K streams benefit; the column map wins at <= 16x16 grids)
+        int m_per_xcd = 0, n_win = 0;
+        int mpx = std::min<int>(4, std::max(1, tiles_m / 8));
+        int nw = 256 / (8 * mpx);
+        bool want_rect = (map_mode == 1) ||
+                         (map_mode < 0 && (int64_t)tiles_m * tiles_n > 256);
+        if (want_rect && map_mode != 0 &&
+            tiles_m % (8 * mpx) == 0 && tiles_n % nw == 0) {
+            m_per_xcd = mpx; n_win = nw;
+        }
         dim3 grid((M / 256) * (N / 256)), block(512);
         hipLaunchKernelGGL(mfma_gemm_bf16_256_kernel, grid, block, 0, stream,
                            (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-                           c.data_ptr<float>(), (int)M, (int)N, (int)K);
+                           c.data_ptr<float>(), (int)M, (int)N, (int)K,
+                           m_per_xcd, n_win);
     } else {
         dim3 grid((M / 128) * (N / 128)), block(256);
         hipLaunchKernelGGL(mfma_gemm_bf16_128_kernel, grid, block, 0, stream,
@@ -366,17 +405,18 @@ torch::Tensor mfma_gemm_bf16(torch::Tensor a, torch::Tensor bt) {
 }
 
 // Pod payload: `iters` GEMM steps on pre-allocated buffers; returns achieved TFLOP/s.
-double burn_gemm(int64_t m, int64_t n, int64_t k, int64_t iters) {
+double burn_gemm(int64_t m, int64_t n, int64_t k, int64_t iters,
+                 int64_t map_mode) {
     check_dims(m, n, k);
     auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
     auto a = torch::randn({m, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
     auto bt = torch::randn({n, k}, opt.dtype(torch::kFloat32)).to(torch::kBFloat16);
     auto c = torch::empty({m, n}, opt.dtype(torch::kFloat32));
     hipStream_t stream = at::hip::getCurrentHIPStream();
-    launch_gemm(a, bt, c, m, n, k);  // warmup
+    launch_gemm(a, bt, c, m, n, k, map_mode);  // warmup
     C10_HIP_CHECK(hipStreamSynchronize(stream));
     auto t0 = std::chrono::steady_clock::now();
-    for (int64_t i = 0; i < iters; ++i) launch_gemm(a, bt, c, m, n, k);
+    for (int64_t i = 0; i < iters; ++i) launch_gemm(a, bt, c, m, n, k, map_mode);
     C10_HIP_CHECK(hipStreamSynchronize(stream));
     auto t1 = std::chrono::steady_clock::now();
     double secs = std::chrono::duration<double>(t1 - t0).count();
@@ -412,10 +452,12 @@ double stream_triad(int64_t n_floats, int64_t iters) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "grove_amd MI355X pod-payload kernels (MFMA bf16 GEMM, HBM stream)";
     m.def("mfma_gemm_bf16", &mfma_gemm_bf16, "C[M,N]=A[M,K]@Bt[N,K]^T (bf16 in, fp32 out)");
-    m.def("burn_gemm", &burn_gemm, "run iters GEMM steps; returns TFLOP/s");
+    m.def("burn_gemm", &burn_gemm, "run iters GEMM steps; returns TFLOP/s",
+          pybind11::arg("m"), pybind11::arg("n"), pybind11::arg("k"),
+          pybind11::arg("iters"), pybind11::arg("map_mode") = -1);
     m.def("burn_gemm_v",
           [](int64_t m_, int64_t n_, int64_t k_, int64_t it, int64_t) {
-              return burn_gemm(m_, n_, k_, it);
+              return burn_gemm(m_, n_, k_, it, -1);
           }, "compat alias for burn_gemm (bk arg ignored)");
     m.def("stream_triad", &stream_triad, "HBM triad; returns GB/s");
     m.def("decode_gemv", &decode_gemv, "y[N]=W[N,K]@x[K] (bf16 in, fp32 out)");

@@ -444,6 +444,9 @@ def main() -> None:
     comm = Comm()
     n_gpus = args.gpus or comm.world
     gang_size = args.gang_size or n_gpus
+    if gang_size > n_gpus * max(1, comm.world):
+        sys.exit(f"--gang-size {gang_size} cannot fit: the bench node advertises "
+                 f"{n_gpus} amd.com/gpu per rank (use --gpus >= gang size)")
 
     if comm.rank != 0:
         _serve_agent(comm)

@@ -65,6 +65,20 @@ def build_app(store: Store, metrics_fn=None,
             if parsed is not None:
                 await plane(scope, receive, send, parsed)
                 return
+            # debug introspection carries cluster state (events, stacks) and can
+            # burn CPU (profile): when authentication is configured, require it
+            # there too — anonymous callers get 403
+            if auth_tokens and scope.get("path", "").startswith("/debug"):
+                from .identity import ANONYMOUS_USER
+                headers = {k.lower(): v for k, v in scope.get("headers") or []}
+                if plane._user(headers) == ANONYMOUS_USER:
+                    body = (b'{"kind":"Status","status":"Failure",'
+                            b'"reason":"Forbidden","code":403}')
+                    await send({"type": "http.response.start", "status": 403,
+                                "headers": [(b"content-type",
+                                             b"application/json")]})
+                    await send({"type": "http.response.body", "body": body})
+                    return
         await aux(scope, receive, send)
 
     app.dataplane = plane  # ApiServer.stop() signals open watch streams

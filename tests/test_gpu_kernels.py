@@ -329,3 +329,4 @@ def test_gpu_chaos_payloads_under_churn():
         assert load_gpuwork().burn_gemm(1024, 1024, 1024, 2) > 10.0
     finally:
         cl.stop()
+

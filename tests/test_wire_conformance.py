@@ -240,3 +240,28 @@ def test_crd_installer_idempotent_over_wire():
         assert rvs2 == rvs, "second install must not rewrite unchanged CRDs"
     finally:
         srv.stop()
+
+
+def test_debug_endpoints_require_auth_when_configured():
+    """ADVICE r1 item 2 follow-through: with bearer auth configured, the
+    introspection surface (/debug/*) rejects anonymous callers with 403 while
+    health stays open; an authenticated caller gets through."""
+    from grove_amd.kubecore.store import Store
+    from grove_amd.kubecore.apiserver import ApiServer
+    import socket
+    import urllib.error
+    import urllib.request
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    srv = ApiServer(Store(), port=port, auth_tokens={"tok1": "agent"}).start()
+    try:
+        with pytest.raises(urllib.error.HTTPError) as exc:
+            urllib.request.urlopen(f"{srv.url}/debug/events", timeout=3)
+        assert exc.value.code == 403
+        req = urllib.request.Request(
+            f"{srv.url}/debug/events", headers={"Authorization": "Bearer tok1"})
+        assert urllib.request.urlopen(req, timeout=3).status == 200
+        assert urllib.request.urlopen(f"{srv.url}/healthz",
+                                      timeout=3).status == 200
+    finally:
+        srv.stop()

@@ -311,10 +311,21 @@ class Cluster:
             if cur is None:
                 out.append(self.store.create(doc))
             else:
-                doc = dict(doc)
-                doc.setdefault("metadata", {})["resourceVersion"] = \
-                    cur["metadata"]["resourceVersion"]
-                out.append(self.store.update(doc))
+                # create-or-update must tolerate racing controller writes (any
+                # status write bumps the shared resourceVersion): re-read + retry
+                from .kubecore.store import ApiError
+                for attempt in range(50):
+                    doc2 = dict(doc)
+                    doc2.setdefault("metadata", {})["resourceVersion"] = \
+                        cur["metadata"]["resourceVersion"]
+                    try:
+                        out.append(self.store.update(doc2))
+                        break
+                    except ApiError as e:
+                        if e.reason != "Conflict" or attempt == 49:
+                            raise
+                        cur = self.store.get(kind, md.get("namespace"),
+                                             md.get("name", ""))
         return out
 
     def delete_pcs(self, name: str, namespace: str = "default") -> None:

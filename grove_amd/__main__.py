@@ -273,6 +273,21 @@ def cmd_apply(args) -> int:
                 n += 1
                 print(f"created {doc.get('kind')}/{doc['metadata']['name']}")
             except urllib.error.HTTPError as e:
+                if e.code == 409:
+                    # kubectl-apply semantics: exists -> strategic-merge PATCH
+                    preq = urllib.request.Request(
+                        f"{url}/{doc['metadata']['name']}",
+                        data=json.dumps(doc).encode(), method="PATCH",
+                        headers={"Content-Type":
+                                 "application/strategic-merge-patch+json"})
+                    try:
+                        urllib.request.urlopen(preq, timeout=10)
+                        n += 1
+                        print(f"configured {doc.get('kind')}/"
+                              f"{doc['metadata']['name']}")
+                        continue
+                    except urllib.error.HTTPError as e2:
+                        e = e2
                 print(f"error {e.code} for {doc.get('kind')}/"
                       f"{doc['metadata'].get('name')}: {e.read().decode()[:200]}")
     return 0 if n else 1

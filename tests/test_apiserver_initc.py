@@ -256,6 +256,10 @@ def test_cli_apply_get_roundtrip(served_cluster, simple1_yaml, tmp_path, capsys)
     assert main(["apply", "-f", str(f), "--server", server]) == 0
     out = capsys.readouterr().out
     assert "created PodCliqueSet/simple1" in out
+    # re-apply = kubectl-apply create-or-patch: second pass PATCHes and reports
+    # "configured" instead of erroring on 409
+    assert main(["apply", "-f", str(f), "--server", server]) == 0
+    assert "configured PodCliqueSet/simple1" in capsys.readouterr().out
 
     deadline = time.time() + 30
     while time.time() < deadline:
